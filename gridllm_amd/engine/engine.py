@@ -1,0 +1,369 @@
+"""The in-process inference engine — the Ollama replacement.
+
+The reference worker shells out to an external Ollama daemon over HTTP
+(reference: client/src/services/OllamaService.ts:97-284). Here the engine is
+in-process: Llama-family models on PyTorch-ROCm with hand-written CDNA4
+kernels (gridllm_amd.ops), a slot-based KV cache in HBM3E, hipGraph-captured
+decode steps (bucketed batch sizes), and a full sampler. Tokens stream to a
+callback — no per-token HTTP/JSON hop (SURVEY.md §3.3 hot-loop fix).
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Callable, Optional, Sequence
+
+import torch
+
+from gridllm_amd.config import EngineConfig
+from gridllm_amd.engine import chat as chat_mod
+from gridllm_amd.engine.sampler import SamplerState, sample
+from gridllm_amd.engine.tokenizer import load_tokenizer
+from gridllm_amd.models.llama import LlamaModel
+from gridllm_amd.models.registry import get_spec
+from gridllm_amd.parallel.tp import TPContext
+from gridllm_amd.types import SamplingOptions
+from gridllm_amd.utils.logging import get_logger
+
+log = get_logger("engine")
+
+_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256)
+
+
+@dataclass
+class GenerationResult:
+    text: str = ""
+    token_ids: list[int] = field(default_factory=list)
+    done_reason: str = "stop"
+    prompt_eval_count: int = 0
+    prompt_eval_duration: int = 0   # ns
+    eval_count: int = 0
+    eval_duration: int = 0          # ns
+    total_duration: int = 0
+    cancelled: bool = False
+
+
+@dataclass
+class _Row:
+    slot: int
+    state: SamplerState
+    decoder: object
+    on_token: Optional[Callable[[str, bool], None]]
+    cancel: Optional[threading.Event]
+    stops: list[str]
+    max_new: int
+    result: GenerationResult
+    last_token: int = 0
+    text_tail: str = ""            # pending window for stop-sequence matching
+
+
+class InferenceEngine:
+    def __init__(self, cfg: Optional[EngineConfig] = None,
+                 tp: Optional[TPContext] = None):
+        self.cfg = cfg or EngineConfig()
+        self.tp = tp or TPContext()
+        dev = self.cfg.device
+        if dev == "auto":
+            dev = "cuda:0" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(dev)
+        self.is_gpu = self.device.type == "cuda"
+        if self.is_gpu and self.tp.size > 1:
+            torch.cuda.set_device(self.device)
+        dtype = self.cfg.dtype
+        if not self.is_gpu and dtype != "fp32":
+            dtype = "fp32"          # CPU reference path runs fp32
+        spec = get_spec(self.cfg.model, self.cfg.max_seq_len)
+        t0 = time.monotonic_ns()
+        # +1 scratch slot for hipGraph warmup/padding rows
+        self.model = LlamaModel(spec, device=str(self.device), dtype=dtype,
+                                tp=self.tp, seed=self.cfg.seed,
+                                kv_slots=self.cfg.max_batch_size + 1,
+                                max_seq=self.cfg.max_seq_len)
+        self.load_duration = time.monotonic_ns() - t0
+        self.spec = spec
+        self.tokenizer = load_tokenizer(self.cfg.weights_path, spec.vocab)
+        self._scratch_slot = self.cfg.max_batch_size  # last slot
+        self._graphs: dict[int, tuple] = {}
+        self._lock = threading.Lock()
+        log.performance("engine ready", model=spec.name,
+                        params=spec.params, device=str(self.device),
+                        tp=self.tp.size, load_ms=self.load_duration // 1_000_000)
+
+    # ------------------------------------------------------------------ sync
+
+    def _sync(self) -> None:
+        if self.is_gpu:
+            torch.cuda.synchronize(self.device)
+
+    # -------------------------------------------------------------- hipGraph
+
+    def _graph_ok(self) -> bool:
+        return self.is_gpu and self.cfg.use_hipgraph and self.tp.size == 1
+
+    def _bucket(self, b: int) -> int:
+        for x in _BUCKETS:
+            if x >= b:
+                return x
+        return b
+
+    def _ensure_graph(self, bucket: int):
+        if bucket in self._graphs:
+            return self._graphs[bucket]
+        tokens = torch.zeros(bucket, dtype=torch.int64, device=self.device)
+        slots = torch.full((bucket,), self._scratch_slot, dtype=torch.int32,
+                           device=self.device)
+        # warmup on a side stream (required before capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.model.forward_decode(tokens, slots)
+        torch.cuda.current_stream().wait_stream(s)
+        self.model.cache.seq_lens[self._scratch_slot] = 0
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            out = self.model.forward_decode(tokens, slots)
+        self.model.cache.seq_lens[self._scratch_slot] = 0
+        self._graphs[bucket] = (g, tokens, slots, out)
+        log.performance("hipGraph captured", bucket=bucket)
+        return self._graphs[bucket]
+
+    def _decode_step(self, token_ids: list[int], slot_ids: list[int]) -> torch.Tensor:
+        """Run one decode step for rows (token, slot). Returns logits [B, V]."""
+        b = len(token_ids)
+        if self._graph_ok():
+            bucket = self._bucket(b)
+            g, tbuf, sbuf, out = self._ensure_graph(bucket)
+            tbuf[:b] = torch.as_tensor(token_ids, dtype=torch.int64,
+                                       device=self.device)
+            if b < bucket:
+                tbuf[b:] = 0
+                sbuf[b:] = self._scratch_slot
+                # keep the scratch slot from overflowing across replays
+                self.model.cache.seq_lens[self._scratch_slot] = 0
+            sbuf[:b] = torch.as_tensor(slot_ids, dtype=torch.int32,
+                                       device=self.device)
+            g.replay()
+            return out[:b]
+        tokens = torch.as_tensor(token_ids, dtype=torch.int64, device=self.device)
+        slots = torch.as_tensor(slot_ids, dtype=torch.int32, device=self.device)
+        return self.model.forward_decode(tokens, slots)
+
+    # ------------------------------------------------------------- sampling
+
+    def _sample_batch(self, logits: torch.Tensor,
+                      rows: Sequence[_Row]) -> list[int]:
+        """Batched sampling; vectorized for the common uniform cases."""
+        states = [r.state for r in rows]
+        if all(s.greedy for s in states):
+            return logits.argmax(dim=-1).tolist()
+        simple = all(
+            s.counts is None and not s.opts.top_p and not s.opts.min_p
+            and not s.opts.top_k and s.gen is None for s in states)
+        if simple:
+            temps = torch.tensor(
+                [s.opts.temperature if s.opts.temperature is not None else 0.8
+                 for s in states], device=logits.device).clamp(min=1e-4)
+            probs = torch.softmax(logits / temps.unsqueeze(1), dim=-1)
+            return torch.multinomial(probs, 1).squeeze(1).tolist()
+        return [sample(logits[i], s) for i, s in enumerate(states)]
+
+    # ------------------------------------------------------------ main loop
+
+    def generate(self, prompt: str, options: Optional[dict] = None,
+                 on_token: Optional[Callable[[str, bool], None]] = None,
+                 cancel: Optional[threading.Event] = None,
+                 raw_stops: Optional[list[str]] = None) -> GenerationResult:
+        return self.generate_batch([(prompt, options or {})],
+                                   on_tokens=[on_token], cancels=[cancel],
+                                   extra_stops=raw_stops)[0]
+
+    def chat(self, messages: list[dict], options: Optional[dict] = None,
+             on_token: Optional[Callable[[str, bool], None]] = None,
+             cancel: Optional[threading.Event] = None) -> GenerationResult:
+        prompt = chat_mod.render_chat(messages)
+        return self.generate(prompt, options, on_token, cancel,
+                             raw_stops=list(chat_mod.CHAT_STOPS))
+
+    def embed(self, inputs: list[str]) -> list[list[float]]:
+        out: list[list[float]] = []
+        with self._lock:
+            for text in inputs:
+                ids = self._encode(text)
+                slot = self.model.cache.alloc()
+                try:
+                    tokens = torch.as_tensor(ids, dtype=torch.int64,
+                                             device=self.device)
+                    _, hidden = self.model.forward_prefill(
+                        tokens, slot, return_hidden=True)
+                    vec = hidden.detach().float().cpu()
+                    norm = vec.norm()
+                    if norm > 0:
+                        vec = vec / norm
+                    out.append(vec.tolist())
+                finally:
+                    self.model.cache.free(slot)
+        return out
+
+    def _encode(self, prompt: str) -> list[int]:
+        ids = self.tokenizer.encode(prompt)
+        limit = self.model.max_seq - 8
+        if len(ids) > limit:
+            ids = ids[-limit:]
+        if not ids:
+            ids = [0]
+        return ids
+
+    def generate_batch(self, requests: list[tuple[str, dict]],
+                       on_tokens: Optional[list] = None,
+                       cancels: Optional[list] = None,
+                       extra_stops: Optional[list[str]] = None
+                       ) -> list[GenerationResult]:
+        """Prefill each request, then batched decode until all complete."""
+        with self._lock:
+            return self._generate_batch(requests, on_tokens, cancels,
+                                        extra_stops)
+
+    def _generate_batch(self, requests, on_tokens, cancels, extra_stops
+                        ) -> list[GenerationResult]:
+        n = len(requests)
+        on_tokens = on_tokens or [None] * n
+        cancels = cancels or [None] * n
+        t_start = time.monotonic_ns()
+        rows: list[_Row] = []
+        results = [GenerationResult() for _ in range(n)]
+
+        # --- prefill (sequential per request; batched decode after)
+        self._sync()
+        t_prefill = time.monotonic_ns()
+        first_logits: list[torch.Tensor] = []
+        for i, (prompt, opts_d) in enumerate(requests):
+            opts = SamplingOptions.from_dict(opts_d)
+            ids = self._encode(prompt)
+            if opts.num_ctx:
+                ids = ids[-max(16, min(opts.num_ctx, self.model.max_seq - 8)):]
+            slot = self.model.cache.alloc()
+            tokens = torch.as_tensor(ids, dtype=torch.int64, device=self.device)
+            logits = self.model.forward_prefill(tokens, slot)
+            first_logits.append(logits)
+            stops = list(opts.stop or [])
+            if extra_stops:
+                stops += extra_stops
+            max_new = opts.num_predict if opts.num_predict and opts.num_predict > 0 \
+                else 128
+            max_new = min(max_new, self.model.max_seq - len(ids) - 1)
+            rows.append(_Row(slot=slot, state=SamplerState.create(opts, self.device),
+                             decoder=self.tokenizer.decode_stream(),
+                             on_token=on_tokens[i], cancel=cancels[i],
+                             stops=stops, max_new=max_new, result=results[i]))
+            results[i].prompt_eval_count = len(ids)
+        self._sync()
+        t_decode = time.monotonic_ns()
+        prefill_ns = t_decode - t_prefill
+        for r in results:
+            r.prompt_eval_duration = prefill_ns // n
+
+        # --- first token from prefill logits
+        active: list[_Row] = []
+        for i, row in enumerate(rows):
+            tok = self._sample_batch(first_logits[i].unsqueeze(0), [row])[0]
+            if self._emit(row, tok) and row.max_new > 1:
+                active.append(row)
+            else:
+                self._retire(row)
+
+        # --- batched decode
+        while active:
+            token_ids = [r.last_token for r in active]
+            slot_ids = [r.slot for r in active]
+            logits = self._decode_step(token_ids, slot_ids)
+            next_ids = self._sample_batch(logits, active)
+            still: list[_Row] = []
+            for r, tok in zip(active, next_ids):
+                if self._emit(r, tok):
+                    still.append(r)
+                else:
+                    self._retire(r)
+            active = still
+        self._sync()
+        t_end = time.monotonic_ns()
+        dec_ns = t_end - t_decode
+        total_evals = max(1, sum(r.eval_count for r in results))
+        for r in results:
+            r.eval_duration = int(dec_ns * (r.eval_count / total_evals))
+            r.total_duration = t_end - t_start
+        return results
+
+    def _emit(self, row: _Row, tok: int) -> bool:
+        """Record a sampled token; returns False when the row is finished."""
+        res = row.result
+        res.token_ids.append(tok)
+        res.eval_count += 1
+        row.last_token = tok
+        row.state.observe(tok, self.spec.vocab)
+        if row.cancel is not None and row.cancel.is_set():
+            res.done_reason = "cancelled"
+            res.cancelled = True
+            return False
+        if tok == self.tokenizer.eos_id:
+            res.done_reason = "stop"
+            return False
+        piece = row.decoder.push(tok)
+        if piece:
+            row.text_tail += piece
+            emit_now = row.text_tail
+            if row.stops:
+                for st in row.stops:
+                    idx = row.text_tail.find(st)
+                    if idx >= 0:
+                        final = row.text_tail[:idx]
+                        if final:
+                            res.text += final
+                            if row.on_token:
+                                row.on_token(final, False)
+                        res.done_reason = "stop"
+                        row.text_tail = ""
+                        return False
+                # hold back a window that could begin a stop sequence
+                hold = max(len(s) for s in row.stops) - 1
+                emit_now = row.text_tail[:-hold] if hold > 0 else row.text_tail
+                row.text_tail = row.text_tail[len(emit_now):]
+            else:
+                row.text_tail = ""
+            if emit_now:
+                res.text += emit_now
+                if row.on_token:
+                    row.on_token(emit_now, False)
+        if res.eval_count >= row.max_new:
+            res.done_reason = "length"
+            return False
+        return True
+
+    def _retire(self, row: _Row) -> None:
+        tail = row.decoder.flush() if hasattr(row.decoder, "flush") else ""
+        final = row.text_tail + tail
+        if final and not row.result.cancelled:
+            row.result.text += final
+            if row.on_token:
+                row.on_token(final, False)
+        row.text_tail = ""
+        if row.on_token:
+            row.on_token("", True)
+        self.model.cache.free(row.slot)
+
+    # ------------------------------------------------------------- metadata
+
+    def model_names(self) -> list[str]:
+        return [self.spec.name]
+
+    def stats(self) -> dict:
+        return {
+            "model": self.spec.name,
+            "params": self.spec.params,
+            "device": str(self.device),
+            "tp": self.tp.size,
+            "kv_slots_free": self.model.cache.n_free,
+            "max_seq": self.model.max_seq,
+        }

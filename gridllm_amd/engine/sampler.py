@@ -1,0 +1,93 @@
+"""Token sampling.
+
+Implements the full option surface the gateway accepts
+(reference contract: server/src/routes/ollama.ts:26-48 and
+client/src/services/OllamaService.ts:108-126): temperature, top_k, top_p,
+min_p, seed, repeat/presence/frequency penalties, stop sequences
+(stop handling lives in the engine loop; this module picks token ids).
+
+Runs on-device with torch ops (rocBLAS-free elementwise/sort paths);
+greedy is a pure argmax. Per-request RNG via a seeded torch.Generator.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+from gridllm_amd.types import SamplingOptions
+
+
+@dataclass
+class SamplerState:
+    opts: SamplingOptions
+    gen: Optional[torch.Generator]
+    device: torch.device
+    # token frequency for penalties (lazy, vocab-sized)
+    counts: Optional[torch.Tensor] = None
+
+    @classmethod
+    def create(cls, opts: SamplingOptions, device: torch.device) -> "SamplerState":
+        gen = None
+        if opts.seed is not None:
+            gen = torch.Generator(device=device)
+            gen.manual_seed(int(opts.seed))
+        return cls(opts=opts, gen=gen, device=device)
+
+    @property
+    def greedy(self) -> bool:
+        t = self.opts.temperature
+        return t is not None and t <= 0.0
+
+    def observe(self, token_id: int, vocab: int) -> None:
+        o = self.opts
+        if (o.repeat_penalty or o.presence_penalty or o.frequency_penalty):
+            if self.counts is None:
+                self.counts = torch.zeros(vocab, dtype=torch.float32,
+                                          device=self.device)
+            self.counts[token_id] += 1
+
+
+def sample(logits: torch.Tensor, state: SamplerState) -> int:
+    """Pick one token id from logits [vocab] (f32)."""
+    o = state.opts
+    if state.counts is not None:
+        seen = state.counts > 0
+        if o.repeat_penalty and o.repeat_penalty != 1.0:
+            pos = logits > 0
+            logits = torch.where(seen & pos, logits / o.repeat_penalty, logits)
+            logits = torch.where(seen & ~pos, logits * o.repeat_penalty, logits)
+        if o.presence_penalty:
+            logits = logits - o.presence_penalty * seen.float()
+        if o.frequency_penalty:
+            logits = logits - o.frequency_penalty * state.counts
+
+    temp = o.temperature if o.temperature is not None else 0.8
+    if temp <= 0.0:
+        return int(torch.argmax(logits).item())
+    logits = logits / temp
+
+    if o.top_k and o.top_k > 0 and o.top_k < logits.shape[-1]:
+        kth = torch.topk(logits, o.top_k).values[-1]
+        logits = torch.where(logits < kth, torch.full_like(logits, float("-inf")),
+                             logits)
+    probs = torch.softmax(logits, dim=-1)
+
+    if o.min_p and 0.0 < o.min_p < 1.0:
+        cutoff = probs.max() * o.min_p
+        probs = torch.where(probs < cutoff, torch.zeros_like(probs), probs)
+
+    if o.top_p and 0.0 < o.top_p < 1.0:
+        sp, si = torch.sort(probs, descending=True)
+        cum = torch.cumsum(sp, dim=-1)
+        keep = cum - sp < o.top_p          # keep tokens until mass reaches top_p
+        sp = torch.where(keep, sp, torch.zeros_like(sp))
+        idx = torch.multinomial(sp, 1, generator=state.gen)
+        return int(si[idx].item())
+
+    s = probs.sum()
+    if not torch.isfinite(s) or s <= 0:
+        return int(torch.argmax(logits).item())
+    return int(torch.multinomial(probs, 1, generator=state.gen).item())

@@ -1,0 +1,115 @@
+"""Tokenization.
+
+Default is a self-contained byte-level tokenizer (ids 0..255 = raw bytes +
+specials) so the framework runs with no downloaded assets (this image has
+no network). If a HuggingFace tokenizer.json path is supplied, the
+`tokenizers` wheel (installed offline) is used instead.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional
+
+
+class ByteTokenizer:
+    BOS = 256
+    EOS = 257
+
+    def __init__(self, vocab_size: int = 512):
+        self.vocab_size = max(vocab_size, 258)
+
+    def encode(self, text: str, bos: bool = True) -> List[int]:
+        ids = list(text.encode("utf-8"))
+        return ([self.BOS] if bos else []) + ids
+
+    def decode(self, ids: List[int]) -> str:
+        data = bytes(i for i in ids if 0 <= i < 256)
+        return data.decode("utf-8", errors="replace")
+
+    def decode_stream(self) -> "ByteStreamDecoder":
+        return ByteStreamDecoder()
+
+    @property
+    def eos_id(self) -> int:
+        return self.EOS
+
+
+class ByteStreamDecoder:
+    """Incremental UTF-8-safe decoding for token streaming."""
+
+    def __init__(self) -> None:
+        self._buf = bytearray()
+
+    def push(self, token_id: int) -> str:
+        if not (0 <= token_id < 256):
+            return ""
+        self._buf.append(token_id)
+        try:
+            out = self._buf.decode("utf-8")
+        except UnicodeDecodeError:
+            if len(self._buf) >= 4:
+                out = self._buf.decode("utf-8", errors="replace")
+            else:
+                return ""
+        self._buf.clear()
+        return out
+
+    def flush(self) -> str:
+        if not self._buf:
+            return ""
+        out = self._buf.decode("utf-8", errors="replace")
+        self._buf.clear()
+        return out
+
+
+class HFTokenizer:
+    def __init__(self, path: str):
+        from tokenizers import Tokenizer  # offline wheel
+        self._tok = Tokenizer.from_file(path)
+        self.vocab_size = self._tok.get_vocab_size()
+        self.eos_id = self.vocab_size - 1
+        for cand in ("<|end_of_text|>", "</s>", "<|eot_id|>"):
+            tid = self._tok.token_to_id(cand)
+            if tid is not None:
+                self.eos_id = tid
+                break
+
+    def encode(self, text: str, bos: bool = True) -> List[int]:
+        return self._tok.encode(text).ids
+
+    def decode(self, ids: List[int]) -> str:
+        return self._tok.decode(ids)
+
+    def decode_stream(self) -> "HFStreamDecoder":
+        return HFStreamDecoder(self._tok)
+
+
+class HFStreamDecoder:
+    def __init__(self, tok) -> None:
+        self._tok = tok
+        self._ids: List[int] = []
+        self._emitted = 0
+
+    def push(self, token_id: int) -> str:
+        self._ids.append(token_id)
+        text = self._tok.decode(self._ids)
+        if text.endswith("�"):
+            return ""
+        out = text[self._emitted:]
+        self._emitted = len(text)
+        return out
+
+    def flush(self) -> str:
+        text = self._tok.decode(self._ids)
+        out = text[self._emitted:]
+        self._emitted = len(text)
+        return out
+
+
+def load_tokenizer(weights_path: str = "", vocab_size: int = 512):
+    if weights_path:
+        tok_json = os.path.join(weights_path, "tokenizer.json")
+        if os.path.exists(tok_json):
+            return HFTokenizer(tok_json)
+    return ByteTokenizer(vocab_size)

@@ -1,0 +1,251 @@
+"""Llama-family transformer, MI355X-first.
+
+Design (no reference analog — the reference delegated all compute to Ollama,
+client/src/services/OllamaService.ts):
+
+* fused weight layout: one QKV GEMM, one gate+up GEMM per layer (library
+  GEMMs via hipBLASLt through torch.nn.functional.linear); everything
+  between GEMMs is a hand-written CDNA4 kernel (gridllm_amd.ops): fused
+  residual-add RMSNorm, RoPE from a host-precomputed table, GQA decode
+  attention with split-S, SwiGLU epilogue, KV append.
+* KV cache: contiguous [slots, Hkv, max_seq, D] bf16 per layer, sized for
+  288 GB HBM3E; slot indirection (slot_map) keeps decode shapes static so
+  the whole decode step is hipGraph-capturable.
+* tensor parallel: head-sharded QKV + row-parallel O / down with one RCCL
+  all-reduce after each (2/layer), KV sharded by head across ranks.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from gridllm_amd import ops
+from gridllm_amd.models.registry import ModelSpec
+from gridllm_amd.parallel.tp import TPContext
+
+_DTYPES = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}
+
+
+@dataclass
+class LayerWeights:
+    wqkv: torch.Tensor       # [(Hq_l + 2*Hkv_l) * D, H]
+    wo: torch.Tensor         # [H, Hq_l * D]
+    wgateup: torch.Tensor    # [2 * F_l, H]
+    wdown: torch.Tensor      # [H, F_l]
+    attn_norm: torch.Tensor  # [H]
+    mlp_norm: torch.Tensor   # [H]
+
+
+class KVCache:
+    """Slot-based KV cache: [slots, Hkv_local, max_seq, D] bf16 per layer."""
+
+    def __init__(self, spec: ModelSpec, slots: int, max_seq: int,
+                 hkv_local: int, device: torch.device, dtype: torch.dtype):
+        self.slots = slots
+        self.max_seq = max_seq
+        shape = (slots, hkv_local, max_seq, spec.head_dim)
+        self.k = [torch.zeros(shape, dtype=dtype, device=device)
+                  for _ in range(spec.layers)]
+        self.v = [torch.zeros(shape, dtype=dtype, device=device)
+                  for _ in range(spec.layers)]
+        self.seq_lens = torch.zeros(slots, dtype=torch.int32, device=device)
+        self._free = list(range(slots))
+
+    def alloc(self) -> int:
+        if not self._free:
+            raise RuntimeError("KV cache: no free slots")
+        s = self._free.pop()
+        self.seq_lens[s] = 0
+        return s
+
+    def free(self, slot: int) -> None:
+        self.seq_lens[slot] = 0
+        self._free.append(slot)
+
+    @property
+    def n_free(self) -> int:
+        return len(self._free)
+
+    def bytes(self) -> int:
+        return sum(t.numel() * t.element_size() for t in self.k) * 2
+
+
+class LlamaModel:
+    def __init__(self, spec: ModelSpec, device: str = "cpu",
+                 dtype: str = "bf16", tp: Optional[TPContext] = None,
+                 seed: int = 0, kv_slots: int = 64, max_seq: int = 0):
+        self.spec = spec
+        self.device = torch.device(device)
+        self.dtype = _DTYPES[dtype]
+        self.tp = tp or TPContext()
+        self.max_seq = max_seq or spec.max_seq_len
+
+        assert spec.heads % self.tp.size == 0
+        assert spec.kv_heads % self.tp.size == 0
+        assert spec.intermediate % self.tp.size == 0
+        self.hq_l = spec.heads // self.tp.size
+        self.hkv_l = spec.kv_heads // self.tp.size
+        self.f_l = spec.intermediate // self.tp.size
+        self.scale = 1.0 / math.sqrt(spec.head_dim)
+
+        self._init_weights(seed)
+        self.cos, self.sin = ops.rope_tables(
+            spec.head_dim, self.max_seq, spec.rope_theta, device=self.device)
+        self.cache = KVCache(spec, kv_slots, self.max_seq, self.hkv_l,
+                             self.device, self.dtype if self.dtype != torch.float32
+                             else torch.float32)
+
+    # -- weights -----------------------------------------------------------
+
+    def _randn(self, gen: torch.Generator, *shape: int,
+               std: float = 0.02) -> torch.Tensor:
+        t = torch.randn(*shape, generator=gen, device=self.device,
+                        dtype=torch.float32) * std
+        return t.to(self.dtype)
+
+    def _init_weights(self, seed: int) -> None:
+        """Random init; TP ranks generate the FULL tensor with the same seed
+        and slice their shard, so any tp degree realizes the same model."""
+        s, tp = self.spec, self.tp
+        d = s.head_dim
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed or 1234)
+
+        self.embed = self._randn(gen, s.vocab, s.hidden)
+        self.final_norm = torch.ones(s.hidden, dtype=self.dtype, device=self.device)
+        if s.tie_embeddings:
+            self.lm_head = self.embed
+        else:
+            self.lm_head = self._randn(gen, s.vocab, s.hidden)
+
+        q0, _ = tp.shard(s.heads)
+        k0, _ = tp.shard(s.kv_heads)
+        f0, _ = tp.shard(s.intermediate)
+        self.layers: list[LayerWeights] = []
+        for _ in range(s.layers):
+            wq = self._randn(gen, s.heads * d, s.hidden)
+            wk = self._randn(gen, s.kv_heads * d, s.hidden)
+            wv = self._randn(gen, s.kv_heads * d, s.hidden)
+            wo_full = self._randn(gen, s.hidden, s.heads * d)
+            wg = self._randn(gen, s.intermediate, s.hidden)
+            wu = self._randn(gen, s.intermediate, s.hidden)
+            wd_full = self._randn(gen, s.hidden, s.intermediate)
+            wqkv = torch.cat([
+                wq[q0 * d:(q0 + self.hq_l) * d],
+                wk[k0 * d:(k0 + self.hkv_l) * d],
+                wv[k0 * d:(k0 + self.hkv_l) * d],
+            ]).contiguous()
+            wo = wo_full[:, q0 * d:(q0 + self.hq_l) * d].contiguous()
+            wgateup = torch.cat([wg[f0:f0 + self.f_l],
+                                 wu[f0:f0 + self.f_l]]).contiguous()
+            wdown = wd_full[:, f0:f0 + self.f_l].contiguous()
+            self.layers.append(LayerWeights(
+                wqkv=wqkv, wo=wo, wgateup=wgateup, wdown=wdown,
+                attn_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device),
+                mlp_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device)))
+
+    def param_bytes(self) -> int:
+        total = self.embed.numel() + self.final_norm.numel()
+        if self.lm_head is not self.embed:
+            total += self.lm_head.numel()
+        for lw in self.layers:
+            for t in (lw.wqkv, lw.wo, lw.wgateup, lw.wdown, lw.attn_norm,
+                      lw.mlp_norm):
+                total += t.numel()
+        return total * self.embed.element_size()
+
+    # -- forward -----------------------------------------------------------
+
+    def _split_qkv(self, qkv: torch.Tensor, n: int):
+        d = self.spec.head_dim
+        q = qkv[:, :self.hq_l * d].view(n, self.hq_l, d)
+        k = qkv[:, self.hq_l * d:(self.hq_l + self.hkv_l) * d].view(n, self.hkv_l, d)
+        v = qkv[:, (self.hq_l + self.hkv_l) * d:].view(n, self.hkv_l, d)
+        return q, k, v
+
+    @torch.no_grad()
+    def forward_prefill(self, tokens: torch.Tensor, slot: int,
+                        start_pos: int = 0,
+                        return_hidden: bool = False) -> torch.Tensor:
+        """Prefill one sequence. tokens: [T] int64 on device.
+        Returns logits [vocab] (f32) of the last position; with
+        return_hidden, returns (logits, mean-pooled final hidden [H] f32)."""
+        s = self.spec
+        t = tokens.shape[0]
+        pos = torch.arange(start_pos, start_pos + t, dtype=torch.int32,
+                           device=self.device)
+        slot_t = torch.full((t,), slot * self.cache.max_seq, dtype=torch.int32,
+                            device=self.device) + pos
+        x = self.embed[tokens]                              # [T, H]
+        resid = x
+        normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
+        for li, lw in enumerate(self.layers):
+            qkv = F.linear(normed, lw.wqkv)
+            q, k, v = self._split_qkv(qkv, t)
+            q, k = ops.rope(q.contiguous(), k.contiguous(), self.cos, self.sin, pos)
+            ops.kv_append(self.cache.k[li], self.cache.v[li], k, v, slot_t)
+            kk = self.cache.k[li][slot, :, :start_pos + t].permute(1, 0, 2).unsqueeze(0)
+            vv = self.cache.v[li][slot, :, :start_pos + t].permute(1, 0, 2).unsqueeze(0)
+            attn = ops.attn_prefill(q.unsqueeze(0), kk, vv, self.scale,
+                                    causal_offset=start_pos)[0]  # [T,Hq_l,D]
+            attn_out = F.linear(attn.reshape(t, -1), lw.wo)
+            self.tp.all_reduce(attn_out)
+            normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,
+                                        residual=resid)
+            gateup = F.linear(normed, lw.wgateup)
+            act = ops.swiglu(gateup)
+            mlp_out = F.linear(act, lw.wdown)
+            self.tp.all_reduce(mlp_out)
+            next_norm = (self.layers[li + 1].attn_norm
+                         if li + 1 < len(self.layers) else self.final_norm)
+            normed, resid = ops.rmsnorm(mlp_out, next_norm, s.norm_eps,
+                                        residual=resid)
+        self.cache.seq_lens[slot] = start_pos + t
+        logits = F.linear(normed[-1:], self.lm_head).float()
+        if return_hidden:
+            return logits[0], normed.float().mean(dim=0)
+        return logits[0]
+
+    @torch.no_grad()
+    def forward_decode(self, tokens: torch.Tensor,
+                       slot_map: torch.Tensor) -> torch.Tensor:
+        """One decode step for a batch. tokens: [B] int64; slot_map: [B] int32.
+        Appends to the cache, advances seq_lens, returns logits [B, vocab] f32.
+        Fully tensor-driven -> hipGraph-capturable."""
+        s = self.spec
+        b = tokens.shape[0]
+        pos = self.cache.seq_lens[slot_map.long()]          # [B] int32
+        slot_t = slot_map * self.cache.max_seq + pos
+        attn_seq = pos + 1
+        x = self.embed[tokens]
+        resid = x
+        normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
+        for li, lw in enumerate(self.layers):
+            qkv = F.linear(normed, lw.wqkv)
+            q, k, v = self._split_qkv(qkv, b)
+            q, k = ops.rope(q.contiguous(), k.contiguous(), self.cos, self.sin, pos)
+            ops.kv_append(self.cache.k[li], self.cache.v[li], k, v, slot_t)
+            attn = ops.attn_decode(q, self.cache.k[li], self.cache.v[li],
+                                   attn_seq, self.scale, slot_map=slot_map)
+            attn_out = F.linear(attn.reshape(b, -1), lw.wo)
+            self.tp.all_reduce(attn_out)
+            normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,
+                                        residual=resid)
+            gateup = F.linear(normed, lw.wgateup)
+            act = ops.swiglu(gateup)
+            mlp_out = F.linear(act, lw.wdown)
+            self.tp.all_reduce(mlp_out)
+            next_norm = (self.layers[li + 1].attn_norm
+                         if li + 1 < len(self.layers) else self.final_norm)
+            normed, resid = ops.rmsnorm(mlp_out, next_norm, s.norm_eps,
+                                        residual=resid)
+        self.cache.seq_lens.index_add_(
+            0, slot_map.long(),
+            torch.ones(b, dtype=torch.int32, device=self.device))
+        logits = F.linear(normed, self.lm_head).float()
+        return logits

@@ -1,0 +1,76 @@
+"""Model architecture registry.
+
+The reference has no model definitions at all (compute lived in Ollama);
+these specs reproduce the Llama-family architectures its deployments serve
+via Ollama model tags (BASELINE.json configs: llama3.2:1b, llama3.2:3b,
+llama3:70b).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, replace
+
+
+@dataclass(frozen=True)
+class ModelSpec:
+    name: str
+    hidden: int
+    layers: int
+    heads: int
+    kv_heads: int
+    head_dim: int
+    intermediate: int
+    vocab: int
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+    max_seq_len: int = 8192
+    tie_embeddings: bool = True
+
+    @property
+    def params(self) -> int:
+        h, f, v = self.hidden, self.intermediate, self.vocab
+        qkv = (self.heads + 2 * self.kv_heads) * self.head_dim * h
+        o = self.heads * self.head_dim * h
+        mlp = 3 * h * f
+        per_layer = qkv + o + mlp + 2 * h
+        emb = v * h * (1 if self.tie_embeddings else 2)
+        return self.layers * per_layer + emb + h
+
+
+_SPECS: dict[str, ModelSpec] = {}
+
+
+def _reg(spec: ModelSpec, *aliases: str) -> None:
+    _SPECS[spec.name] = spec
+    for a in aliases:
+        _SPECS[a] = spec
+
+
+_reg(ModelSpec("llama3.2:1b", hidden=2048, layers=16, heads=32, kv_heads=8,
+               head_dim=64, intermediate=8192, vocab=128256),
+     "llama3.2-1b")
+_reg(ModelSpec("llama3.2:3b", hidden=3072, layers=28, heads=24, kv_heads=8,
+               head_dim=128, intermediate=8192, vocab=128256),
+     "llama3.2", "llama3.2:latest", "llama3.2-3b")
+_reg(ModelSpec("llama3:70b", hidden=8192, layers=80, heads=64, kv_heads=8,
+               head_dim=128, intermediate=28672, vocab=128256,
+               tie_embeddings=False),
+     "llama3.1:70b", "llama3-70b")
+# tiny config for CPU tests and CI
+_reg(ModelSpec("tiny", hidden=256, layers=2, heads=4, kv_heads=2, head_dim=64,
+               intermediate=512, vocab=512, max_seq_len=256),
+     "tiny:latest")
+
+
+def get_spec(name: str, max_seq_len: int = 0) -> ModelSpec:
+    key = name.strip()
+    if key not in _SPECS:
+        raise KeyError(f"unknown model {name!r}; known: {sorted(set(_SPECS))}")
+    spec = _SPECS[key]
+    if max_seq_len and max_seq_len != spec.max_seq_len:
+        spec = replace(spec, max_seq_len=max_seq_len)
+    return spec
+
+
+def list_specs() -> list[str]:
+    return sorted({s.name for s in _SPECS.values()})

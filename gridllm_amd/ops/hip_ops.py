@@ -1,0 +1,150 @@
+"""ctypes binding to the gfx950 kernel library (libgridllm_hip.so).
+
+Launches go onto torch's current HIP stream, so the whole decode step is
+hipGraph-capturable via torch.cuda.CUDAGraph (= hipGraph on ROCm).
+Fails LOUDLY if the library is missing on a GPU machine — there is no
+silent eager fallback on the GPU path.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import Optional
+
+import torch
+
+_LIB_PATH = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                         "libgridllm_hip.so")
+_lib: Optional[ctypes.CDLL] = None
+
+
+class HipExtensionMissing(RuntimeError):
+    pass
+
+
+def _load() -> ctypes.CDLL:
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_LIB_PATH):
+        raise HipExtensionMissing(
+            f"GridLLM HIP kernel library not found at {_LIB_PATH}. "
+            "Build it with: python -m gridllm_amd.ops.build "
+            "(this is required on GPU machines; there is no eager fallback).")
+    lib = ctypes.CDLL(_LIB_PATH)
+    c_vp, c_i, c_l, c_f = ctypes.c_void_p, ctypes.c_int, ctypes.c_long, ctypes.c_float
+    lib.gl_rmsnorm.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_l, c_i, c_f]
+    lib.gl_rmsnorm.restype = c_i
+    lib.gl_rope.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_i, c_i, c_i, c_i]
+    lib.gl_rope.restype = c_i
+    lib.gl_swiglu.argtypes = [c_vp, c_vp, c_vp, c_l, c_i]
+    lib.gl_swiglu.restype = c_i
+    lib.gl_kv_append.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_i, c_i, c_i, c_i]
+    lib.gl_kv_append.restype = c_i
+    lib.gl_attn_decode.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
+                                   c_vp, c_i, c_i, c_i, c_i, c_i, c_i, c_f]
+    lib.gl_attn_decode.restype = c_i
+    lib.gl_device_sync.argtypes = []
+    lib.gl_device_sync.restype = c_i
+    _lib = lib
+    return lib
+
+
+def available() -> bool:
+    return os.path.exists(_LIB_PATH)
+
+
+def _stream() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _check(rc: int, op: str) -> None:
+    if rc != 0:
+        raise RuntimeError(f"HIP kernel {op} failed with code {rc}")
+
+
+def _ptr(t: Optional[torch.Tensor]) -> ctypes.c_void_p:
+    return ctypes.c_void_p(0 if t is None else t.data_ptr())
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float,
+            residual: Optional[torch.Tensor] = None,
+            residual_out: Optional[torch.Tensor] = None,
+            out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """y = rmsnorm(x [+ residual]) * weight; bf16. x: [..., H]."""
+    lib = _load()
+    h = x.shape[-1]
+    n = x.numel() // h
+    if out is None:
+        out = torch.empty_like(x)
+    _check(lib.gl_rmsnorm(_stream(), _ptr(out), _ptr(x), _ptr(weight),
+                          _ptr(residual), _ptr(residual_out), n, h,
+                          ctypes.c_float(eps)), "rmsnorm")
+    return out
+
+
+def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
+         sin: torch.Tensor, pos: torch.Tensor) -> None:
+    """In-place rotary embedding. q: [T,Hq,D], k: [T,Hk,D] bf16;
+    cos/sin: [S,D/2] f32; pos: [T] int32."""
+    lib = _load()
+    t, hq, d = q.shape
+    hk = k.shape[1]
+    _check(lib.gl_rope(_stream(), _ptr(q), _ptr(k), _ptr(cos), _ptr(sin),
+                       _ptr(pos), t, hq, hk, d), "rope")
+
+
+def swiglu(gateup: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out[n,F] = silu(gateup[:, :F]) * gateup[:, F:]; gateup: [..., 2F] bf16."""
+    lib = _load()
+    f2 = gateup.shape[-1]
+    f = f2 // 2
+    n = gateup.numel() // f2
+    if out is None:
+        out = torch.empty(gateup.shape[:-1] + (f,), dtype=gateup.dtype,
+                          device=gateup.device)
+    _check(lib.gl_swiglu(_stream(), _ptr(out), _ptr(gateup), n, f), "swiglu")
+    return out
+
+
+def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k_new: torch.Tensor,
+              v_new: torch.Tensor, slot: torch.Tensor) -> None:
+    """Append rows into [B,Hk,Smax,D] caches. k_new: [N,Hk,D]; slot[N]=b*Smax+pos."""
+    lib = _load()
+    n, hk, d = k_new.shape
+    smax = kcache.shape[2]
+    _check(lib.gl_kv_append(_stream(), _ptr(kcache), _ptr(vcache), _ptr(k_new),
+                            _ptr(v_new), _ptr(slot), n, hk, d, smax), "kv_append")
+
+
+def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+                seqlen: torch.Tensor, scale: float, nsplit: int = 0,
+                slot_map: Optional[torch.Tensor] = None,
+                ws: Optional[torch.Tensor] = None,
+                out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """GQA decode attention. q/out: [B,Hq,D]; caches [Slots,Hk,Smax,D];
+    seqlen [B] int32; slot_map [B] int32 maps batch row -> cache slot
+    (None -> identity). nsplit=0 -> auto."""
+    lib = _load()
+    b, hq, d = q.shape
+    hk, smax = kcache.shape[1], kcache.shape[2]
+    if nsplit <= 0:
+        # fill the chip: >=512 workgroups, but never splits smaller than 256 keys
+        max_useful = max(1, (smax + 255) // 256)
+        nsplit = max(1, min(512 // max(1, b * hk), max_useful, 32))
+    if out is None:
+        out = torch.empty_like(q)
+    if nsplit > 1 and ws is None:
+        g = hq // hk
+        ws = torch.empty(b * hk * g * nsplit * (d + 2), dtype=torch.float32,
+                         device=q.device)
+    _check(lib.gl_attn_decode(_stream(), _ptr(out), _ptr(q), _ptr(kcache),
+                              _ptr(vcache), _ptr(seqlen), _ptr(slot_map),
+                              _ptr(ws), b, hq, hk, d, smax, nsplit,
+                              ctypes.c_float(scale)), "attn_decode")
+    return out
+
+
+def device_sync() -> None:
+    _check(_load().gl_device_sync(), "device_sync")

@@ -1,0 +1,146 @@
+"""HIP kernel numerics vs the plain-PyTorch fp32 reference (gpu-only).
+
+Every kernel is compared against gridllm_amd.ops.reference computed in fp32
+on the same data; tolerances sized for bf16 I/O (~3 decimal digits).
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from gridllm_amd.ops import reference as R  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    from gridllm_amd.ops import hip_ops
+    assert hip_ops.available(), "HIP extension must be built on GPU boxes"
+    return torch.device("cuda:0")
+
+
+def _mx(a, b):
+    return (a.float() - b.float()).abs().max().item()
+
+
+@pytest.mark.parametrize("n,h", [(4, 256), (33, 3072), (128, 8192), (1, 2048)])
+def test_rmsnorm_gpu(dev, n, h):
+    torch.manual_seed(0)
+    from gridllm_amd.ops import hip_ops
+    x = torch.randn(n, h, device=dev).to(torch.bfloat16)
+    w = torch.randn(h, device=dev).to(torch.bfloat16)
+    y = hip_ops.rmsnorm(x, w, 1e-5)
+    ref, _ = R.rmsnorm(x.cpu().float(), w.cpu().float(), 1e-5)
+    assert _mx(y.cpu(), ref) < 0.05
+
+
+def test_rmsnorm_residual_gpu(dev):
+    torch.manual_seed(1)
+    from gridllm_amd.ops import hip_ops
+    n, h = 17, 3072
+    x = torch.randn(n, h, device=dev).to(torch.bfloat16)
+    r = torch.randn(n, h, device=dev).to(torch.bfloat16)
+    w = torch.randn(h, device=dev).to(torch.bfloat16)
+    res_out = torch.empty_like(x)
+    y = hip_ops.rmsnorm(x, w, 1e-5, residual=r, residual_out=res_out)
+    ref_y, ref_res = R.rmsnorm(x.cpu().float(), w.cpu().float(), 1e-5,
+                               residual=r.cpu().float())
+    assert _mx(res_out.cpu(), ref_res) < 0.02
+    assert _mx(y.cpu(), ref_y) < 0.05
+
+
+@pytest.mark.parametrize("t,hq,hk,d", [(7, 24, 8, 128), (64, 32, 8, 64)])
+def test_rope_gpu(dev, t, hq, hk, d):
+    torch.manual_seed(2)
+    from gridllm_amd.ops import hip_ops
+    q = torch.randn(t, hq, d, device=dev).to(torch.bfloat16)
+    k = torch.randn(t, hk, d, device=dev).to(torch.bfloat16)
+    cos, sin = R.rope_tables(d, 256, device=dev)
+    pos = torch.randint(0, 256, (t,), dtype=torch.int32, device=dev)
+    q_ref, k_ref = R.rope(q.cpu().float(), k.cpu().float(),
+                          cos.cpu(), sin.cpu(), pos.cpu())
+    hip_ops.rope(q, k, cos, sin, pos)
+    assert _mx(q.cpu(), q_ref) < 0.05
+    assert _mx(k.cpu(), k_ref) < 0.05
+
+
+def test_swiglu_gpu(dev):
+    torch.manual_seed(3)
+    from gridllm_amd.ops import hip_ops
+    n, f = 65, 8192
+    gu = torch.randn(n, 2 * f, device=dev).to(torch.bfloat16)
+    out = hip_ops.swiglu(gu)
+    ref = R.swiglu(gu.cpu().float())
+    assert _mx(out.cpu(), ref) < 0.1
+
+
+def test_kv_append_gpu(dev):
+    torch.manual_seed(4)
+    from gridllm_amd.ops import hip_ops
+    b, hk, smax, d, n = 3, 8, 64, 128, 5
+    kc = torch.zeros(b, hk, smax, d, device=dev, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    kn = torch.randn(n, hk, d, device=dev).to(torch.bfloat16)
+    vn = torch.randn(n, hk, d, device=dev).to(torch.bfloat16)
+    slot = torch.tensor([0 * smax + 3, 1 * smax + 0, 2 * smax + 63,
+                         0 * smax + 4, 1 * smax + 1],
+                        dtype=torch.int32, device=dev)
+    hip_ops.kv_append(kc, vc, kn, vn, slot)
+    kc_ref = torch.zeros(b, hk, smax, d)
+    vc_ref = torch.zeros_like(kc_ref)
+    R.kv_append(kc_ref, vc_ref, kn.cpu().float(), vn.cpu().float(), slot.cpu())
+    assert _mx(kc.cpu(), kc_ref) < 1e-6  # pure copy, bitwise via bf16
+    assert _mx(vc.cpu(), vc_ref) < 1e-6
+
+
+@pytest.mark.parametrize("b,hq,hk,d,seqs", [
+    (2, 24, 8, 128, [1, 5]),
+    (4, 32, 8, 64, [64, 100, 128, 7]),
+    (1, 64, 8, 128, [1000]),          # forces split-S
+    (2, 8, 8, 128, [300, 17]),        # G=1
+])
+def test_attn_decode_gpu(dev, b, hq, hk, d, seqs):
+    torch.manual_seed(5)
+    from gridllm_amd.ops import hip_ops
+    smax = max(seqs)
+    kc = torch.randn(b, hk, smax, d, device=dev).to(torch.bfloat16)
+    vc = torch.randn(b, hk, smax, d, device=dev).to(torch.bfloat16)
+    q = torch.randn(b, hq, d, device=dev).to(torch.bfloat16)
+    seqlen = torch.tensor(seqs, dtype=torch.int32, device=dev)
+    scale = 1.0 / math.sqrt(d)
+    out = hip_ops.attn_decode(q, kc, vc, seqlen, scale)
+    ref = R.attn_decode(q.cpu().float(), kc.cpu().float(), vc.cpu().float(),
+                        seqlen.cpu(), scale)
+    assert _mx(out.cpu(), ref) < 0.03
+
+
+def test_attn_decode_split_consistency(dev):
+    """nsplit=1 and nsplit=8 must agree (same math, different partitioning)."""
+    torch.manual_seed(6)
+    from gridllm_amd.ops import hip_ops
+    b, hq, hk, d, smax = 2, 24, 8, 128, 2048
+    kc = torch.randn(b, hk, smax, d, device=dev).to(torch.bfloat16)
+    vc = torch.randn(b, hk, smax, d, device=dev).to(torch.bfloat16)
+    q = torch.randn(b, hq, d, device=dev).to(torch.bfloat16)
+    seqlen = torch.tensor([2048, 777], dtype=torch.int32, device=dev)
+    o1 = hip_ops.attn_decode(q, kc, vc, seqlen, 0.0883, nsplit=1)
+    o8 = hip_ops.attn_decode(q, kc, vc, seqlen, 0.0883, nsplit=8)
+    assert _mx(o1.cpu(), o8.cpu()) < 0.01
+
+
+def test_attn_decode_slot_map_gpu(dev):
+    torch.manual_seed(7)
+    from gridllm_amd.ops import hip_ops
+    slots, hq, hk, d, smax = 4, 24, 8, 128, 128
+    kc = torch.randn(slots, hk, smax, d, device=dev).to(torch.bfloat16)
+    vc = torch.randn(slots, hk, smax, d, device=dev).to(torch.bfloat16)
+    q = torch.randn(2, hq, d, device=dev).to(torch.bfloat16)
+    seqlen = torch.tensor([50, 80], dtype=torch.int32, device=dev)
+    slot_map = torch.tensor([3, 1], dtype=torch.int32, device=dev)
+    out = hip_ops.attn_decode(q, kc, vc, seqlen, 0.0883, slot_map=slot_map)
+    ref = R.attn_decode(q.cpu().float(), kc.cpu().float(), vc.cpu().float(),
+                        seqlen.cpu(), 0.0883, slot_map=slot_map.cpu())
+    assert _mx(out.cpu(), ref) < 0.03
