@@ -84,7 +84,10 @@ class InferenceEngine:
         self.load_duration = time.monotonic_ns() - t0
         self.spec = spec
         self.tokenizer = load_tokenizer(self.cfg.weights_path, spec.vocab)
+        # the scratch slot absorbs hipGraph warmup + padding-row traffic and
+        # must never be handed to a real request (determinism bug otherwise)
         self._scratch_slot = self.cfg.max_batch_size  # last slot
+        self.model.cache.reserve(self._scratch_slot)
         self._graphs: dict[int, tuple] = {}
         self._lock = threading.Lock()
         log.performance("engine ready", model=spec.name,

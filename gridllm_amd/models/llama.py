@@ -56,6 +56,11 @@ class KVCache:
         self.seq_lens = torch.zeros(slots, dtype=torch.int32, device=device)
         self._free = list(range(slots))
 
+    def reserve(self, slot: int) -> None:
+        """Remove a slot from the allocatable pool (hipGraph scratch slot)."""
+        if slot in self._free:
+            self._free.remove(slot)
+
     def alloc(self) -> int:
         if not self._free:
             raise RuntimeError("KV cache: no free slots")
