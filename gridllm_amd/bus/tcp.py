@@ -1,0 +1,309 @@
+"""asyncio TCP bus: the multi-process control plane.
+
+NDJSON frames over TCP. The server embeds MemoryBusCore-style state
+(channels / KV with TTL / hashes) and runs inside the scheduler process
+(started by GridLLMServer.start(), like the reference's assumption of a
+reachable Redis — docker-compose.yml:12-30 — but with no external service).
+
+Wire format:
+  client -> server: {"id": n, "op": "...", ...args}
+  server -> client: {"id": n, "ok": true, "value": ...}  (response)
+                    {"push": "<channel>", "payload": {...}}  (pub/sub delivery)
+"""
+
+from __future__ import annotations
+
+import asyncio
+import itertools
+import json
+import time
+from collections import defaultdict
+from typing import Callable, Optional
+
+from gridllm_amd.bus.base import Bus, Handler
+from gridllm_amd.utils.logging import get_logger
+
+log = get_logger("bus")
+
+
+class BusServer:
+    def __init__(self, host: str = "127.0.0.1", port: int = 46379):
+        self.host = host
+        self.port = port
+        self.kv: dict[str, tuple[str, float]] = {}
+        self.hashes: dict[str, dict[str, str]] = defaultdict(dict)
+        self.subs: dict[str, set["_Conn"]] = defaultdict(set)
+        self.conns: set["_Conn"] = set()
+        self._server: Optional[asyncio.AbstractServer] = None
+
+    async def start(self) -> None:
+        self._server = await asyncio.start_server(self._on_conn, self.host,
+                                                  self.port)
+        log.worker("bus server listening", host=self.host, port=self.port)
+
+    async def stop(self) -> None:
+        if self._server:
+            self._server.close()
+            await self._server.wait_closed()
+            self._server = None
+        for conn in list(self.conns):
+            try:
+                conn.writer.close()
+            except Exception:  # noqa: BLE001
+                pass
+        self.conns.clear()
+
+    async def _on_conn(self, reader: asyncio.StreamReader,
+                       writer: asyncio.StreamWriter) -> None:
+        conn = _Conn(self, reader, writer)
+        self.conns.add(conn)
+        try:
+            await conn.run()
+        finally:
+            self.conns.discard(conn)
+
+    # -- state ops ----------------------------------------------------------
+
+    def kv_expired(self, key: str) -> bool:
+        v = self.kv.get(key)
+        if v is None:
+            return True
+        if v[1] and v[1] < time.monotonic():
+            del self.kv[key]
+            return True
+        return False
+
+    async def deliver(self, channel: str, payload: dict) -> None:
+        msg = (json.dumps({"push": channel, "payload": payload}) + "\n").encode()
+        dead = []
+        for conn in list(self.subs.get(channel, ())):
+            try:
+                conn.writer.write(msg)
+            except Exception:  # noqa: BLE001
+                dead.append(conn)
+        for c in dead:
+            await c.drop()
+
+    async def drop_conn(self, conn: "_Conn") -> None:
+        for ch in list(self.subs):
+            self.subs[ch].discard(conn)
+        for wid in conn.identities:
+            log.worker("bus: worker connection lost", worker_id=wid)
+            await self.deliver("worker:disconnected", {"workerId": wid})
+
+
+class _Conn:
+    def __init__(self, server: BusServer, reader: asyncio.StreamReader,
+                 writer: asyncio.StreamWriter):
+        self.s = server
+        self.reader = reader
+        self.writer = writer
+        self.identities: list[str] = []
+        self._dropped = False
+
+    async def drop(self) -> None:
+        if self._dropped:
+            return
+        self._dropped = True
+        try:
+            self.writer.close()
+        except Exception:  # noqa: BLE001
+            pass
+        await self.s.drop_conn(self)
+
+    async def run(self) -> None:
+        try:
+            while True:
+                line = await self.reader.readline()
+                if not line:
+                    break
+                try:
+                    req = json.loads(line)
+                except json.JSONDecodeError:
+                    continue
+                resp = await self._handle(req)
+                if resp is not None:
+                    self.writer.write((json.dumps(resp) + "\n").encode())
+                    await self.writer.drain()
+        except (ConnectionError, asyncio.IncompleteReadError):
+            pass
+        finally:
+            await self.drop()
+
+    async def _handle(self, req: dict) -> Optional[dict]:
+        op = req.get("op")
+        rid = req.get("id")
+        s = self.s
+        value = None
+        if op == "publish":
+            await s.deliver(req["channel"], req.get("payload") or {})
+        elif op == "subscribe":
+            s.subs[req["channel"]].add(self)
+        elif op == "unsubscribe":
+            s.subs[req["channel"]].discard(self)
+        elif op == "identify":
+            self.identities.append(req["worker_id"])
+        elif op == "kv_set":
+            ttl = req.get("ttl_ms") or 0
+            exp = time.monotonic() + ttl / 1000.0 if ttl else 0.0
+            s.kv[req["key"]] = (req["value"], exp)
+        elif op == "kv_get":
+            value = None if s.kv_expired(req["key"]) else s.kv[req["key"]][0]
+        elif op == "kv_del":
+            s.kv.pop(req["key"], None)
+        elif op == "hset":
+            s.hashes[req["key"]][req["field"]] = req["value"]
+        elif op == "hget":
+            value = s.hashes.get(req["key"], {}).get(req["field"])
+        elif op == "hgetall":
+            value = dict(s.hashes.get(req["key"], {}))
+        elif op == "hdel":
+            s.hashes.get(req["key"], {}).pop(req["field"], None)
+        else:
+            return {"id": rid, "ok": False, "error": f"unknown op {op!r}"}
+        if rid is None:
+            return None
+        return {"id": rid, "ok": True, "value": value}
+
+
+class TcpBus(Bus):
+    """Client side. One connection carries requests, responses AND pub/sub
+    pushes (unlike the reference's 3 Redis connections —
+    RedisService.ts:24-35 — an asyncio reader task demuxes safely)."""
+
+    def __init__(self, host: str = "127.0.0.1", port: int = 46379,
+                 on_disconnect: Optional[Callable[[], None]] = None):
+        self.host = host
+        self.port = port
+        self.on_disconnect = on_disconnect
+        self._reader: Optional[asyncio.StreamReader] = None
+        self._writer: Optional[asyncio.StreamWriter] = None
+        self._pending: dict[int, asyncio.Future] = {}
+        self._handlers: dict[str, list[Handler]] = defaultdict(list)
+        self._ids = itertools.count(1)
+        self._reader_task: Optional[asyncio.Task] = None
+        self._identities: list[str] = []
+        self.closed = True
+
+    async def connect(self) -> None:
+        self._reader, self._writer = await asyncio.open_connection(
+            self.host, self.port)
+        self.closed = False
+        self._reader_task = asyncio.get_running_loop().create_task(
+            self._read_loop())
+        # re-establish state after reconnect
+        for wid in self._identities:
+            await self._call("identify", worker_id=wid)
+        for ch in list(self._handlers):
+            if self._handlers[ch]:
+                await self._call("subscribe", channel=ch)
+
+    async def close(self) -> None:
+        self.closed = True
+        if self._reader_task:
+            self._reader_task.cancel()
+            self._reader_task = None
+        if self._writer:
+            try:
+                self._writer.close()
+            except Exception:  # noqa: BLE001
+                pass
+            self._writer = None
+
+    async def _read_loop(self) -> None:
+        assert self._reader is not None
+        try:
+            while True:
+                line = await self._reader.readline()
+                if not line:
+                    break
+                msg = json.loads(line)
+                if "push" in msg:
+                    ch = msg["push"]
+                    for h in list(self._handlers.get(ch, [])):
+                        asyncio.get_running_loop().create_task(
+                            self._safe(h, ch, msg.get("payload") or {}))
+                else:
+                    fut = self._pending.pop(msg.get("id"), None)
+                    if fut and not fut.done():
+                        fut.set_result(msg)
+        except (ConnectionError, json.JSONDecodeError, asyncio.CancelledError):
+            pass
+        finally:
+            for fut in self._pending.values():
+                if not fut.done():
+                    fut.set_exception(ConnectionError("bus connection lost"))
+            self._pending.clear()
+            if not self.closed:
+                self.closed = True
+                if self.on_disconnect:
+                    self.on_disconnect()
+
+    @staticmethod
+    async def _safe(handler: Handler, channel: str, payload: dict) -> None:
+        try:
+            await handler(channel, payload)
+        except Exception:  # noqa: BLE001
+            import traceback
+            traceback.print_exc()
+
+    async def _call(self, op: str, **kw) -> Optional[dict]:
+        if self._writer is None or self.closed:
+            raise ConnectionError("bus not connected")
+        rid = next(self._ids)
+        fut: asyncio.Future = asyncio.get_running_loop().create_future()
+        self._pending[rid] = fut
+        self._writer.write(
+            (json.dumps({"id": rid, "op": op, **kw}) + "\n").encode())
+        await self._writer.drain()
+        resp = await asyncio.wait_for(fut, timeout=10.0)
+        if not resp.get("ok"):
+            raise RuntimeError(f"bus op {op} failed: {resp.get('error')}")
+        return resp.get("value")
+
+    async def identify(self, worker_id: str) -> None:
+        self._identities.append(worker_id)
+        await self._call("identify", worker_id=worker_id)
+
+    async def publish(self, channel: str, payload: dict) -> None:
+        await self._call("publish", channel=channel, payload=payload)
+
+    async def subscribe(self, channel: str, handler: Handler) -> None:
+        first = not self._handlers[channel]
+        self._handlers[channel].append(handler)
+        if first:
+            await self._call("subscribe", channel=channel)
+
+    async def unsubscribe(self, channel: str,
+                          handler: Optional[Handler] = None) -> None:
+        if handler is None:
+            self._handlers[channel] = []
+        else:
+            self._handlers[channel] = [
+                h for h in self._handlers[channel] if h is not handler]
+        if not self._handlers[channel]:
+            try:
+                await self._call("unsubscribe", channel=channel)
+            except ConnectionError:
+                pass
+
+    async def kv_set(self, key: str, value: str, ttl_ms: int = 0) -> None:
+        await self._call("kv_set", key=key, value=value, ttl_ms=ttl_ms)
+
+    async def kv_get(self, key: str) -> Optional[str]:
+        return await self._call("kv_get", key=key)
+
+    async def kv_del(self, key: str) -> None:
+        await self._call("kv_del", key=key)
+
+    async def hset(self, key: str, field: str, value: str) -> None:
+        await self._call("hset", key=key, field=field, value=value)
+
+    async def hget(self, key: str, field: str) -> Optional[str]:
+        return await self._call("hget", key=key, field=field)
+
+    async def hgetall(self, key: str) -> dict[str, str]:
+        return await self._call("hgetall", key=key) or {}
+
+    async def hdel(self, key: str, field: str) -> None:
+        await self._call("hdel", key=key, field=field)

@@ -1,0 +1,139 @@
+"""Request batcher: coalesces concurrent jobs into one batched decode.
+
+The reference could only run 1 job per worker and each job hit Ollama
+serially (client/src/services/WorkerClientService.ts:500-505). Here a worker
+can accept many concurrent jobs; the batcher gathers whatever arrived within
+a small window (or while the previous wave was running) and decodes them as
+ONE batch — the MI355X decode step is weight-read-bound, so batching is
+nearly free throughput.
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+import time
+from concurrent.futures import Future
+from dataclasses import dataclass, field
+from typing import Callable, Optional
+
+from gridllm_amd.engine.engine import GenerationResult, InferenceEngine
+from gridllm_amd.engine import chat as chat_mod
+from gridllm_amd.utils.logging import get_logger
+
+log = get_logger("batcher")
+
+
+@dataclass
+class _Pending:
+    kind: str                      # generate | chat | embedding
+    prompt: str = ""
+    messages: Optional[list] = None
+    inputs: Optional[list[str]] = None
+    options: dict = field(default_factory=dict)
+    on_token: Optional[Callable[[str, bool], None]] = None
+    cancel: Optional[threading.Event] = None
+    future: Future = field(default_factory=Future)
+    extra_stops: Optional[list[str]] = None
+
+
+class Batcher:
+    def __init__(self, engine: InferenceEngine, window_ms: float = 4.0):
+        self.engine = engine
+        self.window_s = window_ms / 1000.0
+        self.q: "queue.Queue[_Pending]" = queue.Queue()
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name="gridllm-batcher")
+        self._thread.start()
+
+    def shutdown(self) -> None:
+        self._stop.set()
+        self.q.put(_Pending(kind="__stop__"))
+        self._thread.join(timeout=5.0)
+
+    # ------------------------------------------------------------ submit API
+
+    def submit_generate(self, prompt: str, options: dict,
+                        on_token=None, cancel=None,
+                        extra_stops=None) -> Future:
+        p = _Pending(kind="generate", prompt=prompt, options=options,
+                     on_token=on_token, cancel=cancel, extra_stops=extra_stops)
+        self.q.put(p)
+        return p.future
+
+    def submit_chat(self, messages: list, options: dict,
+                    on_token=None, cancel=None) -> Future:
+        p = _Pending(kind="chat", messages=messages, options=options,
+                     on_token=on_token, cancel=cancel)
+        self.q.put(p)
+        return p.future
+
+    def submit_embedding(self, inputs: list[str]) -> Future:
+        p = _Pending(kind="embedding", inputs=inputs)
+        self.q.put(p)
+        return p.future
+
+    # ------------------------------------------------------------- the loop
+
+    def _loop(self) -> None:
+        max_batch = max(1, self.engine.cfg.max_batch_size)
+        while not self._stop.is_set():
+            try:
+                first = self.q.get(timeout=0.5)
+            except queue.Empty:
+                continue
+            if first.kind == "__stop__":
+                break
+            wave = [first]
+            deadline = time.monotonic() + self.window_s
+            while len(wave) < max_batch:
+                left = deadline - time.monotonic()
+                if left <= 0:
+                    break
+                try:
+                    nxt = self.q.get(timeout=left)
+                except queue.Empty:
+                    break
+                if nxt.kind == "__stop__":
+                    self._stop.set()
+                    break
+                wave.append(nxt)
+            self._run_wave(wave)
+
+    def _run_wave(self, wave: list[_Pending]) -> None:
+        embeds = [p for p in wave if p.kind == "embedding"]
+        gens = [p for p in wave if p.kind in ("generate", "chat")]
+        for p in embeds:
+            try:
+                p.future.set_result(self.engine.embed(p.inputs or []))
+            except Exception as e:  # noqa: BLE001
+                p.future.set_exception(e)
+        if not gens:
+            return
+        reqs, on_tokens, cancels = [], [], []
+        stops_all: Optional[list[str]] = None
+        for p in gens:
+            if p.kind == "chat":
+                prompt = chat_mod.render_chat(p.messages or [])
+                opts = dict(p.options)
+                opts["stop"] = list(opts.get("stop") or []) + \
+                    list(chat_mod.CHAT_STOPS)
+            else:
+                prompt = p.prompt
+                opts = dict(p.options)
+                if p.extra_stops:
+                    opts["stop"] = list(opts.get("stop") or []) + p.extra_stops
+            reqs.append((prompt, opts))
+            on_tokens.append(p.on_token)
+            cancels.append(p.cancel)
+        try:
+            results = self.engine.generate_batch(reqs, on_tokens=on_tokens,
+                                                 cancels=cancels)
+            for p, r in zip(gens, results):
+                p.future.set_result(r)
+        except Exception as e:  # noqa: BLE001
+            log.logger.exception("wave failed")
+            for p in gens:
+                if not p.future.done():
+                    p.future.set_exception(e)

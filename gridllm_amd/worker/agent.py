@@ -1,0 +1,279 @@
+"""Worker agent: registers a GPU (or TP group) with the control plane and
+executes assigned jobs on the in-process engine.
+
+Reimplements the live worker path of the reference
+(reference: client/src/services/WorkerClientService.ts):
+  * register: hset workers + publish worker:registered (:175-206)
+  * subscribe worker:<id>:job + worker:<id>:reregister (:208-227)
+  * heartbeat every 5 s: publish + TTL key at 2x interval (:311-353)
+  * resource/capability refresh with change-only status publish (:355-440)
+  * job execution dispatched on request_type, streaming chunks published
+    per chunk to job:stream:<id>, completion to job:completed AND
+    job:result:<id> (:497-712)
+Fixes over the reference:
+  * job_cancellation is HONORED (the reference ignored it, :483-491) —
+    the engine's cancel event aborts generation mid-decode.
+  * concurrent jobs are coalesced into one batched decode (engine/batcher).
+  * the engine is in-process — no Ollama HTTP hop, no per-token JSON chain.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import socket
+import threading
+import time
+from typing import Optional
+
+from gridllm_amd.bus.base import Bus
+from gridllm_amd.config import WorkerConfig
+from gridllm_amd.engine.batcher import Batcher
+from gridllm_amd.engine.engine import GenerationResult, InferenceEngine
+from gridllm_amd.types import (Channels, GPUStats, InferenceRequest, Keys,
+                               WorkerInfo, WorkerStatus, now_ms)
+from gridllm_amd.utils.logging import get_logger
+from gridllm_amd.worker import telemetry
+
+log = get_logger("worker")
+
+
+class WorkerAgent:
+    def __init__(self, bus: Bus, engine: InferenceEngine,
+                 cfg: Optional[WorkerConfig] = None,
+                 batcher: Optional[Batcher] = None):
+        self.bus = bus
+        self.engine = engine
+        self.cfg = cfg or WorkerConfig()
+        self.worker_id = self.cfg.worker_id or \
+            f"worker-{socket.gethostname()}-{self.cfg.device}"
+        self.batcher = batcher or Batcher(engine)
+        self.current_jobs = 0
+        self._cancel_events: dict[str, threading.Event] = {}
+        self._tasks: list[asyncio.Task] = []
+        self._job_tasks: set[asyncio.Task] = set()
+        self._last_status: tuple = ()
+        self._started = False
+
+    # ------------------------------------------------------------- lifecycle
+
+    async def start(self) -> None:
+        await self.bus.identify(self.worker_id)
+        await self._register()
+        await self.bus.subscribe(Channels.worker_job(self.worker_id),
+                                 self._h_job_msg)
+        await self.bus.subscribe(Channels.worker_reregister(self.worker_id),
+                                 self._h_reregister)
+        loop = asyncio.get_running_loop()
+        self._tasks.append(loop.create_task(self._heartbeat_loop()))
+        self._tasks.append(loop.create_task(self._resource_loop()))
+        self._started = True
+        log.worker("worker started", worker_id=self.worker_id,
+                   models=self.engine.model_names())
+
+    async def stop(self) -> None:
+        for t in self._tasks:
+            t.cancel()
+        self._tasks.clear()
+        for ev in self._cancel_events.values():
+            ev.set()
+        try:
+            await self.bus.publish(Channels.WORKER_UNREGISTERED,
+                                   {"workerId": self.worker_id})
+        except Exception:  # noqa: BLE001
+            pass
+        self.batcher.shutdown()
+
+    def _info(self) -> WorkerInfo:
+        stats = telemetry.gpu_stats(self.cfg.device)
+        return WorkerInfo(
+            id=self.worker_id,
+            status=(WorkerStatus.BUSY.value
+                    if self.current_jobs >= self.cfg.max_concurrent_tasks
+                    else WorkerStatus.ONLINE.value),
+            models=self.engine.model_names(),
+            max_concurrent_jobs=self.cfg.max_concurrent_tasks,
+            current_jobs=self.current_jobs,
+            performance_tier=telemetry.performance_tier(stats),
+            gpu=stats.to_dict(),
+            tp_degree=self.engine.tp.size,
+        )
+
+    async def _register(self) -> None:
+        info = self._info()
+        await self.bus.hset(Keys.WORKERS, self.worker_id,
+                            json.dumps(info.to_dict()))
+        await self.bus.publish(Channels.WORKER_REGISTERED, info.to_dict())
+
+    async def _h_reregister(self, _ch: str, _payload: dict) -> None:
+        log.worker("re-registration requested", worker_id=self.worker_id)
+        await self._register()
+
+    # ------------------------------------------------------------ heartbeat
+
+    async def _heartbeat_loop(self) -> None:
+        interval = self.cfg.heartbeat_interval_ms / 1000.0
+        while True:
+            try:
+                await self.bus.publish(Channels.WORKER_HEARTBEAT, {
+                    "workerId": self.worker_id,
+                    "status": self._info().status,
+                    "currentJobs": self.current_jobs,
+                    "timestamp": now_ms(),
+                })
+                await self.bus.kv_set(Keys.heartbeat(self.worker_id),
+                                      str(now_ms()),
+                                      ttl_ms=self.cfg.heartbeat_interval_ms * 2)
+            except Exception:  # noqa: BLE001
+                log.logger.warning("heartbeat publish failed")
+            await asyncio.sleep(interval)
+
+    async def _resource_loop(self) -> None:
+        interval = self.cfg.resource_monitor_interval_ms / 1000.0
+        while True:
+            await asyncio.sleep(interval)
+            try:
+                info = self._info()
+                key = (info.status, info.current_jobs, tuple(info.models))
+                if key != self._last_status:
+                    self._last_status = key
+                    await self.bus.publish(Channels.WORKER_STATUS, {
+                        "workerId": self.worker_id,
+                        "status": info.status,
+                        "currentJobs": info.current_jobs,
+                        "models": info.models,
+                        "gpu": info.gpu,
+                    })
+            except Exception:  # noqa: BLE001
+                log.logger.warning("resource publish failed")
+
+    async def _publish_status(self) -> None:
+        info = self._info()
+        await self.bus.publish(Channels.WORKER_STATUS, {
+            "workerId": self.worker_id, "status": info.status,
+            "currentJobs": info.current_jobs,
+        })
+
+    # ------------------------------------------------------------------ jobs
+
+    async def _h_job_msg(self, _ch: str, payload: dict) -> None:
+        mtype = payload.get("type")
+        if mtype == "job_assignment":
+            req = InferenceRequest.from_dict(payload.get("job") or {})
+            if self.current_jobs >= self.cfg.max_concurrent_tasks:
+                # busy rejection (reference :500-505)
+                await self.bus.publish(Channels.JOB_FAILED, {
+                    "jobId": req.id, "workerId": self.worker_id,
+                    "error": "worker busy"})
+                return
+            self.current_jobs += 1
+            await self._publish_status()
+            t = asyncio.get_running_loop().create_task(self._run_job(req))
+            self._job_tasks.add(t)
+            t.add_done_callback(self._job_tasks.discard)
+        elif mtype == "job_cancellation":
+            jid = payload.get("jobId", "")
+            ev = self._cancel_events.get(jid)
+            if ev is not None:
+                ev.set()
+                log.job("job cancellation honored", job_id=jid)
+        else:
+            log.logger.warning(f"unknown job message type {mtype!r}")
+
+    async def _run_job(self, req: InferenceRequest) -> None:
+        loop = asyncio.get_running_loop()
+        cancel = threading.Event()
+        self._cancel_events[req.id] = cancel
+        stream_q: Optional[asyncio.Queue] = asyncio.Queue() if req.stream else None
+        t_start = time.monotonic_ns()
+
+        def on_token(piece: str, done: bool) -> None:
+            if stream_q is not None:
+                loop.call_soon_threadsafe(stream_q.put_nowait, (piece, done))
+
+        pump_task = None
+        if stream_q is not None:
+            pump_task = loop.create_task(self._pump_stream(req.id, stream_q))
+        try:
+            if req.request_type == "embedding":
+                inputs = req.input if req.input is not None else [req.prompt]
+                fut = self.batcher.submit_embedding(inputs)
+                embeddings = await asyncio.wrap_future(fut)
+                payload = {
+                    "jobId": req.id, "workerId": self.worker_id, "done": True,
+                    "model": req.model, "embeddings": embeddings,
+                    "timings": {"total_duration": time.monotonic_ns() - t_start},
+                }
+            else:
+                if req.request_type == "chat" and req.messages:
+                    fut = self.batcher.submit_chat(
+                        req.messages, req.options,
+                        on_token=on_token if req.stream else None,
+                        cancel=cancel)
+                else:
+                    fut = self.batcher.submit_generate(
+                        req.prompt, req.options,
+                        on_token=on_token if req.stream else None,
+                        cancel=cancel)
+                result: GenerationResult = await asyncio.wrap_future(fut)
+                payload = {
+                    "jobId": req.id, "workerId": self.worker_id, "done": True,
+                    "model": req.model,
+                    "response": result.text,
+                    "done_reason": result.done_reason,
+                    "cancelled": result.cancelled,
+                    "timings": {
+                        "total_duration": result.total_duration,
+                        "load_duration": 0,
+                        "prompt_eval_count": result.prompt_eval_count,
+                        "prompt_eval_duration": result.prompt_eval_duration,
+                        "eval_count": result.eval_count,
+                        "eval_duration": result.eval_duration,
+                    },
+                }
+                if req.request_type == "chat":
+                    payload["message"] = {"role": "assistant",
+                                          "content": result.text}
+            if pump_task is not None:
+                await stream_q.put(None)       # sentinel: flush and stop
+                await pump_task
+                pump_task = None
+            await self.bus.publish(Channels.JOB_COMPLETED, {
+                "jobId": req.id, "workerId": self.worker_id})
+            await self.bus.publish(Channels.job_result(req.id), payload)
+            log.job("job completed", job_id=req.id,
+                    eval_count=payload.get("timings", {}).get("eval_count", 0))
+        except Exception as e:  # noqa: BLE001
+            log.logger.exception("job failed")
+            err = {"jobId": req.id, "workerId": self.worker_id,
+                   "error": str(e), "done": True}
+            try:
+                await self.bus.publish(Channels.JOB_FAILED, err)
+                await self.bus.publish(Channels.job_result(req.id), err)
+            except Exception:  # noqa: BLE001
+                pass
+        finally:
+            if pump_task is not None:
+                pump_task.cancel()
+            self._cancel_events.pop(req.id, None)
+            self.current_jobs = max(0, self.current_jobs - 1)
+            try:
+                await self._publish_status()
+            except Exception:  # noqa: BLE001
+                pass
+
+    async def _pump_stream(self, job_id: str, q: asyncio.Queue) -> None:
+        """Publish token chunks to job:stream:<id> (reference :613-626)."""
+        ch = Channels.job_stream(job_id)
+        while True:
+            item = await q.get()
+            if item is None:
+                break
+            piece, done = item
+            await self.bus.publish(ch, {
+                "jobId": job_id, "workerId": self.worker_id,
+                "chunk": {"id": job_id, "response": piece, "done": done},
+                "timestamp": now_ms(),
+            })
+            if done:
+                break

@@ -1,3 +1,6 @@
+import asyncio
+import inspect
+
 import pytest
 import torch
 
@@ -5,6 +8,19 @@ import torch
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires an MI355X GPU (run with -m gpu)")
+    config.addinivalue_line(
+        "markers", "asyncio: coroutine test (run via asyncio.run)")
+
+
+def pytest_pyfunc_call(pyfuncitem):
+    """Minimal async test support (pytest-asyncio is not in this image)."""
+    fn = pyfuncitem.obj
+    if inspect.iscoroutinefunction(fn):
+        kwargs = {n: pyfuncitem.funcargs[n]
+                  for n in pyfuncitem._fixtureinfo.argnames}
+        asyncio.run(asyncio.wait_for(fn(**kwargs), timeout=60))
+        return True
+    return None
 
 
 def pytest_collection_modifyitems(config, items):

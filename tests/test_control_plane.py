@@ -1,0 +1,314 @@
+"""Scheduler + registry semantics on the in-memory bus (fast timers).
+
+Covers the riskiest logic the reference left untested:
+priority ordering, least-loaded selection, retries, orphan requeue-at-front,
+heartbeat-timeout removal, cancellation, streaming channels, crash-resume.
+"""
+
+import asyncio
+import json
+
+import pytest
+
+from gridllm_amd.bus.memory import MemoryBus, MemoryBusCore
+from gridllm_amd.config import ServerConfig
+from gridllm_amd.server.registry import WorkerRegistry
+from gridllm_amd.server.scheduler import JobScheduler
+from gridllm_amd.types import (Channels, InferenceRequest, Keys, WorkerInfo,
+                               now_ms)
+
+FAST = ServerConfig(worker_heartbeat_timeout_ms=300,
+                    worker_cleanup_interval_ms=50,
+                    job_timeout_ms=2000, job_retry_attempts=2,
+                    job_retry_delay_ms=50)
+
+
+class FakeWorker:
+    """Bus-level worker stub: registers, heartbeats, serves jobs."""
+
+    def __init__(self, core, wid, models=("m1",), concurrency=1,
+                 tier="medium", behavior="ok", delay=0.0):
+        self.bus = MemoryBus(core)
+        self.wid = wid
+        self.models = list(models)
+        self.concurrency = concurrency
+        self.tier = tier
+        self.behavior = behavior        # ok | fail | hang
+        self.delay = delay
+        self.jobs_seen = []
+        self.current = 0
+        self._hb_task = None
+
+    async def start(self, heartbeat=True):
+        await self.bus.connect()
+        await self.bus.identify(self.wid)
+        info = WorkerInfo(id=self.wid, models=self.models,
+                          max_concurrent_jobs=self.concurrency,
+                          performance_tier=self.tier)
+        await self.bus.hset(Keys.WORKERS, self.wid, json.dumps(info.to_dict()))
+        await self.bus.publish(Channels.WORKER_REGISTERED, info.to_dict())
+        await self.bus.subscribe(Channels.worker_job(self.wid), self._on_job)
+        if heartbeat:
+            self._hb_task = asyncio.get_running_loop().create_task(self._hb())
+
+    async def _hb(self):
+        while True:
+            await self.bus.publish(Channels.WORKER_HEARTBEAT, {
+                "workerId": self.wid, "status": "online",
+                "currentJobs": self.current})
+            await self.bus.kv_set(Keys.heartbeat(self.wid), str(now_ms()),
+                                  ttl_ms=200)
+            await asyncio.sleep(0.05)
+
+    async def _on_job(self, _ch, payload):
+        if payload.get("type") != "job_assignment":
+            return
+        job = payload["job"]
+        self.jobs_seen.append(job["id"])
+        self.current += 1
+        if self.behavior == "hang":
+            return  # never responds
+        await asyncio.sleep(self.delay)
+        self.current -= 1
+        if self.behavior == "fail":
+            await self.bus.publish(Channels.JOB_FAILED, {
+                "jobId": job["id"], "workerId": self.wid, "error": "boom"})
+        else:
+            await self.bus.publish(Channels.JOB_COMPLETED, {
+                "jobId": job["id"], "workerId": self.wid})
+            await self.bus.publish(Channels.job_result(job["id"]), {
+                "jobId": job["id"], "workerId": self.wid,
+                "response": f"done-{job['id'][:4]}", "done": True})
+
+    async def stop(self):
+        if self._hb_task:
+            self._hb_task.cancel()
+        await self.bus.close()
+
+
+async def make_stack(core, cfg=FAST):
+    bus = MemoryBus(core)
+    await bus.connect()
+    reg = WorkerRegistry(bus, cfg)
+    sched = JobScheduler(bus, reg, cfg)
+    await reg.initialize()
+    await sched.initialize()
+    return bus, reg, sched
+
+
+async def teardown(reg, sched, *workers):
+    for w in workers:
+        await w.stop()
+    await sched.shutdown()
+    await reg.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_register_and_dispatch():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w = FakeWorker(core, "w1")
+    await w.start()
+    await asyncio.sleep(0.05)
+    assert reg.get("w1") is not None
+    req = InferenceRequest(model="m1", prompt="hi", timeout_ms=2000)
+    res = await asyncio.wait_for(sched.submit_and_wait(req), 5)
+    assert res["response"].startswith("done-")
+    assert sched.stats["completed"] == 1
+    assert reg.get("w1").current_jobs == 0
+    await teardown(reg, sched, w)
+
+
+@pytest.mark.asyncio
+async def test_unknown_model_stays_queued_then_404_style():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    req = InferenceRequest(model="nope", prompt="hi", timeout_ms=500)
+    await sched.add_job(req)
+    await asyncio.sleep(0.2)
+    assert sched.queue_position(req.id) == 0
+    await teardown(reg, sched)
+
+
+@pytest.mark.asyncio
+async def test_least_loaded_and_tier_selection():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w1 = FakeWorker(core, "w1", concurrency=4, tier="low")
+    w2 = FakeWorker(core, "w2", concurrency=4, tier="high")
+    await w1.start()
+    await w2.start()
+    await asyncio.sleep(0.05)
+    # equal load -> higher tier (w2) wins
+    sel = sched._select_worker(InferenceRequest(model="m1"))
+    assert sel.id == "w2"
+    await reg.update_job_count("w2", +2)
+    sel = sched._select_worker(InferenceRequest(model="m1"))
+    assert sel.id == "w1"   # least loaded now
+    await teardown(reg, sched, w1, w2)
+
+
+@pytest.mark.asyncio
+async def test_priority_order():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w = FakeWorker(core, "w1", delay=0.03)
+    # enqueue before worker registers so all 3 wait in queue
+    r_low = InferenceRequest(model="m1", priority="low", timeout_ms=3000)
+    r_high = InferenceRequest(model="m1", priority="high", timeout_ms=3000)
+    r_med = InferenceRequest(model="m1", priority="medium", timeout_ms=3000)
+    done = []
+    results = [asyncio.create_task(sched.submit_and_wait(r))
+               for r in (r_low, r_high, r_med)]
+    await asyncio.sleep(0.05)
+    await w.start()
+    await asyncio.wait_for(asyncio.gather(*results), 10)
+    assert w.jobs_seen == [r_high.id, r_med.id, r_low.id]
+    await teardown(reg, sched, w)
+
+
+@pytest.mark.asyncio
+async def test_retry_then_fail():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w = FakeWorker(core, "w1", behavior="fail")
+    await w.start()
+    await asyncio.sleep(0.05)
+    req = InferenceRequest(model="m1", prompt="x", timeout_ms=2000)
+    res = await asyncio.wait_for(sched.submit_and_wait(req), 10)
+    assert "error" in res
+    # initial attempt + 2 retries
+    assert len(w.jobs_seen) == 3
+    assert sched.stats["retries"] == 2
+    assert sched.stats["failed"] == 1
+    await teardown(reg, sched, w)
+
+
+@pytest.mark.asyncio
+async def test_orphan_requeue_on_worker_death():
+    """SIGKILL-equivalent: a hung worker disconnects; its job must be
+    requeued at the FRONT with priority high and complete on a live worker
+    (reference semantics: JobScheduler.ts:553-630)."""
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w1 = FakeWorker(core, "w1", behavior="hang")
+    await w1.start()
+    await asyncio.sleep(0.05)
+    req = InferenceRequest(model="m1", prompt="x", timeout_ms=5000)
+    task = asyncio.create_task(sched.submit_and_wait(req))
+    await asyncio.sleep(0.1)
+    assert req.id in sched.active
+    # worker dies abruptly (bus announces disconnect)
+    await w1.stop()
+    await asyncio.sleep(0.1)
+    assert sched.stats["orphaned"] == 1
+    qreq = sched.queue[0] if sched.queue else None
+    assert qreq is not None and qreq.id == req.id
+    assert qreq.priority == "high"
+    assert qreq.metadata["requeueCount"] == 1
+    # a healthy worker arrives and completes the SAME awaited request
+    w2 = FakeWorker(core, "w2")
+    await w2.start()
+    res = await asyncio.wait_for(task, 10)
+    assert res["response"].startswith("done-")
+    assert res["workerId"] == "w2"
+    await teardown(reg, sched, w2)
+
+
+@pytest.mark.asyncio
+async def test_heartbeat_timeout_removes_worker():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w = FakeWorker(core, "w1")
+    await w.start(heartbeat=False)   # registers but never heartbeats
+    await asyncio.sleep(0.05)
+    assert reg.get("w1") is not None
+    await asyncio.sleep(0.6)         # > heartbeat timeout (300ms) + sweep
+    assert reg.get("w1") is None
+    await teardown(reg, sched, w)
+
+
+@pytest.mark.asyncio
+async def test_cancel_queued_job():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    req = InferenceRequest(model="m1", prompt="x", timeout_ms=5000)
+    task = asyncio.create_task(sched.submit_and_wait(req))
+    await asyncio.sleep(0.05)
+    ok = await sched.cancel_job(req.id)
+    assert ok
+    res = await asyncio.wait_for(task, 5)
+    assert res.get("error") == "cancelled"
+    await teardown(reg, sched)
+
+
+@pytest.mark.asyncio
+async def test_streaming_chunks():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+
+    class StreamWorker(FakeWorker):
+        async def _on_job(self, _ch, payload):
+            if payload.get("type") != "job_assignment":
+                return
+            job = payload["job"]
+            for i, piece in enumerate(["hel", "lo ", "world"]):
+                await self.bus.publish(Channels.job_stream(job["id"]), {
+                    "jobId": job["id"], "workerId": self.wid,
+                    "chunk": {"id": job["id"], "response": piece,
+                              "done": False}})
+            await self.bus.publish(Channels.job_stream(job["id"]), {
+                "jobId": job["id"], "workerId": self.wid,
+                "chunk": {"id": job["id"], "response": "", "done": True}})
+            await self.bus.publish(Channels.JOB_COMPLETED, {
+                "jobId": job["id"], "workerId": self.wid})
+            await self.bus.publish(Channels.job_result(job["id"]), {
+                "jobId": job["id"], "workerId": self.wid,
+                "response": "hello world", "done": True})
+
+    w = StreamWorker(core, "w1")
+    await w.start()
+    await asyncio.sleep(0.05)
+    chunks = []
+    req = InferenceRequest(model="m1", prompt="x", stream=True,
+                           timeout_ms=3000)
+    res = await asyncio.wait_for(
+        sched.submit_streaming(req, lambda c: chunks.append(c)), 5)
+    await asyncio.sleep(0.05)
+    assert res["response"] == "hello world"
+    texts = [c["response"] for c in chunks]
+    assert "".join(texts) == "hello world"
+    assert chunks[-1]["done"] is True
+    await teardown(reg, sched, w)
+
+
+@pytest.mark.asyncio
+async def test_scheduler_crash_resume():
+    """Queue + active jobs persisted on the bus survive a scheduler restart
+    (reference: JobScheduler.ts:82-126)."""
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    req = InferenceRequest(model="m1", prompt="x", timeout_ms=5000)
+    await sched.add_job(req)
+    await asyncio.sleep(0.05)
+    await sched.shutdown()
+    await reg.shutdown()
+    # new scheduler process on the same bus state
+    bus2, reg2, sched2 = await make_stack(core)
+    assert len(sched2.queue) == 1
+    assert sched2.queue[0].id == req.id
+    await teardown(reg2, sched2)
+
+
+@pytest.mark.asyncio
+async def test_registry_rehydration():
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w = FakeWorker(core, "w1")
+    await w.start()
+    await asyncio.sleep(0.05)
+    await sched.shutdown()
+    await reg.shutdown()
+    bus2, reg2, sched2 = await make_stack(core)
+    assert reg2.get("w1") is not None
+    await teardown(reg2, sched2, w)

@@ -1,0 +1,168 @@
+"""WorkerAgent end-to-end against the real (tiny, CPU) engine via the
+scheduler — registration, generate, chat, embedding, streaming, cancellation
+(the reference never honored cancellation: WorkerClientService.ts:483-491)."""
+
+import asyncio
+
+import pytest
+
+from gridllm_amd.bus.memory import MemoryBus, MemoryBusCore
+from gridllm_amd.config import EngineConfig, ServerConfig, WorkerConfig
+from gridllm_amd.engine import InferenceEngine
+from gridllm_amd.server.registry import WorkerRegistry
+from gridllm_amd.server.scheduler import JobScheduler
+from gridllm_amd.types import InferenceRequest
+from gridllm_amd.worker.agent import WorkerAgent
+
+FAST = ServerConfig(worker_heartbeat_timeout_ms=500,
+                    worker_cleanup_interval_ms=100,
+                    job_timeout_ms=10000, job_retry_attempts=1,
+                    job_retry_delay_ms=50)
+
+_ENGINE = None
+
+
+def get_engine():
+    global _ENGINE
+    if _ENGINE is None:
+        _ENGINE = InferenceEngine(EngineConfig(
+            model="tiny", device="cpu", dtype="fp32", max_seq_len=128,
+            max_batch_size=4, use_hipgraph=False))
+    return _ENGINE
+
+
+async def make_world(core):
+    bus_s = MemoryBus(core)
+    await bus_s.connect()
+    reg = WorkerRegistry(bus_s, FAST)
+    sched = JobScheduler(bus_s, reg, FAST)
+    await reg.initialize()
+    await sched.initialize()
+    bus_w = MemoryBus(core)
+    await bus_w.connect()
+    agent = WorkerAgent(bus_w, get_engine(),
+                        WorkerConfig(worker_id="w-test", max_concurrent_tasks=2,
+                                     heartbeat_interval_ms=200))
+    await agent.start()
+    await asyncio.sleep(0.05)
+    return reg, sched, agent
+
+
+async def teardown(reg, sched, agent):
+    await agent.stop()
+    await sched.shutdown()
+    await reg.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_agent_registers_with_real_capabilities():
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    info = reg.get("w-test")
+    assert info is not None
+    assert "tiny" in info.models
+    assert info.max_concurrent_jobs == 2
+    await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_generate_roundtrip():
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    req = InferenceRequest(model="tiny", prompt="hello",
+                           options={"temperature": 0.0, "num_predict": 5},
+                           timeout_ms=30000)
+    res = await asyncio.wait_for(sched.submit_and_wait(req), 30)
+    assert res.get("error") is None
+    assert res["done_reason"] in ("stop", "length")
+    assert res["timings"]["eval_count"] >= 1
+    assert res["timings"]["prompt_eval_count"] >= 1
+    await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_streaming_roundtrip():
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    chunks = []
+    req = InferenceRequest(model="tiny", prompt="stream",
+                           options={"temperature": 0.0, "num_predict": 6},
+                           stream=True, timeout_ms=30000)
+    res = await asyncio.wait_for(
+        sched.submit_streaming(req, lambda c: chunks.append(c)), 30)
+    await asyncio.sleep(0.1)
+    assert res.get("error") is None
+    streamed = "".join(c["response"] for c in chunks)
+    assert streamed == res["response"]
+    assert any(c["done"] for c in chunks)
+    await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_chat_roundtrip():
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    req = InferenceRequest(model="tiny", request_type="chat",
+                           messages=[{"role": "user", "content": "hi"}],
+                           options={"temperature": 0.0, "num_predict": 4},
+                           timeout_ms=30000)
+    res = await asyncio.wait_for(sched.submit_and_wait(req), 30)
+    assert res.get("error") is None
+    assert res["message"]["role"] == "assistant"
+    await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_embedding_roundtrip():
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    req = InferenceRequest(model="tiny", request_type="embedding",
+                           input=["alpha", "beta"], timeout_ms=30000)
+    res = await asyncio.wait_for(sched.submit_and_wait(req), 30)
+    assert res.get("error") is None
+    assert len(res["embeddings"]) == 2
+    await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_cancellation_honored():
+    """cancel_job must abort in-flight generation on the worker."""
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    req = InferenceRequest(model="tiny", prompt="long",
+                           options={"temperature": 0.0, "num_predict": 100000},
+                           timeout_ms=60000)
+    task = asyncio.create_task(sched.submit_and_wait(req))
+    # wait until the job is actually running on the worker
+    for _ in range(100):
+        await asyncio.sleep(0.02)
+        if req.id in agent._cancel_events:
+            break
+    assert req.id in agent._cancel_events
+    ok = await sched.cancel_job(req.id)
+    assert ok
+    res = await asyncio.wait_for(task, 30)
+    assert res.get("error") == "cancelled"
+    # worker must come back idle (generation aborted, slot freed)
+    for _ in range(200):
+        await asyncio.sleep(0.02)
+        if agent.current_jobs == 0:
+            break
+    assert agent.current_jobs == 0
+    assert get_engine().model.cache.n_free == \
+        get_engine().cfg.max_batch_size
+    await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_concurrent_jobs_batched():
+    """Two concurrent jobs coalesce into one engine wave and both finish."""
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    reqs = [InferenceRequest(model="tiny", prompt=p,
+                             options={"temperature": 0.0, "num_predict": 5},
+                             timeout_ms=30000) for p in ("a", "b")]
+    results = await asyncio.wait_for(
+        asyncio.gather(*(sched.submit_and_wait(r) for r in reqs)), 30)
+    assert all(r.get("error") is None for r in results)
+    await teardown(reg, sched, agent)
