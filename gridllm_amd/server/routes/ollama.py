@@ -1,0 +1,280 @@
+"""Ollama-compatible API (reference: server/src/routes/ollama.ts).
+
+Endpoints: /api/generate, /api/chat, /api/tags, /api/embed, /api/embeddings
+(mounted at both /ollama/api/* — the reference's mount point, index.ts:67 —
+and /api/* for drop-in Ollama SDK compatibility).
+
+Wire behavior preserved from the reference:
+  * empty prompt + keep_alive==0 -> synthetic unload response; empty prompt
+    alone -> synthetic load response (ollama.ts:177-214)
+  * streaming = newline-delimited JSON over chunked transfer (:248-303)
+  * /api/tags aggregates worker model lists + gridllm_metadata
+    (:507-571)
+  * 404 when no worker advertises the model (:120-129)
+Difference: /api/chat performs REAL chat templating instead of the
+reference's "role: content" flattening (ollama.ts:366-370) — response
+shapes are identical.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from datetime import datetime, timezone
+from typing import Optional
+
+from fastapi import APIRouter, HTTPException, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from gridllm_amd.server.registry import WorkerRegistry
+from gridllm_amd.server.scheduler import JobScheduler
+from gridllm_amd.types import InferenceRequest
+
+_SAMPLING_KEYS = {
+    "temperature", "top_p", "top_k", "min_p", "seed", "stop", "num_predict",
+    "num_ctx", "repeat_penalty", "presence_penalty", "frequency_penalty",
+}
+
+
+def _now_iso() -> str:
+    return datetime.now(timezone.utc).isoformat().replace("+00:00", "Z")
+
+
+def _clean_options(options: Optional[dict]) -> dict:
+    if not isinstance(options, dict):
+        return {}
+    out = {k: v for k, v in options.items() if k in _SAMPLING_KEYS}
+    stop = out.get("stop")
+    if isinstance(stop, str):
+        out["stop"] = [stop]
+    return out
+
+
+def _timings(res: dict) -> dict:
+    t = res.get("timings") or {}
+    return {
+        "total_duration": int(t.get("total_duration", 0)),
+        "load_duration": int(t.get("load_duration", 0)),
+        "prompt_eval_count": int(t.get("prompt_eval_count", 0)),
+        "prompt_eval_duration": int(t.get("prompt_eval_duration", 0)),
+        "eval_count": int(t.get("eval_count", 0)),
+        "eval_duration": int(t.get("eval_duration", 0)),
+    }
+
+
+def _validate_model(registry: WorkerRegistry, model: str) -> None:
+    if not model or not isinstance(model, str):
+        raise HTTPException(status_code=400, detail="model is required")
+    if not registry.get_workers_by_model(model):
+        raise HTTPException(
+            status_code=404,
+            detail=f"model '{model}' not found — no worker advertises it")
+
+
+def _check_prompt(body: dict) -> str:
+    prompt = body.get("prompt", "")
+    if not isinstance(prompt, str):
+        raise HTTPException(status_code=400, detail="prompt must be a string")
+    if len(prompt) > 100_000:  # reference cap (ollama.ts:19)
+        raise HTTPException(status_code=400, detail="prompt too long (max 100kB)")
+    return prompt
+
+
+def _stream_ndjson(scheduler: JobScheduler, req: InferenceRequest,
+                   to_chunk, to_final):
+    """Shared NDJSON streaming plumbing for generate/chat."""
+
+    async def gen():
+        q: asyncio.Queue = asyncio.Queue()
+
+        def on_chunk(c: dict) -> None:
+            q.put_nowait(("chunk", c))
+
+        async def run():
+            try:
+                res = await scheduler.submit_streaming(req, on_chunk)
+                q.put_nowait(("final", res))
+            except Exception as e:  # noqa: BLE001
+                q.put_nowait(("error", str(e)))
+
+        task = asyncio.get_running_loop().create_task(run())
+        try:
+            while True:
+                kind, item = await q.get()
+                if kind == "chunk":
+                    if item.get("response"):
+                        yield json.dumps(to_chunk(item)) + "\n"
+                    if item.get("done"):
+                        continue  # final payload carries the done record
+                elif kind == "final":
+                    if item.get("error"):
+                        yield json.dumps({"error": item["error"],
+                                          "done": True}) + "\n"
+                    else:
+                        yield json.dumps(to_final(item)) + "\n"
+                    break
+                else:
+                    yield json.dumps({"error": item, "done": True}) + "\n"
+                    break
+        finally:
+            task.cancel()
+
+    return StreamingResponse(gen(), media_type="application/x-ndjson")
+
+
+def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter:
+    r = APIRouter()
+
+    # ------------------------------------------------------------- generate
+
+    @r.post("/api/generate")
+    async def generate(request: Request):
+        body = await request.json()
+        model = body.get("model", "")
+        prompt = _check_prompt(body)
+        stream = bool(body.get("stream", True))
+        keep_alive = body.get("keep_alive")
+
+        # synthetic load/unload (reference ollama.ts:177-214)
+        if prompt == "":
+            _validate_model(registry, model)
+            resp = {"model": model, "created_at": _now_iso(), "response": "",
+                    "done": True}
+            resp["done_reason"] = "unload" if keep_alive in (0, "0") else "load"
+            return JSONResponse(resp)
+
+        _validate_model(registry, model)
+        req = InferenceRequest(
+            model=model, prompt=prompt, request_type="generate",
+            stream=stream, priority=str(body.get("priority", "medium")),
+            timeout_ms=300_000, options=_clean_options(body.get("options")),
+            metadata={k: body[k] for k in
+                      ("suffix", "images", "think", "format", "system",
+                       "template", "raw", "keep_alive", "context")
+                      if k in body},
+        )
+        if body.get("system"):
+            req.prompt = f"{body['system']}\n\n{req.prompt}"
+
+        def chunk_json(c):
+            return {"model": model, "created_at": _now_iso(),
+                    "response": c.get("response", ""), "done": False}
+
+        def final_json(res):
+            return {"model": model, "created_at": _now_iso(), "response": "",
+                    "done": True,
+                    "done_reason": res.get("done_reason", "stop"),
+                    "context": [], **_timings(res)}
+
+        if stream:
+            return _stream_ndjson(scheduler, req, chunk_json, final_json)
+        res = await scheduler.submit_and_wait(req)
+        if res.get("error"):
+            raise HTTPException(status_code=500, detail=res["error"])
+        return JSONResponse({
+            "model": model, "created_at": _now_iso(),
+            "response": res.get("response", ""), "done": True,
+            "done_reason": res.get("done_reason", "stop"), "context": [],
+            **_timings(res)})
+
+    # ----------------------------------------------------------------- chat
+
+    @r.post("/api/chat")
+    async def chat(request: Request):
+        body = await request.json()
+        model = body.get("model", "")
+        messages = body.get("messages")
+        if not isinstance(messages, list) or not messages:
+            raise HTTPException(status_code=400,
+                                detail="messages must be a non-empty array")
+        _validate_model(registry, model)
+        stream = bool(body.get("stream", True))
+        req = InferenceRequest(
+            model=model, messages=messages, request_type="chat",
+            stream=stream, timeout_ms=300_000,
+            options=_clean_options(body.get("options")))
+
+        def chunk_json(c):
+            return {"model": model, "created_at": _now_iso(),
+                    "message": {"role": "assistant",
+                                "content": c.get("response", "")},
+                    "done": False}
+
+        def final_json(res):
+            return {"model": model, "created_at": _now_iso(),
+                    "message": {"role": "assistant", "content": ""},
+                    "done": True,
+                    "done_reason": res.get("done_reason", "stop"),
+                    **_timings(res)}
+
+        if stream:
+            return _stream_ndjson(scheduler, req, chunk_json, final_json)
+        res = await scheduler.submit_and_wait(req)
+        if res.get("error"):
+            raise HTTPException(status_code=500, detail=res["error"])
+        return JSONResponse({
+            "model": model, "created_at": _now_iso(),
+            "message": res.get("message") or
+            {"role": "assistant", "content": res.get("response", "")},
+            "done": True, "done_reason": res.get("done_reason", "stop"),
+            **_timings(res)})
+
+    # ----------------------------------------------------------------- tags
+
+    @r.get("/api/tags")
+    async def tags():
+        models = registry.get_all_models()
+        out = []
+        for name, n_workers in sorted(models.items()):
+            out.append({
+                "name": name, "model": name,
+                "modified_at": _now_iso(),
+                "size": 0, "digest": "",
+                "details": {"parent_model": "", "format": "safetensors",
+                            "family": "llama", "families": ["llama"],
+                            "parameter_size": "", "quantization_level": "BF16"},
+                "gridllm_metadata": {"num_workers_with_model": n_workers},
+            })
+        return JSONResponse({"models": out})
+
+    # ---------------------------------------------------------------- embed
+
+    @r.post("/api/embed")
+    async def embed(request: Request):
+        body = await request.json()
+        model = body.get("model", "")
+        inp = body.get("input")
+        if inp is None:
+            raise HTTPException(status_code=400, detail="input is required")
+        inputs = [inp] if isinstance(inp, str) else list(inp)
+        _validate_model(registry, model)
+        req = InferenceRequest(model=model, request_type="embedding",
+                               input=inputs, timeout_ms=300_000)
+        res = await scheduler.submit_and_wait(req)
+        if res.get("error"):
+            raise HTTPException(status_code=500, detail=res["error"])
+        t = _timings(res)
+        return JSONResponse({
+            "model": model, "embeddings": res.get("embeddings", []),
+            "total_duration": t["total_duration"],
+            "load_duration": t["load_duration"],
+            "prompt_eval_count": t["prompt_eval_count"]})
+
+    @r.post("/api/embeddings")
+    async def embeddings_legacy(request: Request):
+        body = await request.json()
+        model = body.get("model", "")
+        prompt = body.get("prompt")
+        if prompt is None:
+            raise HTTPException(status_code=400, detail="prompt is required")
+        _validate_model(registry, model)
+        req = InferenceRequest(model=model, request_type="embedding",
+                               input=[prompt], timeout_ms=300_000)
+        res = await scheduler.submit_and_wait(req)
+        if res.get("error"):
+            raise HTTPException(status_code=500, detail=res["error"])
+        embs = res.get("embeddings") or [[]]
+        return JSONResponse({"embedding": embs[0]})
+
+    return r

@@ -1,0 +1,241 @@
+"""OpenAI-compatible API (reference: server/src/routes/openai.ts).
+
+/v1/completions, /v1/chat/completions (SSE streaming with `data: ...` frames
+and a terminating `data: [DONE]`, :446-549/:695-789), /v1/models (:822-874).
+Option mapping mirrors :387-411 (max_tokens -> num_predict etc.); chat
+requests run the real chat path (:651-654). response_format json/json_schema
+accepted and forwarded as metadata.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import Optional
+
+from fastapi import APIRouter, HTTPException, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from gridllm_amd.server.registry import WorkerRegistry
+from gridllm_amd.server.scheduler import JobScheduler
+from gridllm_amd.types import InferenceRequest
+
+
+def _map_options(body: dict) -> dict:
+    """OpenAI params -> engine sampling options (reference openai.ts:387-411)."""
+    out: dict = {}
+    if body.get("temperature") is not None:
+        out["temperature"] = body["temperature"]
+    if body.get("top_p") is not None:
+        out["top_p"] = body["top_p"]
+    if body.get("max_tokens") is not None:
+        out["num_predict"] = body["max_tokens"]
+    if body.get("max_completion_tokens") is not None:
+        out["num_predict"] = body["max_completion_tokens"]
+    if body.get("seed") is not None:
+        out["seed"] = body["seed"]
+    if body.get("stop") is not None:
+        stop = body["stop"]
+        out["stop"] = [stop] if isinstance(stop, str) else list(stop)
+    if body.get("presence_penalty") is not None:
+        out["presence_penalty"] = body["presence_penalty"]
+    if body.get("frequency_penalty") is not None:
+        out["frequency_penalty"] = body["frequency_penalty"]
+    return out
+
+
+def _usage(res: dict) -> dict:
+    t = res.get("timings") or {}
+    p = int(t.get("prompt_eval_count", 0))
+    c = int(t.get("eval_count", 0))
+    return {"prompt_tokens": p, "completion_tokens": c, "total_tokens": p + c}
+
+
+def _finish_reason(res: dict) -> str:
+    return {"length": "length", "cancelled": "stop"}.get(
+        res.get("done_reason", "stop"), "stop")
+
+
+def _convert_messages(messages: list) -> list:
+    """OpenAI messages -> engine messages, incl. multimodal text/image split
+    (reference openai.ts:204-243)."""
+    out = []
+    for m in messages:
+        role = m.get("role", "user")
+        content = m.get("content", "")
+        images = []
+        if isinstance(content, list):
+            texts = []
+            for part in content:
+                if not isinstance(part, dict):
+                    continue
+                if part.get("type") == "text":
+                    texts.append(part.get("text", ""))
+                elif part.get("type") == "image_url":
+                    url = (part.get("image_url") or {}).get("url", "")
+                    if url.startswith("data:") and "," in url:
+                        images.append(url.split(",", 1)[1])
+            content = "\n".join(texts)
+        msg = {"role": role, "content": content}
+        if images:
+            msg["images"] = images
+        out.append(msg)
+    return out
+
+
+def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter:
+    r = APIRouter()
+
+    def _validate(model: str) -> None:
+        if not model:
+            raise HTTPException(status_code=400, detail="model is required")
+        if not registry.get_workers_by_model(model):
+            raise HTTPException(status_code=404,
+                                detail=f"model '{model}' not found")
+
+    def _sse(scheduler: JobScheduler, req: InferenceRequest, make_chunk,
+             make_final):
+        async def gen():
+            q: asyncio.Queue = asyncio.Queue()
+
+            def on_chunk(c: dict) -> None:
+                q.put_nowait(("chunk", c))
+
+            async def run():
+                try:
+                    res = await scheduler.submit_streaming(req, on_chunk)
+                    q.put_nowait(("final", res))
+                except Exception as e:  # noqa: BLE001
+                    q.put_nowait(("error", str(e)))
+
+            task = asyncio.get_running_loop().create_task(run())
+            try:
+                while True:
+                    kind, item = await q.get()
+                    if kind == "chunk":
+                        if item.get("response"):
+                            yield f"data: {json.dumps(make_chunk(item))}\n\n"
+                    elif kind == "final":
+                        if not item.get("error"):
+                            yield f"data: {json.dumps(make_final(item))}\n\n"
+                        yield "data: [DONE]\n\n"
+                        break
+                    else:
+                        yield f"data: {json.dumps({'error': item})}\n\n"
+                        yield "data: [DONE]\n\n"
+                        break
+            finally:
+                task.cancel()
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    # ---------------------------------------------------------- completions
+
+    @r.post("/v1/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        model = body.get("model", "")
+        _validate(model)
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):
+            prompt = prompt[0] if prompt else ""
+        stream = bool(body.get("stream", False))
+        echo = bool(body.get("echo", False))
+        cid = f"cmpl-{uuid.uuid4().hex[:24]}"
+        created = int(time.time())
+        req = InferenceRequest(model=model, prompt=prompt,
+                               request_type="generate", stream=stream,
+                               timeout_ms=300_000, options=_map_options(body))
+
+        def make_chunk(c):
+            return {"id": cid, "object": "text_completion", "created": created,
+                    "model": model,
+                    "choices": [{"text": c.get("response", ""), "index": 0,
+                                 "logprobs": None, "finish_reason": None}]}
+
+        def make_final(res):
+            return {"id": cid, "object": "text_completion", "created": created,
+                    "model": model,
+                    "choices": [{"text": "", "index": 0, "logprobs": None,
+                                 "finish_reason": _finish_reason(res)}],
+                    "usage": _usage(res)}
+
+        if stream:
+            return _sse(scheduler, req, make_chunk, make_final)
+        res = await scheduler.submit_and_wait(req)
+        if res.get("error"):
+            raise HTTPException(status_code=500, detail=res["error"])
+        text = res.get("response", "")
+        if echo:
+            text = prompt + text
+        return JSONResponse({
+            "id": cid, "object": "text_completion", "created": created,
+            "model": model,
+            "choices": [{"text": text, "index": 0, "logprobs": None,
+                         "finish_reason": _finish_reason(res)}],
+            "usage": _usage(res)})
+
+    # ----------------------------------------------------- chat completions
+
+    @r.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        body = await request.json()
+        model = body.get("model", "")
+        _validate(model)
+        messages = body.get("messages")
+        if not isinstance(messages, list) or not messages:
+            raise HTTPException(status_code=400,
+                                detail="messages must be a non-empty array")
+        stream = bool(body.get("stream", False))
+        cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        created = int(time.time())
+        metadata = {}
+        rf = body.get("response_format")
+        if isinstance(rf, dict):
+            metadata["format"] = rf.get("type")
+        req = InferenceRequest(model=model,
+                               messages=_convert_messages(messages),
+                               request_type="chat", stream=stream,
+                               timeout_ms=300_000, options=_map_options(body),
+                               metadata=metadata)
+
+        def make_chunk(c):
+            return {"id": cid, "object": "chat.completion.chunk",
+                    "created": created, "model": model,
+                    "choices": [{"index": 0,
+                                 "delta": {"content": c.get("response", "")},
+                                 "finish_reason": None}]}
+
+        def make_final(res):
+            return {"id": cid, "object": "chat.completion.chunk",
+                    "created": created, "model": model,
+                    "choices": [{"index": 0, "delta": {},
+                                 "finish_reason": _finish_reason(res)}],
+                    "usage": _usage(res)}
+
+        if stream:
+            return _sse(scheduler, req, make_chunk, make_final)
+        res = await scheduler.submit_and_wait(req)
+        if res.get("error"):
+            raise HTTPException(status_code=500, detail=res["error"])
+        msg = res.get("message") or {"role": "assistant",
+                                     "content": res.get("response", "")}
+        return JSONResponse({
+            "id": cid, "object": "chat.completion", "created": created,
+            "model": model,
+            "choices": [{"index": 0, "message": msg, "logprobs": None,
+                         "finish_reason": _finish_reason(res)}],
+            "usage": _usage(res)})
+
+    # --------------------------------------------------------------- models
+
+    @r.get("/v1/models")
+    async def models():
+        out = [{"id": name, "object": "model", "created": int(time.time()),
+                "owned_by": "gridllm"}
+               for name in sorted(registry.get_all_models())]
+        return JSONResponse({"object": "list", "data": out})
+
+    return r

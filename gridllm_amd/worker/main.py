@@ -1,0 +1,66 @@
+"""Worker process entrypoint (reference: client/src/index.ts).
+
+One process per MI355X GPU (pin with WORKER_DEVICE / HIP_VISIBLE_DEVICES);
+for TP groups, launch via torch.distributed.run — rank 0 fronts the group
+as a single registered worker, other ranks follow broadcasts
+(gridllm_amd/worker/tp_runner.py).
+
+Reconnects to the bus with exponential backoff
+(reference: WorkerClientService.ts:281-309).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+
+from gridllm_amd.bus.tcp import TcpBus
+from gridllm_amd.config import EngineConfig, WorkerConfig
+from gridllm_amd.engine import InferenceEngine
+from gridllm_amd.utils.logging import get_logger
+from gridllm_amd.worker.agent import WorkerAgent
+
+log = get_logger("worker-main")
+
+
+async def run_worker(wcfg: WorkerConfig, ecfg: EngineConfig) -> None:
+    engine = InferenceEngine(ecfg)
+    disconnected = asyncio.Event()
+    bus = TcpBus(wcfg.bus_host, wcfg.bus_port,
+                 on_disconnect=lambda: disconnected.set())
+    agent = WorkerAgent(bus, engine, wcfg)
+    attempt = 0
+    while True:
+        try:
+            await bus.connect()
+            attempt = 0
+            disconnected.clear()
+            await agent.start()
+            await disconnected.wait()
+            log.worker("bus connection lost; reconnecting",
+                       worker_id=agent.worker_id)
+            for t in agent._tasks:
+                t.cancel()
+            agent._tasks.clear()
+        except (ConnectionError, OSError) as e:
+            attempt += 1
+            if attempt > wcfg.reconnect_max_attempts:
+                log.logger.error("giving up reconnecting to bus")
+                raise
+            delay = min(30.0, wcfg.reconnect_base_delay_ms / 1000.0
+                        * (2 ** attempt))
+            log.worker("reconnect backoff", attempt=attempt, delay_s=delay,
+                       error=str(e))
+            await asyncio.sleep(delay)
+
+
+def main() -> None:
+    wcfg = WorkerConfig.from_env()
+    ecfg = EngineConfig.from_env()
+    if "WORKER_DEVICE" in os.environ:
+        ecfg.device = f"cuda:{wcfg.device}"
+    asyncio.run(run_worker(wcfg, ecfg))
+
+
+if __name__ == "__main__":
+    main()

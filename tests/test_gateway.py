@@ -1,0 +1,246 @@
+"""Gateway wire-format tests: full in-process stack (FastAPI TestClient +
+scheduler + registry + worker agent + tiny CPU engine).
+
+Shape assertions follow the reference's differential oracle
+(reference: tests/integration/integration.ts — same key set / same typeof),
+against golden Ollama/OpenAI response shapes, with no live Ollama needed.
+"""
+
+import asyncio
+import json
+import threading
+
+import pytest
+from fastapi.testclient import TestClient
+
+from gridllm_amd.bus.memory import MemoryBus, MemoryBusCore
+from gridllm_amd.config import EngineConfig, ServerConfig, WorkerConfig
+from gridllm_amd.engine import InferenceEngine
+from gridllm_amd.server.app import create_app
+from gridllm_amd.server.registry import WorkerRegistry
+from gridllm_amd.server.scheduler import JobScheduler
+from gridllm_amd.worker.agent import WorkerAgent
+
+CFG = ServerConfig(worker_heartbeat_timeout_ms=2000,
+                   worker_cleanup_interval_ms=500, job_retry_attempts=1,
+                   job_retry_delay_ms=100)
+
+# golden response shapes (keys + typeof), captured from Ollama/OpenAI docs
+GOLDEN_GENERATE = {
+    "model": str, "created_at": str, "response": str, "done": bool,
+    "done_reason": str, "context": list, "total_duration": int,
+    "load_duration": int, "prompt_eval_count": int,
+    "prompt_eval_duration": int, "eval_count": int, "eval_duration": int,
+}
+GOLDEN_CHAT = {
+    "model": str, "created_at": str, "message": dict, "done": bool,
+    "done_reason": str, "total_duration": int, "load_duration": int,
+    "prompt_eval_count": int, "prompt_eval_duration": int, "eval_count": int,
+    "eval_duration": int,
+}
+GOLDEN_OPENAI_COMPLETION = {
+    "id": str, "object": str, "created": int, "model": str, "choices": list,
+    "usage": dict,
+}
+
+
+def assert_shape(obj: dict, golden: dict) -> None:
+    """Same key set and same type per key (integration.ts:6-35 semantics)."""
+    assert set(obj.keys()) == set(golden.keys()), \
+        f"keys {sorted(obj)} != {sorted(golden)}"
+    for k, t in golden.items():
+        assert isinstance(obj[k], t), f"{k}: {type(obj[k])} != {t}"
+
+
+def build_app():
+    """Whole stack initialized inside the app lifespan, so every coroutine
+    (routes, scheduler tasks, agent tasks) shares the TestClient's loop."""
+    core = MemoryBusCore()
+    bus = MemoryBus(core)
+    registry = WorkerRegistry(bus, CFG)
+    scheduler = JobScheduler(bus, registry, CFG)
+    app = create_app(scheduler, registry)
+    state = {}
+
+    @app.on_event("startup")
+    async def _up():
+        await bus.connect()
+        await registry.initialize()
+        await scheduler.initialize()
+        wbus = MemoryBus(core)
+        await wbus.connect()
+        engine = InferenceEngine(EngineConfig(
+            model="tiny", device="cpu", dtype="fp32", max_seq_len=128,
+            max_batch_size=4, use_hipgraph=False))
+        agent = WorkerAgent(wbus, engine, WorkerConfig(
+            worker_id="w-gw", max_concurrent_tasks=4,
+            heartbeat_interval_ms=500))
+        await agent.start()
+        await asyncio.sleep(0.05)
+        state["agent"] = agent
+
+    @app.on_event("shutdown")
+    async def _down():
+        await state["agent"].stop()
+        await scheduler.shutdown()
+        await registry.shutdown()
+
+    return app
+
+
+@pytest.fixture(scope="module")
+def client():
+    with TestClient(build_app()) as c:
+        yield c
+
+
+def test_generate_non_stream_shape(client):
+    r = client.post("/api/generate", json={
+        "model": "tiny", "prompt": "hello", "stream": False,
+        "options": {"temperature": 0.0, "num_predict": 4}})
+    assert r.status_code == 200, r.text
+    assert_shape(r.json(), GOLDEN_GENERATE)
+    assert r.json()["eval_count"] >= 1
+
+
+def test_generate_streaming_ndjson(client):
+    with client.stream("POST", "/api/generate", json={
+            "model": "tiny", "prompt": "hello", "stream": True,
+            "options": {"temperature": 0.0, "num_predict": 4}}) as r:
+        assert r.status_code == 200
+        lines = [json.loads(l) for l in r.iter_lines() if l]
+    assert lines, "no stream chunks"
+    assert lines[-1]["done"] is True
+    assert "eval_count" in lines[-1]
+    body = "".join(c.get("response", "") for c in lines[:-1])
+    assert isinstance(body, str)
+
+
+def test_generate_unknown_model_404(client):
+    r = client.post("/api/generate", json={"model": "nope", "prompt": "x"})
+    assert r.status_code == 404
+
+
+def test_generate_load_unload_synthetic(client):
+    r = client.post("/api/generate", json={"model": "tiny", "prompt": ""})
+    assert r.status_code == 200
+    assert r.json()["done_reason"] == "load"
+    r = client.post("/api/generate", json={"model": "tiny", "prompt": "",
+                                           "keep_alive": 0})
+    assert r.json()["done_reason"] == "unload"
+
+
+def test_chat_non_stream_shape(client):
+    r = client.post("/api/chat", json={
+        "model": "tiny", "stream": False,
+        "messages": [{"role": "user", "content": "hi"}],
+        "options": {"temperature": 0.0, "num_predict": 4}})
+    assert r.status_code == 200, r.text
+    assert_shape(r.json(), GOLDEN_CHAT)
+    assert r.json()["message"]["role"] == "assistant"
+
+
+def test_tags(client):
+    r = client.get("/api/tags")
+    assert r.status_code == 200
+    models = r.json()["models"]
+    assert any(m["name"] == "tiny" for m in models)
+    m = next(m for m in models if m["name"] == "tiny")
+    assert m["gridllm_metadata"]["num_workers_with_model"] == 1
+    # also available at the reference's mount point
+    r2 = client.get("/ollama/api/tags")
+    assert r2.status_code == 200
+
+
+def test_embed_and_legacy_embeddings(client):
+    r = client.post("/api/embed", json={"model": "tiny", "input": ["a", "b"]})
+    assert r.status_code == 200, r.text
+    assert len(r.json()["embeddings"]) == 2
+    r = client.post("/api/embeddings", json={"model": "tiny", "prompt": "a"})
+    assert r.status_code == 200
+    assert isinstance(r.json()["embedding"], list)
+
+
+def test_openai_completions_shape(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny", "prompt": "hello", "max_tokens": 4,
+        "temperature": 0.0})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert_shape(body, GOLDEN_OPENAI_COMPLETION)
+    assert body["object"] == "text_completion"
+    c = body["choices"][0]
+    assert set(c) == {"text", "index", "logprobs", "finish_reason"}
+    u = body["usage"]
+    assert u["total_tokens"] == u["prompt_tokens"] + u["completion_tokens"]
+
+
+def test_openai_completions_echo(client):
+    r = client.post("/v1/completions", json={
+        "model": "tiny", "prompt": "xyz", "max_tokens": 2,
+        "temperature": 0.0, "echo": True})
+    assert r.json()["choices"][0]["text"].startswith("xyz")
+
+
+def test_openai_chat_shape_and_stream(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny", "max_tokens": 4, "temperature": 0.0,
+        "messages": [{"role": "user", "content": "hi"}]})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    # SSE streaming with [DONE]
+    with client.stream("POST", "/v1/chat/completions", json={
+            "model": "tiny", "max_tokens": 4, "temperature": 0.0,
+            "stream": True,
+            "messages": [{"role": "user", "content": "hi"}]}) as r:
+        assert r.status_code == 200
+        events = [l for l in r.iter_lines() if l.startswith("data: ")]
+    assert events[-1] == "data: [DONE]"
+    first = json.loads(events[0][6:])
+    assert first["object"] == "chat.completion.chunk"
+
+
+def test_openai_multimodal_messages(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny", "max_tokens": 2, "temperature": 0.0,
+        "messages": [{"role": "user", "content": [
+            {"type": "text", "text": "describe"},
+            {"type": "image_url", "image_url": {"url": "data:image/png;base64,AAAA"}},
+        ]}]})
+    assert r.status_code == 200, r.text
+
+
+def test_openai_models(client):
+    r = client.get("/v1/models")
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "list"
+    assert any(m["id"] == "tiny" for m in body["data"])
+
+
+def test_native_inference_api(client):
+    r = client.post("/inference", json={"model": "tiny", "prompt": "x",
+                                        "priority": "high",
+                                        "options": {"temperature": 0.0,
+                                                    "num_predict": 3}})
+    assert r.status_code == 200, r.text
+    assert "response" in r.json()
+    r = client.get("/inference/models")
+    assert any(m["name"] == "tiny" for m in r.json()["models"])
+    r = client.get("/inference/queue")
+    assert "queue" in r.json() and "workers" in r.json()
+    r = client.get("/inference/nonexistent/status")
+    assert r.status_code == 404
+
+
+def test_health_endpoints(client):
+    for path in ("/health", "/health/live", "/health/ready",
+                 "/health/system", "/health/workers", "/health/jobs"):
+        r = client.get(path)
+        assert r.status_code == 200, path
+    r = client.get("/health/workers")
+    assert r.json()["counts"]["total"] == 1
+    r = client.get("/")
+    assert r.json()["name"] == "GridLLM-AMD"
