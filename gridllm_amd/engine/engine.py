@@ -57,6 +57,7 @@ class _Row:
     result: GenerationResult
     last_token: int = 0
     text_tail: str = ""            # pending window for stop-sequence matching
+    force_cancel: bool = False     # TP: rank-0-synchronized cancel decision
 
 
 class InferenceEngine:
@@ -269,6 +270,7 @@ class InferenceEngine:
             r.prompt_eval_duration = prefill_ns // n
 
         # --- first token from prefill logits
+        self._sync_cancel(rows)
         active: list[_Row] = []
         for i, row in enumerate(rows):
             tok = self._sample_batch(first_logits[i].unsqueeze(0), [row])[0]
@@ -283,6 +285,7 @@ class InferenceEngine:
             slot_ids = [r.slot for r in active]
             logits = self._decode_step(token_ids, slot_ids)
             next_ids = self._sample_batch(logits, active)
+            self._sync_cancel(active)
             still: list[_Row] = []
             for r, tok in zip(active, next_ids):
                 if self._emit(r, tok):
@@ -299,6 +302,29 @@ class InferenceEngine:
             r.total_duration = t_end - t_start
         return results
 
+    def _sync_cancel(self, rows: Sequence[_Row]) -> None:
+        """Under TP every rank must retire the same rows at the same step, or
+        the next collective deadlocks (SURVEY.md §7 hard-part 5). Rank 0 owns
+        the cancel events; the decision is broadcast and applied via
+        force_cancel on all ranks (including rank 0 — its own event could
+        flip between broadcast and emit otherwise)."""
+        if not self.tp.enabled:
+            return
+        flags = torch.tensor(
+            [1 if (r.cancel is not None and r.cancel.is_set()) else 0
+             for r in rows],
+            dtype=torch.int32,
+            device=self.device if self.is_gpu else "cpu")
+        self.tp.broadcast(flags, src=0)
+        for r, f in zip(rows, flags.tolist()):
+            if f:
+                r.force_cancel = True
+
+    def _cancelled(self, row: _Row) -> bool:
+        if self.tp.enabled:
+            return row.force_cancel
+        return row.cancel is not None and row.cancel.is_set()
+
     def _emit(self, row: _Row, tok: int) -> bool:
         """Record a sampled token; returns False when the row is finished."""
         res = row.result
@@ -306,7 +332,7 @@ class InferenceEngine:
         res.eval_count += 1
         row.last_token = tok
         row.state.observe(tok, self.spec.vocab)
-        if row.cancel is not None and row.cancel.is_set():
+        if self._cancelled(row):
             res.done_reason = "cancelled"
             res.cancelled = True
             return False
