@@ -59,9 +59,11 @@ def main() -> None:
 
     model = args.model if on_gpu else "tiny"
     batch = args.batch if on_gpu else 4
+    # +1 slot so the TTFT measurement can allocate while the decode batch
+    # holds `batch` slots
     cfg = EngineConfig(model=model, device=device,
                        dtype="bf16" if on_gpu else "fp32",
-                       max_seq_len=args.max_seq, max_batch_size=batch,
+                       max_seq_len=args.max_seq, max_batch_size=batch + 1,
                        use_hipgraph=on_gpu)
     eng = InferenceEngine(cfg)
 

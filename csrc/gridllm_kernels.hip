@@ -209,6 +209,56 @@ k_swiglu(bf16_t* __restrict__ out, const bf16_t* __restrict__ gateup,
 }
 
 // ===========================================================================
+// Fused RoPE + KV-cache write (the per-layer attention prologue).
+//   qkv: [N, (Hq+2*Hk)*D] bf16 — the raw fused-QKV GEMM output (row-dense);
+//   q_out: [N, Hq, D] roped q; k is roped and written straight into kcache,
+//   v copied into vcache (slot[n] = cache_slot*Smax + pos).
+// Replaces {2x contiguous-copy, rope, kv_append} = 4 kernels + one HBM
+// round trip per layer. One workgroup (64 thr) per (token, head).
+// ===========================================================================
+
+__global__ void __launch_bounds__(64)
+k_rope_kv(const bf16_t* __restrict__ qkv, bf16_t* __restrict__ q_out,
+          bf16_t* __restrict__ kc, bf16_t* __restrict__ vc,
+          const float* __restrict__ cost, const float* __restrict__ sint,
+          const int* __restrict__ pos, const int* __restrict__ slot,
+          int Hq, int Hk, int D, int Smax) {
+    const int nh = Hq + Hk;
+    const int n = blockIdx.x / nh;
+    const int h = blockIdx.x % nh;
+    const int W = (Hq + 2 * Hk) * D;
+    const int half = D >> 1;
+    const int p = pos[n];
+    const float* crow = cost + (long)p * half;
+    const float* srow = sint + (long)p * half;
+
+    const bf16_t* src;
+    bf16_t* dst;
+    if (h < Hq) {
+        src = qkv + (long)n * W + h * D;
+        dst = q_out + ((long)n * Hq + h) * D;
+    } else {
+        const int hk = h - Hq;
+        src = qkv + (long)n * W + (Hq + hk) * D;
+        const long sl = slot[n];
+        const long b = sl / Smax, pp = sl % Smax;
+        dst = kc + (((b * Hk + hk) * Smax) + pp) * D;
+        // copy v row (no rope)
+        const bf16_t* vsrc = qkv + (long)n * W + (Hq + Hk + hk) * D;
+        bf16_t* vdst = vc + (((b * Hk + hk) * Smax) + pp) * D;
+        for (int i = threadIdx.x; i < (D >> 1); i += 64)
+            ((u32*)vdst)[i] = ((const u32*)vsrc)[i];
+    }
+    for (int i = threadIdx.x; i < half; i += 64) {
+        float a = bf2f(src[i]);
+        float b = bf2f(src[i + half]);
+        float c = crow[i], s = srow[i];
+        dst[i] = f2bf(a * c - b * s);
+        dst[i + half] = f2bf(a * s + b * c);
+    }
+}
+
+// ===========================================================================
 // KV-cache append.
 //   kc/vc: [B, Hk, Smax, D] bf16;  kn/vn: [N, Hk, D];  slot[N] = b*Smax + pos
 // One workgroup (64 thr) per (token, kv-head); copies both K and V rows.
@@ -248,13 +298,13 @@ k_kv_append(bf16_t* __restrict__ kc, bf16_t* __restrict__ vc,
 
 #define MAXG 8
 
-template <int D>
+template <int D, int G>
 __global__ void __launch_bounds__(256)
 k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
               const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
               const int* __restrict__ seqlen, const int* __restrict__ slot_map,
               float* __restrict__ ws,
-              int B, int Hk, int G, int Smax, int nsplit, float scale) {
+              int B, int Hk, int Smax, int nsplit, float scale) {
     const int split = blockIdx.x % nsplit;
     const int hk = (blockIdx.x / nsplit) % Hk;
     const int b = blockIdx.x / (nsplit * Hk);
@@ -262,9 +312,9 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
 
-    __shared__ float q_lds[MAXG * D];
-    __shared__ float red[4][MAXG * (D > 64 ? D : 64)];
-    __shared__ float wml[4][MAXG][2];         // per-wave m, l
+    __shared__ float q_lds[G * D];
+    __shared__ float red[4][G * (D > 64 ? D : 64)];
+    __shared__ float wml[4][G][2];            // per-wave m, l
 
     const int seq = seqlen[b];
     // split range, 64-aligned chunks
@@ -284,12 +334,12 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     __syncthreads();
 
     // per-wave online softmax state
-    float m[MAXG], l[MAXG];
+    float m[G], l[G];
     // each lane covers d = lane*2 + c*128 (2 f32 per c); NC = D/128 rounded up
     constexpr int NC = (D + 127) / 128;
-    float acc[MAXG][NC][2];
+    float acc[G][NC][2];
 #pragma unroll
-    for (int g = 0; g < MAXG; ++g) {
+    for (int g = 0; g < G; ++g) {
         m[g] = -INFINITY;
         l[g] = 0.f;
 #pragma unroll
@@ -302,9 +352,9 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     for (int kb = s0 + wid * 64; kb < s1; kb += 256) {
         const int j = kb + lane;
         const bool valid = j < s1;
-        float s[MAXG];
+        float s[G];
 #pragma unroll
-        for (int g = 0; g < MAXG; ++g) s[g] = 0.f;
+        for (int g = 0; g < G; ++g) s[g] = 0.f;
         if (valid) {
             const bf16_t* krow = kbase + (long)j * D;
 #pragma unroll
@@ -312,6 +362,7 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
                 uint4 kv = *(const uint4*)(krow + dd);
                 float kf[8];
                 unpack8(kv, kf);
+#pragma unroll
                 for (int g = 0; g < G; ++g) {
                     float a = 0.f;
 #pragma unroll
@@ -321,6 +372,7 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
             }
         }
         // online softmax update per g; store p into LDS (reuse red[] rows)
+#pragma unroll
         for (int g = 0; g < G; ++g) {
             float sv = valid ? s[g] : -INFINITY;
             float bmax = wave_reduce_max(sv);
@@ -350,6 +402,7 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
                     u32 vv = *(const u32*)(vrow + d);
                     float v0 = __uint_as_float((vv & 0xffffu) << 16);
                     float v1 = __uint_as_float(vv & 0xffff0000u);
+#pragma unroll
                     for (int g = 0; g < G; ++g) {
                         float p = red[wid][g * 64 + j2];
                         acc[g][c][0] = fmaf(p, v0, acc[g][c][0]);
@@ -361,11 +414,13 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     }
 
     // cross-wave combine in LDS
+#pragma unroll
     for (int g = 0; g < G; ++g) {
         wml[wid][g][0] = m[g];
         wml[wid][g][1] = l[g];
     }
     __syncthreads();
+#pragma unroll
     for (int g = 0; g < G; ++g) {
         float M = fmaxf(fmaxf(wml[0][g][0], wml[1][g][0]),
                         fmaxf(wml[2][g][0], wml[3][g][0]));
@@ -432,6 +487,26 @@ k_attn_combine(bf16_t* __restrict__ out, const float* __restrict__ ws,
 #define LAUNCH_CHECK() do { hipError_t e = hipGetLastError(); \
     if (e != hipSuccess) return (int)e; } while (0)
 
+template <int D>
+static int launch_attn_decode(hipStream_t st, bf16_t* out, const bf16_t* q,
+                              const bf16_t* kc, const bf16_t* vc,
+                              const int* seqlen, const int* slot_map,
+                              float* ws, int B, int Hk, int G, int Smax,
+                              int nsplit, float scale) {
+    dim3 grid((unsigned)(B * Hk * nsplit)), blk(256);
+    switch (G) {   // compile-time G keeps softmax state in registers
+#define CASE_G(GG) \
+        case GG: k_attn_decode<D, GG><<<grid, blk, 0, st>>>( \
+            out, q, kc, vc, seqlen, slot_map, ws, B, Hk, Smax, nsplit, \
+            scale); break;
+        CASE_G(1) CASE_G(2) CASE_G(3) CASE_G(4)
+        CASE_G(5) CASE_G(6) CASE_G(7) CASE_G(8)
+#undef CASE_G
+        default: return -102;
+    }
+    return 0;
+}
+
 extern "C" {
 
 int gl_rmsnorm(void* stream, void* y, const void* x, const void* w,
@@ -479,27 +554,41 @@ int gl_kv_append(void* stream, void* kc, void* vc, const void* kn,
     return 0;
 }
 
+int gl_rope_kv(void* stream, const void* qkv, void* q_out, void* kc, void* vc,
+               const void* cost, const void* sint, const void* pos,
+               const void* slot, int N, int Hq, int Hk, int D, int Smax) {
+    if (D % 4 != 0 || N == 0) return (D % 4) ? -100 : 0;
+    k_rope_kv<<<dim3((unsigned)(N * (Hq + Hk))), dim3(64), 0,
+                (hipStream_t)stream>>>(
+        (const bf16_t*)qkv, (bf16_t*)q_out, (bf16_t*)kc, (bf16_t*)vc,
+        (const float*)cost, (const float*)sint, (const int*)pos,
+        (const int*)slot, Hq, Hk, D, Smax);
+    LAUNCH_CHECK();
+    return 0;
+}
+
 int gl_attn_decode(void* stream, void* out, const void* q, const void* kc,
                    const void* vc, const void* seqlen, const void* slot_map,
                    void* ws, int B, int Hq,
                    int Hk, int D, int Smax, int nsplit, float scale) {
     const int G = Hq / Hk;
     if (Hq % Hk != 0 || G > MAXG || nsplit < 1) return -100;
-    dim3 grid((unsigned)(B * Hk * nsplit)), blk(256);
     hipStream_t st = (hipStream_t)stream;
+    int rc;
     if (D == 128) {
-        k_attn_decode<128><<<grid, blk, 0, st>>>(
-            (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)kc,
-            (const bf16_t*)vc, (const int*)seqlen, (const int*)slot_map,
-            (float*)ws, B, Hk, G, Smax, nsplit, scale);
+        rc = launch_attn_decode<128>(st, (bf16_t*)out, (const bf16_t*)q,
+                                     (const bf16_t*)kc, (const bf16_t*)vc,
+                                     (const int*)seqlen, (const int*)slot_map,
+                                     (float*)ws, B, Hk, G, Smax, nsplit, scale);
     } else if (D == 64) {
-        k_attn_decode<64><<<grid, blk, 0, st>>>(
-            (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)kc,
-            (const bf16_t*)vc, (const int*)seqlen, (const int*)slot_map,
-            (float*)ws, B, Hk, G, Smax, nsplit, scale);
+        rc = launch_attn_decode<64>(st, (bf16_t*)out, (const bf16_t*)q,
+                                    (const bf16_t*)kc, (const bf16_t*)vc,
+                                    (const int*)seqlen, (const int*)slot_map,
+                                    (float*)ws, B, Hk, G, Smax, nsplit, scale);
     } else {
         return -101;
     }
+    if (rc != 0) return rc;
     LAUNCH_CHECK();
     if (nsplit > 1) {
         k_attn_combine<<<dim3((unsigned)(B * Hq)), dim3(D), 0, st>>>(

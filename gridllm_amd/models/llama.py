@@ -166,13 +166,6 @@ class LlamaModel:
 
     # -- forward -----------------------------------------------------------
 
-    def _split_qkv(self, qkv: torch.Tensor, n: int):
-        d = self.spec.head_dim
-        q = qkv[:, :self.hq_l * d].view(n, self.hq_l, d)
-        k = qkv[:, self.hq_l * d:(self.hq_l + self.hkv_l) * d].view(n, self.hkv_l, d)
-        v = qkv[:, (self.hq_l + self.hkv_l) * d:].view(n, self.hkv_l, d)
-        return q, k, v
-
     @torch.no_grad()
     def forward_prefill(self, tokens: torch.Tensor, slot: int,
                         start_pos: int = 0,
@@ -191,9 +184,9 @@ class LlamaModel:
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
             qkv = F.linear(normed, lw.wqkv)
-            q, k, v = self._split_qkv(qkv, t)
-            q, k = ops.rope(q.contiguous(), k.contiguous(), self.cos, self.sin, pos)
-            ops.kv_append(self.cache.k[li], self.cache.v[li], k, v, slot_t)
+            q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
+                            self.cos, self.sin, pos, slot_t,
+                            self.hq_l, self.hkv_l, self.spec.head_dim)
             kk = self.cache.k[li][slot, :, :start_pos + t].permute(1, 0, 2).unsqueeze(0)
             vv = self.cache.v[li][slot, :, :start_pos + t].permute(1, 0, 2).unsqueeze(0)
             attn = ops.attn_prefill(q.unsqueeze(0), kk, vv, self.scale,
@@ -232,9 +225,9 @@ class LlamaModel:
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
             qkv = F.linear(normed, lw.wqkv)
-            q, k, v = self._split_qkv(qkv, b)
-            q, k = ops.rope(q.contiguous(), k.contiguous(), self.cos, self.sin, pos)
-            ops.kv_append(self.cache.k[li], self.cache.v[li], k, v, slot_t)
+            q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
+                            self.cos, self.sin, pos, slot_t,
+                            self.hq_l, self.hkv_l, self.spec.head_dim)
             attn = ops.attn_decode(q, self.cache.k[li], self.cache.v[li],
                                    attn_seq, self.scale, slot_map=slot_map)
             attn_out = F.linear(attn.reshape(b, -1), lw.wo)

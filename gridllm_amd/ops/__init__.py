@@ -47,6 +47,17 @@ def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
     return reference.rope(q, k, cos, sin, pos)
 
 
+def rope_kv(qkv: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+            cos: torch.Tensor, sin: torch.Tensor, pos: torch.Tensor,
+            slot: torch.Tensor, hq: int, hk: int, d: int) -> torch.Tensor:
+    if _on_gpu(qkv):
+        from gridllm_amd.ops import hip_ops
+        return hip_ops.rope_kv(qkv, kcache, vcache, cos, sin, pos, slot,
+                               hq, hk, d)
+    return reference.rope_kv(qkv, kcache, vcache, cos, sin, pos, slot,
+                             hq, hk, d)
+
+
 def swiglu(gateup: torch.Tensor) -> torch.Tensor:
     if _on_gpu(gateup):
         from gridllm_amd.ops import hip_ops

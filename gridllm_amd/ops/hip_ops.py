@@ -40,6 +40,9 @@ def _load() -> ctypes.CDLL:
     lib.gl_rope.restype = c_i
     lib.gl_swiglu.argtypes = [c_vp, c_vp, c_vp, c_l, c_i]
     lib.gl_swiglu.restype = c_i
+    lib.gl_rope_kv.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
+                               c_vp, c_i, c_i, c_i, c_i, c_i]
+    lib.gl_rope_kv.restype = c_i
     lib.gl_kv_append.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_i, c_i, c_i, c_i]
     lib.gl_kv_append.restype = c_i
     lib.gl_attn_decode.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
@@ -93,6 +96,21 @@ def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
     hk = k.shape[1]
     _check(lib.gl_rope(_stream(), _ptr(q), _ptr(k), _ptr(cos), _ptr(sin),
                        _ptr(pos), t, hq, hk, d), "rope")
+
+
+def rope_kv(qkv: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+            cos: torch.Tensor, sin: torch.Tensor, pos: torch.Tensor,
+            slot: torch.Tensor, hq: int, hk: int, d: int) -> torch.Tensor:
+    """Fused RoPE + KV write from the raw QKV GEMM output.
+    qkv: [N, (hq+2*hk)*d] bf16 (row-dense); returns roped q [N, hq, d]."""
+    lib = _load()
+    n = qkv.shape[0]
+    smax = kcache.shape[2]
+    q_out = torch.empty(n, hq, d, dtype=qkv.dtype, device=qkv.device)
+    _check(lib.gl_rope_kv(_stream(), _ptr(qkv), _ptr(q_out), _ptr(kcache),
+                          _ptr(vcache), _ptr(cos), _ptr(sin), _ptr(pos),
+                          _ptr(slot), n, hq, hk, d, smax), "rope_kv")
+    return q_out
 
 
 def swiglu(gateup: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:

@@ -57,6 +57,20 @@ def swiglu(gateup: torch.Tensor) -> torch.Tensor:
     return (g * torch.sigmoid(g) * u).to(gateup.dtype)
 
 
+def rope_kv(qkv: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
+            cos: torch.Tensor, sin: torch.Tensor, pos: torch.Tensor,
+            slot: torch.Tensor, hq: int, hk: int, d: int) -> torch.Tensor:
+    """Reference of the fused RoPE+KV-write: slice the QKV GEMM output,
+    rope q/k, append k/v to the caches. Returns roped q [N, hq, d]."""
+    n = qkv.shape[0]
+    q = qkv[:, :hq * d].reshape(n, hq, d)
+    k = qkv[:, hq * d:(hq + hk) * d].reshape(n, hk, d)
+    v = qkv[:, (hq + hk) * d:].reshape(n, hk, d).contiguous()
+    q, k = rope(q.contiguous(), k.contiguous(), cos, sin, pos)
+    kv_append(kcache, vcache, k, v, slot)
+    return q
+
+
 def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k_new: torch.Tensor,
               v_new: torch.Tensor, slot: torch.Tensor) -> None:
     """caches [B,Hk,Smax,D]; k_new [N,Hk,D]; slot[N]=b*Smax+pos."""

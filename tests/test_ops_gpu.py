@@ -144,3 +144,26 @@ def test_attn_decode_slot_map_gpu(dev):
     ref = R.attn_decode(q.cpu().float(), kc.cpu().float(), vc.cpu().float(),
                         seqlen.cpu(), 0.0883, slot_map=slot_map.cpu())
     assert _mx(out.cpu(), ref) < 0.03
+
+
+@pytest.mark.parametrize("n,hq,hk,d", [(5, 24, 8, 128), (16, 32, 8, 64)])
+def test_rope_kv_fused_gpu(dev, n, hq, hk, d):
+    """Fused rope+kv-write vs the composed reference (strided qkv input)."""
+    torch.manual_seed(8)
+    from gridllm_amd.ops import hip_ops
+    smax, slots = 64, 3
+    qkv = torch.randn(n, (hq + 2 * hk) * d, device=dev).to(torch.bfloat16)
+    kc = torch.zeros(slots, hk, smax, d, device=dev, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    cos, sin = R.rope_tables(d, 128, device=dev)
+    pos = torch.randint(0, 120, (n,), dtype=torch.int32, device=dev)
+    slot = (torch.randint(0, slots, (n,), dtype=torch.int32, device=dev) * smax
+            + torch.arange(n, dtype=torch.int32, device=dev))  # unique cells
+    q = hip_ops.rope_kv(qkv, kc, vc, cos, sin, pos, slot, hq, hk, d)
+    kc_ref = torch.zeros(slots, hk, smax, d)
+    vc_ref = torch.zeros_like(kc_ref)
+    q_ref = R.rope_kv(qkv.cpu().float(), kc_ref, vc_ref, cos.cpu(), sin.cpu(),
+                      pos.cpu(), slot.cpu(), hq, hk, d)
+    assert _mx(q.cpu(), q_ref) < 0.05
+    assert _mx(kc.cpu(), kc_ref) < 0.05
+    assert _mx(vc.cpu(), vc_ref) < 1e-6
