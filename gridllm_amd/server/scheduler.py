@@ -156,30 +156,36 @@ class JobScheduler:
         if not self.queue:
             return
         self.queue.sort(key=lambda r: -_PRI_RANK.get(r.priority, 2))
-        remaining: list[InferenceRequest] = []
-        changed = False
+        assigned: set[str] = set()
         for req in list(self.queue):
             worker = self._select_worker(req)
             if worker is None:
-                remaining.append(req)
                 continue
-            ok = await self._assign(req, worker)
-            if ok:
-                changed = True
-            else:
-                remaining.append(req)
-        if changed:
-            self.queue = remaining
+            if await self._assign(req, worker):
+                assigned.add(req.id)
+        if assigned:
+            # filter by id — add_job may have appended to self.queue while
+            # _assign awaited (a snapshot-replace here would drop those jobs)
+            self.queue = [r for r in self.queue if r.id not in assigned]
             await self._persist_queue()
 
     def _select_worker(self, req: InferenceRequest):
-        """Least-loaded, tier tiebreak (reference: :317-360)."""
-        cands = self.registry.get_available_workers_by_model(req.model)
-        # exclude workers that already hold an active job assigned but
-        # not yet reflected (belt and braces; current_jobs covers it)
+        """Least-loaded, tier tiebreak (reference: :317-360). Load = max of
+        the registry's view and the scheduler's own active-job count — a
+        worker heartbeat carrying a stale (lower) currentJobs must not cause
+        over-assignment."""
+        sched_load: dict[str, int] = {}
+        for a in self.active.values():
+            sched_load[a.worker_id] = sched_load.get(a.worker_id, 0) + 1
+
+        def load(w) -> int:
+            return max(w.current_jobs, sched_load.get(w.id, 0))
+
+        cands = [w for w in self.registry.get_available_workers_by_model(req.model)
+                 if load(w) < w.max_concurrent_jobs]
         if not cands:
             return None
-        cands.sort(key=lambda w: (w.current_jobs,
+        cands.sort(key=lambda w: (load(w),
                                   -_TIER_RANK.get(w.performance_tier, 2)))
         return cands[0]
 

@@ -1,0 +1,209 @@
+"""Full multi-process end-to-end: real server process (TCP bus + uvicorn
+HTTP) + 2 real worker processes (tiny CPU engines), driven over HTTP.
+
+This is the reference's CI scenario (redis+server+client
+full-stack) but self-contained, plus the chaos test the reference never had:
+SIGKILL a busy worker and assert the orphaned request completes on the
+survivor (SURVEY.md §3.5 semantics).
+"""
+
+import json
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+
+import httpx
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+WORKER_SRC = r"""
+import asyncio, os, sys
+from gridllm_amd.bus.tcp import TcpBus
+from gridllm_amd.config import EngineConfig, WorkerConfig
+from gridllm_amd.engine import InferenceEngine
+from gridllm_amd.worker.agent import WorkerAgent
+
+async def main():
+    wcfg = WorkerConfig(worker_id=os.environ["WORKER_ID"],
+                        bus_port=int(os.environ["BUS_PORT"]),
+                        max_concurrent_tasks=2, heartbeat_interval_ms=500)
+    ecfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                        max_seq_len=128, max_batch_size=4, use_hipgraph=False)
+    slow = float(os.environ.get("WORKER_SLOW", "0"))
+    engine = InferenceEngine(ecfg)
+    if slow > 0:  # make decode slow so chaos test can kill mid-job
+        orig = engine._decode_step
+        import time as _t
+        def slow_step(tokens, slots):
+            _t.sleep(slow)
+            return orig(tokens, slots)
+        engine._decode_step = slow_step
+    bus = TcpBus("127.0.0.1", wcfg.bus_port)
+    await bus.connect()
+    agent = WorkerAgent(bus, engine, wcfg)
+    await agent.start()
+    print("WORKER_READY", flush=True)
+    await asyncio.Event().wait()
+
+asyncio.run(main())
+"""
+
+SERVER_SRC = r"""
+import asyncio, os
+import uvicorn
+from gridllm_amd.config import ServerConfig
+from gridllm_amd.server.app import GridLLMServer
+
+cfg = ServerConfig(port=int(os.environ["HTTP_PORT"]),
+                   bus_port=int(os.environ["BUS_PORT"]),
+                   worker_heartbeat_timeout_ms=2000,
+                   worker_cleanup_interval_ms=300,
+                   job_retry_attempts=1, job_retry_delay_ms=200)
+server = GridLLMServer(cfg)
+app = server.app
+
+@app.on_event("startup")
+async def _s():
+    await server.start()
+    print("SERVER_READY", flush=True)
+
+uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error")
+"""
+
+
+def _spawn(src: str, env: dict, ready_tag: str, timeout=60) -> subprocess.Popen:
+    proc = subprocess.Popen(
+        [sys.executable, "-u", "-c", src],
+        env={**os.environ, **env, "PYTHONPATH": REPO},
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        cwd=REPO)
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        line = proc.stdout.readline()
+        if ready_tag in line:
+            # keep draining stdout forever or the child blocks on a full pipe
+            import threading
+
+            def _drain(p=proc):
+                try:
+                    for _ in p.stdout:
+                        pass
+                except Exception:  # noqa: BLE001
+                    pass
+
+            threading.Thread(target=_drain, daemon=True).start()
+            return proc
+        if proc.poll() is not None:
+            raise RuntimeError(f"process died:\n{proc.stdout.read()}")
+    proc.kill()
+    raise TimeoutError(f"no {ready_tag} within {timeout}s")
+
+
+@pytest.fixture(scope="module")
+def cluster():
+    bus_port, http_port = _free_port(), _free_port()
+    env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port)}
+    server = _spawn(SERVER_SRC, env, "SERVER_READY")
+    workers = {}
+    for wid, slow in (("w-a", "0.15"), ("w-b", "0")):
+        workers[wid] = _spawn(WORKER_SRC,
+                              {**env, "WORKER_ID": wid, "WORKER_SLOW": slow},
+                              "WORKER_READY")
+    base = f"http://127.0.0.1:{http_port}"
+    # wait for registration
+    with httpx.Client(base_url=base, timeout=30) as c:
+        for _ in range(100):
+            r = c.get("/health/workers")
+            if r.status_code == 200 and r.json()["counts"]["total"] == 2:
+                break
+            time.sleep(0.1)
+        else:
+            raise TimeoutError("workers never registered")
+    yield {"base": base, "server": server, "workers": workers}
+    for p in list(workers.values()) + [server]:
+        if p.poll() is None:
+            p.send_signal(signal.SIGKILL)
+    for p in list(workers.values()) + [server]:
+        p.wait(timeout=10)
+
+
+def test_http_generate_across_processes(cluster):
+    with httpx.Client(base_url=cluster["base"], timeout=60) as c:
+        r = c.post("/api/generate", json={
+            "model": "tiny", "prompt": "hello world", "stream": False,
+            "options": {"temperature": 0.0, "num_predict": 4}})
+        assert r.status_code == 200, r.text
+        assert r.json()["done"] is True
+        assert r.json()["eval_count"] >= 1
+
+
+def test_http_streaming_across_processes(cluster):
+    with httpx.Client(base_url=cluster["base"], timeout=60) as c:
+        with c.stream("POST", "/api/generate", json={
+                "model": "tiny", "prompt": "stream", "stream": True,
+                "options": {"temperature": 0.0, "num_predict": 4}}) as r:
+            assert r.status_code == 200
+            lines = [json.loads(l) for l in r.iter_lines() if l]
+        assert lines[-1]["done"] is True
+
+
+def test_parallel_requests_use_both_workers(cluster):
+    import concurrent.futures as cf
+
+    def one(i):
+        with httpx.Client(base_url=cluster["base"], timeout=120) as c:
+            r = c.post("/api/generate", json={
+                "model": "tiny", "prompt": f"req {i}", "stream": False,
+                "options": {"temperature": 0.0, "num_predict": 3}})
+            return r.status_code
+
+    with cf.ThreadPoolExecutor(8) as ex:
+        codes = list(ex.map(one, range(8)))
+    assert codes == [200] * 8
+
+
+def test_chaos_sigkill_busy_worker_orphan_requeue(cluster):
+    """Kill the slow worker mid-job: the job must be orphaned, requeued at
+    the front, and completed by the surviving worker — the ORIGINAL awaited
+    HTTP request succeeds (reference guarantee: JobScheduler.ts:553-630)."""
+    import concurrent.futures as cf
+    slow = cluster["workers"]["w-a"]
+
+    def long_req():
+        with httpx.Client(base_url=cluster["base"], timeout=180) as c:
+            return c.post("/api/generate", json={
+                "model": "tiny", "prompt": "chaos", "stream": False,
+                "priority": "high",
+                "options": {"temperature": 0.0, "num_predict": 30}})
+
+    with cf.ThreadPoolExecutor(4) as ex:
+        # saturate both workers so the slow one is certainly busy
+        futs = [ex.submit(long_req) for _ in range(4)]
+        time.sleep(1.5)                    # jobs are mid-decode now
+        slow.send_signal(signal.SIGKILL)   # hard kill
+        slow.wait(timeout=10)
+        results = [f.result() for f in futs]
+    assert all(r.status_code == 200 for r in results), \
+        [r.text[:200] for r in results]
+    # the dead worker must be gone from the registry
+    with httpx.Client(base_url=cluster["base"], timeout=30) as c:
+        for _ in range(100):
+            names = [w["id"] for w in
+                     c.get("/health/workers").json()["workers"]]
+            if "w-a" not in names:
+                break
+            time.sleep(0.2)
+        assert "w-a" not in names
