@@ -480,6 +480,108 @@ k_attn_combine(bf16_t* __restrict__ out, const float* __restrict__ ws,
 }
 
 // ===========================================================================
+// Skinny-M weight-streaming GEMM:  out[M,N] = x[M,K] @ W[N,K]^T   (bf16 I/O)
+//
+// The decode-path projection GEMMs (M = batch <= 256, W = torch linear
+// weight, row-major [N,K]) are HBM-bound on W. hipBLASLt reaches only ~40%
+// of the roofline on these shapes (profiles/r01_decode_profile.md), so this
+// kernel streams W once with non-temporal loads straight into MFMA B
+// fragments — both operands are k-contiguous, so fragments load directly
+// from HBM/L2 with no LDS staging (guide: "GEMV / M<=16 decode weights:
+// load straight to VGPRs"; generalized to M<=256 via 16-row MFMA tiles).
+//
+// mfma_f32_16x16x32_bf16 lane mapping used here:
+//   A[i][k]: lane l holds rows i = l&15, k = (l>>4)*8 + 0..7  (8 bf16 = 16 B)
+//   B[k][j]: lane l holds col  j = l&15, same k chunk
+//   D[i][j]: col j = l&15, row i = (l>>4)*4 + reg   (guide §3)
+// Block = 4 waves; each wave owns 16 output columns (BN = 64 per block);
+// grid = (N/64) * splitk, k-slices 32-aligned; splitk > 1 accumulates f32
+// partials in ws, combined by k_gemm_combine.
+// Constraints: N % 64 == 0, K % 32 == 0, M <= 16*MT.
+// ===========================================================================
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+template <int MT>
+__global__ void __launch_bounds__(256)
+k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
+              const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+              int M, int N, int K, int splitk, int kslice) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int ks = blockIdx.x % splitk;
+    const int ntile = blockIdx.x / splitk;
+    const int n0 = ntile * 64 + wid * 16;       // this wave's column base
+    const int k0 = ks * kslice;
+    const int k1 = min(K, k0 + kslice);
+    const int fr = lane & 15;                   // fragment row (A) / col (B)
+    const int kg = (lane >> 4) * 8;             // k sub-chunk
+
+    f32x4 acc[MT];
+#pragma unroll
+    for (int t = 0; t < MT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    const bf16_t* wrow = w + (long)(n0 + fr) * K + kg;
+    const bf16_t* xrow = x + (long)fr * K + kg;
+
+    for (int k = k0; k < k1; k += 32) {
+        bf16x8 b = __builtin_nontemporal_load((const bf16x8*)(wrow + k));
+#pragma unroll
+        for (int t = 0; t < MT; ++t) {
+            const int m = t * 16 + fr;
+            bf16x8 a = {};
+            if (m < M) a = *(const bf16x8*)(xrow + (long)t * 16 * K + k);
+            acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
+                                                             0, 0, 0);
+        }
+    }
+
+    const int orow = (lane >> 4) * 4;           // D rows this lane holds
+    const int ocol = n0 + fr;
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int m = t * 16 + orow + r;
+            if (m < M) {
+                if (splitk == 1) {
+                    out[(long)m * N + ocol] = f2bf(acc[t][r]);
+                } else {
+                    ws[((long)ks * M + m) * N + ocol] = acc[t][r];
+                }
+            }
+        }
+    }
+}
+
+// Combine split-K f32 partials -> bf16 out. Grid-stride over M*N/8 uint4s.
+__global__ void __launch_bounds__(256)
+k_gemm_combine(bf16_t* __restrict__ out, const float* __restrict__ ws,
+               long mn, int splitk) {
+    for (long i = blockIdx.x * 256L + threadIdx.x; i * 4 < mn;
+         i += gridDim.x * 256L) {
+        float4 s = ((const float4*)ws)[i];
+        for (int k = 1; k < splitk; ++k) {
+            float4 v = ((const float4*)(ws + (long)k * mn))[i];
+            s.x += v.x; s.y += v.y; s.z += v.z; s.w += v.w;
+        }
+        u32 lo = (u32)f2bf(s.x) | ((u32)f2bf(s.y) << 16);
+        u32 hi = (u32)f2bf(s.z) | ((u32)f2bf(s.w) << 16);
+        ((uint2*)out)[i] = make_uint2(lo, hi);
+    }
+}
+
+template <int MT>
+static void launch_gemm_skinny(hipStream_t st, bf16_t* out, float* ws,
+                               const bf16_t* x, const bf16_t* w, int M, int N,
+                               int K, int splitk, int kslice) {
+    dim3 grid((unsigned)((N / 64) * splitk)), blk(256);
+    k_gemm_skinny<MT><<<grid, blk, 0, st>>>(out, ws, x, w, M, N, K, splitk,
+                                            kslice);
+}
+
+// ===========================================================================
 // Host-callable entry points (extern "C", explicit stream — ctypes-bound).
 // All return hipError_t as int (0 == success); callers must check.
 // ===========================================================================
@@ -593,6 +695,32 @@ int gl_attn_decode(void* stream, void* out, const void* q, const void* kc,
     if (nsplit > 1) {
         k_attn_combine<<<dim3((unsigned)(B * Hq)), dim3(D), 0, st>>>(
             (bf16_t*)out, (const float*)ws, Hq, D, nsplit);
+        LAUNCH_CHECK();
+    }
+    return 0;
+}
+
+int gl_gemm_skinny(void* stream, void* out, void* ws, const void* x,
+                   const void* w, int M, int N, int K, int splitk,
+                   int kslice) {
+    if (N % 64 != 0 || K % 32 != 0 || kslice % 32 != 0) return -100;
+    hipStream_t st = (hipStream_t)stream;
+    const int mt = (M + 15) / 16;
+#define CASE_MT(T) else if (mt <= T) launch_gemm_skinny<T>( \
+        st, (bf16_t*)out, (float*)ws, (const bf16_t*)x, (const bf16_t*)w, \
+        M, N, K, splitk, kslice)
+    if (false) {}
+    CASE_MT(1); CASE_MT(2); CASE_MT(3); CASE_MT(4); CASE_MT(6); CASE_MT(8);
+    CASE_MT(12); CASE_MT(16);
+    else return -103;
+#undef CASE_MT
+    LAUNCH_CHECK();
+    if (splitk > 1) {
+        long mn = (long)M * N;
+        long chunks = (mn + 3) / 4;
+        int grid = (int)min((chunks + 255) / 256, (long)2048);
+        k_gemm_combine<<<dim3(grid), dim3(256), 0, st>>>(
+            (bf16_t*)out, (const float*)ws, mn, splitk);
         LAUNCH_CHECK();
     }
     return 0;

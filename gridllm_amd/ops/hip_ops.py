@@ -48,6 +48,9 @@ def _load() -> ctypes.CDLL:
     lib.gl_attn_decode.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
                                    c_vp, c_i, c_i, c_i, c_i, c_i, c_i, c_f]
     lib.gl_attn_decode.restype = c_i
+    lib.gl_gemm_skinny.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp,
+                                   c_i, c_i, c_i, c_i, c_i]
+    lib.gl_gemm_skinny.restype = c_i
     lib.gl_device_sync.argtypes = []
     lib.gl_device_sync.restype = c_i
     _lib = lib
@@ -161,6 +164,32 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
                               _ptr(vcache), _ptr(seqlen), _ptr(slot_map),
                               _ptr(ws), b, hq, hk, d, smax, nsplit,
                               ctypes.c_float(scale)), "attn_decode")
+    return out
+
+
+def gemm_skinny_supported(m: int, n: int, k: int) -> bool:
+    return m <= 256 and n % 64 == 0 and k % 32 == 0
+
+
+def gemm_skinny(x: torch.Tensor, w: torch.Tensor,
+                out: Optional[torch.Tensor] = None,
+                splitk: int = 0) -> torch.Tensor:
+    """out[M,N] = x[M,K] @ w[N,K]^T, bf16, M<=256 (decode projections).
+    Streams w once (nt loads) into MFMA fragments; split-K sized to fill
+    256 CUs with f32 partial slabs + combine."""
+    lib = _load()
+    m, k = x.shape
+    n = w.shape[0]
+    if splitk <= 0:
+        splitk = max(1, min(512 // max(1, n // 64), k // 512, 16))
+    kslice = ((k + splitk - 1) // splitk + 31) // 32 * 32
+    if out is None:
+        out = torch.empty(m, n, dtype=x.dtype, device=x.device)
+    ws = None
+    if splitk > 1:
+        ws = torch.empty(splitk * m * n, dtype=torch.float32, device=x.device)
+    _check(lib.gl_gemm_skinny(_stream(), _ptr(out), _ptr(ws), _ptr(x), _ptr(w),
+                              m, n, k, splitk, kslice), "gemm_skinny")
     return out
 
 

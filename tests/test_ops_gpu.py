@@ -167,3 +167,29 @@ def test_rope_kv_fused_gpu(dev, n, hq, hk, d):
     assert _mx(q.cpu(), q_ref) < 0.05
     assert _mx(kc.cpu(), kc_ref) < 0.05
     assert _mx(vc.cpu(), vc_ref) < 1e-6
+
+
+@pytest.mark.parametrize("m,n,k", [
+    (1, 3072, 3072), (13, 5120, 3072), (64, 16384, 3072),
+    (64, 3072, 8192), (127, 3072, 3072), (256, 128256, 3072),
+])
+def test_gemm_skinny_gpu(dev, m, n, k):
+    """Skinny GEMM vs fp32 matmul of the same bf16 operands."""
+    torch.manual_seed(9)
+    from gridllm_amd.ops import hip_ops
+    x = (torch.randn(m, k, device=dev) * 0.1).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) * 0.1).to(torch.bfloat16)
+    out = hip_ops.gemm_skinny(x, w)
+    ref = x.float() @ w.float().t()
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.02 * ref.abs().max().item(), err
+
+
+def test_gemm_skinny_splitk_consistent(dev):
+    torch.manual_seed(10)
+    from gridllm_amd.ops import hip_ops
+    x = (torch.randn(64, 4096, device=dev) * 0.1).to(torch.bfloat16)
+    w = (torch.randn(1024, 4096, device=dev) * 0.1).to(torch.bfloat16)
+    o1 = hip_ops.gemm_skinny(x, w, splitk=1)
+    o8 = hip_ops.gemm_skinny(x, w, splitk=8)
+    assert (o1.float() - o8.float()).abs().max().item() < 0.05
