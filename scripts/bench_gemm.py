@@ -1,6 +1,7 @@
 """Micro-benchmark: skinny-M GEMM (ours) vs hipBLASLt (torch F.linear) on the
 llama3.2-3B decode projection shapes. Within-probe interleaved A/B."""
-import sys, time
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 import torch.nn.functional as F
 from gridllm_amd.ops import hip_ops
