@@ -525,15 +525,53 @@ k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
     const bf16_t* wrow = w + (long)(n0 + fr) * K + kg;
     const bf16_t* xrow = x + (long)fr * K + kg;
 
-    for (int k = k0; k < k1; k += 32) {
-        bf16x8 b = __builtin_nontemporal_load((const bf16x8*)(wrow + k));
+    // U-deep non-temporal prefetch ring on the weight stream: one wave keeps
+    // U*1KiB of W in flight so a 1-wave/SIMD launch still covers the ~900-cy
+    // HBM latency. Static ring indices (full unroll) — a runtime index would
+    // spill the ring to scratch (guide §5.4 rule 20).
+    constexpr int U = 8;
+    bf16x8 bbuf[U];
 #pragma unroll
-        for (int t = 0; t < MT; ++t) {
-            const int m = t * 16 + fr;
-            bf16x8 a = {};
-            if (m < M) a = *(const bf16x8*)(xrow + (long)t * 16 * K + k);
-            acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
-                                                             0, 0, 0);
+    for (int u = 0; u < U; ++u) {
+        const int k = k0 + u * 32;
+        if (k < k1)
+            bbuf[u] = __builtin_nontemporal_load((const bf16x8*)(wrow + k));
+    }
+    int k = k0;
+    for (; k + U * 32 <= k1; k += U * 32) {
+#pragma unroll
+        for (int u = 0; u < U; ++u) {
+            const bf16x8 b = bbuf[u];
+            const int kn = k + (u + U) * 32;
+            if (kn < k1)
+                bbuf[u] = __builtin_nontemporal_load(
+                    (const bf16x8*)(wrow + kn));
+#pragma unroll
+            for (int t = 0; t < MT; ++t) {
+                const int m = t * 16 + fr;
+                bf16x8 a = {};
+                if (m < M)
+                    a = *(const bf16x8*)(xrow + (long)t * 16 * K + k + u * 32);
+                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
+                                                                 0, 0, 0);
+            }
+        }
+    }
+    // tail: < U chunks remain, their bbuf entries were prefetched above
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+        const int kk = k + u * 32;
+        if (kk < k1) {
+            const bf16x8 b = bbuf[u];
+#pragma unroll
+            for (int t = 0; t < MT; ++t) {
+                const int m = t * 16 + fr;
+                bf16x8 a = {};
+                if (m < M)
+                    a = *(const bf16x8*)(xrow + (long)t * 16 * K + kk);
+                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
+                                                                 0, 0, 0);
+            }
         }
     }
 

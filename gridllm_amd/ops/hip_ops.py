@@ -181,7 +181,7 @@ def gemm_skinny(x: torch.Tensor, w: torch.Tensor,
     m, k = x.shape
     n = w.shape[0]
     if splitk <= 0:
-        splitk = max(1, min(512 // max(1, n // 64), k // 512, 16))
+        splitk = max(1, min(512 // max(1, n // 64), k // 256, 16))
     kslice = ((k + splitk - 1) // splitk + 31) // 32 * 32
     if out is None:
         out = torch.empty(m, n, dtype=x.dtype, device=x.device)
