@@ -504,7 +504,9 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 template <int MT>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(256, 2)   // 2 waves/SIMD: frees VGPRs so
+                                            // the scheduler keeps the load
+                                            // pipeline deep (not vmcnt(0))
 k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
               const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
               int M, int N, int K, int splitk, int kslice) {
@@ -522,38 +524,74 @@ k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
 #pragma unroll
     for (int t = 0; t < MT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
+    // EVERY load in the hot loop is branchless: a lane-divergent guard
+    // around a load makes hipcc branch per element and drain vmcnt(0) each
+    // time (guide §5 trap (c)) — instead, rows m >= M read a CLAMPED valid
+    // address; their garbage feeds only output rows the epilogue never
+    // stores (MFMA row i depends only on A row i). B prefetches past the
+    // slice end re-read the last chunk.
     const bf16_t* wrow = w + (long)(n0 + fr) * K + kg;
-    const bf16_t* xrow = x + (long)fr * K + kg;
+    const bf16_t* xp[MT];
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+        const int m = min(t * 16 + fr, M - 1);
+        xp[t] = x + (long)m * K + kg;
+    }
+    const int klast = k1 - 32;                  // k1-k0 >= 32, both 32-aligned
 
     // U-deep non-temporal prefetch ring on the weight stream: one wave keeps
-    // U*1KiB of W in flight so a 1-wave/SIMD launch still covers the ~900-cy
+    // U KiB of W in flight so a 1-wave/SIMD launch still covers the ~900-cy
     // HBM latency. Static ring indices (full unroll) — a runtime index would
     // spill the ring to scratch (guide §5.4 rule 20).
     constexpr int U = 8;
     bf16x8 bbuf[U];
 #pragma unroll
     for (int u = 0; u < U; ++u) {
-        const int k = k0 + u * 32;
-        if (k < k1)
-            bbuf[u] = __builtin_nontemporal_load((const bf16x8*)(wrow + k));
+        bbuf[u] = __builtin_nontemporal_load(
+            (const bf16x8*)(wrow + min(k0 + u * 32, klast)));
     }
+    // A fragments are double-buffered one u-step ahead (parity-static ring;
+    // runtime ring indices would go to scratch — §5.4 rule 20).
+    bf16x8 av0[MT], av1[MT];
+#pragma unroll
+    for (int t = 0; t < MT; ++t)
+        av0[t] = *(const bf16x8*)(xp[t] + k0);
     int k = k0;
     for (; k + U * 32 <= k1; k += U * 32) {
 #pragma unroll
-        for (int u = 0; u < U; ++u) {
-            const bf16x8 b = bbuf[u];
-            const int kn = k + (u + U) * 32;
-            if (kn < k1)
+        for (int u = 0; u < U; u += 2) {
+            {
+                const bf16x8 b = bbuf[u];
                 bbuf[u] = __builtin_nontemporal_load(
-                    (const bf16x8*)(wrow + kn));
+                    (const bf16x8*)(wrow + min(k + (u + U) * 32, klast)));
 #pragma unroll
-            for (int t = 0; t < MT; ++t) {
-                const int m = t * 16 + fr;
-                bf16x8 a = {};
-                if (m < M)
-                    a = *(const bf16x8*)(xrow + (long)t * 16 * K + k + u * 32);
-                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
-                                                                 0, 0, 0);
+                for (int t = 0; t < MT; ++t)
+                    av1[t] = *(const bf16x8*)(
+                        xp[t] + min(k + (u + 1) * 32, klast));
+#pragma unroll
+                for (int t = 0; t < MT; ++t)
+                    acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        av0[t], b, acc[t], 0, 0, 0);
+                // pin emission order: this step's loads batch AHEAD of the
+                // MFMAs (which consume the previous step's fragments), so
+                // waits become counted instead of vmcnt(0) per use
+                __builtin_amdgcn_sched_group_barrier(0x20, MT + 1, 0);
+                __builtin_amdgcn_sched_group_barrier(0x8, MT, 0);
+            }
+            {
+                const bf16x8 b = bbuf[u + 1];
+                bbuf[u + 1] = __builtin_nontemporal_load(
+                    (const bf16x8*)(wrow + min(k + (u + 1 + U) * 32, klast)));
+#pragma unroll
+                for (int t = 0; t < MT; ++t)
+                    av0[t] = *(const bf16x8*)(
+                        xp[t] + min(k + (u + 2) * 32, klast));
+#pragma unroll
+                for (int t = 0; t < MT; ++t)
+                    acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        av1[t], b, acc[t], 0, 0, 0);
+                __builtin_amdgcn_sched_group_barrier(0x20, MT + 1, 0);
+                __builtin_amdgcn_sched_group_barrier(0x8, MT, 0);
             }
         }
     }
@@ -565,10 +603,7 @@ k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
             const bf16x8 b = bbuf[u];
 #pragma unroll
             for (int t = 0; t < MT; ++t) {
-                const int m = t * 16 + fr;
-                bf16x8 a = {};
-                if (m < M)
-                    a = *(const bf16x8*)(xrow + (long)t * 16 * K + kk);
+                const bf16x8 a = *(const bf16x8*)(xp[t] + kk);
                 acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
                                                                  0, 0, 0);
             }
