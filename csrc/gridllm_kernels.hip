@@ -526,17 +526,11 @@ k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
 
     // EVERY load in the hot loop is branchless: a lane-divergent guard
     // around a load makes hipcc branch per element and drain vmcnt(0) each
-    // time (guide §5 trap (c)) — instead, rows m >= M read a CLAMPED valid
-    // address; their garbage feeds only output rows the epilogue never
-    // stores (MFMA row i depends only on A row i). B prefetches past the
-    // slice end re-read the last chunk.
+    // time (guide §5 trap (c)) — rows m >= M read a CLAMPED valid address;
+    // their garbage feeds only output rows the epilogue never stores (MFMA
+    // row i depends only on A row i). B prefetches past the slice end
+    // re-read the last chunk.
     const bf16_t* wrow = w + (long)(n0 + fr) * K + kg;
-    const bf16_t* xp[MT];
-#pragma unroll
-    for (int t = 0; t < MT; ++t) {
-        const int m = min(t * 16 + fr, M - 1);
-        xp[t] = x + (long)m * K + kg;
-    }
     const int klast = k1 - 32;                  // k1-k0 >= 32, both 32-aligned
 
     // U-deep non-temporal prefetch ring on the weight stream: one wave keeps
@@ -550,63 +544,142 @@ k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
         bbuf[u] = __builtin_nontemporal_load(
             (const bf16x8*)(wrow + min(k0 + u * 32, klast)));
     }
-    // A fragments are double-buffered one u-step ahead (parity-static ring;
-    // runtime ring indices would go to scratch — §5.4 rule 20).
-    bf16x8 av0[MT], av1[MT];
+
+    if constexpr (MT == 1) {
+        // M <= 16: all 16 fragment rows clamp to few x rows — direct loads
+        // are broadcast-cheap, no staging needed.
+        const bf16_t* xp = x + (long)min(fr, M - 1) * K + kg;
+        int k = k0;
+        for (; k + U * 32 <= k1; k += U * 32) {
 #pragma unroll
-    for (int t = 0; t < MT; ++t)
-        av0[t] = *(const bf16x8*)(xp[t] + k0);
-    int k = k0;
-    for (; k + U * 32 <= k1; k += U * 32) {
-#pragma unroll
-        for (int u = 0; u < U; u += 2) {
-            {
+            for (int u = 0; u < U; ++u) {
                 const bf16x8 b = bbuf[u];
                 bbuf[u] = __builtin_nontemporal_load(
                     (const bf16x8*)(wrow + min(k + (u + U) * 32, klast)));
-#pragma unroll
-                for (int t = 0; t < MT; ++t)
-                    av1[t] = *(const bf16x8*)(
-                        xp[t] + min(k + (u + 1) * 32, klast));
-#pragma unroll
-                for (int t = 0; t < MT; ++t)
-                    acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        av0[t], b, acc[t], 0, 0, 0);
-                // pin emission order: this step's loads batch AHEAD of the
-                // MFMAs (which consume the previous step's fragments), so
-                // waits become counted instead of vmcnt(0) per use
-                __builtin_amdgcn_sched_group_barrier(0x20, MT + 1, 0);
-                __builtin_amdgcn_sched_group_barrier(0x8, MT, 0);
-            }
-            {
-                const bf16x8 b = bbuf[u + 1];
-                bbuf[u + 1] = __builtin_nontemporal_load(
-                    (const bf16x8*)(wrow + min(k + (u + 1 + U) * 32, klast)));
-#pragma unroll
-                for (int t = 0; t < MT; ++t)
-                    av0[t] = *(const bf16x8*)(
-                        xp[t] + min(k + (u + 2) * 32, klast));
-#pragma unroll
-                for (int t = 0; t < MT; ++t)
-                    acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        av1[t], b, acc[t], 0, 0, 0);
-                __builtin_amdgcn_sched_group_barrier(0x20, MT + 1, 0);
-                __builtin_amdgcn_sched_group_barrier(0x8, MT, 0);
+                const bf16x8 a = *(const bf16x8*)(xp + k + u * 32);
+                acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[0],
+                                                                 0, 0, 0);
+                __builtin_amdgcn_sched_group_barrier(0x20, 2, 0);
+                __builtin_amdgcn_sched_group_barrier(0x8, 1, 0);
             }
         }
-    }
-    // tail: < U chunks remain, their bbuf entries were prefetched above
 #pragma unroll
-    for (int u = 0; u < U; ++u) {
-        const int kk = k + u * 32;
-        if (kk < k1) {
-            const bf16x8 b = bbuf[u];
-#pragma unroll
-            for (int t = 0; t < MT; ++t) {
-                const bf16x8 a = *(const bf16x8*)(xp[t] + kk);
-                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
-                                                                 0, 0, 0);
+        for (int u = 0; u < U; ++u) {
+            const int kk = k + u * 32;
+            if (kk < k1) {
+                const bf16x8 a = *(const bf16x8*)(xp + kk);
+                acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a, bbuf[u], acc[0], 0, 0, 0);
             }
+        }
+    } else {
+        // M > 16: a direct A-fragment load would touch 64 cache lines per
+        // instruction (16 rows x 6-16 KB apart — the "fragment-shaped x
+        // loads" anti-pattern, guide §5). Stage x k-panels through LDS with
+        // coalesced full-line loads; fragments come back via XOR-swizzled
+        // ds_read_b128 (row&15)<<4 — conflict-free since a 16-lane group
+        // reads rows 0..15 (T2 note). Ping-pong panels, unrolled in PAIRS so
+        // every B-ring slot index is static (rule 20); the B ring's plain
+        // VGPR loads survive __syncthreads (no glds in flight).
+        constexpr int PANEL = 128;              // k per panel (4 MFMA chunks)
+        constexpr int ROWS = MT * 16;
+        __shared__ __attribute__((aligned(16))) bf16_t xs[2 * ROWS * PANEL];
+        const int nchunks = ROWS * (PANEL / 8); // 16B chunks per panel
+        const int swz = (fr & 15) << 4;
+
+        auto stage = [&](int kp, int buf) {
+            // thread c covers chunk (row = c/16, kc = c%16): 16 lanes read
+            // one row's 256 B contiguously; ds_write_b128 8-lane groups hit
+            // 8 distinct swizzled slots.
+            for (int c = threadIdx.x; c < nchunks; c += 256) {
+                const int row = c >> 4;
+                const int kc = c & 15;
+                const bf16_t* src = x + (long)min(row, M - 1) * K
+                                    + min((long)kp + kc * 8, (long)K - 8);
+                uint4 v = *(const uint4*)src;
+                const int byte = ((row * PANEL + kc * 8) * 2)
+                                 ^ ((row & 15) << 4);
+                *(uint4*)((char*)(xs + (long)buf * ROWS * PANEL) + byte) = v;
+            }
+        };
+
+        const char* xb0 = (const char*)xs;
+        const char* xb1 = (const char*)(xs + (long)ROWS * PANEL);
+        auto panel_mfma = [&](const char* xbase, const bf16x8 (&bb)[4]) {
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+#pragma unroll
+                for (int t = 0; t < MT; ++t) {
+                    const int row = t * 16 + fr;
+                    const int byte = ((row * PANEL + u * 32 + kg) * 2) ^ swz;
+                    const bf16x8 a = *(const bf16x8*)(xbase + byte);
+                    acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a, bb[u], acc[t], 0, 0, 0);
+                }
+            }
+        };
+
+        // bbuf[0..3] = panel-even chunks, bbuf[4..7] = panel-odd chunks
+        bf16x8 beven[4], bodd[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+            beven[u] = bbuf[u];
+            bodd[u] = bbuf[u + 4];
+        }
+        stage(k0, 0);
+        __syncthreads();
+        int k = k0;
+        const int kfull = k0 + ((k1 - k0) / (2 * PANEL)) * (2 * PANEL);
+        for (; k < kfull; k += 2 * PANEL) {
+            // ---- even panel [k, k+PANEL): compute from xs buf0 / beven
+            stage(k + PANEL, 1);
+            {
+                bf16x8 cur[4];
+#pragma unroll
+                for (int u = 0; u < 4; ++u) {
+                    cur[u] = beven[u];
+                    beven[u] = __builtin_nontemporal_load((const bf16x8*)(
+                        wrow + min(k + 2 * PANEL + u * 32, klast)));
+                }
+                panel_mfma(xb0, cur);
+            }
+            __syncthreads();
+            // ---- odd panel [k+PANEL, k+2*PANEL): xs buf1 / bodd
+            if (k + 2 * PANEL < k1)
+                stage(k + 2 * PANEL, 0);
+            {
+                bf16x8 cur[4];
+#pragma unroll
+                for (int u = 0; u < 4; ++u) {
+                    cur[u] = bodd[u];
+                    bodd[u] = __builtin_nontemporal_load((const bf16x8*)(
+                        wrow + min(k + 3 * PANEL + u * 32, klast)));
+                }
+                panel_mfma(xb1, cur);
+            }
+            __syncthreads();
+        }
+        // ---- ragged tail: < 2 panels left (cold path, guarded chunks)
+        int buf = 0;
+        for (; k < k1; k += PANEL, buf ^= 1) {
+            stage(k, buf);
+            __syncthreads();
+            const int ch = min(PANEL, k1 - k) >> 5;
+            const char* xbase = buf ? xb1 : xb0;
+            const bf16_t* wk = wrow + k;
+            for (int u = 0; u < ch; ++u) {
+                const bf16x8 b = __builtin_nontemporal_load(
+                    (const bf16x8*)(wk + u * 32));
+#pragma unroll
+                for (int t = 0; t < MT; ++t) {
+                    const int row = t * 16 + fr;
+                    const int byte = ((row * PANEL + u * 32 + kg) * 2) ^ swz;
+                    const bf16x8 a = *(const bf16x8*)(xbase + byte);
+                    acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a, b, acc[t], 0, 0, 0);
+                }
+            }
+            __syncthreads();
         }
     }
 
