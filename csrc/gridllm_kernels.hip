@@ -504,12 +504,16 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 template <int MT>
-__global__ void __launch_bounds__(256, 2)   // 2 waves/SIMD: frees VGPRs so
-                                            // the scheduler keeps the load
-                                            // pipeline deep (not vmcnt(0))
+__global__ void __launch_bounds__(256, 2)
 k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
               const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
               int M, int N, int K, int splitk, int kslice) {
+    // Occupancy-first design: the grid is (N/64) * splitk blocks with
+    // kslice a small multiple of 256, so 2-8 blocks share each CU and TLP
+    // hides the weight-stream latency (one wave per SIMD cannot — measured).
+    // Per 256-k sub-slice the body is straight-line: issue all 8 B chunks
+    // (nt, 16 B/lane, k-contiguous), stage the x panel into LDS coalesced
+    // (XOR-swizzled), barrier, 8 x MT MFMAs off ds_read_b128 fragments.
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
     const int ks = blockIdx.x % splitk;
@@ -520,166 +524,66 @@ k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
     const int fr = lane & 15;                   // fragment row (A) / col (B)
     const int kg = (lane >> 4) * 8;             // k sub-chunk
 
+    constexpr int KS = 256;                     // k per staged panel
+    constexpr int ROWS = MT * 16;
+    __shared__ __attribute__((aligned(16))) bf16_t xs[ROWS * KS];
+    const int swz = (fr & 15) << 4;
+
     f32x4 acc[MT];
 #pragma unroll
     for (int t = 0; t < MT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-    // EVERY load in the hot loop is branchless: a lane-divergent guard
-    // around a load makes hipcc branch per element and drain vmcnt(0) each
-    // time (guide §5 trap (c)) — rows m >= M read a CLAMPED valid address;
-    // their garbage feeds only output rows the epilogue never stores (MFMA
-    // row i depends only on A row i). B prefetches past the slice end
-    // re-read the last chunk.
     const bf16_t* wrow = w + (long)(n0 + fr) * K + kg;
-    const int klast = k1 - 32;                  // k1-k0 >= 32, both 32-aligned
 
-    // U-deep non-temporal prefetch ring on the weight stream: one wave keeps
-    // U KiB of W in flight so a 1-wave/SIMD launch still covers the ~900-cy
-    // HBM latency. Static ring indices (full unroll) — a runtime index would
-    // spill the ring to scratch (guide §5.4 rule 20).
-    constexpr int U = 8;
-    bf16x8 bbuf[U];
+    for (int kp = k0; kp < k1; kp += KS) {
+        const int nch = min(KS, k1 - kp) >> 5;  // 32-k chunks (8 when full)
+        // all B chunks of this panel in flight first (branchless clamp)
+        const int kcl = k1 - 32;
+        bf16x8 bb[8];
 #pragma unroll
-    for (int u = 0; u < U; ++u) {
-        bbuf[u] = __builtin_nontemporal_load(
-            (const bf16x8*)(wrow + min(k0 + u * 32, klast)));
-    }
-
-    if constexpr (MT == 1) {
-        // M <= 16: all 16 fragment rows clamp to few x rows — direct loads
-        // are broadcast-cheap, no staging needed.
-        const bf16_t* xp = x + (long)min(fr, M - 1) * K + kg;
-        int k = k0;
-        for (; k + U * 32 <= k1; k += U * 32) {
+        for (int u = 0; u < 8; ++u)
+            bb[u] = __builtin_nontemporal_load(
+                (const bf16x8*)(wrow + min(kp + u * 32, kcl)));
+        // stage x[*, kp:kp+KS] -> LDS: thread c covers (row=c/32, kc=c%32);
+        // 32 lanes read one row's 512 B contiguously; swizzled ds_write_b128
+        if (kp > k0) __syncthreads();           // previous panel fully read
 #pragma unroll
-            for (int u = 0; u < U; ++u) {
-                const bf16x8 b = bbuf[u];
-                bbuf[u] = __builtin_nontemporal_load(
-                    (const bf16x8*)(wrow + min(k + (u + U) * 32, klast)));
-                const bf16x8 a = *(const bf16x8*)(xp + k + u * 32);
-                acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[0],
-                                                                 0, 0, 0);
-                __builtin_amdgcn_sched_group_barrier(0x20, 2, 0);
-                __builtin_amdgcn_sched_group_barrier(0x8, 1, 0);
-            }
+        for (int i = 0; i < MT * 2; ++i) {
+            const int c = (int)threadIdx.x + i * 256;
+            const int row = c >> 5;
+            const int kc = c & 31;
+            const uint4 v = *(const uint4*)(
+                x + (long)min(row, M - 1) * K
+                + min((long)kp + kc * 8, (long)K - 8));
+            const int byte = ((row * KS + kc * 8) * 2) ^ ((row & 15) << 4);
+            *(uint4*)((char*)xs + byte) = v;
         }
+        __syncthreads();
+        // 8 chunks x MT MFMAs; full panels take the unguarded fast path
+        if (nch == 8) {
 #pragma unroll
-        for (int u = 0; u < U; ++u) {
-            const int kk = k + u * 32;
-            if (kk < k1) {
-                const bf16x8 a = *(const bf16x8*)(xp + kk);
-                acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a, bbuf[u], acc[0], 0, 0, 0);
-            }
-        }
-    } else {
-        // M > 16: a direct A-fragment load would touch 64 cache lines per
-        // instruction (16 rows x 6-16 KB apart — the "fragment-shaped x
-        // loads" anti-pattern, guide §5). Stage x k-panels through LDS with
-        // coalesced full-line loads; fragments come back via XOR-swizzled
-        // ds_read_b128 (row&15)<<4 — conflict-free since a 16-lane group
-        // reads rows 0..15 (T2 note). Ping-pong panels, unrolled in PAIRS so
-        // every B-ring slot index is static (rule 20); the B ring's plain
-        // VGPR loads survive __syncthreads (no glds in flight).
-        constexpr int PANEL = 128;              // k per panel (4 MFMA chunks)
-        constexpr int ROWS = MT * 16;
-        __shared__ __attribute__((aligned(16))) bf16_t xs[2 * ROWS * PANEL];
-        const int nchunks = ROWS * (PANEL / 8); // 16B chunks per panel
-        const int swz = (fr & 15) << 4;
-
-        auto stage = [&](int kp, int buf) {
-            // thread c covers chunk (row = c/16, kc = c%16): 16 lanes read
-            // one row's 256 B contiguously; ds_write_b128 8-lane groups hit
-            // 8 distinct swizzled slots.
-            for (int c = threadIdx.x; c < nchunks; c += 256) {
-                const int row = c >> 4;
-                const int kc = c & 15;
-                const bf16_t* src = x + (long)min(row, M - 1) * K
-                                    + min((long)kp + kc * 8, (long)K - 8);
-                uint4 v = *(const uint4*)src;
-                const int byte = ((row * PANEL + kc * 8) * 2)
-                                 ^ ((row & 15) << 4);
-                *(uint4*)((char*)(xs + (long)buf * ROWS * PANEL) + byte) = v;
-            }
-        };
-
-        const char* xb0 = (const char*)xs;
-        const char* xb1 = (const char*)(xs + (long)ROWS * PANEL);
-        auto panel_mfma = [&](const char* xbase, const bf16x8 (&bb)[4]) {
-#pragma unroll
-            for (int u = 0; u < 4; ++u) {
+            for (int u = 0; u < 8; ++u) {
 #pragma unroll
                 for (int t = 0; t < MT; ++t) {
                     const int row = t * 16 + fr;
-                    const int byte = ((row * PANEL + u * 32 + kg) * 2) ^ swz;
-                    const bf16x8 a = *(const bf16x8*)(xbase + byte);
+                    const int byte = ((row * KS + u * 32 + kg) * 2) ^ swz;
+                    const bf16x8 a = *(const bf16x8*)((const char*)xs + byte);
                     acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a, bb[u], acc[t], 0, 0, 0);
                 }
             }
-        };
-
-        // bbuf[0..3] = panel-even chunks, bbuf[4..7] = panel-odd chunks
-        bf16x8 beven[4], bodd[4];
-#pragma unroll
-        for (int u = 0; u < 4; ++u) {
-            beven[u] = bbuf[u];
-            bodd[u] = bbuf[u + 4];
-        }
-        stage(k0, 0);
-        __syncthreads();
-        int k = k0;
-        const int kfull = k0 + ((k1 - k0) / (2 * PANEL)) * (2 * PANEL);
-        for (; k < kfull; k += 2 * PANEL) {
-            // ---- even panel [k, k+PANEL): compute from xs buf0 / beven
-            stage(k + PANEL, 1);
-            {
-                bf16x8 cur[4];
-#pragma unroll
-                for (int u = 0; u < 4; ++u) {
-                    cur[u] = beven[u];
-                    beven[u] = __builtin_nontemporal_load((const bf16x8*)(
-                        wrow + min(k + 2 * PANEL + u * 32, klast)));
-                }
-                panel_mfma(xb0, cur);
-            }
-            __syncthreads();
-            // ---- odd panel [k+PANEL, k+2*PANEL): xs buf1 / bodd
-            if (k + 2 * PANEL < k1)
-                stage(k + 2 * PANEL, 0);
-            {
-                bf16x8 cur[4];
-#pragma unroll
-                for (int u = 0; u < 4; ++u) {
-                    cur[u] = bodd[u];
-                    bodd[u] = __builtin_nontemporal_load((const bf16x8*)(
-                        wrow + min(k + 3 * PANEL + u * 32, klast)));
-                }
-                panel_mfma(xb1, cur);
-            }
-            __syncthreads();
-        }
-        // ---- ragged tail: < 2 panels left (cold path, guarded chunks)
-        int buf = 0;
-        for (; k < k1; k += PANEL, buf ^= 1) {
-            stage(k, buf);
-            __syncthreads();
-            const int ch = min(PANEL, k1 - k) >> 5;
-            const char* xbase = buf ? xb1 : xb0;
-            const bf16_t* wk = wrow + k;
-            for (int u = 0; u < ch; ++u) {
-                const bf16x8 b = __builtin_nontemporal_load(
-                    (const bf16x8*)(wk + u * 32));
+        } else {
+            for (int u = 0; u < nch; ++u) {
+                const bf16x8 b = bb[u];
 #pragma unroll
                 for (int t = 0; t < MT; ++t) {
                     const int row = t * 16 + fr;
-                    const int byte = ((row * PANEL + u * 32 + kg) * 2) ^ swz;
-                    const bf16x8 a = *(const bf16x8*)(xbase + byte);
+                    const int byte = ((row * KS + u * 32 + kg) * 2) ^ swz;
+                    const bf16x8 a = *(const bf16x8*)((const char*)xs + byte);
                     acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a, b, acc[t], 0, 0, 0);
                 }
             }
-            __syncthreads();
         }
     }
 
