@@ -180,9 +180,13 @@ def gemm_skinny(x: torch.Tensor, w: torch.Tensor,
     lib = _load()
     m, k = x.shape
     n = w.shape[0]
+    ntiles = max(1, n // 64)
     if splitk <= 0:
-        splitk = max(1, min(512 // max(1, n // 64), k // 256, 16))
-    kslice = ((k + splitk - 1) // splitk + 31) // 32 * 32
+        # target ~2048 blocks; kslice is a multiple of the kernel's 256-k
+        # panel and every split is useful (no idle blocks)
+        splitk = max(1, min(k // 256, 2048 // ntiles))
+    kslice = ((k + splitk - 1) // splitk + 255) // 256 * 256
+    splitk = (k + kslice - 1) // kslice
     if out is None:
         out = torch.empty(m, n, dtype=x.dtype, device=x.device)
     ws = None
