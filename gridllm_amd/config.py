@@ -119,6 +119,7 @@ class EngineConfig:
     kv_cache_budget_mb: int = 0                # 0 -> auto from free HBM
     tp_degree: int = 1
     use_hipgraph: bool = True                  # capture decode step in a hipGraph
+    autotune: bool = True                      # per-shape GEMM dispatch tuning
     weights_path: str = ""                     # safetensors dir; empty -> random init
     seed: int = 0
 
@@ -132,6 +133,7 @@ class EngineConfig:
             max_batch_size=_env_int("ENGINE_MAX_BATCH", 64, 1, 4096),
             tp_degree=_env_int("ENGINE_TP", 1, 1, 8),
             use_hipgraph=_env_int("ENGINE_HIPGRAPH", 1, 0, 1) == 1,
+            autotune=_env_int("ENGINE_AUTOTUNE", 1, 0, 1) == 1,
             weights_path=_env_str("ENGINE_WEIGHTS", ""),
             seed=_env_int("ENGINE_SEED", 0),
         )

@@ -90,6 +90,12 @@ class InferenceEngine:
         self._scratch_slot = self.cfg.max_batch_size  # last slot
         self.model.cache.reserve(self._scratch_slot)
         self._graphs: dict[int, tuple] = {}
+        if self.is_gpu and self.cfg.autotune:
+            from gridllm_amd.ops.autotune import tune_model_shapes
+            buckets = [b for b in _BUCKETS if b <= self.cfg.max_batch_size]
+            if self.cfg.max_batch_size not in buckets:
+                buckets.append(self._bucket(self.cfg.max_batch_size))
+            tune_model_shapes(self.model, sorted(set(buckets)))
         self._lock = threading.Lock()
         log.performance("engine ready", model=spec.name,
                         params=spec.params, device=str(self.device),

@@ -22,7 +22,9 @@ from dataclasses import dataclass
 from typing import Optional
 
 import torch
-import torch.nn.functional as F
+import torch.nn.functional as F  # noqa: F401
+
+from gridllm_amd.ops.autotune import linear as _linear
 
 from gridllm_amd import ops
 from gridllm_amd.models.registry import ModelSpec
@@ -183,7 +185,7 @@ class LlamaModel:
         resid = x
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
-            qkv = F.linear(normed, lw.wqkv)
+            qkv = _linear(normed, lw.wqkv)
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
                             self.cos, self.sin, pos, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)
@@ -191,20 +193,20 @@ class LlamaModel:
             vv = self.cache.v[li][slot, :, :start_pos + t].permute(1, 0, 2).unsqueeze(0)
             attn = ops.attn_prefill(q.unsqueeze(0), kk, vv, self.scale,
                                     causal_offset=start_pos)[0]  # [T,Hq_l,D]
-            attn_out = F.linear(attn.reshape(t, -1), lw.wo)
+            attn_out = _linear(attn.reshape(t, -1), lw.wo)
             self.tp.all_reduce(attn_out)
             normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,
                                         residual=resid)
-            gateup = F.linear(normed, lw.wgateup)
+            gateup = _linear(normed, lw.wgateup)
             act = ops.swiglu(gateup)
-            mlp_out = F.linear(act, lw.wdown)
+            mlp_out = _linear(act, lw.wdown)
             self.tp.all_reduce(mlp_out)
             next_norm = (self.layers[li + 1].attn_norm
                          if li + 1 < len(self.layers) else self.final_norm)
             normed, resid = ops.rmsnorm(mlp_out, next_norm, s.norm_eps,
                                         residual=resid)
         self.cache.seq_lens[slot] = start_pos + t
-        logits = F.linear(normed[-1:], self.lm_head).float()
+        logits = _linear(normed[-1:], self.lm_head).float()
         if return_hidden:
             return logits[0], normed.float().mean(dim=0)
         return logits[0]
@@ -224,19 +226,19 @@ class LlamaModel:
         resid = x
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
-            qkv = F.linear(normed, lw.wqkv)
+            qkv = _linear(normed, lw.wqkv)
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
                             self.cos, self.sin, pos, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)
             attn = ops.attn_decode(q, self.cache.k[li], self.cache.v[li],
                                    attn_seq, self.scale, slot_map=slot_map)
-            attn_out = F.linear(attn.reshape(b, -1), lw.wo)
+            attn_out = _linear(attn.reshape(b, -1), lw.wo)
             self.tp.all_reduce(attn_out)
             normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,
                                         residual=resid)
-            gateup = F.linear(normed, lw.wgateup)
+            gateup = _linear(normed, lw.wgateup)
             act = ops.swiglu(gateup)
-            mlp_out = F.linear(act, lw.wdown)
+            mlp_out = _linear(act, lw.wdown)
             self.tp.all_reduce(mlp_out)
             next_norm = (self.layers[li + 1].attn_norm
                          if li + 1 < len(self.layers) else self.final_norm)
@@ -245,5 +247,5 @@ class LlamaModel:
         self.cache.seq_lens.index_add_(
             0, slot_map.long(),
             torch.ones(b, dtype=torch.int32, device=self.device))
-        logits = F.linear(normed, self.lm_head).float()
+        logits = _linear(normed, self.lm_head).float()
         return logits

@@ -1,0 +1,98 @@
+"""Per-shape GEMM dispatch autotuner.
+
+For each decode projection shape (M, N, K) the engine will hit, measure
+hipBLASLt (torch F.linear) against the in-house skinny GEMM at several
+split-K choices, and cache the winner. ops.linear() then dispatches through
+the table — never slower than the library, faster where the hand-written
+weight-streaming kernel wins. (PyTorch TunableOp only tunes WITHIN
+hipBLASLt; measured +2% on these shapes — profiles/r01_decode_profile.md.)
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from gridllm_amd.ops import hip_ops
+from gridllm_amd.utils.logging import get_logger
+
+log = get_logger("autotune")
+
+# (m, n, k) -> ("blas", 0) | ("ours", splitk)
+_CHOICE: dict[tuple[int, int, int], tuple[str, int]] = {}
+_SPLITK_CANDIDATES = (0, 2, 4, 6, 8, 12, 16, 24, 32)
+
+
+def _time_fn(fn, iters: int = 20) -> float:
+    for _ in range(4):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.monotonic()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.monotonic() - t0) / iters
+
+
+def tune_shape(m: int, n: int, k: int, device) -> tuple[str, int]:
+    key = (m, n, k)
+    if key in _CHOICE:
+        return _CHOICE[key]
+    x = torch.randn(m, k, device=device).to(torch.bfloat16) * 0.1
+    w = torch.randn(n, k, device=device).to(torch.bfloat16) * 0.1
+    best = ("blas", 0)
+    t_best = _time_fn(lambda: F.linear(x, w))
+    if hip_ops.gemm_skinny_supported(m, n, k):
+        seen = set()
+        for sk in _SPLITK_CANDIDATES:
+            if sk > max(1, k // 256):
+                continue
+            if sk in seen:
+                continue
+            seen.add(sk)
+            try:
+                t = _time_fn(lambda: hip_ops.gemm_skinny(x, w, splitk=sk))
+            except RuntimeError:
+                continue
+            if t < t_best * 0.97:           # require a real margin
+                t_best = t
+                best = ("ours", sk)
+    _CHOICE[key] = best
+    return best
+
+
+def choice(m: int, n: int, k: int) -> Optional[tuple[str, int]]:
+    return _CHOICE.get((m, n, k))
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Dispatch x @ w.T through the tuned table (GPU bf16 2-D x only)."""
+    if x.dim() == 2 and x.is_cuda and x.dtype == torch.bfloat16:
+        c = _CHOICE.get((x.shape[0], w.shape[0], w.shape[1]))
+        if c is not None and c[0] == "ours":
+            return hip_ops.gemm_skinny(x, w, splitk=c[1])
+    return F.linear(x, w)
+
+
+def tune_model_shapes(model, buckets) -> None:
+    """Pre-tune every decode GEMM shape for the given batch buckets."""
+    s = model.spec
+    d = s.head_dim
+    shapes = [
+        ((s.heads // model.tp.size + 2 * (s.kv_heads // model.tp.size)) * d,
+         s.hidden),                               # qkv
+        (s.hidden, (s.heads // model.tp.size) * d),   # o
+        (2 * s.intermediate // model.tp.size, s.hidden),  # gateup
+        (s.hidden, s.intermediate // model.tp.size),      # down
+        (s.vocab, s.hidden),                      # lm_head
+    ]
+    t0 = time.monotonic()
+    for m in buckets:
+        for n, k in shapes:
+            tune_shape(m, n, k, model.device)
+    picked = sum(1 for v in _CHOICE.values() if v[0] == "ours")
+    log.performance("gemm autotune done", shapes=len(_CHOICE),
+                    ours=picked, ms=int((time.monotonic() - t0) * 1000))
