@@ -38,12 +38,22 @@ class _Pending:
 
 
 class Batcher:
-    def __init__(self, engine: InferenceEngine, window_ms: float = 4.0):
+    """Continuous batching by default: requests join the live decode batch
+    at the next step boundary and finished rows leave it, so one long
+    generation never blocks new arrivals (TTFT = prefill + one step).
+    The TP lockstep path (worker/tp_runner.TPBatcher) overrides with
+    wave mode: every rank must see identical work, so rank 0 broadcasts
+    whole waves and runs each to completion."""
+
+    def __init__(self, engine: InferenceEngine, window_ms: float = 4.0,
+                 continuous: bool = True):
         self.engine = engine
         self.window_s = window_ms / 1000.0
+        self.continuous = continuous and engine.tp.size == 1
         self.q: "queue.Queue[_Pending]" = queue.Queue()
         self._stop = threading.Event()
-        self._thread = threading.Thread(target=self._loop, daemon=True,
+        target = self._loop_continuous if self.continuous else self._loop
+        self._thread = threading.Thread(target=target, daemon=True,
                                         name="gridllm-batcher")
         self._thread.start()
 
@@ -74,7 +84,89 @@ class Batcher:
         self.q.put(p)
         return p.future
 
-    # ------------------------------------------------------------- the loop
+    # ------------------------------------------------ continuous batching
+
+    def _prep(self, p: _Pending) -> tuple[str, dict]:
+        """Render the prompt + merge stop sequences for a generate/chat row."""
+        if p.kind == "chat":
+            prompt = chat_mod.render_chat(p.messages or [])
+            opts = dict(p.options)
+            opts["stop"] = list(opts.get("stop") or []) + \
+                list(chat_mod.CHAT_STOPS)
+        else:
+            prompt = p.prompt
+            opts = dict(p.options)
+            if p.extra_stops:
+                opts["stop"] = list(opts.get("stop") or []) + p.extra_stops
+        return prompt, opts
+
+    def _loop_continuous(self) -> None:
+        max_batch = max(1, self.engine.cfg.max_batch_size)
+        live: list[tuple[_Pending, object]] = []   # (pending, engine row)
+        while True:
+            # -- admit: block only when idle, otherwise drain without waiting
+            arrivals: list[_Pending] = []
+            if not live:
+                try:
+                    arrivals.append(self.q.get(timeout=0.5))
+                except queue.Empty:
+                    if self._stop.is_set():
+                        break
+                    continue
+            while (len(live) + len(arrivals) < max_batch
+                   and self.engine.model.cache.n_free > len(arrivals)):
+                try:
+                    arrivals.append(self.q.get_nowait())
+                except queue.Empty:
+                    break
+            stop_now = False
+            for p in arrivals:
+                if p.kind == "__stop__":
+                    stop_now = True
+                    continue
+                if p.kind == "embedding":
+                    try:
+                        p.future.set_result(self.engine.embed(p.inputs or []))
+                    except Exception as e:  # noqa: BLE001
+                        p.future.set_exception(e)
+                    continue
+                prompt, opts = self._prep(p)
+                try:
+                    row = self.engine.start_row(prompt, opts, p.on_token,
+                                                p.cancel)
+                except Exception as e:  # noqa: BLE001
+                    log.logger.exception("admission failed")
+                    p.future.set_exception(e)
+                    continue
+                if row.done:
+                    p.future.set_result(row.result)
+                else:
+                    live.append((p, row))
+            # -- one decode step for everything in flight
+            if live:
+                try:
+                    self.engine.step_rows([row for _, row in live])
+                except Exception as e:  # noqa: BLE001
+                    log.logger.exception("decode step failed")
+                    for p, _ in live:
+                        if not p.future.done():
+                            p.future.set_exception(e)
+                    live = []
+                still: list[tuple[_Pending, object]] = []
+                for p, row in live:
+                    if row.done:
+                        p.future.set_result(row.result)
+                    else:
+                        still.append((p, row))
+                live = still
+            if stop_now or (self._stop.is_set() and not live):
+                for p, _ in live:
+                    if not p.future.done():
+                        p.future.set_exception(
+                            RuntimeError("batcher shut down"))
+                break
+
+    # ------------------------------------------------------ wave loop (TP)
 
     def _loop(self) -> None:
         max_batch = max(1, self.engine.cfg.max_batch_size)

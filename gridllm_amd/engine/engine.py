@@ -58,6 +58,9 @@ class _Row:
     last_token: int = 0
     text_tail: str = ""            # pending window for stop-sequence matching
     force_cancel: bool = False     # TP: rank-0-synchronized cancel decision
+    done: bool = False
+    t_start_ns: int = 0            # wall clock at admission
+    t_decode_ns: int = 0           # wall clock when prefill+first token done
 
 
 class InferenceEngine:
@@ -241,72 +244,79 @@ class InferenceEngine:
         n = len(requests)
         on_tokens = on_tokens or [None] * n
         cancels = cancels or [None] * n
-        t_start = time.monotonic_ns()
-        rows: list[_Row] = []
-        results = [GenerationResult() for _ in range(n)]
-
-        # --- prefill (sequential per request; batched decode after)
-        self._sync()
-        t_prefill = time.monotonic_ns()
-        first_logits: list[torch.Tensor] = []
-        for i, (prompt, opts_d) in enumerate(requests):
-            opts = SamplingOptions.from_dict(opts_d)
-            ids = self._encode(prompt)
-            if opts.num_ctx:
-                ids = ids[-max(16, min(opts.num_ctx, self.model.max_seq - 8)):]
-            slot = self.model.cache.alloc()
-            tokens = torch.as_tensor(ids, dtype=torch.int64, device=self.device)
-            logits = self.model.forward_prefill(tokens, slot)
-            first_logits.append(logits)
-            stops = list(opts.stop or [])
-            if extra_stops:
-                stops += extra_stops
-            max_new = opts.num_predict if opts.num_predict and opts.num_predict > 0 \
-                else 128
-            max_new = min(max_new, self.model.max_seq - len(ids) - 1)
-            rows.append(_Row(slot=slot, state=SamplerState.create(opts, self.device),
-                             decoder=self.tokenizer.decode_stream(),
-                             on_token=on_tokens[i], cancel=cancels[i],
-                             stops=stops, max_new=max_new, result=results[i]))
-            results[i].prompt_eval_count = len(ids)
-        self._sync()
-        t_decode = time.monotonic_ns()
-        prefill_ns = t_decode - t_prefill
-        for r in results:
-            r.prompt_eval_duration = prefill_ns // n
-
-        # --- first token from prefill logits
-        self._sync_cancel(rows)
-        active: list[_Row] = []
-        for i, row in enumerate(rows):
-            tok = self._sample_batch(first_logits[i].unsqueeze(0), [row])[0]
-            if self._emit(row, tok) and row.max_new > 1:
-                active.append(row)
-            else:
-                self._retire(row)
-
-        # --- batched decode
+        rows = [self._start_row(prompt, opts_d, on_tokens[i], cancels[i],
+                                extra_stops)
+                for i, (prompt, opts_d) in enumerate(requests)]
+        active = [r for r in rows if not r.done]
         while active:
-            token_ids = [r.last_token for r in active]
-            slot_ids = [r.slot for r in active]
-            logits = self._decode_step(token_ids, slot_ids)
-            next_ids = self._sample_batch(logits, active)
-            self._sync_cancel(active)
-            still: list[_Row] = []
-            for r, tok in zip(active, next_ids):
-                if self._emit(r, tok):
-                    still.append(r)
-                else:
-                    self._retire(r)
-            active = still
-        self._sync()
-        t_end = time.monotonic_ns()
-        dec_ns = t_end - t_decode
-        total_evals = max(1, sum(r.eval_count for r in results))
-        for r in results:
-            r.eval_duration = int(dec_ns * (r.eval_count / total_evals))
-            r.total_duration = t_end - t_start
-        return results
+            active = self._step_rows(active)
+        return [r.result for r in rows]
+
+    # -------------------------------------------------- row-stepping API
+    # (continuous batching: the batcher admits rows into a live decode
+    #  loop; generate_batch above is the run-to-completion wrapper used
+    #  by the TP lockstep path and tests)
+
+    def start_row(self, prompt: str, opts_d: dict,
+                  on_token: Optional[Callable[[str, bool], None]] = None,
+                  cancel: Optional[threading.Event] = None,
+                  extra_stops: Optional[list[str]] = None) -> _Row:
+        with self._lock:
+            return self._start_row(prompt, opts_d, on_token, cancel,
+                                   extra_stops)
+
+    def step_rows(self, rows: list["_Row"]) -> list["_Row"]:
+        with self._lock:
+            return self._step_rows(rows)
+
+    def _start_row(self, prompt, opts_d, on_token, cancel,
+                   extra_stops) -> _Row:
+        """Prefill one request, sample+emit its first token. The returned
+        row is retired already (row.done) if it finished on token 1."""
+        t0 = time.monotonic_ns()
+        opts = SamplingOptions.from_dict(opts_d)
+        ids = self._encode(prompt)
+        if opts.num_ctx:
+            ids = ids[-max(16, min(opts.num_ctx, self.model.max_seq - 8)):]
+        slot = self.model.cache.alloc()
+        tokens = torch.as_tensor(ids, dtype=torch.int64, device=self.device)
+        logits = self.model.forward_prefill(tokens, slot)
+        stops = list(opts.stop or [])
+        if extra_stops:
+            stops += extra_stops
+        max_new = opts.num_predict if opts.num_predict and opts.num_predict > 0 \
+            else 128
+        max_new = min(max_new, self.model.max_seq - len(ids) - 1)
+        row = _Row(slot=slot, state=SamplerState.create(opts, self.device),
+                   decoder=self.tokenizer.decode_stream(),
+                   on_token=on_token, cancel=cancel,
+                   stops=stops, max_new=max_new, result=GenerationResult(),
+                   t_start_ns=t0)
+        row.result.prompt_eval_count = len(ids)
+        self._sync_cancel([row])
+        tok = self._sample_batch(logits.unsqueeze(0), [row])[0]  # syncs
+        t1 = time.monotonic_ns()
+        row.result.prompt_eval_duration = t1 - t0
+        row.t_decode_ns = t1
+        if self._emit(row, tok) and row.max_new > 1:
+            return row
+        self._retire(row)
+        return row
+
+    def _step_rows(self, active: list["_Row"]) -> list["_Row"]:
+        """One batched decode step; returns the rows still running."""
+        token_ids = [r.last_token for r in active]
+        slot_ids = [r.slot for r in active]
+        logits = self._decode_step(token_ids, slot_ids)
+        next_ids = self._sample_batch(logits, active)
+        self._sync_cancel(active)
+        still: list[_Row] = []
+        for r, tok in zip(active, next_ids):
+            if self._emit(r, tok):
+                still.append(r)
+            else:
+                self._retire(r)
+        return still
 
     def _sync_cancel(self, rows: Sequence[_Row]) -> None:
         """Under TP every rank must retire the same rows at the same step, or
@@ -384,6 +394,10 @@ class InferenceEngine:
             if row.on_token:
                 row.on_token(final, False)
         row.text_tail = ""
+        t_end = time.monotonic_ns()
+        row.result.eval_duration = max(0, t_end - row.t_decode_ns)
+        row.result.total_duration = t_end - row.t_start_ns
+        row.done = True
         if row.on_token:
             row.on_token("", True)
         self.model.cache.free(row.slot)

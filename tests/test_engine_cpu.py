@@ -101,3 +101,48 @@ def test_num_predict_length_cap(engine):
     assert r.eval_count <= 5
     if r.eval_count == 5 and r.done_reason != "stop":
         assert r.done_reason == "length"
+
+
+def test_continuous_batcher_interleaves():
+    """A short request submitted while a long one is mid-decode must join the
+    live batch and finish first (continuous batching, not wave batching)."""
+    from gridllm_amd.engine.batcher import Batcher
+    cfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                       max_seq_len=128, max_batch_size=4, use_hipgraph=False)
+    eng = InferenceEngine(cfg)
+    b = Batcher(eng)
+    assert b.continuous
+    try:
+        order = []
+        started = threading.Event()
+
+        def long_cb(piece, done):
+            if piece:
+                started.set()
+            if done:
+                order.append("long")
+
+        long_fut = b.submit_generate(
+            "long", {"temperature": 0.0, "num_predict": 80},
+            on_token=long_cb)
+        assert started.wait(timeout=10)   # long row is mid-decode now
+        short_fut = b.submit_generate(
+            "short", {"temperature": 0.0, "num_predict": 2},
+            on_token=lambda piece, done: done and order.append("short"))
+        r_short = short_fut.result(timeout=30)
+        r_long = long_fut.result(timeout=30)
+        assert r_short.eval_count <= 2
+        assert r_long.eval_count == 80
+        assert order == ["short", "long"]
+    finally:
+        b.shutdown()
+
+
+def test_row_timings_populated():
+    cfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                       max_seq_len=128, max_batch_size=2, use_hipgraph=False)
+    eng = InferenceEngine(cfg)
+    r = eng.generate("timing", {"temperature": 0.0, "num_predict": 4})
+    assert r.prompt_eval_duration > 0
+    assert r.eval_duration >= 0
+    assert r.total_duration >= r.prompt_eval_duration
