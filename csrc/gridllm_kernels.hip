@@ -622,6 +622,186 @@ k_gemm_combine(bf16_t* __restrict__ out, const float* __restrict__ ws,
     }
 }
 
+// ===========================================================================
+// GQA prefill attention — flash-style on MFMA (bf16), reads K/V straight
+// from the slot cache (no permuted host-side views).
+//   q:   [T, Hq, D] bf16 (roped)
+//   kc/vc: [slots, Hk, Smax, D] bf16
+//   out: [T, Hq, D] bf16
+// Workgroup = 4 waves; wave w owns 32 q-rows (q0 + w*32) of ONE q-head;
+// K/V walked in 32-row tiles with online softmax (guide App. B prefill).
+//
+// Layout trick (guide §5.5 "swapped QK^T"): S^T = mfma(A=K, B=Q) puts a
+// whole q-row's scores in ONE lane (C col = lane&31 = q), so the softmax
+// row-reduce is 15 in-register fmax + one shfl_xor(32) — no LDS. P is then
+// routed to the PV MFMA's B-operand layout through a per-wave LDS scratch
+// (paired b32 writes -> b128 fragment reads, same-wave lgkmcnt wait only),
+// and O^T = mfma(A=V^T, B=P) accumulates with V staged TRANSPOSED in LDS
+// (transpose once per tile, cooperative; A-fragments then read b128-clean).
+// mfma_f32_32x32x16_bf16 maps (guide §3):
+//   A[i][k]: lane l holds i = l&31, k = (l>>5)*8 + 0..7
+//   B[k][j]: lane l holds j = l&31, same k chunk
+//   C/D[i][j]: col j = l&31, row i = (reg&3) + 8*(reg>>2) + 4*(l>>5)
+// Constraints: D in {64, 128}; Hq % Hk == 0.
+// ===========================================================================
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+template <int D>
+__global__ void __launch_bounds__(256, 2)
+k_attn_prefill(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
+               const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
+               int T, int Hq, int G, int Hk, int Smax, int slot,
+               int s_total, int causal_off, float scale) {
+    constexpr int NS = D / 16;     // 16-wide d slices (QK^T fragments)
+    constexpr int NT = D / 32;     // 32-wide d tiles (output / V^T)
+    constexpr int VST = 40;        // padded k-stride of V^T tile (16B-mult)
+
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int hi = lane >> 5;
+    const int ln = lane & 31;
+    const int hq = blockIdx.y;
+    const int hk = hq / G;
+    const int q0 = blockIdx.x * 128;
+    const int qw = q0 + wid * 32;
+
+    __shared__ __attribute__((aligned(16))) bf16_t vt[D * VST];
+    __shared__ __attribute__((aligned(16))) bf16_t plds[4][2 * 64 * 8];
+
+    const bf16_t* kbase = kc + ((long)slot * Hk + hk) * (long)Smax * D;
+    const bf16_t* vbase = vc + ((long)slot * Hk + hk) * (long)Smax * D;
+
+    // Q fragments: lane holds Q[qw+ln][s*16 + hi*8 + 0..7] (B operand)
+    bf16x8 qf[NS];
+    {
+        const bf16_t* qrow =
+            q + ((long)min(qw + ln, T - 1) * Hq + hq) * D + hi * 8;
+#pragma unroll
+        for (int s = 0; s < NS; ++s) qf[s] = *(const bf16x8*)(qrow + s * 16);
+    }
+
+    const int q_abs = causal_off + qw + ln;   // this lane's absolute q pos
+    float m = -INFINITY, l = 0.f;
+    f32x16 acc[NT];
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
+
+    // causal upper bound for the workgroup's LAST q-row
+    const int kmax = min(s_total, causal_off + q0 + 128);
+
+    for (int k0 = 0; k0 < kmax; k0 += 32) {
+        // ---- stage V^T tile (cooperative transpose) --------------------
+        __syncthreads();               // previous tile's vt readers done
+        {
+            constexpr int CPR = D / 8;            // b128 chunks per V row
+            for (int c = threadIdx.x; c < 32 * CPR; c += 256) {
+                const int kr = c / CPR;
+                const int d8 = (c % CPR) * 8;
+                uint4 v = *(const uint4*)(
+                    vbase + (long)min(k0 + kr, s_total - 1) * D + d8);
+                const bf16_t* pv = (const bf16_t*)&v;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) vt[(d8 + j) * VST + kr] = pv[j];
+            }
+        }
+        __syncthreads();
+
+        // ---- S^T = K·Q^T (K fragments straight from HBM/L2) ------------
+        f32x16 s_acc;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) s_acc[r] = 0.f;
+        {
+            const bf16_t* krow =
+                kbase + (long)min(k0 + ln, s_total - 1) * D + hi * 8;
+#pragma unroll
+            for (int s = 0; s < NS; ++s) {
+                const bf16x8 kf = *(const bf16x8*)(krow + s * 16);
+                s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    kf, qf[s], s_acc, 0, 0, 0);
+            }
+        }
+
+        // ---- mask + online softmax (lane owns one q-row) ---------------
+        float sv[16], tmax = -INFINITY;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+            const int kg = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            const bool ok = (kg < s_total) && (kg <= q_abs);
+            sv[r] = ok ? s_acc[r] * scale : -INFINITY;
+            tmax = fmaxf(tmax, sv[r]);
+        }
+        tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
+        const float mn = fmaxf(m, tmax);
+        const float alpha = (m == -INFINITY) ? 0.f : __expf(m - mn);
+        float p[16], psum = 0.f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+            p[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mn);
+            psum += p[r];
+        }
+        psum += __shfl_xor(psum, 32, WAVE);
+        l = l * alpha + psum;
+        m = mn;
+
+        // ---- P -> PV B-fragment layout via per-wave LDS scratch --------
+        // (kpos, q=ln) lands where slice sidx = kpos>>4, lane
+        // lr = ln + 32*((kpos>>3)&1), word i = kpos&7 will read it; regs
+        // pair (r, r+1) = (kpos, kpos+1) -> one packed b32 write.
+        bf16_t* ps = plds[wid];
+#pragma unroll
+        for (int r = 0; r < 16; r += 2) {
+            const int kpos = (r & 3) + 8 * (r >> 2) + 4 * hi;
+            const int sidx = kpos >> 4;
+            const int lr = ln + 32 * ((kpos >> 3) & 1);
+            u32 w = (u32)f2bf(p[r]) | ((u32)f2bf(p[r + 1]) << 16);
+            *(u32*)(ps + (sidx * 64 + lr) * 8 + (kpos & 7)) = w;
+        }
+        // rescale output accumulators while the P writes land
+#pragma unroll
+        for (int t = 0; t < NT; ++t)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) acc[t][r] *= alpha;
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+        // ---- O^T += V^T · P --------------------------------------------
+        bf16x8 pf[2];
+#pragma unroll
+        for (int s = 0; s < 2; ++s)
+            pf[s] = *(const bf16x8*)(ps + (s * 64 + lane) * 8);
+#pragma unroll
+        for (int t = 0; t < NT; ++t) {
+            const bf16_t* vrow = vt + (t * 32 + ln) * VST + hi * 8;
+#pragma unroll
+            for (int s = 0; s < 2; ++s) {
+                const bf16x8 vf = *(const bf16x8*)(vrow + s * 16);
+                acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    vf, pf[s], acc[t], 0, 0, 0);
+            }
+        }
+    }
+
+    // ---- normalize + store: lane holds O^T[*][q=ln] — one q-row --------
+    if (qw + ln < T) {
+        const float inv = (l > 0.f) ? 1.f / l : 0.f;
+        bf16_t* orow = out + ((long)(qw + ln) * Hq + hq) * D;
+#pragma unroll
+        for (int t = 0; t < NT; ++t) {
+#pragma unroll
+            for (int g = 0; g < 4; ++g) {
+                const int d = t * 32 + 8 * g + 4 * hi;
+                u32 lo = (u32)f2bf(acc[t][4 * g] * inv) |
+                         ((u32)f2bf(acc[t][4 * g + 1] * inv) << 16);
+                u32 h2 = (u32)f2bf(acc[t][4 * g + 2] * inv) |
+                         ((u32)f2bf(acc[t][4 * g + 3] * inv) << 16);
+                *(uint2*)(orow + d) = make_uint2(lo, h2);
+            }
+        }
+    }
+}
+
 template <int MT>
 static void launch_gemm_skinny(hipStream_t st, bf16_t* out, float* ws,
                                const bf16_t* x, const bf16_t* w, int M, int N,
@@ -773,6 +953,29 @@ int gl_gemm_skinny(void* stream, void* out, void* ws, const void* x,
             (bf16_t*)out, (const float*)ws, mn, splitk);
         LAUNCH_CHECK();
     }
+    return 0;
+}
+
+int gl_attn_prefill(void* stream, void* out, const void* q, const void* kc,
+                    const void* vc, int T, int Hq, int Hk, int D, int Smax,
+                    int slot, int s_total, int causal_off, float scale) {
+    hipStream_t st = (hipStream_t)stream;
+    if (Hq % Hk) return -110;
+    const int G = Hq / Hk;
+    dim3 grid((unsigned)((T + 127) / 128), (unsigned)Hq), blk(256);
+    if (D == 128)
+        k_attn_prefill<128><<<grid, blk, 0, st>>>(
+            (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)kc,
+            (const bf16_t*)vc, T, Hq, G, Hk, Smax, slot, s_total,
+            causal_off, scale);
+    else if (D == 64)
+        k_attn_prefill<64><<<grid, blk, 0, st>>>(
+            (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)kc,
+            (const bf16_t*)vc, T, Hq, G, Hk, Smax, slot, s_total,
+            causal_off, scale);
+    else
+        return -111;
+    LAUNCH_CHECK();
     return 0;
 }
 

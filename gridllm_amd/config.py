@@ -60,6 +60,8 @@ class ServerConfig:
     max_concurrent_jobs_per_worker: int = 1     # ref default 1
     status_log_interval_s: int = 60
     log_level: str = "info"
+    rate_limit_max_requests: int = 0            # 0 = unlimited (ref: built
+    rate_limit_window_ms: int = 900_000         # but unmounted; we mount it)
 
     @classmethod
     def from_env(cls) -> "ServerConfig":
@@ -75,6 +77,8 @@ class ServerConfig:
             job_retry_delay_ms=_env_int("JOB_RETRY_DELAY", 5_000, 0),
             max_concurrent_jobs_per_worker=_env_int("MAX_CONCURRENT_JOBS_PER_WORKER", 1, 1, 4096),
             log_level=_env_str("LOG_LEVEL", "info", ["debug", "info", "warn", "error"]),
+            rate_limit_max_requests=_env_int("RATE_LIMIT_MAX_REQUESTS", 0, 0),
+            rate_limit_window_ms=_env_int("RATE_LIMIT_WINDOW_MS", 900_000, 1000),
         )
 
 

@@ -189,10 +189,9 @@ class LlamaModel:
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
                             self.cos, self.sin, pos, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)
-            kk = self.cache.k[li][slot, :, :start_pos + t].permute(1, 0, 2).unsqueeze(0)
-            vv = self.cache.v[li][slot, :, :start_pos + t].permute(1, 0, 2).unsqueeze(0)
-            attn = ops.attn_prefill(q.unsqueeze(0), kk, vv, self.scale,
-                                    causal_offset=start_pos)[0]  # [T,Hq_l,D]
+            attn = ops.attn_prefill_cache(
+                q, self.cache.k[li], self.cache.v[li], slot, start_pos + t,
+                start_pos, self.scale)                           # [T,Hq_l,D]
             attn_out = _linear(attn.reshape(t, -1), lw.wo)
             self.tp.all_reduce(attn_out)
             normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,

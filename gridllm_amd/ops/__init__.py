@@ -16,7 +16,7 @@ from gridllm_amd.ops.hip_ops import HipExtensionMissing, available as hip_availa
 
 __all__ = [
     "rmsnorm", "rope", "rope_tables", "swiglu", "kv_append", "attn_decode",
-    "attn_prefill", "hip_available", "HipExtensionMissing",
+    "attn_prefill", "attn_prefill_cache", "hip_available", "HipExtensionMissing",
 ]
 
 rope_tables = reference.rope_tables
@@ -87,11 +87,30 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
 
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  scale: float, causal_offset: int = 0) -> torch.Tensor:
-    # v1: chunked-math path via hipBLASLt matmuls on GPU as well; a
-    # hand-written flash prefill kernel replaces this on the GPU path next.
+    # view-based fallback (CPU reference path / odd head dims); the GPU
+    # cache path goes through attn_prefill_cache below.
     if _on_gpu(q):
         return _attn_prefill_chunked(q, k, v, scale, causal_offset)
     return reference.attn_prefill(q, k, v, scale, causal_offset)
+
+
+def attn_prefill_cache(q: torch.Tensor, kcache: torch.Tensor,
+                       vcache: torch.Tensor, slot: int, s_total: int,
+                       causal_offset: int, scale: float) -> torch.Tensor:
+    """GQA prefill for one sequence against its cache slot.
+    q: [T,Hq,D]; kcache/vcache: [Slots,Hk,Smax,D]; returns [T,Hq,D]."""
+    if (_on_gpu(q) and q.dtype == torch.bfloat16
+            and q.shape[-1] in (64, 128)):
+        from gridllm_amd.ops import hip_ops
+        return hip_ops.attn_prefill_cache(q, kcache, vcache, slot, s_total,
+                                          causal_offset, scale)
+    kk = kcache[slot, :, :s_total].permute(1, 0, 2).unsqueeze(0)
+    vv = vcache[slot, :, :s_total].permute(1, 0, 2).unsqueeze(0)
+    if _on_gpu(q):
+        return _attn_prefill_chunked(q.unsqueeze(0), kk, vv, scale,
+                                     causal_offset)[0]
+    return reference.attn_prefill(q.unsqueeze(0), kk, vv, scale,
+                                  causal_offset)[0]
 
 
 def _attn_prefill_chunked(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,

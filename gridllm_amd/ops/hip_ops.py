@@ -167,6 +167,25 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
     return out
 
 
+def attn_prefill_cache(q: torch.Tensor, kcache: torch.Tensor,
+                       vcache: torch.Tensor, slot: int, s_total: int,
+                       causal_offset: int, scale: float,
+                       out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Flash-style MFMA GQA prefill straight off the slot cache.
+    q/out: [T,Hq,D] bf16; caches [Slots,Hk,Smax,D]; keys 0..s_total-1 of
+    `slot` attendable, causally masked at position causal_offset+t."""
+    lib = _load()
+    t, hq, d = q.shape
+    hk, smax = kcache.shape[1], kcache.shape[2]
+    if out is None:
+        out = torch.empty_like(q)
+    _check(lib.gl_attn_prefill(_stream(), _ptr(out), _ptr(q), _ptr(kcache),
+                               _ptr(vcache), t, hq, hk, d, smax, slot,
+                               s_total, causal_offset,
+                               ctypes.c_float(scale)), "attn_prefill")
+    return out
+
+
 def gemm_skinny_supported(m: int, n: int, k: int) -> bool:
     return m <= 256 and n % 64 == 0 and k % 32 == 0
 

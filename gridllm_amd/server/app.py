@@ -28,8 +28,14 @@ from gridllm_amd.utils.logging import get_logger
 log = get_logger("server")
 
 
-def create_app(scheduler: JobScheduler, registry: WorkerRegistry) -> FastAPI:
+def create_app(scheduler: JobScheduler, registry: WorkerRegistry,
+               cfg: Optional[ServerConfig] = None) -> FastAPI:
     app = FastAPI(title="GridLLM-AMD", version=__version__)
+
+    if cfg is not None:
+        from gridllm_amd.server.middleware import mount_rate_limiter
+        mount_rate_limiter(app, cfg.rate_limit_max_requests,
+                           cfg.rate_limit_window_ms)
 
     ollama_r = ollama_routes.build_router(scheduler, registry)
     app.include_router(ollama_r, prefix="/ollama")  # reference mount point
@@ -68,7 +74,7 @@ class GridLLMServer:
         self.bus = TcpBus(self.cfg.bus_host, self.cfg.bus_port)
         self.registry = WorkerRegistry(self.bus, self.cfg)
         self.scheduler = JobScheduler(self.bus, self.registry, self.cfg)
-        self.app = create_app(self.scheduler, self.registry)
+        self.app = create_app(self.scheduler, self.registry, self.cfg)
         self._status_task: Optional[asyncio.Task] = None
 
     async def start(self) -> None:

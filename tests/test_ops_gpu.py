@@ -193,3 +193,28 @@ def test_gemm_skinny_splitk_consistent(dev):
     o1 = hip_ops.gemm_skinny(x, w, splitk=1)
     o8 = hip_ops.gemm_skinny(x, w, splitk=8)
     assert (o1.float() - o8.float()).abs().max().item() < 0.05
+
+
+@pytest.mark.parametrize("t,hq,hk,d,cau", [
+    (128, 24, 8, 128, 0),     # llama3.2-3B shape, aligned T
+    (77, 24, 8, 128, 0),      # ragged T (masked rows / partial tiles)
+    (200, 24, 8, 128, 33),    # cached prefix: causal offset, s_total%32!=0
+    (256, 32, 8, 64, 0),      # D=64 path, G=4
+    (31, 8, 8, 128, 5),       # MHA (G=1), tiny T
+])
+def test_attn_prefill_mfma_gpu(dev, t, hq, hk, d, cau):
+    """MFMA flash prefill vs fp32 torch reference, straight off the cache."""
+    torch.manual_seed(t * 7 + cau)
+    from gridllm_amd.ops import hip_ops
+    smax, slots, slot = 512, 3, 1
+    s_total = cau + t
+    q = (torch.randn(t, hq, d, device=dev) * 0.5).to(torch.bfloat16)
+    kc = (torch.randn(slots, hk, smax, d, device=dev) * 0.5).to(torch.bfloat16)
+    vc = (torch.randn(slots, hk, smax, d, device=dev) * 0.5).to(torch.bfloat16)
+    scale = 1.0 / math.sqrt(d)
+    out = hip_ops.attn_prefill_cache(q, kc, vc, slot, s_total, cau, scale)
+    kk = kc[slot, :, :s_total].permute(1, 0, 2).unsqueeze(0)
+    vv = vc[slot, :, :s_total].permute(1, 0, 2).unsqueeze(0)
+    ref = R.attn_prefill(q.unsqueeze(0).cpu().float(), kk.cpu().float(),
+                         vv.cpu().float(), scale, causal_offset=cau)[0]
+    assert _mx(out.cpu(), ref) < 0.03
