@@ -312,3 +312,32 @@ async def test_registry_rehydration():
     bus2, reg2, sched2 = await make_stack(core)
     assert reg2.get("w1") is not None
     await teardown(reg2, sched2, w)
+
+
+@pytest.mark.asyncio
+async def test_mixed_fleet_model_aware_routing():
+    """Heterogeneous fleet (BASELINE config #5): workers advertising
+    different models; each request lands only on a worker holding its
+    model, and a multi-model worker serves both."""
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w_small = FakeWorker(core, "w-3b", models=("llama3.2:3b",))
+    w_big = FakeWorker(core, "w-70b", models=("llama3:70b",))
+    w_both = FakeWorker(core, "w-mixed", models=("llama3.2:3b", "llama3:70b"),
+                        concurrency=2)
+    for w in (w_small, w_big, w_both):
+        await w.start()
+    await asyncio.sleep(0.1)
+
+    reqs = []
+    for i in range(6):
+        model = "llama3.2:3b" if i % 2 == 0 else "llama3:70b"
+        reqs.append(InferenceRequest(model=model, prompt=f"p{i}"))
+    results = await asyncio.wait_for(
+        asyncio.gather(*(sched.submit_and_wait(r) for r in reqs)), 20)
+    assert all(r.get("response") for r in results)
+    # every 3b job went to a 3b-capable worker, 70b likewise
+    assert set(w_small.jobs_seen) <= {r.id for r in reqs if r.model == "llama3.2:3b"}
+    assert set(w_big.jobs_seen) <= {r.id for r in reqs if r.model == "llama3:70b"}
+    assert not (set(w_small.jobs_seen) & set(w_big.jobs_seen))
+    await teardown(reg, sched, w_small, w_big, w_both)

@@ -244,3 +244,38 @@ def test_health_endpoints(client):
     assert r.json()["counts"]["total"] == 1
     r = client.get("/")
     assert r.json()["name"] == "GridLLM-AMD"
+
+
+def test_rate_limiter_mounted():
+    """The reference shipped a fixed-window limiter but never mounted it
+    (client/src/middleware/rateLimiter.ts); ours is mounted when enabled."""
+    from gridllm_amd.server.middleware import FixedWindowLimiter
+    lim = FixedWindowLimiter(max_requests=2, window_ms=1000)
+    ok1, rem1, _ = lim.check("a", now_ms=0)
+    ok2, rem2, _ = lim.check("a", now_ms=10)
+    ok3, rem3, _ = lim.check("a", now_ms=20)
+    assert (ok1, ok2, ok3) == (True, True, False)
+    assert (rem1, rem2, rem3) == (1, 0, 0)
+    ok4, _, _ = lim.check("a", now_ms=1500)   # window rolled
+    assert ok4
+    ok_b, _, _ = lim.check("b", now_ms=25)    # independent client
+    assert ok_b
+
+
+def test_rate_limiter_http_429():
+    cfg = ServerConfig(rate_limit_max_requests=2, rate_limit_window_ms=60_000)
+    core = MemoryBusCore()
+    bus = MemoryBus(core)
+    from gridllm_amd.server.registry import WorkerRegistry as _WR
+    from gridllm_amd.server.scheduler import JobScheduler as _JS
+    registry = _WR(bus, cfg)
+    scheduler = _JS(bus, registry, cfg)
+    app = create_app(scheduler, registry, cfg)
+    with TestClient(app) as c:
+        assert c.get("/").status_code == 200
+        r2 = c.get("/")
+        assert r2.status_code == 200
+        assert r2.headers["X-RateLimit-Remaining"] == "0"
+        r3 = c.get("/")
+        assert r3.status_code == 429
+        assert c.get("/health").status_code == 200   # health exempt
