@@ -66,6 +66,19 @@ def tune_shape(m: int, n: int, k: int, device) -> tuple[str, int]:
             if t < t_best * 0.97:           # require a real margin
                 t_best = t
                 best = ("ours", sk)
+    if hip_ops.gemm_proj_supported(m, n, k):
+        # latency-bound regime: (N/64)*splitk ~ 0.5-1x the 256 CUs
+        rule = max(1, min(k // 256, 256 // max(1, n // 64)))
+        for sk in sorted({1, rule, rule * 2, max(1, rule // 2)}):
+            if sk > max(1, k // 64):
+                continue
+            try:
+                t = _time_fn(lambda w: hip_ops.gemm_proj(x, w, splitk=sk), ws)
+            except RuntimeError:
+                continue
+            if t < t_best * 0.97:
+                t_best = t
+                best = ("proj", sk)
     del ws
     _CHOICE[key] = best
     return best
@@ -79,8 +92,11 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Dispatch x @ w.T through the tuned table (GPU bf16 2-D x only)."""
     if x.dim() == 2 and x.is_cuda and x.dtype == torch.bfloat16:
         c = _CHOICE.get((x.shape[0], w.shape[0], w.shape[1]))
-        if c is not None and c[0] == "ours":
-            return hip_ops.gemm_skinny(x, w, splitk=c[1])
+        if c is not None:
+            if c[0] == "ours":
+                return hip_ops.gemm_skinny(x, w, splitk=c[1])
+            if c[0] == "proj":
+                return hip_ops.gemm_proj(x, w, splitk=c[1])
     return F.linear(x, w)
 
 

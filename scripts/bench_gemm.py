@@ -1,46 +1,69 @@
-"""Micro-benchmark: skinny-M GEMM (ours) vs hipBLASLt (torch F.linear) on the
-llama3.2-3B decode projection shapes. Within-probe interleaved A/B."""
+"""Micro-benchmark: decode projection GEMMs — hipBLASLt vs gemm_skinny vs
+gemm_proj on the llama3.2-3B shapes, HBM-cold (weights cycled past the
+256 MB Infinity Cache, as in-situ decode streams them).
+
+Usage: python scripts/bench_gemm.py [M] [--splitk-sweep]
+"""
 import os, sys, time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 import torch.nn.functional as F
 from gridllm_amd.ops import hip_ops
 
-def bench(fn, iters=200):
-    for _ in range(20):
-        fn()
+
+def bench_cold(fn, ws, iters=60):
+    for i in range(8):
+        fn(ws[i % len(ws)])
     torch.cuda.synchronize()
     t0 = time.monotonic()
-    for _ in range(iters):
-        fn()
+    for i in range(iters):
+        fn(ws[i % len(ws)])
     torch.cuda.synchronize()
     return (time.monotonic() - t0) / iters * 1e6  # us
 
+
 def main():
     dev = "cuda:0"
-    M = int(sys.argv[1]) if len(sys.argv) > 1 else 64
+    M = int(sys.argv[1]) if len(sys.argv) > 1 else 256
+    sweep = "--splitk-sweep" in sys.argv
     shapes = [("qkv", 3072, 5120), ("o", 3072, 3072),
               ("gateup", 3072, 16384), ("down", 8192, 3072),
               ("lm_head", 3072, 128256)]
     print(f"M={M}")
-    tot_ours = tot_blas = 0.0
+    tot = {"blas": 0.0, "skinny": 0.0, "proj": 0.0}
     for name, K, N in shapes:
         x = (torch.randn(M, K, device=dev) * 0.1).to(torch.bfloat16)
-        w = (torch.randn(N, K, device=dev) * 0.1).to(torch.bfloat16)
+        wbytes = N * K * 2
+        nc = min(6, max(2, (320 << 20) // wbytes))
+        ws = [(torch.randn(N, K, device=dev) * 0.1).to(torch.bfloat16)
+              for _ in range(nc)]
+        w = ws[0]
         ref = (x.float() @ w.float().t())
-        ours = hip_ops.gemm_skinny(x, w)
-        err = (ours.float() - ref).abs().max().item()
         scale = ref.abs().max().item()
-        gb = N * K * 2 / 1e9
-        t_ours = bench(lambda: hip_ops.gemm_skinny(x, w))
-        t_blas = bench(lambda: F.linear(x, w))
-        ok = err < 0.02 * scale
-        print(f"{name:8s} K={K:6d} N={N:6d} W={gb*1000:7.1f}MB  "
-              f"ours {t_ours:7.1f}us ({gb/t_ours*1e6/1e3:5.2f} TB/s)  "
-              f"blas {t_blas:7.1f}us ({gb/t_blas*1e6/1e3:5.2f} TB/s)  "
-              f"err {err:.4f}/{scale:.1f} {'OK' if ok else 'FAIL'}")
-        tot_ours += t_ours; tot_blas += t_blas
-    print(f"layer-sum ours {tot_ours:.1f}us  blas {tot_blas:.1f}us")
+        gb = wbytes / 1e9
+        res = {}
+        res["blas"] = (bench_cold(lambda w: F.linear(x, w), ws), 0.0)
+        if hip_ops.gemm_skinny_supported(M, N, K):
+            err = (hip_ops.gemm_skinny(x, w).float() - ref).abs().max().item()
+            res["skinny"] = (bench_cold(lambda w: hip_ops.gemm_skinny(x, w), ws), err)
+        if hip_ops.gemm_proj_supported(M, N, K):
+            err = (hip_ops.gemm_proj(x, w).float() - ref).abs().max().item()
+            res["proj"] = (bench_cold(lambda w: hip_ops.gemm_proj(x, w), ws), err)
+            if sweep:
+                for sk in (1, 2, 3, 4, 6, 8, 12):
+                    if sk > max(1, K // 256):
+                        continue
+                    t = bench_cold(lambda w: hip_ops.gemm_proj(x, w, splitk=sk), ws)
+                    print(f"    {name} proj splitk={sk:2d}: {t:7.1f}us "
+                          f"({gb/t*1e6/1e3:5.2f} TB/s, grid={(N//64)*sk})")
+        line = f"{name:8s} K={K:6d} N={N:6d} W={gb*1000:7.1f}MB "
+        for kname, (t, err) in res.items():
+            ok = "OK" if err < 0.02 * scale + 1e-6 else f"FAIL({err:.3f})"
+            line += f" {kname} {t:7.1f}us ({gb/t*1e6/1e3:5.2f} TB/s) {ok} "
+            tot[kname] = tot.get(kname, 0.0) + t
+        print(line)
+    print("layer-sum:", {k: round(v, 1) for k, v in tot.items() if v})
+
 
 if __name__ == "__main__":
     main()

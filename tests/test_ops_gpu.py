@@ -218,3 +218,30 @@ def test_attn_prefill_mfma_gpu(dev, t, hq, hk, d, cau):
     ref = R.attn_prefill(q.unsqueeze(0).cpu().float(), kk.cpu().float(),
                          vv.cpu().float(), scale, causal_offset=cau)[0]
     assert _mx(out.cpu(), ref) < 0.03
+
+
+@pytest.mark.parametrize("m,n,k", [
+    (256, 5120, 3072), (256, 3072, 8192), (128, 3072, 3072),
+    (200, 16384, 3072),       # ragged M
+    (512, 3072, 3072), (65, 128, 64),
+])
+def test_gemm_proj_gpu(dev, m, n, k):
+    """Projection GEMM (M in (64,512]) vs fp32 matmul of the same operands."""
+    torch.manual_seed(11)
+    from gridllm_amd.ops import hip_ops
+    x = (torch.randn(m, k, device=dev) * 0.1).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) * 0.1).to(torch.bfloat16)
+    out = hip_ops.gemm_proj(x, w)
+    ref = x.float() @ w.float().t()
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.02 * ref.abs().max().item() + 1e-6, err
+
+
+def test_gemm_proj_splitk_consistent(dev):
+    torch.manual_seed(12)
+    from gridllm_amd.ops import hip_ops
+    x = (torch.randn(256, 8192, device=dev) * 0.1).to(torch.bfloat16)
+    w = (torch.randn(1024, 8192, device=dev) * 0.1).to(torch.bfloat16)
+    o1 = hip_ops.gemm_proj(x, w, splitk=1)
+    o8 = hip_ops.gemm_proj(x, w, splitk=8)
+    assert (o1.float() - o8.float()).abs().max().item() < 0.05
