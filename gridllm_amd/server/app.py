@@ -37,6 +37,15 @@ def create_app(scheduler: JobScheduler, registry: WorkerRegistry,
         mount_rate_limiter(app, cfg.rate_limit_max_requests,
                            cfg.rate_limit_window_ms)
 
+    from gridllm_amd.server import metrics as metrics_mod
+    metrics_mod.init_metrics(scheduler, registry)
+
+    @app.get("/metrics")
+    async def metrics():
+        from fastapi import Response
+        body, ctype = metrics_mod.render()
+        return Response(content=body, media_type=ctype)
+
     ollama_r = ollama_routes.build_router(scheduler, registry)
     app.include_router(ollama_r, prefix="/ollama")  # reference mount point
     app.include_router(ollama_r)                    # drop-in Ollama compat

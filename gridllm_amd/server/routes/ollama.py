@@ -86,9 +86,15 @@ def _stream_ndjson(scheduler: JobScheduler, req: InferenceRequest,
     """Shared NDJSON streaming plumbing for generate/chat."""
 
     async def gen():
+        from gridllm_amd.server import metrics
         q: asyncio.Queue = asyncio.Queue()
+        t0 = time.monotonic()
+        first = [True]
 
         def on_chunk(c: dict) -> None:
+            if first[0]:
+                first[0] = False
+                metrics.observe_ttft(time.monotonic() - t0)
             q.put_nowait(("chunk", c))
 
         async def run():
@@ -112,6 +118,7 @@ def _stream_ndjson(scheduler: JobScheduler, req: InferenceRequest,
                         yield json.dumps({"error": item["error"],
                                           "done": True}) + "\n"
                     else:
+                        metrics.add_tokens(item.get("eval_count"))
                         yield json.dumps(to_final(item)) + "\n"
                     break
                 else:

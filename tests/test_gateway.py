@@ -279,3 +279,15 @@ def test_rate_limiter_http_429():
         r3 = c.get("/")
         assert r3.status_code == 429
         assert c.get("/health").status_code == 200   # health exempt
+
+
+def test_metrics_endpoint(client):
+    client.post("/api/generate", json={
+        "model": "tiny", "prompt": "metrics", "stream": False,
+        "options": {"temperature": 0.0, "num_predict": 3}})
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    assert "gridllm_jobs_completed_total" in body
+    assert "gridllm_workers" in body
+    assert "gridllm_ttft_seconds" in body
