@@ -85,6 +85,18 @@ class InferenceEngine:
                                 tp=self.tp, seed=self.cfg.seed,
                                 kv_slots=self.cfg.max_batch_size + 1,
                                 max_seq=self.cfg.max_seq_len)
+        if self.cfg.weights_path:
+            import glob as _glob
+            import os as _os
+            if _glob.glob(_os.path.join(self.cfg.weights_path,
+                                        "*.safetensors")):
+                n = self.model.load_safetensors(self.cfg.weights_path)
+                log.performance("weights loaded", tensors=n,
+                                path=self.cfg.weights_path)
+            else:
+                log.logger.warning(
+                    "weights_path %r has no .safetensors; random init",
+                    self.cfg.weights_path)
         self.load_duration = time.monotonic_ns() - t0
         self.spec = spec
         self.tokenizer = load_tokenizer(self.cfg.weights_path, spec.vocab)

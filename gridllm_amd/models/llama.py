@@ -156,6 +156,83 @@ class LlamaModel:
                 attn_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device),
                 mlp_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device)))
 
+    def load_safetensors(self, path: str) -> int:
+        """Load HF-layout llama weights (model.safetensors or sharded
+        model-*-of-*.safetensors in `path`) into the fused TP-sharded
+        layout. Each rank reads only its shard slice. Returns the number
+        of tensors consumed. (the engine-side analog of
+        checkpoint/resume — the reference had none, weights lived inside
+        Ollama.)"""
+        import glob as _glob
+        import os
+        from safetensors import safe_open
+
+        files = sorted(_glob.glob(os.path.join(path, "*.safetensors")))
+        if not files:
+            raise FileNotFoundError(f"no .safetensors under {path!r}")
+        tensors: dict[str, torch.Tensor] = {}
+        want = self._hf_key_set()
+        for f in files:
+            with safe_open(f, framework="pt") as sf:
+                for k in sf.keys():
+                    if k in want:
+                        tensors[k] = sf.get_tensor(k)
+        missing = want - set(tensors)
+        if missing:
+            raise KeyError(f"missing weights: {sorted(missing)[:5]} "
+                           f"(+{max(0, len(missing) - 5)} more)")
+        self._apply_hf(tensors)
+        return len(tensors)
+
+    def _hf_key_set(self) -> set:
+        s = self.spec
+        keys = {"model.embed_tokens.weight", "model.norm.weight"}
+        if not s.tie_embeddings:
+            keys.add("lm_head.weight")
+        for i in range(s.layers):
+            p = f"model.layers.{i}."
+            keys |= {p + "self_attn.q_proj.weight",
+                     p + "self_attn.k_proj.weight",
+                     p + "self_attn.v_proj.weight",
+                     p + "self_attn.o_proj.weight",
+                     p + "mlp.gate_proj.weight", p + "mlp.up_proj.weight",
+                     p + "mlp.down_proj.weight",
+                     p + "input_layernorm.weight",
+                     p + "post_attention_layernorm.weight"}
+        return keys
+
+    def _apply_hf(self, t: dict) -> None:
+        s, tp, d = self.spec, self.tp, self.spec.head_dim
+        dt, dev = self.dtype, self.device
+
+        def cv(x: torch.Tensor) -> torch.Tensor:
+            return x.to(device=dev, dtype=dt).contiguous()
+
+        self.embed = cv(t["model.embed_tokens.weight"])
+        self.final_norm = cv(t["model.norm.weight"])
+        self.lm_head = self.embed if s.tie_embeddings \
+            else cv(t["lm_head.weight"])
+        q0, _ = tp.shard(s.heads)
+        k0, _ = tp.shard(s.kv_heads)
+        f0, _ = tp.shard(s.intermediate)
+        for i, lw in enumerate(self.layers):
+            p = f"model.layers.{i}."
+            wq = t[p + "self_attn.q_proj.weight"]
+            wk = t[p + "self_attn.k_proj.weight"]
+            wv = t[p + "self_attn.v_proj.weight"]
+            lw.wqkv = cv(torch.cat([
+                wq[q0 * d:(q0 + self.hq_l) * d],
+                wk[k0 * d:(k0 + self.hkv_l) * d],
+                wv[k0 * d:(k0 + self.hkv_l) * d]]))
+            lw.wo = cv(t[p + "self_attn.o_proj.weight"]
+                       [:, q0 * d:(q0 + self.hq_l) * d])
+            lw.wgateup = cv(torch.cat([
+                t[p + "mlp.gate_proj.weight"][f0:f0 + self.f_l],
+                t[p + "mlp.up_proj.weight"][f0:f0 + self.f_l]]))
+            lw.wdown = cv(t[p + "mlp.down_proj.weight"][:, f0:f0 + self.f_l])
+            lw.attn_norm = cv(t[p + "input_layernorm.weight"])
+            lw.mlp_norm = cv(t[p + "post_attention_layernorm.weight"])
+
     def param_bytes(self) -> int:
         total = self.embed.numel() + self.final_norm.numel()
         if self.lm_head is not self.embed:

@@ -146,3 +146,58 @@ def test_row_timings_populated():
     assert r.prompt_eval_duration > 0
     assert r.eval_duration >= 0
     assert r.total_duration >= r.prompt_eval_duration
+
+
+def test_safetensors_weight_loading(tmp_path):
+    """HF-layout llama weights round-trip into the fused layout, and a
+    TP=2 load sees the same model as TP=1 (shard slicing)."""
+    import safetensors.torch as st
+    from gridllm_amd.models.llama import LlamaModel
+    from gridllm_amd.models.registry import get_spec
+    from gridllm_amd.parallel.tp import TPContext
+
+    spec = get_spec("tiny", 64)
+    d = spec.head_dim
+    gen = torch.Generator().manual_seed(7)
+
+    def rnd(*shape):
+        return torch.randn(*shape, generator=gen) * 0.05
+
+    weights = {"model.embed_tokens.weight": rnd(spec.vocab, spec.hidden),
+               "model.norm.weight": rnd(spec.hidden)}
+    for i in range(spec.layers):
+        p = f"model.layers.{i}."
+        weights[p + "self_attn.q_proj.weight"] = rnd(spec.heads * d, spec.hidden)
+        weights[p + "self_attn.k_proj.weight"] = rnd(spec.kv_heads * d, spec.hidden)
+        weights[p + "self_attn.v_proj.weight"] = rnd(spec.kv_heads * d, spec.hidden)
+        weights[p + "self_attn.o_proj.weight"] = rnd(spec.hidden, spec.heads * d)
+        weights[p + "mlp.gate_proj.weight"] = rnd(spec.intermediate, spec.hidden)
+        weights[p + "mlp.up_proj.weight"] = rnd(spec.intermediate, spec.hidden)
+        weights[p + "mlp.down_proj.weight"] = rnd(spec.hidden, spec.intermediate)
+        weights[p + "input_layernorm.weight"] = rnd(spec.hidden)
+        weights[p + "post_attention_layernorm.weight"] = rnd(spec.hidden)
+    st.save_file(weights, str(tmp_path / "model.safetensors"))
+
+    m = LlamaModel(spec, device="cpu", dtype="fp32", kv_slots=2, max_seq=64)
+    n = m.load_safetensors(str(tmp_path))
+    assert n == len(weights)
+    assert torch.equal(m.embed, weights["model.embed_tokens.weight"])
+    lw = m.layers[0]
+    assert torch.equal(lw.wqkv, torch.cat([
+        weights["model.layers.0.self_attn.q_proj.weight"],
+        weights["model.layers.0.self_attn.k_proj.weight"],
+        weights["model.layers.0.self_attn.v_proj.weight"]]))
+    assert torch.equal(lw.wdown,
+                       weights["model.layers.0.mlp.down_proj.weight"])
+
+    # TP=2 rank shards: concatenating both ranks' wgateup gate halves
+    # reconstructs the full gate matrix (no collectives during load)
+    shards = []
+    for r in (0, 1):
+        mr = LlamaModel(spec, device="cpu", dtype="fp32", kv_slots=2,
+                        max_seq=64, tp=TPContext(rank=r, size=2))
+        mr.load_safetensors(str(tmp_path))
+        f_l = spec.intermediate // 2
+        shards.append(mr.layers[0].wgateup[:f_l])
+    assert torch.equal(torch.cat(shards),
+                       weights["model.layers.0.mlp.gate_proj.weight"])
