@@ -391,22 +391,37 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         }
         // wave-local LDS visibility (no cross-wave traffic here)
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        // accumulate P·V: lane covers d = lane*2 + c*128
+        // accumulate P·V: lane covers d = lane*2 + c*128. 8-row batches
+        // with all 8 loads issued up front — the row-at-a-time form is a
+        // serial L2-latency chain (measured 3.2 TB/s; this is the fix).
         const int jmax = min(64, s1 - kb);
-        for (int j2 = 0; j2 < jmax; ++j2) {
-            const bf16_t* vrow = vbase + (long)(kb + j2) * D;
+        for (int jb = 0; jb < jmax; jb += 8) {
+            u32 vv[8][NC];
 #pragma unroll
-            for (int c = 0; c < NC; ++c) {
-                const int d = lane * 2 + c * 128;
-                if (d < D) {
-                    u32 vv = *(const u32*)(vrow + d);
-                    float v0 = __uint_as_float((vv & 0xffffu) << 16);
-                    float v1 = __uint_as_float(vv & 0xffff0000u);
+            for (int u = 0; u < 8; ++u) {
+                const long vrow = (long)(kb + min(jb + u, jmax - 1)) * D;
 #pragma unroll
-                    for (int g = 0; g < G; ++g) {
-                        float p = red[wid][g * 64 + j2];
-                        acc[g][c][0] = fmaf(p, v0, acc[g][c][0]);
-                        acc[g][c][1] = fmaf(p, v1, acc[g][c][1]);
+                for (int c = 0; c < NC; ++c) {
+                    const int d = lane * 2 + c * 128;
+                    vv[u][c] = (d < D)
+                        ? *(const u32*)(vbase + vrow + d) : 0u;
+                }
+            }
+#pragma unroll
+            for (int u = 0; u < 8; ++u) {
+                if (jb + u >= jmax) break;
+#pragma unroll
+                for (int c = 0; c < NC; ++c) {
+                    const int d = lane * 2 + c * 128;
+                    if (d < D) {
+                        float v0 = __uint_as_float((vv[u][c] & 0xffffu) << 16);
+                        float v1 = __uint_as_float(vv[u][c] & 0xffff0000u);
+#pragma unroll
+                        for (int g = 0; g < G; ++g) {
+                            float p = red[wid][g * 64 + jb + u];
+                            acc[g][c][0] = fmaf(p, v0, acc[g][c][0]);
+                            acc[g][c][1] = fmaf(p, v1, acc[g][c][1]);
+                        }
                     }
                 }
             }
@@ -623,7 +638,15 @@ k_gemm_skinny(bf16_t* __restrict__ out, float* __restrict__ ws,
 // Constraints: N % 128 == 0 (64 at MT=4), K % 64 == 0, kslice % 64 == 0.
 // ===========================================================================
 
-template <int MT, int BN, int NB>   // MT: 16-row x-tiles/wave; NB: LDS bufs
+DEV_INLINE float silu_mul(float g, float u) {
+    return g * (1.f / (1.f + __expf(-g))) * u;
+}
+
+// EPI: 0 = plain store; 1 = fused SwiGLU epilogue — W rows are interleaved
+// in 128-row groups [64 gate | 64 up] so each BN=128 block holds matching
+// gate/up column pairs and writes out[M, N/2] = silu(g)*u directly (the
+// separate k_swiglu pass and its gateup HBM round trip disappear).
+template <int MT, int BN, int NB, int EPI = 0>
 __global__ void __launch_bounds__(512)
 k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
             const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
@@ -739,6 +762,21 @@ k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
 
     // ---- store: lane holds D[i][j], i = (l>>4)*4 + r, j = fr ------------
     const int orow0 = (lane >> 4) * 4;
+    if (EPI == 1 && splitk == 1) {
+#pragma unroll
+        for (int t = 0; t < MT; ++t)
+#pragma unroll
+            for (int c = 0; c < BN / 32; ++c)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int m = (wid * MT + t) * 16 + orow0 + r;
+                    if (m < M)
+                        out[(long)m * (N >> 1) + (n0 >> 1) + c * 16 + fr] =
+                            f2bf(silu_mul(acc[t][c][r],
+                                          acc[t][c + BN / 32][r]));
+                }
+        return;
+    }
 #pragma unroll
     for (int t = 0; t < MT; ++t) {
 #pragma unroll
@@ -758,13 +796,63 @@ k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
     }
 }
 
-template <int MT, int BN, int NB>
+template <int MT, int BN, int NB, int EPI = 0>
 static void launch_gemm_proj(hipStream_t st, bf16_t* out, float* ws,
                              const bf16_t* x, const bf16_t* w, int M, int N,
                              int K, int splitk, int kslice) {
     dim3 grid((unsigned)((N / BN) * splitk)), blk(512);
-    k_gemm_proj<MT, BN, NB><<<grid, blk, 0, st>>>(out, ws, x, w, M, N, K,
-                                                  splitk, kslice);
+    k_gemm_proj<MT, BN, NB, EPI><<<grid, blk, 0, st>>>(out, ws, x, w, M, N,
+                                                       K, splitk, kslice);
+}
+
+// Combine split-K slabs of an INTERLEAVED gate/up projection with the
+// SwiGLU epilogue applied during the combine: out[M, N/2] bf16.
+__global__ void __launch_bounds__(256)
+k_gemm_combine_swiglu(bf16_t* __restrict__ out, const float* __restrict__ ws,
+                      int M, int N, int splitk) {
+    const int no = N >> 1;
+    const long mn = (long)M * N;
+    const long total = (long)M * (no >> 1);       // u32 (2-col) units
+    for (long i = blockIdx.x * 256L + threadIdx.x; i < total;
+         i += gridDim.x * 256L) {
+        const long m = i / (no >> 1);
+        const int j = (int)(i % (no >> 1)) * 2;   // even out col
+        const int grp = j >> 6, win = j & 63;
+        const long gidx = m * N + grp * 128 + win;
+        float g0 = 0.f, g1 = 0.f, u0 = 0.f, u1 = 0.f;
+        for (int k = 0; k < splitk; ++k) {
+            const float* p = ws + (long)k * mn + gidx;
+            g0 += p[0]; g1 += p[1]; u0 += p[64]; u1 += p[65];
+        }
+        u32 lo = (u32)f2bf(silu_mul(g0, u0)) |
+                 ((u32)f2bf(silu_mul(g1, u1)) << 16);
+        *(u32*)(out + m * (long)no + j) = lo;
+    }
+}
+
+// SwiGLU over an INTERLEAVED gate/up activation (128-col groups
+// [64 gate | 64 up]) — the library-GEMM fallback path for interleaved
+// weights: out[n, F] from gu[n, 2F].
+__global__ void __launch_bounds__(256)
+k_swiglu_il(bf16_t* __restrict__ out, const bf16_t* __restrict__ gu,
+            long n, int F) {
+    const int fc = F >> 3;                        // 8-col units per out row
+    const long total = n * fc;
+    for (long idx = blockIdx.x * 256L + threadIdx.x; idx < total;
+         idx += gridDim.x * 256L) {
+        const long r = idx / fc;
+        const int j = (int)(idx % fc) * 8;
+        const int grp = j >> 6, win = j & 63;
+        const bf16_t* base = gu + r * 2L * F + grp * 128 + win;
+        uint4 gv = *(const uint4*)base;
+        uint4 uv = *(const uint4*)(base + 64);
+        float g[8], u[8];
+        unpack8(gv, g);
+        unpack8(uv, u);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) g[i] = silu_mul(g[i], u[i]);
+        ((uint4*)(out + r * (long)F))[j >> 3] = pack8(g);
+    }
 }
 
 // Combine split-K f32 partials -> bf16 out. Grid-stride over M*N/8 uint4s.
@@ -1120,13 +1208,31 @@ int gl_gemm_skinny(void* stream, void* out, void* ws, const void* x,
 
 int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
                  const void* w, int M, int N, int K, int splitk,
-                 int kslice) {
+                 int kslice, int fuse_swiglu) {
     if (K % 64 != 0 || kslice % 64 != 0) return -100;
     if (M > 512) return -104;
     hipStream_t st = (hipStream_t)stream;
     const int mt = (M + 127) / 128;     // 16-row x-tiles per wave (8 waves)
     if (mt <= 2 && N % 128 != 0) return -100;
     if (mt > 2 && N % 64 != 0) return -100;
+    if (fuse_swiglu && mt > 2) return -105;   // needs BN=128 gate/up pairs
+    if (fuse_swiglu) {
+        if (mt <= 1)
+            launch_gemm_proj<1, 128, 3, 1>(st, (bf16_t*)out, (float*)ws,
+                (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk, kslice);
+        else
+            launch_gemm_proj<2, 128, 3, 1>(st, (bf16_t*)out, (float*)ws,
+                (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk, kslice);
+        LAUNCH_CHECK();
+        if (splitk > 1) {
+            long chunks = ((long)M * (N >> 2) + 255) / 256;
+            int grid = (int)min(chunks, (long)2048);
+            k_gemm_combine_swiglu<<<dim3(grid), dim3(256), 0, st>>>(
+                (bf16_t*)out, (const float*)ws, M, N, splitk);
+            LAUNCH_CHECK();
+        }
+        return 0;
+    }
     if (mt <= 1)
         launch_gemm_proj<1, 128, 3>(st, (bf16_t*)out, (float*)ws,
             (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk, kslice);
@@ -1145,6 +1251,17 @@ int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
             (bf16_t*)out, (const float*)ws, mn, splitk);
         LAUNCH_CHECK();
     }
+    return 0;
+}
+
+int gl_swiglu_il(void* stream, void* out, const void* gu, long n, int f) {
+    if (f % 64 != 0) return -100;
+    hipStream_t st = (hipStream_t)stream;
+    long total = n * (f >> 3);
+    int grid = (int)min((total + 255) / 256, (long)4096);
+    k_swiglu_il<<<dim3(grid), dim3(256), 0, st>>>(
+        (bf16_t*)out, (const bf16_t*)gu, n, f);
+    LAUNCH_CHECK();
     return 0;
 }
 

@@ -97,6 +97,8 @@ class InferenceEngine:
                 log.logger.warning(
                     "weights_path %r has no .safetensors; random init",
                     self.cfg.weights_path)
+        if self.is_gpu and dtype == "bf16":
+            self.model.interleave_gateup()   # fused-SwiGLU projection layout
         self.load_duration = time.monotonic_ns() - t0
         self.spec = spec
         self.tokenizer = load_tokenizer(self.cfg.weights_path, spec.vocab)

@@ -25,6 +25,7 @@ import torch
 import torch.nn.functional as F  # noqa: F401
 
 from gridllm_amd.ops.autotune import linear as _linear
+from gridllm_amd.ops.autotune import linear_gateup as _linear_gateup
 
 from gridllm_amd import ops
 from gridllm_amd.models.registry import ModelSpec
@@ -100,6 +101,7 @@ class LlamaModel:
         self.f_l = spec.intermediate // self.tp.size
         self.scale = 1.0 / math.sqrt(spec.head_dim)
 
+        self.gateup_il = False
         self._init_weights(seed)
         self.cos, self.sin = ops.rope_tables(
             spec.head_dim, self.max_seq, spec.rope_theta, device=self.device)
@@ -156,6 +158,28 @@ class LlamaModel:
                 attn_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device),
                 mlp_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device)))
 
+    def interleave_gateup(self) -> None:
+        """Reorder each wgateup's rows into 128-row groups [64 gate |
+        64 up] so the fused-SwiGLU projection kernel (ops.gemm_proj with
+        fuse_swiglu) finds matching gate/up column pairs in every BN=128
+        block. GPU-only layout change; forward dispatches accordingly."""
+        if self.gateup_il or self.f_l % 64 != 0:
+            return
+        f = self.f_l
+        g = torch.arange(f)
+        perm = torch.empty(2 * f, dtype=torch.int64)
+        perm[(g // 64) * 128 + g % 64] = g          # gate rows
+        perm[(g // 64) * 128 + g % 64 + 64] = f + g  # up rows
+        perm = perm.to(self.device)
+        for lw in self.layers:
+            lw.wgateup = lw.wgateup[perm].contiguous()
+        self.gateup_il = True
+
+    def _mlp_act(self, normed: torch.Tensor, lw: LayerWeights) -> torch.Tensor:
+        if self.gateup_il:
+            return _linear_gateup(normed, lw.wgateup)
+        return ops.swiglu(_linear(normed, lw.wgateup))
+
     def load_safetensors(self, path: str) -> int:
         """Load HF-layout llama weights (model.safetensors or sharded
         model-*-of-*.safetensors in `path`) into the fused TP-sharded
@@ -167,6 +191,7 @@ class LlamaModel:
         import os
         from safetensors import safe_open
 
+        self.gateup_il = False          # loads land in the plain layout
         files = sorted(_glob.glob(os.path.join(path, "*.safetensors")))
         if not files:
             raise FileNotFoundError(f"no .safetensors under {path!r}")
@@ -273,8 +298,7 @@ class LlamaModel:
             self.tp.all_reduce(attn_out)
             normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,
                                         residual=resid)
-            gateup = _linear(normed, lw.wgateup)
-            act = ops.swiglu(gateup)
+            act = self._mlp_act(normed, lw)
             mlp_out = _linear(act, lw.wdown)
             self.tp.all_reduce(mlp_out)
             next_norm = (self.layers[li + 1].attn_norm
@@ -312,8 +336,7 @@ class LlamaModel:
             self.tp.all_reduce(attn_out)
             normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,
                                         residual=resid)
-            gateup = _linear(normed, lw.wgateup)
-            act = ops.swiglu(gateup)
+            act = self._mlp_act(normed, lw)
             mlp_out = _linear(act, lw.wdown)
             self.tp.all_reduce(mlp_out)
             next_norm = (self.layers[li + 1].attn_norm

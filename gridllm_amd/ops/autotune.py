@@ -23,6 +23,9 @@ log = get_logger("autotune")
 
 # (m, n, k) -> ("blas", 0) | ("ours", splitk)
 _CHOICE: dict[tuple[int, int, int], tuple[str, int]] = {}
+# gate/up projection with fused SwiGLU (n = 2F interleaved rows):
+# (m, n, k) -> ("blas", 0) | ("proj", splitk)
+_CHOICE_GU: dict[tuple[int, int, int], tuple[str, int]] = {}
 _SPLITK_CANDIDATES = (0, 2, 4, 6, 8, 12, 16, 24, 32)
 
 
@@ -84,6 +87,46 @@ def tune_shape(m: int, n: int, k: int, device) -> tuple[str, int]:
     return best
 
 
+def tune_gateup(m: int, n: int, k: int, device) -> tuple[str, int]:
+    """Tune the interleaved gate/up projection (n = 2F rows): library GEMM
+    + interleaved-SwiGLU pass vs the fused-epilogue projection kernel."""
+    key = (m, n, k)
+    if key in _CHOICE_GU:
+        return _CHOICE_GU[key]
+    x = torch.randn(m, k, device=device).to(torch.bfloat16) * 0.1
+    wbytes = n * k * 2
+    ncopies = min(6, max(2, (320 << 20) // max(1, wbytes)))
+    ws = [torch.randn(n, k, device=device).to(torch.bfloat16) * 0.1
+          for _ in range(ncopies)]
+    best = ("blas", 0)
+    t_best = _time_fn(lambda w: hip_ops.swiglu_il(F.linear(x, w)), ws)
+    if m <= 256 and n % 128 == 0 and hip_ops.gemm_proj_supported(m, n, k):
+        rule = max(1, min(k // 256, 256 // max(1, n // 128)))
+        for sk in sorted({1, rule, rule * 2, max(1, rule // 2)}):
+            if sk > max(1, k // 64):
+                continue
+            try:
+                t = _time_fn(lambda w: hip_ops.gemm_proj(
+                    x, w, splitk=sk, fuse_swiglu=True), ws)
+            except RuntimeError:
+                continue
+            if t < t_best * 0.97:
+                t_best = t
+                best = ("proj", sk)
+    del ws
+    _CHOICE_GU[key] = best
+    return best
+
+
+def linear_gateup(x: torch.Tensor, w_il: torch.Tensor) -> torch.Tensor:
+    """Fused gate/up projection + SwiGLU on an INTERLEAVED weight:
+    returns [M, F] activated. Dispatches through the tuned table."""
+    c = _CHOICE_GU.get((x.shape[0], w_il.shape[0], w_il.shape[1]))
+    if c is not None and c[0] == "proj":
+        return hip_ops.gemm_proj(x, w_il, splitk=c[1], fuse_swiglu=True)
+    return hip_ops.swiglu_il(F.linear(x, w_il))
+
+
 def choice(m: int, n: int, k: int) -> Optional[tuple[str, int]]:
     return _CHOICE.get((m, n, k))
 
@@ -108,14 +151,18 @@ def tune_model_shapes(model, buckets) -> None:
         ((s.heads // model.tp.size + 2 * (s.kv_heads // model.tp.size)) * d,
          s.hidden),                               # qkv
         (s.hidden, (s.heads // model.tp.size) * d),   # o
-        (2 * s.intermediate // model.tp.size, s.hidden),  # gateup
         (s.hidden, s.intermediate // model.tp.size),      # down
         (s.vocab, s.hidden),                      # lm_head
     ]
+    gu_shape = (2 * s.intermediate // model.tp.size, s.hidden)
     t0 = time.monotonic()
     for m in buckets:
         for n, k in shapes:
             tune_shape(m, n, k, model.device)
+        if getattr(model, "gateup_il", False):
+            tune_gateup(m, gu_shape[0], gu_shape[1], model.device)
+        else:
+            tune_shape(m, gu_shape[0], gu_shape[1], model.device)
     picked = sum(1 for v in _CHOICE.values() if v[0] == "ours")
     log.performance("gemm autotune done", shapes=len(_CHOICE),
                     ours=picked, ms=int((time.monotonic() - t0) * 1000))
