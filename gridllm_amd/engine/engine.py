@@ -147,8 +147,12 @@ class InferenceEngine:
         with torch.cuda.stream(s):
             for _ in range(2):
                 self.model.forward_decode(tokens, slots)
+                # reset BETWEEN iterations: each warmup advances the scratch
+                # slot by `bucket` positions, and at bucket >= max_seq the
+                # second pass would write past the slot's KV region (the
+                # b=1024 x max_seq=1024 GPU fault)
+                self.model.cache.seq_lens[self._scratch_slot] = 0
         torch.cuda.current_stream().wait_stream(s)
-        self.model.cache.seq_lens[self._scratch_slot] = 0
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             out = self.model.forward_decode(tokens, slots)

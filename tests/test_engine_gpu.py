@@ -133,3 +133,17 @@ def test_continuous_batcher_gpu():
         assert solo.token_ids == r_short.token_ids
     finally:
         b.shutdown()
+
+
+def test_graph_warmup_bucket_at_max_seq():
+    """Regression: graph warmup runs forward_decode twice on the scratch
+    slot; without a reset between passes, bucket >= max_seq pushed the
+    second pass's KV write past the slot's region (b1024 GPU fault)."""
+    eng = InferenceEngine(EngineConfig(
+        model="llama3.2:1b", device="cuda:0", dtype="bf16",
+        max_seq_len=16, max_batch_size=16, use_hipgraph=True,
+        autotune=False))
+    reqs = [("a", {"temperature": 0.0, "num_predict": 2})] * 16
+    results = eng.generate_batch(reqs)
+    assert all(r.eval_count >= 1 for r in results)
+    assert eng.model.cache.seq_lens[eng._scratch_slot].item() == 0
