@@ -333,18 +333,26 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     }
     __syncthreads();
 
-    // per-wave online softmax state
+    // per-wave online softmax state.
+    // P.V layout: the wave reads RPI = 64/(D/8) V-rows per b128-load
+    // iteration — lane covers row-group rg = lane/DV and d = (lane%DV)*8
+    // + 0..7 — so a 64-key block takes 64/RPI wide-load iterations instead
+    // of 64 row-at-a-time u32 iterations (which measured as a serial
+    // L2-latency chain, 3.2 TB/s). Each lane holds PARTIAL sums over its
+    // row group; one shfl_xor ladder at the end totals them.
     float m[G], l[G];
-    // each lane covers d = lane*2 + c*128 (2 f32 per c); NC = D/128 rounded up
-    constexpr int NC = (D + 127) / 128;
-    float acc[G][NC][2];
+    constexpr int DV = D / 8;          // lanes per V row
+    constexpr int RPI = 64 / DV;       // V rows per iteration
+    float facc[G][8];
 #pragma unroll
     for (int g = 0; g < G; ++g) {
         m[g] = -INFINITY;
         l[g] = 0.f;
 #pragma unroll
-        for (int c = 0; c < NC; ++c) acc[g][c][0] = acc[g][c][1] = 0.f;
+        for (int i = 0; i < 8; ++i) facc[g][i] = 0.f;
     }
+    const int rg = lane / DV;
+    const int d8 = (lane % DV) * 8;
 
     const bf16_t* kbase = kc + ((long)bc * Hk + hk) * Smax * D;
     const bf16_t* vbase = vc + ((long)bc * Hk + hk) * Smax * D;
@@ -382,51 +390,40 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
             float p = (sv == -INFINITY) ? 0.f : __expf(sv - mn);
             l[g] = l[g] * alpha + wave_reduce_sum(p);
 #pragma unroll
-            for (int c = 0; c < NC; ++c) {
-                acc[g][c][0] *= alpha;
-                acc[g][c][1] *= alpha;
-            }
+            for (int i = 0; i < 8; ++i) facc[g][i] *= alpha;
             m[g] = mn;
             red[wid][g * 64 + lane] = p;      // per-wave scratch: p for 64 keys
         }
         // wave-local LDS visibility (no cross-wave traffic here)
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        // accumulate P·V: lane covers d = lane*2 + c*128. 8-row batches
-        // with all 8 loads issued up front — the row-at-a-time form is a
-        // serial L2-latency chain (measured 3.2 TB/s; this is the fix).
+        // accumulate P·V: RPI rows per iteration, one b128 load per lane
         const int jmax = min(64, s1 - kb);
-        for (int jb = 0; jb < jmax; jb += 8) {
-            u32 vv[8][NC];
+#pragma unroll 4
+        for (int jb = 0; jb < jmax; jb += RPI) {
+            const int j2 = jb + rg;
+            const uint4 vv = *(const uint4*)(
+                vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
+            float vf[8];
+            unpack8(vv, vf);
 #pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                const long vrow = (long)(kb + min(jb + u, jmax - 1)) * D;
+            for (int g = 0; g < G; ++g) {
+                const float p = (j2 < jmax) ? red[wid][g * 64 + j2] : 0.f;
 #pragma unroll
-                for (int c = 0; c < NC; ++c) {
-                    const int d = lane * 2 + c * 128;
-                    vv[u][c] = (d < D)
-                        ? *(const u32*)(vbase + vrow + d) : 0u;
-                }
-            }
-#pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                if (jb + u >= jmax) break;
-#pragma unroll
-                for (int c = 0; c < NC; ++c) {
-                    const int d = lane * 2 + c * 128;
-                    if (d < D) {
-                        float v0 = __uint_as_float((vv[u][c] & 0xffffu) << 16);
-                        float v1 = __uint_as_float(vv[u][c] & 0xffff0000u);
-#pragma unroll
-                        for (int g = 0; g < G; ++g) {
-                            float p = red[wid][g * 64 + jb + u];
-                            acc[g][c][0] = fmaf(p, v0, acc[g][c][0]);
-                            acc[g][c][1] = fmaf(p, v1, acc[g][c][1]);
-                        }
-                    }
-                }
+                for (int i = 0; i < 8; ++i)
+                    facc[g][i] = fmaf(p, vf[i], facc[g][i]);
             }
         }
     }
+
+    // total each lane's row-group partials: after the ladder every lane
+    // holds the full sum for its d = d8 + 0..7 slice
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int off = 32; off >= DV; off >>= 1)
+                facc[g][i] += __shfl_xor(facc[g][i], off, WAVE);
 
     // cross-wave combine in LDS
 #pragma unroll
@@ -440,14 +437,10 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         float M = fmaxf(fmaxf(wml[0][g][0], wml[1][g][0]),
                         fmaxf(wml[2][g][0], wml[3][g][0]));
         float f = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - M);
+        if (lane < DV)
 #pragma unroll
-        for (int c = 0; c < NC; ++c) {
-            const int d = lane * 2 + c * 128;
-            if (d < D) {
-                red[wid][g * D + d] = acc[g][c][0] * f;
-                red[wid][g * D + d + 1] = acc[g][c][1] * f;
-            }
-        }
+            for (int i = 0; i < 8; ++i)
+                red[wid][g * D + d8 + i] = facc[g][i] * f;
         if (lane == g % 64) wml[wid][g][1] = l[g] * f;  // scaled l (one lane writes)
     }
     __syncthreads();

@@ -52,6 +52,10 @@ _reg(ModelSpec("llama3.2:1b", hidden=2048, layers=16, heads=32, kv_heads=8,
 _reg(ModelSpec("llama3.2:3b", hidden=3072, layers=28, heads=24, kv_heads=8,
                head_dim=128, intermediate=8192, vocab=128256),
      "llama3.2", "llama3.2:latest", "llama3.2-3b")
+_reg(ModelSpec("llama3:8b", hidden=4096, layers=32, heads=32, kv_heads=8,
+               head_dim=128, intermediate=14336, vocab=128256,
+               tie_embeddings=False),
+     "llama3.1:8b", "llama3-8b", "llama3", "llama3:latest")
 _reg(ModelSpec("llama3:70b", hidden=8192, layers=80, heads=64, kv_heads=8,
                head_dim=128, intermediate=28672, vocab=128256,
                tie_embeddings=False),

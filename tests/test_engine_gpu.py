@@ -117,11 +117,16 @@ def test_continuous_batcher_gpu():
         autotune=False))
     b = Batcher(eng)
     try:
-        started = threading.Event()
+        import time as _t
+        free0 = eng.model.cache.n_free
         long_fut = b.submit_generate(
-            "the quick brown fox", {"temperature": 0.0, "num_predict": 64},
-            on_token=lambda piece, done: piece and started.set())
-        assert started.wait(timeout=60)
+            "the quick brown fox", {"temperature": 0.0, "num_predict": 64})
+        # admitted = a KV slot is held (text pieces may be empty with
+        # random-weight tokens, so don't wait on on_token)
+        deadline = _t.monotonic() + 60
+        while eng.model.cache.n_free == free0 and _t.monotonic() < deadline:
+            _t.sleep(0.05)
+        assert eng.model.cache.n_free < free0, "long row never admitted"
         short_fut = b.submit_generate(
             "hi", {"temperature": 0.0, "num_predict": 4})
         r_short = short_fut.result(timeout=120)
