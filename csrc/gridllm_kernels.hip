@@ -299,7 +299,7 @@ k_kv_append(bf16_t* __restrict__ kc, bf16_t* __restrict__ vc,
 #define MAXG 8
 
 template <int D, int G>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(256, 3)
 k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
               const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
               const int* __restrict__ seqlen, const int* __restrict__ slot_map,
@@ -398,7 +398,6 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         // accumulate P·V: RPI rows per iteration, one b128 load per lane
         const int jmax = min(64, s1 - kb);
-#pragma unroll 4
         for (int jb = 0; jb < jmax; jb += RPI) {
             const int j2 = jb + rg;
             const uint4 vv = *(const uint4*)(
@@ -462,6 +461,149 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
             if (d == 0) {
                 wsp[D] = M;
                 wsp[D + 1] = L;
+            }
+        }
+    }
+}
+
+// ===========================================================================
+// GQA decode attention, ONE WAVE per (b, hk[, split]) — the large-batch
+// variant. At serving batches B*Hk >= ~2048 the chip holds every block
+// resident (16 waves/CU), so per-wave independence beats the 4-wave
+// cooperative block: no q-staging barrier, no cross-wave LDS combine, no
+// idle waves when seq/64 < 4. Same math as k_attn_decode; V phase reads
+// RPI rows per b128 iteration with an end shfl ladder.
+// ===========================================================================
+
+template <int D, int G>
+__global__ void __launch_bounds__(64, 3)
+k_attn_decode1(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
+               const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
+               const int* __restrict__ seqlen,
+               const int* __restrict__ slot_map,
+               float* __restrict__ ws,
+               int B, int Hk, int Smax, int nsplit, float scale) {
+    const int split = blockIdx.x % nsplit;
+    const int hk = (blockIdx.x / nsplit) % Hk;
+    const int b = blockIdx.x / (nsplit * Hk);
+    const int bc = slot_map ? slot_map[b] : b;
+    const int lane = threadIdx.x;
+
+    __shared__ float q_lds[G * D];
+    __shared__ float p_lds[G * 64];
+
+    const int seq = seqlen[b];
+    const int chunks = (seq + 63) >> 6;
+    const int per = (chunks + nsplit - 1) / nsplit;
+    const int s0 = (split * per) << 6;
+    const int s1 = min(seq, (split * per + per) << 6);
+
+    const int Hq = Hk * G;
+    for (int i = lane; i < G * D; i += 64)
+        q_lds[i] = bf2f(q[((long)b * Hq + hk * (long)G + i / D) * D
+                          + i % D]) * scale;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    float m[G], l[G];
+    constexpr int DV = D / 8;
+    constexpr int RPI = 64 / DV;
+    float facc[G][8];
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+        m[g] = -INFINITY;
+        l[g] = 0.f;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) facc[g][i] = 0.f;
+    }
+    const int rg = lane / DV;
+    const int d8 = (lane % DV) * 8;
+
+    const bf16_t* kbase = kc + ((long)bc * Hk + hk) * (long)Smax * D;
+    const bf16_t* vbase = vc + ((long)bc * Hk + hk) * (long)Smax * D;
+
+    for (int kb = s0; kb < s1; kb += 64) {
+        const int j = kb + lane;
+        const bool valid = j < s1;
+        float s[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) s[g] = 0.f;
+        if (valid) {
+            const bf16_t* krow = kbase + (long)j * D;
+#pragma unroll
+            for (int dd = 0; dd < D; dd += 8) {
+                uint4 kv = *(const uint4*)(krow + dd);
+                float kf[8];
+                unpack8(kv, kf);
+#pragma unroll
+                for (int g = 0; g < G; ++g) {
+                    float a = 0.f;
+#pragma unroll
+                    for (int i = 0; i < 8; ++i)
+                        a += kf[i] * q_lds[g * D + dd + i];
+                    s[g] += a;
+                }
+            }
+        }
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+            float sv = valid ? s[g] : -INFINITY;
+            float bmax = wave_reduce_max(sv);
+            float mn = fmaxf(m[g], bmax);
+            float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
+            float p = (sv == -INFINITY) ? 0.f : __expf(sv - mn);
+            l[g] = l[g] * alpha + wave_reduce_sum(p);
+#pragma unroll
+            for (int i = 0; i < 8; ++i) facc[g][i] *= alpha;
+            m[g] = mn;
+            p_lds[g * 64 + lane] = p;
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        // P.V: RPI rows per b128 iteration; p read from same-wave LDS
+        // (an LDS hop, not shfl: it bounds how far the scheduler can
+        // interleave the K and V phases — shfl spilled 1.5 KB here)
+        const int jmax = min(64, s1 - kb);
+        for (int jb = 0; jb < jmax; jb += RPI) {
+            const int j2 = jb + rg;
+            const uint4 vv = *(const uint4*)(
+                vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
+            float vf[8];
+            unpack8(vv, vf);
+#pragma unroll
+            for (int g = 0; g < G; ++g) {
+                const float p = (j2 < jmax) ? p_lds[g * 64 + j2] : 0.f;
+#pragma unroll
+                for (int i = 0; i < 8; ++i)
+                    facc[g][i] = fmaf(p, vf[i], facc[g][i]);
+            }
+        }
+    }
+
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int off = 32; off >= DV; off >>= 1)
+                facc[g][i] += __shfl_xor(facc[g][i], off, WAVE);
+
+    if (lane < DV) {
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+            if (nsplit == 1) {
+                const float L = l[g];
+#pragma unroll
+                for (int i = 0; i < 8; ++i)
+                    out[((long)b * Hq + hk * (long)G + g) * D + d8 + i] =
+                        f2bf(L > 0.f ? facc[g][i] / L : 0.f);
+            } else {
+                float* wsp = ws + ((((long)b * Hk + hk) * G + g) * nsplit
+                                   + split) * (D + 2);
+#pragma unroll
+                for (int i = 0; i < 8; ++i) wsp[d8 + i] = facc[g][i];
+                if (lane == 0) {
+                    wsp[D] = m[g];
+                    wsp[D + 1] = l[g];
+                }
             }
         }
     }
@@ -1068,12 +1210,16 @@ static int launch_attn_decode(hipStream_t st, bf16_t* out, const bf16_t* q,
                               const int* seqlen, const int* slot_map,
                               float* ws, int B, int Hk, int G, int Smax,
                               int nsplit, float scale) {
-    dim3 grid((unsigned)(B * Hk * nsplit)), blk(256);
+    dim3 grid((unsigned)(B * Hk * nsplit));
+    const bool onewave = (B * Hk * nsplit) >= 2048;   // grid fills the chip
     switch (G) {   // compile-time G keeps softmax state in registers
 #define CASE_G(GG) \
-        case GG: k_attn_decode<D, GG><<<grid, blk, 0, st>>>( \
-            out, q, kc, vc, seqlen, slot_map, ws, B, Hk, Smax, nsplit, \
-            scale); break;
+        case GG: if (onewave) k_attn_decode1<D, GG><<<grid, dim3(64), 0, st>>>( \
+                out, q, kc, vc, seqlen, slot_map, ws, B, Hk, Smax, nsplit, \
+                scale); \
+            else k_attn_decode<D, GG><<<grid, dim3(256), 0, st>>>( \
+                out, q, kc, vc, seqlen, slot_map, ws, B, Hk, Smax, nsplit, \
+                scale); break;
         CASE_G(1) CASE_G(2) CASE_G(3) CASE_G(4)
         CASE_G(5) CASE_G(6) CASE_G(7) CASE_G(8)
 #undef CASE_G

@@ -191,7 +191,8 @@ class WorkerRegistry:
         (reference: WorkerRegistry.ts:413-419)."""
         return [w for w in self.workers.values()
                 if model in w.models
-                and w.status != WorkerStatus.OFFLINE.value
+                and w.status not in (WorkerStatus.OFFLINE.value,
+                                     WorkerStatus.PAUSED.value)
                 and w.current_jobs < w.max_concurrent_jobs]
 
     def get_all_models(self) -> dict[str, int]:

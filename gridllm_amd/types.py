@@ -39,6 +39,7 @@ class WorkerStatus(str, Enum):
     ONLINE = "online"
     BUSY = "busy"
     OFFLINE = "offline"
+    PAUSED = "paused"            # maintenance: registered, not schedulable
 
 
 class RequestType(str, Enum):
@@ -262,6 +263,12 @@ class Channels:
     @staticmethod
     def worker_reregister(worker_id: str) -> str:
         return f"worker:{worker_id}:reregister"
+
+    BROADCAST = "broadcast"      # fleet-wide control verbs
+
+    @staticmethod
+    def worker_control(worker_id: str) -> str:
+        return f"worker:{worker_id}:control"
 
     @staticmethod
     def job_result(job_id: str) -> str:

@@ -54,6 +54,8 @@ class WorkerAgent:
         self._job_tasks: set[asyncio.Task] = set()
         self._last_status: tuple = ()
         self._started = False
+        self.paused = False
+        self.draining = False
 
     # ------------------------------------------------------------- lifecycle
 
@@ -64,6 +66,9 @@ class WorkerAgent:
                                  self._h_job_msg)
         await self.bus.subscribe(Channels.worker_reregister(self.worker_id),
                                  self._h_reregister)
+        await self.bus.subscribe(Channels.worker_control(self.worker_id),
+                                 self._h_control)
+        await self.bus.subscribe(Channels.BROADCAST, self._h_control)
         loop = asyncio.get_running_loop()
         self._tasks.append(loop.create_task(self._heartbeat_loop()))
         self._tasks.append(loop.create_task(self._resource_loop()))
@@ -86,11 +91,15 @@ class WorkerAgent:
 
     def _info(self) -> WorkerInfo:
         stats = telemetry.gpu_stats(self.cfg.device)
+        if self.paused:
+            status = WorkerStatus.PAUSED.value
+        elif self.current_jobs >= self.cfg.max_concurrent_tasks:
+            status = WorkerStatus.BUSY.value
+        else:
+            status = WorkerStatus.ONLINE.value
         return WorkerInfo(
             id=self.worker_id,
-            status=(WorkerStatus.BUSY.value
-                    if self.current_jobs >= self.cfg.max_concurrent_tasks
-                    else WorkerStatus.ONLINE.value),
+            status=status,
             models=self.engine.model_names(),
             max_concurrent_jobs=self.cfg.max_concurrent_tasks,
             current_jobs=self.current_jobs,
@@ -154,12 +163,46 @@ class WorkerAgent:
             "currentJobs": info.current_jobs,
         })
 
+    # --------------------------------------------------------------- control
+
+    async def _h_control(self, _ch: str, payload: dict) -> None:
+        """Maintenance verbs (pause/resume/drain) — adopted from the
+        reference's dead broker path (BrokerClientService.ts:568-657,
+        SURVEY.md §2.2-C6) onto the live worker."""
+        verb = payload.get("type")
+        if verb == "pause":
+            self.paused = True
+            await self._publish_status()
+            log.worker("paused", worker_id=self.worker_id)
+        elif verb == "resume":
+            self.paused = False
+            self.draining = False
+            await self._publish_status()
+            log.worker("resumed", worker_id=self.worker_id)
+        elif verb == "drain":
+            self.paused = True
+            self.draining = True
+            await self._publish_status()
+            log.worker("draining", worker_id=self.worker_id,
+                       in_flight=len(self._job_tasks))
+            if self._job_tasks:
+                await asyncio.gather(*list(self._job_tasks),
+                                     return_exceptions=True)
+            await self.bus.publish(Channels.WORKER_UNREGISTERED,
+                                   {"workerId": self.worker_id})
+            log.worker("drained", worker_id=self.worker_id)
+
     # ------------------------------------------------------------------ jobs
 
     async def _h_job_msg(self, _ch: str, payload: dict) -> None:
         mtype = payload.get("type")
         if mtype == "job_assignment":
             req = InferenceRequest.from_dict(payload.get("job") or {})
+            if self.paused:
+                await self.bus.publish(Channels.JOB_FAILED, {
+                    "jobId": req.id, "workerId": self.worker_id,
+                    "error": "worker paused"})
+                return
             if self.current_jobs >= self.cfg.max_concurrent_tasks:
                 # busy rejection (reference :500-505)
                 await self.bus.publish(Channels.JOB_FAILED, {

@@ -101,10 +101,16 @@ def test_kv_append_gpu(dev):
     (4, 32, 8, 64, [64, 100, 128, 7]),
     (1, 64, 8, 128, [1000]),          # forces split-S
     (2, 8, 8, 128, [300, 17]),        # G=1
+    (256, 24, 8, 128, None),          # large batch -> one-wave kernel
+    (300, 32, 8, 64, None),           # one-wave, D=64, G=4
 ])
 def test_attn_decode_gpu(dev, b, hq, hk, d, seqs):
     torch.manual_seed(5)
     from gridllm_amd.ops import hip_ops
+    if seqs is None:
+        seqs = [int(x) for x in
+                torch.randint(1, 200, (b,), generator=torch.Generator()
+                              .manual_seed(b)).tolist()]
     smax = max(seqs)
     kc = torch.randn(b, hk, smax, d, device=dev).to(torch.bfloat16)
     vc = torch.randn(b, hk, smax, d, device=dev).to(torch.bfloat16)
@@ -268,7 +274,7 @@ def test_gemm_proj_fused_swiglu_gpu(dev):
 
     out = hip_ops.gemm_proj(x, w_il, fuse_swiglu=True)
     assert out.shape == (m, f)
-    assert _mx(out, ref.cpu()) < 0.02 * ref.abs().max().item() + 1e-6
+    assert _mx(out.cpu(), ref.cpu()) < 0.02 * ref.abs().max().item() + 1e-6
 
     o1 = hip_ops.gemm_proj(x, w_il, splitk=1, fuse_swiglu=True)
     o4 = hip_ops.gemm_proj(x, w_il, splitk=4, fuse_swiglu=True)
@@ -276,4 +282,4 @@ def test_gemm_proj_fused_swiglu_gpu(dev):
 
     # the library-GEMM fallback path over the same interleaved weight
     o_blas = hip_ops.swiglu_il(torch.nn.functional.linear(x, w_il))
-    assert _mx(o_blas, ref.cpu()) < 0.02 * ref.abs().max().item() + 1e-6
+    assert _mx(o_blas.cpu(), ref.cpu()) < 0.02 * ref.abs().max().item() + 1e-6

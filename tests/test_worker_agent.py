@@ -166,3 +166,42 @@ async def test_concurrent_jobs_batched():
         asyncio.gather(*(sched.submit_and_wait(r) for r in reqs)), 30)
     assert all(r.get("error") is None for r in results)
     await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_pause_resume_drain_verbs():
+    """Maintenance verbs adopted from the reference's dead broker path:
+    paused workers reject assignments and leave the schedulable pool;
+    drain finishes in-flight work then unregisters."""
+    from gridllm_amd.types import Channels
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    bus = MemoryBus(core)
+    await bus.connect()
+    wid = agent.worker_id
+    try:
+        await bus.publish(Channels.worker_control(wid), {"type": "pause"})
+        await asyncio.sleep(0.1)
+        assert agent.paused
+        assert agent._info().status == "paused"
+        assert not reg.get_available_workers_by_model("tiny")
+
+        await bus.publish(Channels.worker_control(wid), {"type": "resume"})
+        await asyncio.sleep(0.1)
+        assert not agent.paused
+        req = InferenceRequest(model="tiny", prompt="after resume",
+                               options={"temperature": 0.0, "num_predict": 2})
+        res = await asyncio.wait_for(sched.submit_and_wait(req), 30)
+        assert res.get("response") is not None
+
+        events = []
+
+        async def on_unreg(ch, p):
+            events.append(p)
+        await bus.subscribe(Channels.WORKER_UNREGISTERED, on_unreg)
+        await bus.publish(Channels.BROADCAST, {"type": "drain"})
+        await asyncio.sleep(0.3)
+        assert agent.draining
+        assert any(e.get("workerId") == wid for e in events)
+    finally:
+        await teardown(reg, sched, agent)
