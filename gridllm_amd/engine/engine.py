@@ -29,7 +29,7 @@ from gridllm_amd.utils.logging import get_logger
 
 log = get_logger("engine")
 
-_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256)
+_BUCKETS = (1, 2, 4, 8, 16, 32, 64, 128, 256, 384, 512, 768, 1024)
 
 
 @dataclass
