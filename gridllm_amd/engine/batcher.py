@@ -46,10 +46,13 @@ class Batcher:
     whole waves and runs each to completion."""
 
     def __init__(self, engine: InferenceEngine, window_ms: float = 4.0,
-                 continuous: bool = True):
+                 continuous: bool = True, admit_per_step: int = 32):
         self.engine = engine
         self.window_s = window_ms / 1000.0
         self.continuous = continuous and engine.tp.size == 1
+        # bound sequential prefills between decode steps: a burst of new
+        # prompts must not stall in-flight rows for the whole burst
+        self.admit_per_step = max(1, admit_per_step)
         self.q: "queue.Queue[_Pending]" = queue.Queue()
         self._stop = threading.Event()
         target = self._loop_continuous if self.continuous else self._loop
@@ -114,6 +117,7 @@ class Batcher:
                         break
                     continue
             while (len(live) + len(arrivals) < max_batch
+                   and len(arrivals) < self.admit_per_step
                    and self.engine.model.cache.n_free > len(arrivals)):
                 try:
                     arrivals.append(self.q.get_nowait())

@@ -1,0 +1,214 @@
+#!/usr/bin/env python3
+"""Serving-path load test: real server process + real worker process, driven
+through HTTP /api/generate streaming — measures the BASELINE metric on the
+FULL path (gateway -> scheduler -> bus -> worker -> engine -> stream back),
+not just the engine loop.
+
+Usage: python scripts/loadtest.py [--model llama3.2:3b] [--concurrency 128]
+         [--requests 256] [--num-predict 64] [--device cuda:0]
+Prints one JSON line: tokens/s (node), TTFT p50/p95, request stats.
+"""
+
+import argparse
+import asyncio
+import json
+import os
+import signal
+import socket
+import statistics
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+SERVER_SRC = r"""
+import asyncio, os
+import uvicorn
+from gridllm_amd.config import ServerConfig
+from gridllm_amd.server.app import GridLLMServer
+
+cfg = ServerConfig(port=int(os.environ["HTTP_PORT"]),
+                   bus_port=int(os.environ["BUS_PORT"]))
+server = GridLLMServer(cfg)
+app = server.app
+
+@app.on_event("startup")
+async def _s():
+    await server.start()
+    print("SERVER_READY", flush=True)
+
+uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error")
+"""
+
+WORKER_SRC = r"""
+import asyncio, os
+from gridllm_amd.bus.tcp import TcpBus
+from gridllm_amd.config import EngineConfig, WorkerConfig
+from gridllm_amd.engine import InferenceEngine
+from gridllm_amd.worker.agent import WorkerAgent
+
+async def main():
+    conc = int(os.environ["LT_CONC"])
+    wcfg = WorkerConfig(worker_id="lt-worker",
+                        bus_port=int(os.environ["BUS_PORT"]),
+                        max_concurrent_tasks=conc)
+    ecfg = EngineConfig(model=os.environ["LT_MODEL"],
+                        device=os.environ["LT_DEVICE"],
+                        dtype=os.environ["LT_DTYPE"],
+                        max_seq_len=int(os.environ.get("LT_MAXSEQ", "1024")),
+                        max_batch_size=conc,
+                        use_hipgraph=os.environ["LT_DEVICE"].startswith("cuda"))
+    engine = InferenceEngine(ecfg)
+    bus = TcpBus("127.0.0.1", wcfg.bus_port)
+    await bus.connect()
+    agent = WorkerAgent(bus, engine, wcfg)
+    await agent.start()
+    print("WORKER_READY", flush=True)
+    await asyncio.Event().wait()
+
+asyncio.run(main())
+"""
+
+
+def _spawn(src, env, tag, timeout=300):
+    proc = subprocess.Popen([sys.executable, "-u", "-c", src],
+                            env={**os.environ, **env, "PYTHONPATH": REPO},
+                            stdout=subprocess.PIPE,
+                            stderr=subprocess.STDOUT, text=True, cwd=REPO)
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        line = proc.stdout.readline()
+        if tag in line:
+            import threading
+
+            def _drain(p=proc):
+                try:
+                    for _ in p.stdout:
+                        pass
+                except Exception:  # noqa: BLE001
+                    pass
+            threading.Thread(target=_drain, daemon=True).start()
+            return proc
+        if proc.poll() is not None:
+            raise RuntimeError(f"process died:\n{proc.stdout.read()}")
+    proc.kill()
+    raise TimeoutError(f"no {tag} within {timeout}s")
+
+
+async def run_clients(base, args):
+    import aiohttp
+    sem = asyncio.Semaphore(args.concurrency)
+    ttfts, tokens, errors = [], [], [0]
+    conn = aiohttp.TCPConnector(limit=args.concurrency + 16)
+    async with aiohttp.ClientSession(base_url=base, connector=conn) as sess:
+
+        async def one(i):
+            async with sem:
+                t0 = time.monotonic()
+                first = None
+                n = 0
+                try:
+                    async with sess.post("/api/generate", json={
+                            "model": args.model,
+                            "prompt": f"request {i}: " + "lorem " * args.prompt_words,
+                            "stream": True,
+                            "options": {"temperature": 0.8, "seed": i,
+                                        "num_predict": args.num_predict}},
+                            timeout=aiohttp.ClientTimeout(total=600)) as resp:
+                        async for raw in resp.content:
+                            if not raw.strip():
+                                continue
+                            c = json.loads(raw)
+                            if first is None and c.get("response"):
+                                first = time.monotonic() - t0
+                            if c.get("done"):
+                                n = c.get("eval_count", 0)
+                except Exception:  # noqa: BLE001
+                    errors[0] += 1
+                    return
+                if first is not None:
+                    ttfts.append(first * 1000)
+                tokens.append(n)
+
+        t0 = time.monotonic()
+        await asyncio.gather(*(one(i) for i in range(args.requests)))
+        wall = time.monotonic() - t0
+    return wall, ttfts, tokens, errors[0]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama3.2:3b")
+    ap.add_argument("--device", default="cuda:0")
+    ap.add_argument("--dtype", default="bf16")
+    ap.add_argument("--concurrency", type=int, default=128)
+    ap.add_argument("--requests", type=int, default=256)
+    ap.add_argument("--num-predict", type=int, default=64)
+    ap.add_argument("--prompt-words", type=int, default=24)
+    ap.add_argument("--max-seq", type=int, default=1024)
+    args = ap.parse_args()
+
+    import torch
+    if not torch.cuda.is_available() and args.device.startswith("cuda"):
+        args.device, args.dtype, args.model = "cpu", "fp32", "tiny"
+        args.concurrency, args.requests = 4, 16
+
+    bus_port, http_port = _free_port(), _free_port()
+    env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port),
+           "LT_CONC": str(args.concurrency), "LT_MODEL": args.model,
+           "LT_DEVICE": args.device, "LT_DTYPE": args.dtype,
+           "LT_MAXSEQ": str(args.max_seq)}
+    server = _spawn(SERVER_SRC, env, "SERVER_READY")
+    worker = _spawn(WORKER_SRC, env, "WORKER_READY")
+    try:
+        base = f"http://127.0.0.1:{http_port}"
+        # registration wait
+        import httpx
+        with httpx.Client(base_url=base, timeout=30) as c:
+            for _ in range(200):
+                r = c.get("/health/workers")
+                if r.status_code == 200 and r.json()["counts"]["total"] >= 1:
+                    break
+                time.sleep(0.1)
+        # small warmup wave
+        wa = argparse.Namespace(**vars(args))
+        wa.requests = min(args.concurrency, 16)
+        asyncio.run(run_clients(base, wa))
+        wall, ttfts, tokens, errs = asyncio.run(run_clients(base, args))
+        out = {
+            "metric": "output tokens/sec (node), /api/generate streaming",
+            "value": round(sum(tokens) / wall, 1),
+            "unit": "tokens/s",
+            "requests": args.requests,
+            "concurrency": args.concurrency,
+            "num_predict": args.num_predict,
+            "errors": errs,
+            "wall_s": round(wall, 2),
+            "ttft_p50_ms": round(statistics.median(ttfts), 2) if ttfts else None,
+            "ttft_p95_ms": round(sorted(ttfts)[int(len(ttfts) * 0.95)], 2)
+            if ttfts else None,
+            "model": args.model,
+            "device": args.device,
+        }
+        print(json.dumps(out))
+    finally:
+        for p in (worker, server):
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in (worker, server):
+            p.wait(timeout=10)
+
+
+if __name__ == "__main__":
+    main()
