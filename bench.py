@@ -28,10 +28,10 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--batch", type=int, default=256, help="sequences per GPU")
+    ap.add_argument("--batch", type=int, default=1024, help="sequences per GPU")
     ap.add_argument("--model", default="llama3.2:3b")
     ap.add_argument("--prompt-len", type=int, default=128)
-    ap.add_argument("--max-seq", type=int, default=2048)
+    ap.add_argument("--max-seq", type=int, default=1024)
     ap.add_argument("--device", default="")
     args = ap.parse_args()
 
