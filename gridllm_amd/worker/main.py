@@ -23,12 +23,54 @@ from gridllm_amd.worker.agent import WorkerAgent
 log = get_logger("worker-main")
 
 
+def build_health_app(agent: WorkerAgent):
+    """Worker-local health/status HTTP app (reference: client/src/index.ts
+    :53-91 + client/src/routes/health.ts:8-59), with real GPU telemetry."""
+    from fastapi import FastAPI
+    app = FastAPI(title="GridLLM-AMD worker")
+
+    @app.get("/")
+    async def root():
+        info = agent._info()
+        return {"name": "GridLLM-AMD worker", "workerId": agent.worker_id,
+                "status": info.status, "models": info.models,
+                "currentJobs": info.current_jobs}
+
+    @app.get("/health")
+    @app.get("/health/live")
+    async def health():
+        return {"status": "healthy", "workerId": agent.worker_id}
+
+    @app.get("/health/ready")
+    async def ready():
+        ok = agent._started and not agent.paused
+        return {"status": "ready" if ok else "not_ready",
+                "paused": agent.paused}
+
+    @app.get("/health/system")
+    async def system():
+        info = agent._info()
+        return {"workerId": agent.worker_id, "gpu": info.gpu,
+                "engine": agent.engine.stats()}
+
+    @app.get("/worker/status")
+    async def status():
+        return agent._info().to_dict()
+
+    return app
+
+
 async def run_worker(wcfg: WorkerConfig, ecfg: EngineConfig) -> None:
     engine = InferenceEngine(ecfg)
     disconnected = asyncio.Event()
     bus = TcpBus(wcfg.bus_host, wcfg.bus_port,
                  on_disconnect=lambda: disconnected.set())
     agent = WorkerAgent(bus, engine, wcfg)
+    if wcfg.health_port > 0:
+        import uvicorn
+        ucfg = uvicorn.Config(build_health_app(agent), host="0.0.0.0",
+                              port=wcfg.health_port, log_level="error")
+        asyncio.get_running_loop().create_task(uvicorn.Server(ucfg).serve())
     attempt = 0
     while True:
         try:

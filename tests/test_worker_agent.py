@@ -205,3 +205,29 @@ async def test_pause_resume_drain_verbs():
         assert any(e.get("workerId") == wid for e in events)
     finally:
         await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_worker_health_app():
+    """Worker-local health endpoints (reference client/src/routes/health.ts)
+    serve status + GPU telemetry for the live agent."""
+    from httpx import ASGITransport, AsyncClient
+    from gridllm_amd.worker.main import build_health_app
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    try:
+        app = build_health_app(agent)
+        async with AsyncClient(transport=ASGITransport(app=app),
+                               base_url="http://t") as c:
+            r = await c.get("/health")
+            assert r.status_code == 200 and r.json()["status"] == "healthy"
+            r = await c.get("/health/ready")
+            assert r.json()["status"] == "ready"
+            r = await c.get("/worker/status")
+            body = r.json()
+            assert body["id"] == agent.worker_id
+            assert "gpu" in body
+            r = await c.get("/health/system")
+            assert "engine" in r.json()
+    finally:
+        await teardown(reg, sched, agent)
