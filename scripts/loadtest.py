@@ -140,6 +140,9 @@ async def run_clients(base, args):
                 if first is not None:
                     ttfts.append(first * 1000)
                 tokens.append(n)
+                if len(tokens) % 64 == 0:
+                    print(f"  {len(tokens)} requests done",
+                          file=sys.stderr, flush=True)
 
         t0 = time.monotonic()
         await asyncio.gather(*(one(i) for i in range(args.requests)))
@@ -160,7 +163,7 @@ def main():
     args = ap.parse_args()
 
     import torch
-    if not torch.cuda.is_available() and args.device.startswith("cuda"):
+    if not torch.cuda.is_available():
         args.device, args.dtype, args.model = "cpu", "fp32", "tiny"
         args.concurrency, args.requests = 4, 16
 
@@ -169,7 +172,9 @@ def main():
            "LT_CONC": str(args.concurrency), "LT_MODEL": args.model,
            "LT_DEVICE": args.device, "LT_DTYPE": args.dtype,
            "LT_MAXSEQ": str(args.max_seq)}
+    print("spawning server...", file=sys.stderr, flush=True)
     server = _spawn(SERVER_SRC, env, "SERVER_READY")
+    print("spawning worker...", file=sys.stderr, flush=True)
     worker = _spawn(WORKER_SRC, env, "WORKER_READY")
     try:
         base = f"http://127.0.0.1:{http_port}"
@@ -181,10 +186,12 @@ def main():
                 if r.status_code == 200 and r.json()["counts"]["total"] >= 1:
                     break
                 time.sleep(0.1)
+        print("workers registered; warmup...", file=sys.stderr, flush=True)
         # small warmup wave
         wa = argparse.Namespace(**vars(args))
         wa.requests = min(args.concurrency, 16)
         asyncio.run(run_clients(base, wa))
+        print("warmup done; main wave...", file=sys.stderr, flush=True)
         wall, ttfts, tokens, errs = asyncio.run(run_clients(base, args))
         out = {
             "metric": "output tokens/sec (node), /api/generate streaming",
