@@ -266,7 +266,15 @@ class TcpBus(Bus):
         await self._call("identify", worker_id=worker_id)
 
     async def publish(self, channel: str, payload: dict) -> None:
-        await self._call("publish", channel=channel, payload=payload)
+        """Fire-and-forget, like Redis pub/sub (at-most-once; the control
+        plane compensates with orphan-requeue — SURVEY.md §2.3). Awaiting a
+        per-message ack round-trip capped token streaming at ~2.6k msg/s."""
+        if self._writer is None or self.closed:
+            raise ConnectionError("bus not connected")
+        self._writer.write(
+            (json.dumps({"op": "publish", "channel": channel,
+                         "payload": payload}) + "\n").encode())
+        await self._writer.drain()
 
     async def subscribe(self, channel: str, handler: Handler) -> None:
         first = not self._handlers[channel]

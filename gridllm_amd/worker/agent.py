@@ -306,13 +306,26 @@ class WorkerAgent:
                 pass
 
     async def _pump_stream(self, job_id: str, q: asyncio.Queue) -> None:
-        """Publish token chunks to job:stream:<id> (reference :613-626)."""
+        """Publish token chunks to job:stream:<id> (reference :613-626).
+        Adaptive coalescing: whatever pieces accumulated while the previous
+        publish was in flight go out as ONE chunk — per-token messages cap
+        the node at the bus message rate, not the engine rate."""
         ch = Channels.job_stream(job_id)
         while True:
             item = await q.get()
             if item is None:
                 break
             piece, done = item
+            while not done:
+                try:
+                    nxt = q.get_nowait()
+                except asyncio.QueueEmpty:
+                    break
+                if nxt is None:
+                    done = True
+                    break
+                piece += nxt[0]
+                done = nxt[1]
             await self.bus.publish(ch, {
                 "jobId": job_id, "workerId": self.worker_id,
                 "chunk": {"id": job_id, "response": piece, "done": done},
