@@ -187,9 +187,10 @@ def main():
                     break
                 time.sleep(0.1)
         print("workers registered; warmup...", file=sys.stderr, flush=True)
-        # small warmup wave
+        # full-concurrency warmup: captures every hipGraph bucket the main
+        # wave will hit, so one-time capture cost stays out of the numbers
         wa = argparse.Namespace(**vars(args))
-        wa.requests = min(args.concurrency, 16)
+        wa.requests = args.concurrency * 2
         asyncio.run(run_clients(base, wa))
         print("warmup done; main wave...", file=sys.stderr, flush=True)
         wall, ttfts, tokens, errs = asyncio.run(run_clients(base, args))
