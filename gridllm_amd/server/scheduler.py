@@ -43,6 +43,8 @@ class JobScheduler:
         self.registry = registry
         self.cfg = cfg or ServerConfig()
         self.queue: list[InferenceRequest] = []
+        self._queue_dirty = False
+        self._persist_task: Optional[asyncio.Task] = None
         self.active: dict[str, JobAssignment] = {}
         self._timeout_tasks: dict[str, asyncio.Task] = {}
         self._retry_tasks: set[asyncio.Task] = set()
@@ -81,6 +83,15 @@ class JobScheduler:
             t.cancel()
         self._tasks.clear()
         self._timeout_tasks.clear()
+        if self._persist_task is not None:
+            self._persist_task.cancel()
+        if self._queue_dirty:            # final synchronous snapshot
+            try:
+                await self.bus.kv_set(
+                    Keys.JOB_QUEUE,
+                    json.dumps([r.to_dict() for r in self.queue]))
+            except Exception:  # noqa: BLE001
+                pass
 
     async def _load_existing(self) -> None:
         """Crash-resume: re-load queue and active jobs persisted on the bus
@@ -109,8 +120,26 @@ class JobScheduler:
     # -------------------------------------------------------------- enqueue
 
     async def _persist_queue(self) -> None:
-        await self.bus.kv_set(Keys.JOB_QUEUE,
-                              json.dumps([r.to_dict() for r in self.queue]))
+        """Debounced: the reference re-serialized the whole queue to Redis
+        on EVERY enqueue/dequeue (JobScheduler.ts:656) — O(n^2) bytes under
+        load. Mark dirty and flush at most every 50 ms; crash-resume only
+        needs a recent snapshot (orphan-requeue covers the gap)."""
+        self._queue_dirty = True
+        if self._persist_task is None or self._persist_task.done():
+            self._persist_task = asyncio.get_running_loop().create_task(
+                self._flush_queue())
+
+    async def _flush_queue(self) -> None:
+        await asyncio.sleep(0.05)
+        while self._queue_dirty:
+            self._queue_dirty = False
+            try:
+                await self.bus.kv_set(
+                    Keys.JOB_QUEUE,
+                    json.dumps([r.to_dict() for r in self.queue]))
+            except Exception:  # noqa: BLE001
+                log.logger.warning("queue persist failed")
+                return
 
     async def add_job(self, req: InferenceRequest, front: bool = False) -> None:
         if front:
