@@ -92,6 +92,8 @@ class WorkerConfig:
     bus_port: int = 46379
     heartbeat_interval_ms: int = 5_000          # ref hardcodes 5 s (config:94)
     resource_monitor_interval_ms: int = 10_000
+    stream_flush_ms: int = 25      # min interval between stream publishes
+                                   # per job (first token flushes at once)
     max_concurrent_tasks: int = 1
     device: int = 0                             # HIP device index this worker pins
     reconnect_max_attempts: int = 10
@@ -106,6 +108,7 @@ class WorkerConfig:
             bus_port=_env_int("BUS_PORT", 46379, 1, 65535),
             heartbeat_interval_ms=_env_int("HEARTBEAT_INTERVAL", 5_000, 200),
             resource_monitor_interval_ms=_env_int("RESOURCE_MONITOR_INTERVAL", 10_000, 500),
+            stream_flush_ms=_env_int("STREAM_FLUSH_MS", 25, 0),
             max_concurrent_tasks=_env_int("MAX_CONCURRENT_TASKS", 1, 1, 4096),
             device=_env_int("WORKER_DEVICE", 0, 0, 63),
         )
