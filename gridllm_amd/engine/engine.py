@@ -190,15 +190,9 @@ class InferenceEngine:
         states = [r.state for r in rows]
         if all(s.greedy for s in states):
             return logits.argmax(dim=-1).tolist()
-        simple = all(
-            s.counts is None and not s.opts.top_p and not s.opts.min_p
-            and not s.opts.top_k and s.gen is None for s in states)
-        if simple:
-            temps = torch.tensor(
-                [s.opts.temperature if s.opts.temperature is not None else 0.8
-                 for s in states], device=logits.device).clamp(min=1e-4)
-            probs = torch.softmax(logits / temps.unsqueeze(1), dim=-1)
-            return torch.multinomial(probs, 1).squeeze(1).tolist()
+        if all(s.counts is None for s in states):
+            from gridllm_amd.engine.sampler import sample_batch_vectorized
+            return sample_batch_vectorized(logits, states)
         return [sample(logits[i], s) for i, s in enumerate(states)]
 
     # ------------------------------------------------------------ main loop

@@ -91,3 +91,48 @@ def sample(logits: torch.Tensor, state: SamplerState) -> int:
     if not torch.isfinite(s) or s <= 0:
         return int(torch.argmax(logits).item())
     return int(torch.multinomial(probs, 1, generator=state.gen).item())
+
+
+def sample_batch_vectorized(logits: torch.Tensor,
+                            states: list["SamplerState"]) -> list[int]:
+    """Batched sampling for penalty-free rows: batched top_k/top_p/min_p
+    filtering + per-row Gumbel noise (seeded rows use their own generator,
+    issued without a sync) + ONE argmax + ONE host copy. The per-row
+    `sample()` loop costs a GPU sync per ROW per step — at serving batch
+    sizes that multiplies the decode step several-fold."""
+    dev = logits.device
+    b, v = logits.shape
+    temps = torch.tensor(
+        [max(s.opts.temperature if s.opts.temperature is not None else 0.8,
+             1e-4) for s in states], device=dev).unsqueeze(1)
+    x = logits / temps
+    need_filter = any((s.opts.top_k and 0 < s.opts.top_k < v)
+                      or (s.opts.top_p and 0.0 < s.opts.top_p < 1.0)
+                      or (s.opts.min_p and 0.0 < s.opts.min_p < 1.0)
+                      for s in states)
+    if need_filter:
+        probs = torch.softmax(x, -1)
+        sp, _ = torch.sort(probs, -1, descending=True)
+        thr = torch.zeros(b, 1, device=dev)
+        kidx = torch.tensor(
+            [int(s.opts.top_k) - 1 if s.opts.top_k and 0 < s.opts.top_k < v
+             else v - 1 for s in states], device=dev).unsqueeze(1)
+        thr = torch.maximum(thr, sp.gather(1, kidx))
+        mps = torch.tensor(
+            [s.opts.min_p if s.opts.min_p and 0 < s.opts.min_p < 1 else 0.0
+             for s in states], device=dev).unsqueeze(1)
+        thr = torch.maximum(thr, sp[:, :1] * mps)
+        tps = torch.tensor(
+            [s.opts.top_p if s.opts.top_p and 0 < s.opts.top_p < 1 else 1.0
+             for s in states], device=dev).unsqueeze(1)
+        cum = torch.cumsum(sp, -1)
+        cnt = ((cum - sp) < tps).sum(1, keepdim=True).clamp(min=1)
+        thr = torch.maximum(thr, sp.gather(1, cnt - 1))
+        x = torch.where(probs >= thr, x, torch.full_like(x, float("-inf")))
+    noise = torch.zeros_like(x)
+    for i, s in enumerate(states):
+        if not s.greedy:
+            u = torch.rand(v, generator=s.gen, device=dev)
+            inner = (-torch.log(u.clamp_min(1e-20))).clamp_min(1e-20)
+            noise[i] = -torch.log(inner)
+    return (x + noise).argmax(-1).tolist()
