@@ -329,7 +329,7 @@ class WorkerAgent:
         flush_s = max(0.002, self.cfg.stream_flush_ms / 1000.0)
         sent_any = False
         while True:
-            await asyncio.sleep(flush_s if sent_any else 0.002)
+            await asyncio.sleep(flush_s if sent_any else 0.005)
             with lk:
                 pieces = buf[:]
                 buf.clear()

@@ -82,6 +82,9 @@ asyncio.run(main())
 
 
 def _spawn(src, env, tag, timeout=300):
+    logf = None
+    if os.environ.get("LT_DEBUG"):
+        logf = open(f"gpurun_out/lt_{tag.lower()}.log", "w")
     proc = subprocess.Popen([sys.executable, "-u", "-c", src],
                             env={**os.environ, **env, "PYTHONPATH": REPO},
                             stdout=subprocess.PIPE,
@@ -89,13 +92,18 @@ def _spawn(src, env, tag, timeout=300):
     t0 = time.time()
     while time.time() - t0 < timeout:
         line = proc.stdout.readline()
+        if logf and line:
+            logf.write(line)
+            logf.flush()
         if tag in line:
             import threading
 
-            def _drain(p=proc):
+            def _drain(p=proc, lf=logf):
                 try:
-                    for _ in p.stdout:
-                        pass
+                    for ln in p.stdout:
+                        if lf:
+                            lf.write(ln)
+                            lf.flush()
                 except Exception:  # noqa: BLE001
                     pass
             threading.Thread(target=_drain, daemon=True).start()
