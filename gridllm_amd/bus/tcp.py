@@ -37,8 +37,13 @@ class BusServer:
         self._server: Optional[asyncio.AbstractServer] = None
 
     async def start(self) -> None:
+        # 32 MiB line limit: a persisted job-queue snapshot of a few
+        # hundred requests crosses asyncio's 64 KiB readline default and
+        # LimitOverrunError silently drops the connection (the serving
+        # stall at >=192-way concurrency)
         self._server = await asyncio.start_server(self._on_conn, self.host,
-                                                  self.port)
+                                                  self.port,
+                                                  limit=32 * 1024 * 1024)
         log.worker("bus server listening", host=self.host, port=self.port)
 
     async def stop(self) -> None:
@@ -187,7 +192,7 @@ class TcpBus(Bus):
 
     async def connect(self) -> None:
         self._reader, self._writer = await asyncio.open_connection(
-            self.host, self.port)
+            self.host, self.port, limit=32 * 1024 * 1024)
         self.closed = False
         self._reader_task = asyncio.get_running_loop().create_task(
             self._read_loop())

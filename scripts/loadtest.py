@@ -173,7 +173,8 @@ def main():
     import torch
     if not torch.cuda.is_available():
         args.device, args.dtype, args.model = "cpu", "fp32", "tiny"
-        args.concurrency, args.requests = 4, 16
+        if not os.environ.get("LT_KEEP_CONC"):
+            args.concurrency, args.requests = 4, 16
 
     bus_port, http_port = _free_port(), _free_port()
     env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port),

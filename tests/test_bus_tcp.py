@@ -90,3 +90,35 @@ async def test_client_disconnect_callback():
     await c.connect()
     await srv.stop()   # server goes away
     await asyncio.wait_for(flag.wait(), 2)
+
+
+@pytest.mark.asyncio
+async def test_large_payload_roundtrip():
+    """Regression: a >64 KiB payload (e.g. a persisted queue snapshot of a
+    few hundred jobs) must not trip asyncio's readline limit and drop the
+    connection (the >=192-way serving stall)."""
+    from gridllm_amd.bus.tcp import BusServer, TcpBus
+    import socket as _s
+    srv_sock = _s.socket(); srv_sock.bind(("127.0.0.1", 0))
+    port = srv_sock.getsockname()[1]; srv_sock.close()
+    server = BusServer("127.0.0.1", port)
+    await server.start()
+    bus = TcpBus("127.0.0.1", port)
+    await bus.connect()
+    try:
+        big = "x" * (300 * 1024)
+        await bus.kv_set("bigkey", big)
+        got = await bus.kv_get("bigkey")
+        assert got == big
+        seen = asyncio.Event()
+
+        async def h(ch, p):
+            assert len(p["blob"]) == 200 * 1024
+            seen.set()
+        await bus.subscribe("bigchan", h)
+        await bus.publish("bigchan", {"blob": "y" * (200 * 1024)})
+        await asyncio.wait_for(seen.wait(), 10)
+        assert not bus.closed
+    finally:
+        await bus.close()
+        await server.stop()
