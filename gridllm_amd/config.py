@@ -10,7 +10,7 @@ reference's Joi schema throw (server/src/config/index.ts:45-49).
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

@@ -17,7 +17,7 @@ from concurrent.futures import Future
 from dataclasses import dataclass, field
 from typing import Callable, Optional
 
-from gridllm_amd.engine.engine import GenerationResult, InferenceEngine
+from gridllm_amd.engine.engine import InferenceEngine
 from gridllm_amd.engine import chat as chat_mod
 from gridllm_amd.utils.logging import get_logger
 

@@ -9,7 +9,7 @@ no network). If a HuggingFace tokenizer.json path is supplied, the
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+from typing import List
 
 
 class ByteTokenizer:

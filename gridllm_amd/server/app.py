@@ -14,7 +14,6 @@ from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse
 
 from gridllm_amd import __version__
-from gridllm_amd.bus.memory import MemoryBus, MemoryBusCore
 from gridllm_amd.bus.tcp import BusServer, TcpBus
 from gridllm_amd.config import ServerConfig
 from gridllm_amd.server.registry import WorkerRegistry

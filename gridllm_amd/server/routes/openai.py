@@ -13,7 +13,6 @@ import asyncio
 import json
 import time
 import uuid
-from typing import Optional
 
 from fastapi import APIRouter, HTTPException, Request
 from fastapi.responses import JSONResponse, StreamingResponse

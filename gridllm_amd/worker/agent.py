@@ -30,7 +30,7 @@ from gridllm_amd.bus.base import Bus
 from gridllm_amd.config import WorkerConfig
 from gridllm_amd.engine.batcher import Batcher
 from gridllm_amd.engine.engine import GenerationResult, InferenceEngine
-from gridllm_amd.types import (Channels, GPUStats, InferenceRequest, Keys,
+from gridllm_amd.types import (Channels, InferenceRequest, Keys,
                                WorkerInfo, WorkerStatus, now_ms)
 from gridllm_amd.utils.logging import get_logger
 from gridllm_amd.worker import telemetry

@@ -21,7 +21,6 @@ from __future__ import annotations
 
 import asyncio
 import random
-from typing import Optional
 
 import torch.distributed as dist
 
