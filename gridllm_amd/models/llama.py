@@ -3,11 +3,14 @@
 Design (no reference analog — the reference delegated all compute to Ollama,
 client/src/services/OllamaService.ts):
 
-* fused weight layout: one QKV GEMM, one gate+up GEMM per layer (library
-  GEMMs via hipBLASLt through torch.nn.functional.linear); everything
-  between GEMMs is a hand-written CDNA4 kernel (gridllm_amd.ops): fused
-  residual-add RMSNorm, RoPE from a host-precomputed table, GQA decode
-  attention with split-S, SwiGLU epilogue, KV append.
+* fused weight layout: one QKV GEMM, one gate+up GEMM per layer. Decode
+  GEMMs dispatch through a per-shape autotuner (ops.autotune: hipBLASLt vs
+  the in-house skinny / glds-pipelined projection kernels); the gate+up
+  projection runs with a fused-SwiGLU epilogue on an interleaved weight
+  layout. Everything between GEMMs is a hand-written CDNA4 kernel
+  (gridllm_amd.ops): fused residual-add RMSNorm, fused RoPE+KV-write, MFMA
+  flash prefill, GQA decode attention (4-wave + one-wave large-batch
+  variants, split-S for long sequences).
 * KV cache: contiguous [slots, Hkv, max_seq, D] bf16 per layer, sized for
   288 GB HBM3E; slot indirection (slot_map) keeps decode shapes static so
   the whole decode step is hipGraph-capturable.
