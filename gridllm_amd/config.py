@@ -127,6 +127,7 @@ class EngineConfig:
     tp_degree: int = 1
     use_hipgraph: bool = True                  # capture decode step in a hipGraph
     autotune: bool = True                      # per-shape GEMM dispatch tuning
+    context_shift: bool = True                 # slide the KV window past max_seq
     weights_path: str = ""                     # safetensors dir; empty -> random init
     seed: int = 0
 
@@ -141,6 +142,7 @@ class EngineConfig:
             tp_degree=_env_int("ENGINE_TP", 1, 1, 8),
             use_hipgraph=_env_int("ENGINE_HIPGRAPH", 1, 0, 1) == 1,
             autotune=_env_int("ENGINE_AUTOTUNE", 1, 0, 1) == 1,
+            context_shift=_env_int("ENGINE_CONTEXT_SHIFT", 1, 0, 1) == 1,
             weights_path=_env_str("ENGINE_WEIGHTS", ""),
             seed=_env_int("ENGINE_SEED", 0),
         )

@@ -61,6 +61,7 @@ class _Row:
     done: bool = False
     t_start_ns: int = 0            # wall clock at admission
     t_decode_ns: int = 0           # wall clock when prefill+first token done
+    cur_len: int = 0               # storage length in the KV slot
 
 
 class InferenceEngine:
@@ -298,12 +299,17 @@ class InferenceEngine:
             stops += extra_stops
         max_new = opts.num_predict if opts.num_predict and opts.num_predict > 0 \
             else 128
-        max_new = min(max_new, self.model.max_seq - len(ids) - 1)
+        if self.cfg.context_shift:
+            # context shift slides the KV window, so generation is bounded
+            # by the rope table (original absolute positions), not max_seq
+            max_new = min(max_new, self.model.rope_len - len(ids) - 1)
+        else:
+            max_new = min(max_new, self.model.max_seq - len(ids) - 1)
         row = _Row(slot=slot, state=SamplerState.create(opts, self.device),
                    decoder=self.tokenizer.decode_stream(),
                    on_token=on_token, cancel=cancel,
                    stops=stops, max_new=max_new, result=GenerationResult(),
-                   t_start_ns=t0)
+                   t_start_ns=t0, cur_len=len(ids))
         row.result.prompt_eval_count = len(ids)
         self._sync_cancel([row])
         tok = self._sample_batch(logits.unsqueeze(0), [row])[0]  # syncs
@@ -317,6 +323,10 @@ class InferenceEngine:
 
     def _step_rows(self, active: list["_Row"]) -> list["_Row"]:
         """One batched decode step; returns the rows still running."""
+        if self.cfg.context_shift:
+            for r in active:
+                if r.cur_len >= self.model.max_seq:
+                    r.cur_len = self.model.shift_slot(r.slot)
         token_ids = [r.last_token for r in active]
         slot_ids = [r.slot for r in active]
         logits = self._decode_step(token_ids, slot_ids)
@@ -324,6 +334,7 @@ class InferenceEngine:
         self._sync_cancel(active)
         still: list[_Row] = []
         for r, tok in zip(active, next_ids):
+            r.cur_len += 1
             if self._emit(r, tok):
                 still.append(r)
             else:

@@ -60,6 +60,11 @@ class KVCache:
         self.v = [torch.zeros(shape, dtype=dtype, device=device)
                   for _ in range(spec.layers)]
         self.seq_lens = torch.zeros(slots, dtype=torch.int32, device=device)
+        # context shift: per-slot rope-position offset — after a shift the
+        # kept K entries stay roped at their ORIGINAL absolute positions,
+        # so new tokens must rope at storage_pos + rope_off (read inside
+        # the hipGraph; plain tensor -> capture-safe)
+        self.rope_off = torch.zeros(slots, dtype=torch.int32, device=device)
         self._free = list(range(slots))
 
     def reserve(self, slot: int) -> None:
@@ -76,6 +81,7 @@ class KVCache:
 
     def free(self, slot: int) -> None:
         self.seq_lens[slot] = 0
+        self.rope_off[slot] = 0
         self._free.append(slot)
 
     @property
@@ -106,8 +112,11 @@ class LlamaModel:
 
         self.gateup_il = False
         self._init_weights(seed)
+        # tables cover 4x the window: context shift keeps original absolute
+        # rope positions, so a long generation's positions exceed max_seq
+        self.rope_len = 4 * self.max_seq
         self.cos, self.sin = ops.rope_tables(
-            spec.head_dim, self.max_seq, spec.rope_theta, device=self.device)
+            spec.head_dim, self.rope_len, spec.rope_theta, device=self.device)
         self.cache = KVCache(spec, kv_slots, self.max_seq, self.hkv_l,
                              self.device, self.dtype if self.dtype != torch.float32
                              else torch.float32)
@@ -271,6 +280,23 @@ class LlamaModel:
                 total += t.numel()
         return total * self.embed.element_size()
 
+    def shift_slot(self, slot: int) -> int:
+        """Context shift (the reference delegated this to Ollama/llama.cpp's
+        num_ctx handling): drop the oldest half of the slot's KV, keep the
+        recent half AT ITS ORIGINAL ROPE POSITIONS (rope_off grows by the
+        drop count, so relative distances are exact). Returns the new
+        storage length. Eager, per-row, rare — no kernel needed."""
+        keep = self.max_seq // 2
+        drop = self.max_seq - keep
+        for li in range(self.spec.layers):
+            k = self.cache.k[li]
+            v = self.cache.v[li]
+            k[slot, :, :keep] = k[slot, :, drop:].clone()
+            v[slot, :, :keep] = v[slot, :, drop:].clone()
+        self.cache.seq_lens[slot] = keep
+        self.cache.rope_off[slot] += drop
+        return keep
+
     # -- forward -----------------------------------------------------------
 
     @torch.no_grad()
@@ -325,13 +351,15 @@ class LlamaModel:
         pos = self.cache.seq_lens[slot_map.long()]          # [B] int32
         slot_t = slot_map * self.cache.max_seq + pos
         attn_seq = pos + 1
+        # rope at the ABSOLUTE position (shifted slots have rope_off > 0)
+        pos_rope = pos + self.cache.rope_off[slot_map.long()]
         x = self.embed[tokens]
         resid = x
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
             qkv = _linear(normed, lw.wqkv)
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
-                            self.cos, self.sin, pos, slot_t,
+                            self.cos, self.sin, pos_rope, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)
             attn = ops.attn_decode(q, self.cache.k[li], self.cache.v[li],
                                    attn_seq, self.scale, slot_map=slot_map)

@@ -201,3 +201,49 @@ def test_safetensors_weight_loading(tmp_path):
         shards.append(mr.layers[0].wgateup[:f_l])
     assert torch.equal(torch.cat(shards),
                        weights["model.layers.0.mlp.gate_proj.weight"])
+
+
+def test_context_shift_generates_past_max_seq():
+    """Generation continues past the KV window (the reference delegated
+    context handling to Ollama; this is our llama.cpp-style shift)."""
+    cfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                       max_seq_len=32, max_batch_size=2, use_hipgraph=False)
+    eng = InferenceEngine(cfg)
+    r = eng.generate("abcdefgh", {"temperature": 0.0, "num_predict": 60})
+    assert r.eval_count == 60          # not stopped at the 32-token window
+    assert r.done_reason in ("length", "stop")
+    # the slot was freed and its rope offset cleared
+    assert int(eng.model.cache.rope_off.max().item()) == 0
+
+
+def test_context_shift_matches_fresh_prefill():
+    """Rope geometry after a shift: with ONE layer, K/V of a token depend
+    only on that token (no cross-token mixing before attention), so a
+    shifted cache (kept tokens at ORIGINAL rope positions, rope_off > 0)
+    must equal a fresh position-0 prefill of the same tokens — rope
+    attention depends only on relative positions. (With >1 layer the kept
+    K/V legitimately encode the dropped context; shift is lossy there.)"""
+    import torch as T
+    from gridllm_amd.models.llama import LlamaModel
+    from gridllm_amd.models.registry import ModelSpec
+
+    spec = ModelSpec("shift-test", hidden=128, layers=1, heads=4,
+                     kv_heads=2, head_dim=32, intermediate=256, vocab=256,
+                     max_seq_len=32)
+    ma = LlamaModel(spec, device="cpu", dtype="fp32", kv_slots=2, max_seq=32)
+    mb = LlamaModel(spec, device="cpu", dtype="fp32", kv_slots=2, max_seq=32)
+
+    T.manual_seed(0)
+    toks = T.randint(0, spec.vocab, (40,))
+    sa = ma.cache.alloc()
+    ma.forward_prefill(toks[:32], sa)
+    keep = ma.shift_slot(sa)           # keeps toks[16:32] at rope pos 16..31
+    assert keep == 16
+    assert int(ma.cache.rope_off[sa].item()) == 16
+    sb = mb.cache.alloc()
+    mb.forward_prefill(toks[16:32], sb)   # same tokens at rope pos 0..15
+    for t in toks[32:36]:
+        la = ma.forward_decode(t.view(1), T.tensor([sa], dtype=T.int32))
+        lb = mb.forward_decode(t.view(1), T.tensor([sb], dtype=T.int32))
+        assert (la - lb).abs().max().item() < 5e-3, \
+            (la - lb).abs().max().item()
