@@ -82,6 +82,23 @@ def _ptr(t: Optional[torch.Tensor]) -> ctypes.c_void_p:
     return ctypes.c_void_p(0 if t is None else t.data_ptr())
 
 
+# Debug validation (the sanitizer analog for a ctypes-bound kernel layer,
+# SURVEY.md §5.2): GRIDLLM_VALIDATE=1 makes every wrapper range-check its
+# index tensors BEFORE launching — a device sync per call, debug only.
+# (A slot_t overflow of exactly the kind this catches cost a GPU box:
+# the bucket>=max_seq graph-warmup fault.)
+_VALIDATE = os.environ.get("GRIDLLM_VALIDATE", "") == "1"
+
+
+def _check_range(name: str, t: torch.Tensor, hi: int) -> None:
+    if not _VALIDATE or t.numel() == 0:
+        return
+    mx, mn = int(t.max().item()), int(t.min().item())
+    if mn < 0 or mx >= hi:
+        raise RuntimeError(
+            f"GRIDLLM_VALIDATE: {name} out of range [{mn},{mx}] vs [0,{hi})")
+
+
 def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float,
             residual: Optional[torch.Tensor] = None,
             residual_out: Optional[torch.Tensor] = None,
@@ -117,6 +134,8 @@ def rope_kv(qkv: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
     lib = _load()
     n = qkv.shape[0]
     smax = kcache.shape[2]
+    _check_range("rope_kv.slot", slot, kcache.shape[0] * smax)
+    _check_range("rope_kv.pos", pos, cos.shape[0])
     q_out = torch.empty(n, hq, d, dtype=qkv.dtype, device=qkv.device)
     _check(lib.gl_rope_kv(_stream(), _ptr(qkv), _ptr(q_out), _ptr(kcache),
                           _ptr(vcache), _ptr(cos), _ptr(sin), _ptr(pos),
@@ -143,6 +162,7 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k_new: torch.Tensor,
     lib = _load()
     n, hk, d = k_new.shape
     smax = kcache.shape[2]
+    _check_range("kv_append.slot", slot, kcache.shape[0] * smax)
     _check(lib.gl_kv_append(_stream(), _ptr(kcache), _ptr(vcache), _ptr(k_new),
                             _ptr(v_new), _ptr(slot), n, hk, d, smax), "kv_append")
 
@@ -162,6 +182,9 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
         # fill the chip: >=512 workgroups, but never splits smaller than 256 keys
         max_useful = max(1, (smax + 255) // 256)
         nsplit = max(1, min(512 // max(1, b * hk), max_useful, 32))
+    _check_range("attn_decode.seqlen", seqlen, smax + 1)
+    if slot_map is not None:
+        _check_range("attn_decode.slot_map", slot_map, kcache.shape[0])
     if out is None:
         out = torch.empty_like(q)
     if nsplit > 1 and ws is None:

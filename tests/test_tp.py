@@ -36,6 +36,12 @@ def _tp_worker(rank: int, world: int, port: int, q):
                 torch.tensor([t], dtype=torch.int64),
                 torch.tensor([slot], dtype=torch.int32))
             decs.append(lg[0])
+        # context shift is per-rank-deterministic: each rank slides its own
+        # KV shard; decode afterwards must still match TP=1
+        model.shift_slot(slot)
+        lg = model.forward_decode(torch.tensor([9], dtype=torch.int64),
+                                  torch.tensor([slot], dtype=torch.int32))
+        decs.append(lg[0])
         if rank == 0:
             q.put(("ok", logits.detach().tolist(), torch.stack(decs).detach().tolist()))
         dist.destroy_process_group()
@@ -58,6 +64,10 @@ def _single():
         lg = model.forward_decode(torch.tensor([t], dtype=torch.int64),
                                   torch.tensor([slot], dtype=torch.int32))
         decs.append(lg[0])
+    model.shift_slot(slot)
+    lg = model.forward_decode(torch.tensor([9], dtype=torch.int64),
+                              torch.tensor([slot], dtype=torch.int32))
+    decs.append(lg[0])
     return logits, torch.stack(decs)
 
 
