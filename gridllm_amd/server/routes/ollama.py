@@ -245,6 +245,65 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
             })
         return JSONResponse({"models": out})
 
+    # ------------------------------------------------- version / show / ps
+    # (beyond reference parity: the reference proxied only
+    # generate/chat/tags/embed — SURVEY.md §2.2-S5 — but Ollama SDKs probe
+    # these on connect, so a drop-in gateway should answer them)
+
+    @r.get("/api/version")
+    async def version():
+        from gridllm_amd import __version__
+        return JSONResponse({"version": __version__})
+
+    @r.post("/api/show")
+    async def show(request: Request):
+        body = await request.json()
+        model = body.get("model") or body.get("name") or ""
+        _validate_model(registry, model)
+        try:
+            from gridllm_amd.models.registry import get_spec
+            spec = get_spec(model)
+            info = {
+                "general.architecture": "llama",
+                "general.parameter_count": spec.params,
+                "llama.block_count": spec.layers,
+                "llama.embedding_length": spec.hidden,
+                "llama.attention.head_count": spec.heads,
+                "llama.attention.head_count_kv": spec.kv_heads,
+                "llama.feed_forward_length": spec.intermediate,
+                "llama.context_length": spec.max_seq_len,
+                "llama.vocab_size": spec.vocab,
+                "llama.rope.freq_base": spec.rope_theta,
+            }
+        except KeyError:
+            info = {}
+        return JSONResponse({
+            "modelfile": "", "parameters": "", "template": "",
+            "details": {"parent_model": "", "format": "safetensors",
+                        "family": "llama", "families": ["llama"],
+                        "parameter_size": "", "quantization_level": "BF16"},
+            "model_info": info,
+            "capabilities": ["completion", "chat", "embedding"],
+        })
+
+    @r.get("/api/ps")
+    async def ps():
+        out = []
+        for w in registry.all_workers():
+            for name in w.models:
+                out.append({
+                    "name": name, "model": name,
+                    "size": 0, "digest": "",
+                    "details": {"family": "llama",
+                                "quantization_level": "BF16"},
+                    "expires_at": _now_iso(),
+                    "size_vram": (w.gpu or {}).get("memory_used_mb", 0)
+                    * 1024 * 1024,
+                    "gridllm_metadata": {"worker_id": w.id,
+                                         "status": w.status},
+                })
+        return JSONResponse({"models": out})
+
     # ---------------------------------------------------------------- embed
 
     @r.post("/api/embed")

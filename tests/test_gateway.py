@@ -291,3 +291,17 @@ def test_metrics_endpoint(client):
     assert "gridllm_jobs_completed_total" in body
     assert "gridllm_workers" in body
     assert "gridllm_ttft_seconds" in body
+
+
+def test_version_show_ps(client):
+    r = client.get("/api/version")
+    assert r.status_code == 200 and "version" in r.json()
+    r = client.post("/api/show", json={"model": "tiny"})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["model_info"]["general.architecture"] == "llama"
+    assert "completion" in body["capabilities"]
+    assert client.post("/api/show", json={"model": "nope"}).status_code == 404
+    r = client.get("/api/ps")
+    assert r.status_code == 200
+    assert any(m["model"] == "tiny" for m in r.json()["models"])
