@@ -124,28 +124,34 @@ class Batcher:
                 except queue.Empty:
                     break
             stop_now = False
+            gens: list[_Pending] = []
             for p in arrivals:
                 if p.kind == "__stop__":
                     stop_now = True
-                    continue
-                if p.kind == "embedding":
+                elif p.kind == "embedding":
                     try:
                         p.future.set_result(self.engine.embed(p.inputs or []))
                     except Exception as e:  # noqa: BLE001
                         p.future.set_exception(e)
-                    continue
-                prompt, opts = self._prep(p)
-                try:
-                    row = self.engine.start_row(prompt, opts, p.on_token,
-                                                p.cancel)
-                except Exception as e:  # noqa: BLE001
-                    log.logger.exception("admission failed")
-                    p.future.set_exception(e)
-                    continue
-                if row.done:
-                    p.future.set_result(row.result)
                 else:
-                    live.append((p, row))
+                    gens.append(p)
+            if gens:
+                items = []
+                for p in gens:
+                    prompt, opts = self._prep(p)
+                    items.append((prompt, opts, p.on_token, p.cancel, None))
+                try:
+                    new_rows = self.engine.start_rows(items)
+                except Exception as e:  # noqa: BLE001
+                    log.logger.exception("batched admission failed")
+                    for p in gens:
+                        p.future.set_exception(e)
+                    new_rows = []
+                for p, row in zip(gens, new_rows):
+                    if row.done:
+                        p.future.set_result(row.result)
+                    else:
+                        live.append((p, row))
             # -- one decode step for everything in flight
             if live:
                 try:

@@ -257,9 +257,9 @@ class InferenceEngine:
         n = len(requests)
         on_tokens = on_tokens or [None] * n
         cancels = cancels or [None] * n
-        rows = [self._start_row(prompt, opts_d, on_tokens[i], cancels[i],
-                                extra_stops)
-                for i, (prompt, opts_d) in enumerate(requests)]
+        rows = self._start_rows(
+            [(prompt, opts_d, on_tokens[i], cancels[i], extra_stops)
+             for i, (prompt, opts_d) in enumerate(requests)])
         active = [r for r in rows if not r.done]
         while active:
             active = self._step_rows(active)
@@ -275,51 +275,78 @@ class InferenceEngine:
                   cancel: Optional[threading.Event] = None,
                   extra_stops: Optional[list[str]] = None) -> _Row:
         with self._lock:
-            return self._start_row(prompt, opts_d, on_token, cancel,
-                                   extra_stops)
+            return self._start_rows([(prompt, opts_d, on_token, cancel,
+                                      extra_stops)])[0]
+
+    def start_rows(self, items: list[tuple]) -> list["_Row"]:
+        """Admit a BATCH of requests: one concatenated prefill pass (GEMMs
+        at M = sum of prompt lengths instead of per-prompt launches), then
+        batched first-token sampling. items: (prompt, opts_d, on_token,
+        cancel, extra_stops) tuples."""
+        with self._lock:
+            return self._start_rows(items)
 
     def step_rows(self, rows: list["_Row"]) -> list["_Row"]:
         with self._lock:
             return self._step_rows(rows)
 
-    def _start_row(self, prompt, opts_d, on_token, cancel,
-                   extra_stops) -> _Row:
-        """Prefill one request, sample+emit its first token. The returned
-        row is retired already (row.done) if it finished on token 1."""
+    def _start_rows(self, items: list[tuple]) -> list["_Row"]:
         t0 = time.monotonic_ns()
-        opts = SamplingOptions.from_dict(opts_d)
-        ids = self._encode(prompt)
-        if opts.num_ctx:
-            ids = ids[-max(16, min(opts.num_ctx, self.model.max_seq - 8)):]
-        slot = self.model.cache.alloc()
-        tokens = torch.as_tensor(ids, dtype=torch.int64, device=self.device)
-        logits = self.model.forward_prefill(tokens, slot)
-        stops = list(opts.stop or [])
-        if extra_stops:
-            stops += extra_stops
-        max_new = opts.num_predict if opts.num_predict and opts.num_predict > 0 \
-            else 128
-        if self.cfg.context_shift:
-            # context shift slides the KV window, so generation is bounded
-            # by the rope table (original absolute positions), not max_seq
-            max_new = min(max_new, self.model.rope_len - len(ids) - 1)
+        rows: list[_Row] = []
+        token_lists: list[torch.Tensor] = []
+        slots: list[int] = []
+        try:
+            for prompt, opts_d, on_token, cancel, extra_stops in items:
+                opts = SamplingOptions.from_dict(opts_d)
+                ids = self._encode(prompt)
+                if opts.num_ctx:
+                    ids = ids[-max(16, min(opts.num_ctx,
+                                           self.model.max_seq - 8)):]
+                slot = self.model.cache.alloc()
+                slots.append(slot)
+                token_lists.append(torch.as_tensor(
+                    ids, dtype=torch.int64, device=self.device))
+                stops = list(opts.stop or [])
+                if extra_stops:
+                    stops += extra_stops
+                max_new = opts.num_predict \
+                    if opts.num_predict and opts.num_predict > 0 else 128
+                if self.cfg.context_shift:
+                    # the KV window slides, so generation is bounded by the
+                    # rope table (original absolute positions), not max_seq
+                    max_new = min(max_new,
+                                  self.model.rope_len - len(ids) - 1)
+                else:
+                    max_new = min(max_new,
+                                  self.model.max_seq - len(ids) - 1)
+                row = _Row(slot=slot,
+                           state=SamplerState.create(opts, self.device),
+                           decoder=self.tokenizer.decode_stream(),
+                           on_token=on_token, cancel=cancel,
+                           stops=stops, max_new=max_new,
+                           result=GenerationResult(),
+                           t_start_ns=t0, cur_len=len(ids))
+                row.result.prompt_eval_count = len(ids)
+                rows.append(row)
+        except Exception:
+            for sl in slots:          # release what this batch allocated
+                self.model.cache.free(sl)
+            raise
+        if len(rows) == 1:
+            logits = self.model.forward_prefill(token_lists[0],
+                                                slots[0]).unsqueeze(0)
         else:
-            max_new = min(max_new, self.model.max_seq - len(ids) - 1)
-        row = _Row(slot=slot, state=SamplerState.create(opts, self.device),
-                   decoder=self.tokenizer.decode_stream(),
-                   on_token=on_token, cancel=cancel,
-                   stops=stops, max_new=max_new, result=GenerationResult(),
-                   t_start_ns=t0, cur_len=len(ids))
-        row.result.prompt_eval_count = len(ids)
-        self._sync_cancel([row])
-        tok = self._sample_batch(logits.unsqueeze(0), [row])[0]  # syncs
+            logits = self.model.forward_prefill_batch(token_lists, slots)
+        self._sync_cancel(rows)
+        toks = self._sample_batch(logits, rows)           # syncs
         t1 = time.monotonic_ns()
-        row.result.prompt_eval_duration = t1 - t0
-        row.t_decode_ns = t1
-        if self._emit(row, tok) and row.max_new > 1:
-            return row
-        self._retire(row)
-        return row
+        per = (t1 - t0) // max(1, len(rows))
+        for row, tok in zip(rows, toks):
+            row.result.prompt_eval_duration = per
+            row.t_decode_ns = t1
+            if not (self._emit(row, tok) and row.max_new > 1):
+                self._retire(row)
+        return rows
 
     def _step_rows(self, active: list["_Row"]) -> list["_Row"]:
         """One batched decode step; returns the rows still running."""

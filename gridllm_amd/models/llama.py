@@ -341,6 +341,57 @@ class LlamaModel:
         return logits[0]
 
     @torch.no_grad()
+    def forward_prefill_batch(self, token_lists: list[torch.Tensor],
+                              slots: list[int]) -> torch.Tensor:
+        """Prefill SEVERAL fresh sequences in one pass: tokens are
+        concatenated so every GEMM runs at M = sum(T_i) instead of k tiny
+        launches (the admission-burst TTFT cost under serving load);
+        attention stays per-sequence against each slot. Returns last-token
+        logits [k, vocab] f32."""
+        s = self.spec
+        lens = [int(t.shape[0]) for t in token_lists]
+        bounds = [0]
+        for ln in lens:
+            bounds.append(bounds[-1] + ln)
+        tokens = torch.cat(token_lists)
+        pos = torch.cat([torch.arange(ln, dtype=torch.int32,
+                                      device=self.device) for ln in lens])
+        slot_t = torch.cat([
+            torch.full((ln,), sl * self.cache.max_seq, dtype=torch.int32,
+                       device=self.device)
+            + torch.arange(ln, dtype=torch.int32, device=self.device)
+            for ln, sl in zip(lens, slots)])
+        x = self.embed[tokens]
+        resid = x
+        normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
+        for li, lw in enumerate(self.layers):
+            qkv = _linear(normed, lw.wqkv)
+            q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
+                            self.cos, self.sin, pos, slot_t,
+                            self.hq_l, self.hkv_l, self.spec.head_dim)
+            attn = torch.empty_like(q)
+            for i, sl in enumerate(slots):
+                attn[bounds[i]:bounds[i + 1]] = ops.attn_prefill_cache(
+                    q[bounds[i]:bounds[i + 1]], self.cache.k[li],
+                    self.cache.v[li], sl, lens[i], 0, self.scale)
+            attn_out = _linear(attn.reshape(bounds[-1], -1), lw.wo)
+            self.tp.all_reduce(attn_out)
+            normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,
+                                        residual=resid)
+            act = self._mlp_act(normed, lw)
+            mlp_out = _linear(act, lw.wdown)
+            self.tp.all_reduce(mlp_out)
+            next_norm = (self.layers[li + 1].attn_norm
+                         if li + 1 < len(self.layers) else self.final_norm)
+            normed, resid = ops.rmsnorm(mlp_out, next_norm, s.norm_eps,
+                                        residual=resid)
+        for ln, sl in zip(lens, slots):
+            self.cache.seq_lens[sl] = ln
+        lasts = torch.tensor([b - 1 for b in bounds[1:]], dtype=torch.long,
+                             device=self.device)
+        return _linear(normed[lasts], self.lm_head).float()
+
+    @torch.no_grad()
     def forward_decode(self, tokens: torch.Tensor,
                        slot_map: torch.Tensor) -> torch.Tensor:
         """One decode step for a batch. tokens: [B] int64; slot_map: [B] int32.

@@ -247,3 +247,38 @@ def test_context_shift_matches_fresh_prefill():
         lb = mb.forward_decode(t.view(1), T.tensor([sb], dtype=T.int32))
         assert (la - lb).abs().max().item() < 5e-3, \
             (la - lb).abs().max().item()
+
+
+def test_batched_prefill_matches_sequential():
+    """forward_prefill_batch (one concatenated pass, per-sequence
+    attention) must produce the same first-token logits as per-sequence
+    prefill — the admission-burst path is numerically the same model."""
+    import torch as T
+    from gridllm_amd.models.llama import LlamaModel
+    from gridllm_amd.models.registry import get_spec
+
+    spec = get_spec("tiny", 64)
+    m1 = LlamaModel(spec, device="cpu", dtype="fp32", kv_slots=6, max_seq=64)
+    m2 = LlamaModel(spec, device="cpu", dtype="fp32", kv_slots=6, max_seq=64)
+    T.manual_seed(1)
+    prompts = [T.randint(0, spec.vocab, (n,)) for n in (5, 11, 3)]
+
+    seq_logits = []
+    seq_slots = []
+    for p in prompts:
+        sl = m1.cache.alloc()
+        seq_slots.append(sl)
+        seq_logits.append(m1.forward_prefill(p, sl))
+    bat_slots = [m2.cache.alloc() for _ in prompts]
+    bat_logits = m2.forward_prefill_batch(prompts, bat_slots)
+
+    for i in range(len(prompts)):
+        d = (seq_logits[i] - bat_logits[i]).abs().max().item()
+        assert d < 1e-4, d
+    # caches identical too (KV written at the same positions)
+    for li in range(spec.layers):
+        for i in range(len(prompts)):
+            n = prompts[i].shape[0]
+            dk = (m1.cache.k[li][seq_slots[i], :, :n]
+                  - m2.cache.k[li][bat_slots[i], :, :n]).abs().max().item()
+            assert dk < 1e-4, dk
