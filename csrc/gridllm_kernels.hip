@@ -7,11 +7,14 @@
 //
 // Design notes (see /opt/skills guides):
 //  * wave64; all bf16 I/O is loaded vectorized (uint4 = 16 B = 8 bf16),
-//    compute in f32.
-//  * memory-bound ops target the HBM3E roofline (~6.3 TB/s achievable).
-//  * attention decode: one workgroup per (batch, kv_head, split) so small
-//    batches still fill 256 CUs; online softmax per wave, LDS combine,
-//    optional f32 partials + combine kernel for split-S.
+//    compute in f32; MFMA (mfma_f32_16x16x32/32x32x16_bf16) for the
+//    matmul-shaped kernels (prefill attention, projection GEMMs).
+//  * memory-bound ops target the HBM3E roofline (~6.3 TB/s achievable);
+//    the M<=512 decode projections are per-CU load-path/latency-bound, so
+//    gemm_proj pipelines both operands through LDS by global_load_lds.
+//  * attention decode: 4-wave cooperative workgroup per (batch, kv_head,
+//    split) for small batches (LDS combine, split-S partials), one-wave
+//    workgroups once B*Hk alone fills the chip.
 //  * exported as extern "C" with an explicit hipStream_t so launches are
 //    capturable in hipGraphs from the Python host layer (ctypes binding —
 //    no torch C++ ABI dependency, no hipify, no CUDA shims).
