@@ -784,18 +784,19 @@ DEV_INLINE float silu_mul(float g, float u) {
 // in 128-row groups [64 gate | 64 up] so each BN=128 block holds matching
 // gate/up column pairs and writes out[M, N/2] = silu(g)*u directly (the
 // separate k_swiglu pass and its gateup HBM round trip disappear).
-template <int MT, int BN, int NB, int EPI = 0>
-__global__ void __launch_bounds__(512)
+template <int MT, int BN, int NB, int EPI = 0, int WAVES = 8, int BK = 64>
+__global__ void __launch_bounds__(WAVES * 64)
 k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
             const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
             int M, int N, int K, int splitk, int kslice) {
-    constexpr int R = 128 * MT;         // x rows staged (>= M)
-    constexpr int BK = 64;
-    constexpr int XI = R / 64;          // x glds per wave per tile
-    constexpr int WI = BN / 64;         // w glds per wave per tile
+    constexpr int R = WAVES * 16 * MT;  // x rows staged (>= M)
+    constexpr int XI = R * BK / (512 * WAVES);   // x glds per wave per tile
+    constexpr int WI = BN * BK / (512 * WAVES);  // w glds per wave per tile
     constexpr int G = XI + WI;
     constexpr int XB = R * BK;          // x region elems per buffer
     constexpr int TB = XB + BN * BK;    // elems per buffer
+    constexpr int RPG = 512 / BK;       // rows per glds instruction
+    constexpr int SWZ = BK / 8 - 1;     // XOR swizzle mask (stay in-row)
 
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -816,23 +817,23 @@ k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
         for (int c = 0; c < BN / 16; ++c)
             acc[t][c] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-    // per-lane glds source precompute: lane covers (sub-row l>>3, 16B l&7)
-    const int lrow = lane >> 3;
-    const int lkb = (lane & 7) * 16;
-    // x instr jj covers tile rows [jj*8, jj*8+8)
+    // per-lane glds source: lane covers (sub-row, 16 B column) of a 1 KiB
+    // piece = RPG rows of BK*2 bytes
+    const int lrow = lane / (BK / 8);
+    const int lkb = (lane % (BK / 8)) * 16;
     const char* xsrc[XI];
 #pragma unroll
     for (int i = 0; i < XI; ++i) {
-        const int row = (wid * XI + i) * 8 + lrow;
+        const int row = (wid * XI + i) * RPG + lrow;
         xsrc[i] = (const char*)(x + (long)min(row, M - 1) * K) +
-                  (lkb ^ ((row & 7) << 4));
+                  (lkb ^ ((row & SWZ) << 4));
     }
     const char* wsrc[WI];
 #pragma unroll
     for (int i = 0; i < WI; ++i) {
-        const int row = (wid * WI + i) * 8 + lrow;
+        const int row = (wid * WI + i) * RPG + lrow;
         wsrc[i] = (const char*)(w + (long)(n0 + row) * K) +
-                  (lkb ^ ((row & 7) << 4));
+                  (lkb ^ ((row & SWZ) << 4));
     }
 
     auto issue = [&](int buf, int kt) {
@@ -857,8 +858,11 @@ k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
                 16, 0, 0);
     };
 
-    issue(0, 0);
-    if (NB > 2 && nt > 1) issue(1, 1);
+    // pre-issue NB-1 tiles: the steady-state wait leaves NB-2 tiles in
+    // flight, which only holds if that many were launched ahead
+#pragma unroll
+    for (int u = 0; u < NB - 1; ++u)
+        if (u < nt) issue(u, u);
 
     for (int kt = 0; kt < nt; ++kt) {
         const int cur = kt % NB;
@@ -877,19 +881,19 @@ k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
         const bf16_t* xb = lds + cur * TB;
         const bf16_t* wb = lds + cur * TB + XB;
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < BK / 32; ++kc) {
             bf16x8 b[BN / 16];
 #pragma unroll
             for (int c = 0; c < BN / 16; ++c) {
                 const int row = c * 16 + fr;
                 b[c] = *(const bf16x8*)((const char*)(wb + row * BK) +
-                    (((kc * 32 + kg) * 2) ^ ((row & 7) << 4)));
+                    (((kc * 32 + kg) * 2) ^ ((row & SWZ) << 4)));
             }
 #pragma unroll
             for (int t = 0; t < MT; ++t) {
                 const int row = (wid * MT + t) * 16 + fr;
                 const bf16x8 a = *(const bf16x8*)((const char*)(xb + row * BK) +
-                    (((kc * 32 + kg) * 2) ^ ((row & 7) << 4)));
+                    (((kc * 32 + kg) * 2) ^ ((row & SWZ) << 4)));
 #pragma unroll
                 for (int c = 0; c < BN / 16; ++c)
                     acc[t][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -934,13 +938,13 @@ k_gemm_proj(bf16_t* __restrict__ out, float* __restrict__ ws,
     }
 }
 
-template <int MT, int BN, int NB, int EPI = 0>
+template <int MT, int BN, int NB, int EPI = 0, int WAVES = 8, int BK = 64>
 static void launch_gemm_proj(hipStream_t st, bf16_t* out, float* ws,
                              const bf16_t* x, const bf16_t* w, int M, int N,
                              int K, int splitk, int kslice) {
-    dim3 grid((unsigned)((N / BN) * splitk)), blk(512);
-    k_gemm_proj<MT, BN, NB, EPI><<<grid, blk, 0, st>>>(out, ws, x, w, M, N,
-                                                       K, splitk, kslice);
+    dim3 grid((unsigned)((N / BN) * splitk)), blk(WAVES * 64);
+    k_gemm_proj<MT, BN, NB, EPI, WAVES, BK><<<grid, blk, 0, st>>>(
+        out, ws, x, w, M, N, K, splitk, kslice);
 }
 
 // Combine split-K slabs of an INTERLEAVED gate/up projection with the
@@ -1350,10 +1354,26 @@ int gl_gemm_skinny(void* stream, void* out, void* ws, const void* x,
 
 int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
                  const void* w, int M, int N, int K, int splitk,
-                 int kslice, int fuse_swiglu) {
+                 int kslice, int fuse_swiglu, int variant) {
     if (K % 64 != 0 || kslice % 64 != 0) return -100;
     if (M > 512) return -104;
     hipStream_t st = (hipStream_t)stream;
+    if (variant == 1) {
+        // wide variant: 16 waves x BN=256 x BK=32 — halves the per-CU
+        // x-row traffic by amortizing over twice the columns
+        if (M > 256 || N % 256 != 0 || fuse_swiglu) return -106;
+        launch_gemm_proj<1, 256, 4, 0, 16, 32>(st, (bf16_t*)out, (float*)ws,
+            (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk, kslice);
+        LAUNCH_CHECK();
+        if (splitk > 1) {
+            long mn = (long)M * N;
+            int grid = (int)min(((mn + 3) / 4 + 255) / 256, (long)2048);
+            k_gemm_combine<<<dim3(grid), dim3(256), 0, st>>>(
+                (bf16_t*)out, (const float*)ws, mn, splitk);
+            LAUNCH_CHECK();
+        }
+        return 0;
+    }
     const int mt = (M + 127) / 128;     // 16-row x-tiles per wave (8 waves)
     if (mt <= 2 && N % 128 != 0) return -100;
     if (mt > 2 && N % 64 != 0) return -100;

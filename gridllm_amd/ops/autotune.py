@@ -82,6 +82,19 @@ def tune_shape(m: int, n: int, k: int, device) -> tuple[str, int]:
             if t < t_best * 0.97:
                 t_best = t
                 best = ("proj", sk)
+    if hip_ops.gemm_proj_wide_supported(m, n, k):
+        rule = max(1, min(k // 256, 256 // max(1, n // 256)))
+        for sk in sorted({1, rule, rule * 2}):
+            if sk > max(1, k // 64):
+                continue
+            try:
+                t = _time_fn(lambda w: hip_ops.gemm_proj_wide(x, w, splitk=sk),
+                             ws)
+            except RuntimeError:
+                continue
+            if t < t_best * 0.97:
+                t_best = t
+                best = ("projw", sk)
     del ws
     _CHOICE[key] = best
     return best
@@ -140,6 +153,8 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
                 return hip_ops.gemm_skinny(x, w, splitk=c[1])
             if c[0] == "proj":
                 return hip_ops.gemm_proj(x, w, splitk=c[1])
+            if c[0] == "projw":
+                return hip_ops.gemm_proj_wide(x, w, splitk=c[1])
     return F.linear(x, w)
 
 

@@ -52,7 +52,7 @@ def _load() -> ctypes.CDLL:
                                    c_i, c_i, c_i, c_i, c_i]
     lib.gl_gemm_skinny.restype = c_i
     lib.gl_gemm_proj.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp,
-                                 c_i, c_i, c_i, c_i, c_i, c_i]
+                                 c_i, c_i, c_i, c_i, c_i, c_i, c_i]
     lib.gl_gemm_proj.restype = c_i
     lib.gl_attn_prefill.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_i, c_i,
                                     c_i, c_i, c_i, c_i, c_i, c_i, c_f]
@@ -279,7 +279,35 @@ def gemm_proj(x: torch.Tensor, w: torch.Tensor,
         ws = torch.empty(splitk * m * n, dtype=torch.float32, device=x.device)
     _check(lib.gl_gemm_proj(_stream(), _ptr(out), _ptr(ws), _ptr(x), _ptr(w),
                             m, n, k, splitk, kslice,
-                            1 if fuse_swiglu else 0), "gemm_proj")
+                            1 if fuse_swiglu else 0, 0), "gemm_proj")
+    return out
+
+
+def gemm_proj_wide_supported(m: int, n: int, k: int) -> bool:
+    return 64 < m <= 256 and n % 256 == 0 and k % 64 == 0
+
+
+def gemm_proj_wide(x: torch.Tensor, w: torch.Tensor,
+                   out: Optional[torch.Tensor] = None,
+                   splitk: int = 0) -> torch.Tensor:
+    """16-wave x BN=256 x BK=32 projection variant: amortizes the staged
+    x rows over twice the columns (per-CU load-path relief). The autotuner
+    decides in-situ whether it beats the 8-wave tile / hipBLASLt."""
+    lib = _load()
+    m, k = x.shape
+    n = w.shape[0]
+    ntiles = max(1, n // 256)
+    if splitk <= 0:
+        splitk = max(1, min(k // 256, (256 + ntiles - 1) // ntiles))
+    kslice = ((k + splitk - 1) // splitk + 63) // 64 * 64
+    splitk = (k + kslice - 1) // kslice
+    if out is None:
+        out = torch.empty(m, n, dtype=x.dtype, device=x.device)
+    ws = None
+    if splitk > 1:
+        ws = torch.empty(splitk * m * n, dtype=torch.float32, device=x.device)
+    _check(lib.gl_gemm_proj(_stream(), _ptr(out), _ptr(ws), _ptr(x), _ptr(w),
+                            m, n, k, splitk, kslice, 0, 1), "gemm_proj_wide")
     return out
 
 

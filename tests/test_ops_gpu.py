@@ -283,3 +283,22 @@ def test_gemm_proj_fused_swiglu_gpu(dev):
     # the library-GEMM fallback path over the same interleaved weight
     o_blas = hip_ops.swiglu_il(torch.nn.functional.linear(x, w_il))
     assert _mx(o_blas.cpu(), ref.cpu()) < 0.02 * ref.abs().max().item() + 1e-6
+
+
+@pytest.mark.parametrize("m,n,k", [
+    (256, 5120, 3072), (128, 3072, 8192), (200, 16384, 3072),
+    (65, 256, 64),
+])
+def test_gemm_proj_wide_gpu(dev, m, n, k):
+    """16-wave BN=256 variant vs fp32 matmul + splitk consistency."""
+    torch.manual_seed(14)
+    from gridllm_amd.ops import hip_ops
+    x = (torch.randn(m, k, device=dev) * 0.1).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) * 0.1).to(torch.bfloat16)
+    out = hip_ops.gemm_proj_wide(x, w)
+    ref = x.float() @ w.float().t()
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.02 * ref.abs().max().item() + 1e-6, err
+    o1 = hip_ops.gemm_proj_wide(x, w, splitk=1)
+    o4 = hip_ops.gemm_proj_wide(x, w, splitk=4)
+    assert (o1.float() - o4.float()).abs().max().item() < 0.05
