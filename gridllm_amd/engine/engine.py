@@ -191,10 +191,8 @@ class InferenceEngine:
         states = [r.state for r in rows]
         if all(s.greedy for s in states):
             return logits.argmax(dim=-1).tolist()
-        if all(s.counts is None for s in states):
-            from gridllm_amd.engine.sampler import sample_batch_vectorized
-            return sample_batch_vectorized(logits, states)
-        return [sample(logits[i], s) for i, s in enumerate(states)]
+        from gridllm_amd.engine.sampler import sample_batch_vectorized
+        return sample_batch_vectorized(logits, states)
 
     # ------------------------------------------------------------ main loop
 

@@ -95,13 +95,32 @@ def sample(logits: torch.Tensor, state: SamplerState) -> int:
 
 def sample_batch_vectorized(logits: torch.Tensor,
                             states: list["SamplerState"]) -> list[int]:
-    """Batched sampling for penalty-free rows: batched top_k/top_p/min_p
-    filtering + per-row Gumbel noise (seeded rows use their own generator,
-    issued without a sync) + ONE argmax + ONE host copy. The per-row
-    `sample()` loop costs a GPU sync per ROW per step — at serving batch
-    sizes that multiplies the decode step several-fold."""
+    """Batched sampling: penalties (per-row tensor ops, no syncs) ->
+    temperature -> batched top_k/top_p/min_p filtering -> per-row Gumbel
+    noise (seeded rows use their own generator, issued without a sync) ->
+    ONE argmax + ONE host copy. The per-row `sample()` loop costs a GPU
+    sync per ROW per step — at serving batch sizes that multiplies the
+    decode step several-fold."""
     dev = logits.device
     b, v = logits.shape
+    # penalties on RAW logits (before temperature), matching sample().
+    # NOTE: mutates `logits` rows in place — callers pass per-step decode
+    # outputs that are fully rewritten before their next use.
+    for i, s in enumerate(states):
+        if s.counts is None:
+            continue
+        o = s.opts
+        seen = s.counts > 0
+        xi = logits[i]
+        if o.repeat_penalty and o.repeat_penalty != 1.0:
+            pos = xi > 0
+            xi = torch.where(seen & pos, xi / o.repeat_penalty, xi)
+            xi = torch.where(seen & ~pos, xi * o.repeat_penalty, xi)
+        if o.presence_penalty:
+            xi = xi - o.presence_penalty * seen.float()
+        if o.frequency_penalty:
+            xi = xi - o.frequency_penalty * s.counts
+        logits[i] = xi
     temps = torch.tensor(
         [max(s.opts.temperature if s.opts.temperature is not None else 0.8,
              1e-4) for s in states], device=dev).unsqueeze(1)

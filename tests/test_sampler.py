@@ -110,3 +110,20 @@ def test_vectorized_mixed_greedy():
     out = sample_batch_vectorized(
         lg, [make_state(temperature=0.0), make_state(temperature=1.0, seed=3)])
     assert out[0] == 7
+
+
+def test_vectorized_penalties_match_single_row():
+    """The batched penalty math must agree with the single-row sampler
+    (greedy rows make the comparison deterministic)."""
+    from gridllm_amd.engine.sampler import sample_batch_vectorized
+    lg = logits_fixed()
+    for kw in ({"repeat_penalty": 1e9}, {"frequency_penalty": 2.0},
+               {"presence_penalty": 1.5}):
+        s1 = make_state(temperature=0.0, **kw)
+        s2 = make_state(temperature=0.0, **kw)
+        for _ in range(3):
+            s1.observe(7, 64)
+            s2.observe(7, 64)
+        single = sample(lg.clone(), s1)
+        batched = sample_batch_vectorized(lg.clone().unsqueeze(0), [s2])[0]
+        assert single == batched, (kw, single, batched)
