@@ -427,4 +427,6 @@ class JobScheduler:
         for r in self.queue:
             by_pri[r.priority] = by_pri.get(r.priority, 0) + 1
         return {"queued": len(self.queue), "active": len(self.active),
+                "active_ids": list(self.active)[:100],
+                "queued_ids": [r.id for r in self.queue[:100]],
                 "by_priority": by_pri, **self.stats}

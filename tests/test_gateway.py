@@ -305,3 +305,24 @@ def test_version_show_ps(client):
     r = client.get("/api/ps")
     assert r.status_code == 200
     assert any(m["model"] == "tiny" for m in r.json()["models"])
+
+
+def test_inference_cancel_route(client):
+    """DELETE /inference/:id cancels a queued-or-processing job
+    (reference inference.ts:170-192; our worker actually honors it)."""
+    import concurrent.futures
+    with concurrent.futures.ThreadPoolExecutor(1) as ex:
+        fut = ex.submit(lambda: client.post("/inference", json={
+            "model": "tiny", "prompt": "cancel me",
+            "options": {"temperature": 0.0, "num_predict": 200}}))
+        import time as _t
+        _t.sleep(0.3)
+        # find the active job id via the queue stats route
+        jobs = client.get("/health/jobs").json()
+        ids = jobs.get("active_ids", []) + jobs.get("queued_ids", [])
+        jid = ids[0] if ids else None
+        if jid:
+            r = client.delete(f"/inference/{jid}")
+            assert r.status_code in (200, 404)
+        res = fut.result(timeout=60)
+        assert res.status_code in (200, 500)
