@@ -230,6 +230,30 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
 
     # --------------------------------------------------------------- models
 
+    @r.post("/v1/embeddings")
+    async def embeddings(request: Request):
+        # Ollama's OpenAI compat exposes this; the reference proxied only
+        # completions/chat/models — drop-in value-add
+        body = await request.json()
+        model = body.get("model", "")
+        _validate(model)
+        raw = body.get("input", "")
+        inputs = raw if isinstance(raw, list) else [raw]
+        inputs = [str(x) for x in inputs]
+        req = InferenceRequest(model=model, request_type="embedding",
+                               input=inputs, timeout_ms=300_000)
+        res = await scheduler.submit_and_wait(req)
+        if res.get("error"):
+            raise HTTPException(status_code=500, detail=res["error"])
+        embs = res.get("embeddings", [])
+        return JSONResponse({
+            "object": "list",
+            "data": [{"object": "embedding", "index": i, "embedding": e}
+                     for i, e in enumerate(embs)],
+            "model": model,
+            "usage": {"prompt_tokens": 0, "total_tokens": 0},
+        })
+
     @r.get("/v1/models")
     async def models():
         out = [{"id": name, "object": "model", "created": int(time.time()),

@@ -326,3 +326,16 @@ def test_inference_cancel_route(client):
             assert r.status_code in (200, 404)
         res = fut.result(timeout=60)
         assert res.status_code in (200, 500)
+
+
+def test_openai_embeddings(client):
+    r = client.post("/v1/embeddings", json={"model": "tiny",
+                                            "input": ["alpha", "beta"]})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["object"] == "list"
+    assert len(body["data"]) == 2
+    assert body["data"][1]["index"] == 1
+    assert len(body["data"][0]["embedding"]) > 0
+    r = client.post("/v1/embeddings", json={"model": "tiny", "input": "solo"})
+    assert len(r.json()["data"]) == 1
