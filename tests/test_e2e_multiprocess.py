@@ -132,7 +132,8 @@ def cluster():
             time.sleep(0.1)
         else:
             raise TimeoutError("workers never registered")
-    yield {"base": base, "server": server, "workers": workers}
+    yield {"base": base, "server": server, "workers": workers,
+           "bus_port": bus_port}
     for p in list(workers.values()) + [server]:
         if p.poll() is None:
             p.send_signal(signal.SIGKILL)
@@ -207,3 +208,46 @@ def test_chaos_sigkill_busy_worker_orphan_requeue(cluster):
                 break
             time.sleep(0.2)
         assert "w-a" not in names
+
+
+def test_chaos_sigkill_during_streaming(cluster):
+    """SIGKILL a worker mid-STREAM: the orphaned job requeues and the
+    ORIGINAL streaming HTTP response keeps going until a final done record
+    arrives from the survivor (the job:result subscriber
+    survives the worker)."""
+    # w-a was killed by the previous test on some orderings; use whatever
+    # workers remain and spawn a fresh disposable one to kill
+    victim = _spawn(WORKER_SRC,
+                    {"BUS_PORT": str(cluster["bus_port"]),
+                     "WORKER_ID": "w-victim", "WORKER_SLOW": "0.2"},
+                    "WORKER_READY")
+    try:
+        time.sleep(1.0)
+        import concurrent.futures as cf
+
+        def stream_req(i):
+            lines = []
+            with httpx.Client(base_url=cluster["base"], timeout=180) as c:
+                with c.stream("POST", "/api/generate", json={
+                        "model": "tiny", "prompt": f"s{i}", "stream": True,
+                        "options": {"temperature": 0.0,
+                                    "num_predict": 40}}) as r:
+                    for ln in r.iter_lines():
+                        if ln:
+                            lines.append(json.loads(ln))
+            return lines
+
+        with cf.ThreadPoolExecutor(6) as ex:
+            futs = [ex.submit(stream_req, i) for i in range(6)]
+            time.sleep(2.0)                     # streams are mid-decode
+            victim.send_signal(signal.SIGKILL)
+            victim.wait(timeout=10)
+            results = [f.result() for f in futs]
+        for lines in results:
+            assert lines, "stream produced nothing"
+            assert lines[-1].get("done") is True
+            assert not lines[-1].get("error"), lines[-1]
+    finally:
+        if victim.poll() is None:
+            victim.send_signal(signal.SIGKILL)
+            victim.wait(timeout=10)
