@@ -453,7 +453,11 @@ class InferenceEngine:
     # ------------------------------------------------------------- metadata
 
     def model_names(self) -> list[str]:
-        return [self.spec.name]
+        from gridllm_amd.models.registry import names_for
+        try:
+            return names_for(self.spec.name)
+        except KeyError:
+            return [self.spec.name]
 
     def stats(self) -> dict:
         return {

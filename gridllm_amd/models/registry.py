@@ -78,3 +78,16 @@ def get_spec(name: str, max_seq_len: int = 0) -> ModelSpec:
 
 def list_specs() -> list[str]:
     return sorted({s.name for s in _SPECS.values()})
+
+
+def register_spec(spec: ModelSpec, *aliases: str) -> None:
+    """Register a custom architecture at runtime (tests, custom configs)."""
+    _reg(spec, *aliases)
+
+
+def names_for(name: str) -> list[str]:
+    """Canonical name + every alias of a model — what a worker should
+    advertise so clients can use any Ollama-style spelling (the reference
+    advertised Ollama's full tag list, WorkerClientService.ts:129-154)."""
+    spec = get_spec(name)
+    return sorted({k for k, v in _SPECS.items() if v.name == spec.name})

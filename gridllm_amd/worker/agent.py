@@ -41,13 +41,28 @@ log = get_logger("worker")
 class WorkerAgent:
     def __init__(self, bus: Bus, engine: InferenceEngine,
                  cfg: Optional[WorkerConfig] = None,
-                 batcher: Optional[Batcher] = None):
+                 batcher: Optional[Batcher] = None,
+                 extra_engines: Optional[list[InferenceEngine]] = None):
+        """One worker can serve SEVERAL models (the reference advertised
+        Ollama's whole tag list, WorkerClientService.ts:129-154): pass
+        extra_engines and jobs dispatch to the engine whose model matches.
+        Each engine gets its own continuous batcher."""
         self.bus = bus
         self.engine = engine
         self.cfg = cfg or WorkerConfig()
         self.worker_id = self.cfg.worker_id or \
             f"worker-{socket.gethostname()}-{self.cfg.device}"
         self.batcher = batcher or Batcher(engine)
+        # model name (incl. aliases) -> (engine, batcher)
+        self._by_model: dict[str, tuple[InferenceEngine, Batcher]] = {}
+        for name in engine.model_names():
+            self._by_model[name] = (engine, self.batcher)
+        self._extra_batchers: list[Batcher] = []
+        for eng in (extra_engines or []):
+            b = Batcher(eng)
+            self._extra_batchers.append(b)
+            for name in eng.model_names():
+                self._by_model.setdefault(name, (eng, b))
         self.current_jobs = 0
         self._cancel_events: dict[str, threading.Event] = {}
         self._tasks: list[asyncio.Task] = []
@@ -88,6 +103,8 @@ class WorkerAgent:
         except Exception:  # noqa: BLE001
             pass
         self.batcher.shutdown()
+        for b in self._extra_batchers:
+            b.shutdown()
 
     def _info(self) -> WorkerInfo:
         stats = telemetry.gpu_stats(self.cfg.device)
@@ -100,7 +117,7 @@ class WorkerAgent:
         return WorkerInfo(
             id=self.worker_id,
             status=status,
-            models=self.engine.model_names(),
+            models=sorted(self._by_model),
             max_concurrent_jobs=self.cfg.max_concurrent_tasks,
             current_jobs=self.current_jobs,
             performance_tier=telemetry.performance_tier(stats),
@@ -250,9 +267,15 @@ class WorkerAgent:
             pump_task = loop.create_task(
                 self._pump_stream(req.id, buf, lk, done_ev))
         try:
+            pair = self._by_model.get(req.model)
+            if pair is None:
+                raise RuntimeError(
+                    f"model {req.model!r} not served by this worker "
+                    f"(has: {sorted(self._by_model)})")
+            _, batcher = pair
             if req.request_type == "embedding":
                 inputs = req.input if req.input is not None else [req.prompt]
-                fut = self.batcher.submit_embedding(inputs)
+                fut = batcher.submit_embedding(inputs)
                 embeddings = await asyncio.wrap_future(fut)
                 payload = {
                     "jobId": req.id, "workerId": self.worker_id, "done": True,
@@ -261,12 +284,12 @@ class WorkerAgent:
                 }
             else:
                 if req.request_type == "chat" and req.messages:
-                    fut = self.batcher.submit_chat(
+                    fut = batcher.submit_chat(
                         req.messages, req.options,
                         on_token=on_token if req.stream else None,
                         cancel=cancel)
                 else:
-                    fut = self.batcher.submit_generate(
+                    fut = batcher.submit_generate(
                         req.prompt, req.options,
                         on_token=on_token if req.stream else None,
                         cancel=cancel)

@@ -62,10 +62,20 @@ def build_health_app(agent: WorkerAgent):
 
 async def run_worker(wcfg: WorkerConfig, ecfg: EngineConfig) -> None:
     engine = InferenceEngine(ecfg)
+    # one worker can serve several models on the same GPU (288 GB HBM):
+    # ENGINE_MODELS="llama3.2:3b,llama3:8b" adds engines beyond ENGINE_MODEL
+    extra = []
+    more = [m.strip() for m in
+            os.environ.get("ENGINE_MODELS", "").split(",") if m.strip()]
+    for name in more:
+        if name == ecfg.model:
+            continue
+        import dataclasses
+        extra.append(InferenceEngine(dataclasses.replace(ecfg, model=name)))
     disconnected = asyncio.Event()
     bus = TcpBus(wcfg.bus_host, wcfg.bus_port,
                  on_disconnect=lambda: disconnected.set())
-    agent = WorkerAgent(bus, engine, wcfg)
+    agent = WorkerAgent(bus, engine, wcfg, extra_engines=extra)
     if wcfg.health_port > 0:
         import uvicorn
         ucfg = uvicorn.Config(build_health_app(agent), host="0.0.0.0",

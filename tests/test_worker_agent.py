@@ -231,3 +231,50 @@ async def test_worker_health_app():
             assert "engine" in r.json()
     finally:
         await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_multi_model_worker():
+    """ONE worker serving several models (the reference advertised Ollama's
+    whole tag list): jobs dispatch to the engine matching req.model, and
+    aliases are advertised so any Ollama-style spelling routes."""
+    from gridllm_amd.models.registry import ModelSpec, register_spec
+    register_spec(ModelSpec("tiny2", hidden=128, layers=1, heads=2,
+                            kv_heads=1, head_dim=64, intermediate=256,
+                            vocab=300, max_seq_len=128),
+                  "tiny2:latest")
+    core = MemoryBusCore()
+    bus_s = MemoryBus(core)
+    await bus_s.connect()
+    reg = WorkerRegistry(bus_s, FAST)
+    sched = JobScheduler(bus_s, reg, FAST)
+    await reg.initialize()
+    await sched.initialize()
+    bus_w = MemoryBus(core)
+    await bus_w.connect()
+    e2 = InferenceEngine(EngineConfig(
+        model="tiny2", device="cpu", dtype="fp32", max_seq_len=64,
+        max_batch_size=2, use_hipgraph=False))
+    agent = WorkerAgent(bus_w, get_engine(),
+                        WorkerConfig(worker_id="w-multi",
+                                     max_concurrent_tasks=4,
+                                     heartbeat_interval_ms=500),
+                        extra_engines=[e2])
+    await agent.start()
+    await asyncio.sleep(0.05)
+    try:
+        info = reg.get("w-multi")
+        assert "tiny" in info.models and "tiny2" in info.models
+        assert "tiny2:latest" in info.models        # alias advertised
+        r1 = await asyncio.wait_for(sched.submit_and_wait(
+            InferenceRequest(model="tiny", prompt="a",
+                             options={"temperature": 0.0,
+                                      "num_predict": 3})), 30)
+        r2 = await asyncio.wait_for(sched.submit_and_wait(
+            InferenceRequest(model="tiny2:latest", prompt="b",
+                             options={"temperature": 0.0,
+                                      "num_predict": 3})), 30)
+        assert r1.get("response") is not None
+        assert r2.get("response") is not None and not r2.get("error")
+    finally:
+        await teardown(reg, sched, agent)
