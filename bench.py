@@ -67,13 +67,19 @@ def main() -> None:
                        use_hipgraph=on_gpu)
     eng = InferenceEngine(cfg)
 
-    # --- synthetic prompts: prefill `batch` slots with prompt_len random tokens
+    # --- synthetic prompts: prefill `batch` slots with prompt_len random
+    # tokens (batched prefill: chunks of 64 sequences per pass)
     torch.manual_seed(1000 + rank)
     slots = [eng.model.cache.alloc() for _ in range(batch)]
     vocab = eng.spec.vocab
-    for s in slots:
-        ids = torch.randint(0, vocab, (args.prompt_len,), device=eng.device)
-        eng.model.forward_prefill(ids, s)
+    for i in range(0, batch, 64):
+        chunk = slots[i:i + 64]
+        toks = [torch.randint(0, vocab, (args.prompt_len,), device=eng.device)
+                for _ in chunk]
+        if len(chunk) > 1:
+            eng.model.forward_prefill_batch(toks, chunk)
+        else:
+            eng.model.forward_prefill(toks[0], chunk[0])
 
     opts = SamplingOptions.from_dict({"temperature": 0.0})
 
