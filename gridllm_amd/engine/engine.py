@@ -213,22 +213,30 @@ class InferenceEngine:
 
     def embed(self, inputs: list[str]) -> list[list[float]]:
         out: list[list[float]] = []
-        with self._lock:
-            for text in inputs:
-                ids = self._encode(text)
-                slot = self.model.cache.alloc()
-                try:
-                    tokens = torch.as_tensor(ids, dtype=torch.int64,
-                                             device=self.device)
-                    _, hidden = self.model.forward_prefill(
-                        tokens, slot, return_hidden=True)
-                    vec = hidden.detach().float().cpu()
-                    norm = vec.norm()
-                    if norm > 0:
-                        vec = vec / norm
-                    out.append(vec.tolist())
-                finally:
-                    self.model.cache.free(slot)
+        for i0 in range(0, len(inputs), 16):      # bound KV-slot pressure
+            chunk = inputs[i0:i0 + 16]
+            with self._lock:
+                token_lists = [torch.as_tensor(self._encode(t),
+                                               dtype=torch.int64,
+                                               device=self.device)
+                               for t in chunk]
+                nslots = min(len(chunk),
+                             max(1, self.model.cache.n_free))
+                hiddens = []
+                for j0 in range(0, len(chunk), nslots):
+                    part = token_lists[j0:j0 + nslots]
+                    slots = [self.model.cache.alloc() for _ in part]
+                    try:
+                        _, h = self.model.forward_prefill_batch(
+                            part, slots, return_hidden=True)
+                        hiddens.append(h)
+                    finally:
+                        for sl in slots:
+                            self.model.cache.free(sl)
+                hidden = torch.cat(hiddens)
+            for vec in hidden.detach().cpu():
+                n = vec.norm()
+                out.append((vec / n if n > 0 else vec).tolist())
         return out
 
     def _encode(self, prompt: str) -> list[int]:

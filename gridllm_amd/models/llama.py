@@ -342,12 +342,14 @@ class LlamaModel:
 
     @torch.no_grad()
     def forward_prefill_batch(self, token_lists: list[torch.Tensor],
-                              slots: list[int]) -> torch.Tensor:
+                              slots: list[int],
+                              return_hidden: bool = False) -> torch.Tensor:
         """Prefill SEVERAL fresh sequences in one pass: tokens are
         concatenated so every GEMM runs at M = sum(T_i) instead of k tiny
         launches (the admission-burst TTFT cost under serving load);
         attention stays per-sequence against each slot. Returns last-token
-        logits [k, vocab] f32."""
+        logits [k, vocab] f32; with return_hidden, (logits, mean-pooled
+        final hidden [k, H] f32) — the embedding path."""
         s = self.spec
         lens = [int(t.shape[0]) for t in token_lists]
         bounds = [0]
@@ -389,7 +391,13 @@ class LlamaModel:
             self.cache.seq_lens[sl] = ln
         lasts = torch.tensor([b - 1 for b in bounds[1:]], dtype=torch.long,
                              device=self.device)
-        return _linear(normed[lasts], self.lm_head).float()
+        logits = _linear(normed[lasts], self.lm_head).float()
+        if return_hidden:
+            hidden = torch.stack([
+                normed[bounds[i]:bounds[i + 1]].float().mean(dim=0)
+                for i in range(len(token_lists))])
+            return logits, hidden
+        return logits
 
     @torch.no_grad()
     def forward_decode(self, tokens: torch.Tensor,
