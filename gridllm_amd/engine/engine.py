@@ -62,6 +62,7 @@ class _Row:
     t_start_ns: int = 0            # wall clock at admission
     t_decode_ns: int = 0           # wall clock when prefill+first token done
     cur_len: int = 0               # storage length in the KV slot
+    constraint: object = None      # grammar automaton (format="json")
 
 
 class InferenceEngine:
@@ -189,10 +190,27 @@ class InferenceEngine:
                       rows: Sequence[_Row]) -> list[int]:
         """Batched sampling; vectorized for the common uniform cases."""
         states = [r.state for r in rows]
+        for i, r in enumerate(rows):
+            if r.constraint is not None:
+                self._apply_constraint_mask(logits[i], r.constraint)
         if all(s.greedy for s in states):
             return logits.argmax(dim=-1).tolist()
         from gridllm_amd.engine.sampler import sample_batch_vectorized
         return sample_batch_vectorized(logits, states)
+
+    def _apply_constraint_mask(self, row_logits: torch.Tensor,
+                               constraint) -> None:
+        """Mask row logits to the automaton's legal next bytes (+EOS when a
+        complete value has been emitted). Byte tokenizer: ids 0..255 are
+        raw bytes. In-place; no host sync."""
+        allowed, eos_ok = constraint.allowed()
+        idx = list(allowed)
+        if eos_ok:
+            idx.append(self.tokenizer.eos_id)
+        keep = torch.full_like(row_logits, float("-inf"))
+        ix = torch.tensor(idx, dtype=torch.long, device=row_logits.device)
+        keep[ix] = row_logits[ix]
+        row_logits.copy_(keep)
 
     # ------------------------------------------------------------ main loop
 
@@ -325,13 +343,21 @@ class InferenceEngine:
                 else:
                     max_new = min(max_new,
                                   self.model.max_seq - len(ids) - 1)
+                constraint = None
+                if opts.format == "json" and self.tokenizer.vocab_size and \
+                        getattr(self.tokenizer, "BOS", None) is not None:
+                    # byte-level tokenizer -> exact byte-grammar masking
+                    from gridllm_amd.engine.constraints import \
+                        JsonByteConstraint
+                    constraint = JsonByteConstraint()
                 row = _Row(slot=slot,
                            state=SamplerState.create(opts, self.device),
                            decoder=self.tokenizer.decode_stream(),
                            on_token=on_token, cancel=cancel,
                            stops=stops, max_new=max_new,
                            result=GenerationResult(),
-                           t_start_ns=t0, cur_len=len(ids))
+                           t_start_ns=t0, cur_len=len(ids),
+                           constraint=constraint)
                 row.result.prompt_eval_count = len(ids)
                 rows.append(row)
         except Exception:
@@ -404,6 +430,12 @@ class InferenceEngine:
         res.eval_count += 1
         row.last_token = tok
         row.state.observe(tok, self.spec.vocab)
+        if row.constraint is not None and 0 <= tok < 256:
+            try:
+                row.constraint.push(tok)
+            except ValueError:
+                res.done_reason = "stop"   # automaton refused (shouldn't
+                return False               # happen with masking in place)
         if self._cancelled(row):
             res.done_reason = "cancelled"
             res.cancelled = True

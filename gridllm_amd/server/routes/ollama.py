@@ -152,10 +152,14 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
             return JSONResponse(resp)
 
         _validate_model(registry, model)
+        opts = _clean_options(body.get("options"))
+        if body.get("format") == "json":
+            opts["format"] = "json"    # engine enforces via a byte-level
+                                       # JSON automaton (engine/constraints)
         req = InferenceRequest(
             model=model, prompt=prompt, request_type="generate",
             stream=stream, priority=str(body.get("priority", "medium")),
-            timeout_ms=300_000, options=_clean_options(body.get("options")),
+            timeout_ms=300_000, options=opts,
             metadata={k: body[k] for k in
                       ("suffix", "images", "think", "format", "system",
                        "template", "raw", "keep_alive", "context")

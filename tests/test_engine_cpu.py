@@ -282,3 +282,27 @@ def test_batched_prefill_matches_sequential():
             dk = (m1.cache.k[li][seq_slots[i], :, :n]
                   - m2.cache.k[li][bat_slots[i], :, :n]).abs().max().item()
             assert dk < 1e-4, dk
+
+
+def test_format_json_constrained_output():
+    """format="json": the byte-level automaton masks sampling so emitted
+    text is always a syntactically valid JSON prefix, and EOS is only
+    reachable at a complete top-level value."""
+    from gridllm_amd.engine.constraints import JsonByteConstraint
+    cfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                       max_seq_len=128, max_batch_size=2, use_hipgraph=False)
+    eng = InferenceEngine(cfg)
+    for seed in (1, 2, 3):
+        r = eng.generate("give me json",
+                         {"temperature": 1.0, "seed": seed,
+                          "num_predict": 48, "format": "json"})
+        # replay the emitted bytes through a fresh automaton: every byte
+        # must be legal in sequence (i.e. the text is a valid JSON prefix)
+        c = JsonByteConstraint()
+        for by in r.text.encode():
+            allowed, _ = c.allowed()
+            assert by in allowed, (r.text, chr(by))
+            c.push(by)
+        if r.done_reason == "stop" and r.token_ids[-1] == eng.tokenizer.eos_id:
+            _, eos_ok = c.allowed()
+            assert eos_ok, r.text

@@ -339,3 +339,20 @@ def test_openai_embeddings(client):
     assert len(body["data"][0]["embedding"]) > 0
     r = client.post("/v1/embeddings", json={"model": "tiny", "input": "solo"})
     assert len(r.json()["data"]) == 1
+
+
+def test_generate_format_json(client):
+    """Ollama's format="json" is ENFORCED by the engine (the reference
+    just forwarded the flag to Ollama)."""
+    r = client.post("/api/generate", json={
+        "model": "tiny", "prompt": "produce json", "stream": False,
+        "format": "json",
+        "options": {"temperature": 1.0, "seed": 5, "num_predict": 32}})
+    assert r.status_code == 200, r.text
+    text = r.json()["response"]
+    from gridllm_amd.engine.constraints import JsonByteConstraint
+    c = JsonByteConstraint()
+    for by in text.encode():
+        allowed, _ = c.allowed()
+        assert by in allowed, (text, chr(by))
+        c.push(by)
