@@ -201,10 +201,12 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                                 detail="messages must be a non-empty array")
         _validate_model(registry, model)
         stream = bool(body.get("stream", True))
+        opts = _clean_options(body.get("options"))
+        if body.get("format") == "json":
+            opts["format"] = "json"
         req = InferenceRequest(
             model=model, messages=messages, request_type="chat",
-            stream=stream, timeout_ms=300_000,
-            options=_clean_options(body.get("options")))
+            stream=stream, timeout_ms=300_000, options=opts)
 
         def chunk_json(c):
             return {"model": model, "created_at": _now_iso(),
