@@ -176,7 +176,7 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
             return {"model": model, "created_at": _now_iso(), "response": "",
                     "done": True,
                     "done_reason": res.get("done_reason", "stop"),
-                    "context": [], **_timings(res)}
+                    "context": res.get("context") or [], **_timings(res)}
 
         if stream:
             return _stream_ndjson(scheduler, req, chunk_json, final_json)
@@ -186,7 +186,8 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         return JSONResponse({
             "model": model, "created_at": _now_iso(),
             "response": res.get("response", ""), "done": True,
-            "done_reason": res.get("done_reason", "stop"), "context": [],
+            "done_reason": res.get("done_reason", "stop"),
+            "context": res.get("context") or [],
             **_timings(res)})
 
     # ----------------------------------------------------------------- chat

@@ -272,7 +272,18 @@ class WorkerAgent:
                 raise RuntimeError(
                     f"model {req.model!r} not served by this worker "
                     f"(has: {sorted(self._by_model)})")
-            _, batcher = pair
+            engine, batcher = pair
+            # Ollama conversation continuation: `context` carries the prior
+            # exchange's token ids; prepend their text to the prompt
+            # (byte tokenizer -> exact roundtrip)
+            ctx_ids = (req.metadata or {}).get("context")
+            if ctx_ids and req.request_type != "embedding":
+                try:
+                    prefix = engine.tokenizer.decode(
+                        [int(t) for t in ctx_ids])
+                    req.prompt = prefix + req.prompt
+                except Exception:  # noqa: BLE001
+                    pass
             if req.request_type == "embedding":
                 inputs = req.input if req.input is not None else [req.prompt]
                 fut = batcher.submit_embedding(inputs)
@@ -300,6 +311,11 @@ class WorkerAgent:
                     "response": result.text,
                     "done_reason": result.done_reason,
                     "cancelled": result.cancelled,
+                    # Ollama context = ids encoding the WHOLE conversation
+                    # (req.prompt already carries the decoded prior context)
+                    "context": engine.tokenizer.encode(
+                        req.prompt + result.text, bos=False)
+                    if req.request_type != "chat" else [],
                     "timings": {
                         "total_duration": result.total_duration,
                         "load_duration": 0,

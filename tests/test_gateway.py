@@ -356,3 +356,22 @@ def test_generate_format_json(client):
         allowed, _ = c.allowed()
         assert by in allowed, (text, chr(by))
         c.push(by)
+
+
+def test_generate_context_roundtrip(client):
+    """Ollama conversation continuation: the returned `context` ids feed the
+    next request and deterministically extend the same conversation."""
+    opts = {"temperature": 0.0, "num_predict": 6}
+    r1 = client.post("/api/generate", json={
+        "model": "tiny", "prompt": "hello", "stream": False,
+        "options": opts}).json()
+    assert r1["context"], r1
+    r2 = client.post("/api/generate", json={
+        "model": "tiny", "prompt": " again", "stream": False,
+        "context": r1["context"], "options": opts}).json()
+    assert r2["response"] is not None and r2["context"]
+    # equivalent to one concatenated prompt
+    r3 = client.post("/api/generate", json={
+        "model": "tiny", "prompt": "hello" + r1["response"] + " again",
+        "stream": False, "options": opts}).json()
+    assert r2["response"] == r3["response"], (r2["response"], r3["response"])
