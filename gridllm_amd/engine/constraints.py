@@ -156,6 +156,8 @@ class JsonByteConstraint:
         if m == EXP_SIGNED:
             self.mode = EXP
             return
+        if m == EXP:
+            return                                     # digit: stay
         if m == AFTER:
             if b in _WS:
                 return
