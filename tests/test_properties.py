@@ -77,3 +77,41 @@ def test_sampler_respects_filters(seed, temp, top_k, top_p):
     s2 = SamplerState.create(opts, torch.device("cpu"))
     tok2 = sample_batch_vectorized(lg.clone().unsqueeze(0), [s2])[0]
     assert tok2 in topk_ids
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.text(max_size=40))
+def test_stream_decoder_reassembles_utf8(text):
+    """Pushing a text's bytes one at a time through the streaming decoder
+    must reproduce the text exactly (multi-byte code points are held back
+    until complete)."""
+    from gridllm_amd.engine.tokenizer import ByteTokenizer
+    tok = ByteTokenizer()
+    dec = tok.decode_stream()
+    out = []
+    for by in text.encode("utf-8"):
+        out.append(dec.push(by))
+    out.append(dec.flush())
+    assert "".join(out) == text
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.sampled_from(["generate", "chat", "embedding"]),
+       st.sampled_from(["high", "medium", "low"]),
+       st.text(max_size=20), st.booleans(),
+       st.integers(min_value=0, max_value=2**31 - 1))
+def test_request_wire_roundtrip(rtype, pri, prompt, stream, seed):
+    """InferenceRequest survives the bus (to_dict -> from_dict identity on
+    every field the protocol carries)."""
+    from gridllm_amd.types import InferenceRequest
+    req = InferenceRequest(model="m", prompt=prompt, request_type=rtype,
+                           priority=pri, stream=stream,
+                           options={"seed": seed, "temperature": 0.5},
+                           metadata={"context": [1, 2, 3]})
+    back = InferenceRequest.from_dict(req.to_dict())
+    assert back.id == req.id
+    assert back.model == req.model and back.prompt == req.prompt
+    assert back.request_type == rtype and back.priority == pri
+    assert back.stream == stream
+    assert back.options == req.options
+    assert back.metadata == req.metadata
