@@ -304,7 +304,7 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                     "details": {"family": "llama",
                                 "quantization_level": "BF16"},
                     "expires_at": _now_iso(),
-                    "size_vram": (w.gpu or {}).get("memory_used_mb", 0)
+                    "size_vram": (w.gpu or {}).get("hbm_used_mb", 0)
                     * 1024 * 1024,
                     "gridllm_metadata": {"worker_id": w.id,
                                          "status": w.status},
