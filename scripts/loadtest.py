@@ -60,7 +60,8 @@ from gridllm_amd.worker.agent import WorkerAgent
 
 async def main():
     conc = int(os.environ["LT_CONC"])
-    wcfg = WorkerConfig(worker_id="lt-worker",
+    wid = os.environ.get("LT_WID", "lt-worker")
+    wcfg = WorkerConfig(worker_id=wid,
                         bus_port=int(os.environ["BUS_PORT"]),
                         max_concurrent_tasks=conc)
     ecfg = EngineConfig(model=os.environ["LT_MODEL"],
@@ -168,6 +169,8 @@ def main():
     ap.add_argument("--num-predict", type=int, default=64)
     ap.add_argument("--prompt-words", type=int, default=24)
     ap.add_argument("--max-seq", type=int, default=1024)
+    ap.add_argument("--workers", type=int, default=1,
+                    help="DP workers; worker i pins cuda:i")
     args = ap.parse_args()
 
     import torch
@@ -183,8 +186,15 @@ def main():
            "LT_MAXSEQ": str(args.max_seq)}
     print("spawning server...", file=sys.stderr, flush=True)
     server = _spawn(SERVER_SRC, env, "SERVER_READY")
-    print("spawning worker...", file=sys.stderr, flush=True)
-    worker = _spawn(WORKER_SRC, env, "WORKER_READY")
+    workers = []
+    for i in range(args.workers):
+        wenv = dict(env)
+        wenv["LT_WID"] = f"lt-worker-{i}"
+        if args.device.startswith("cuda"):
+            wenv["LT_DEVICE"] = f"cuda:{i}"
+        print(f"spawning worker {i}...", file=sys.stderr, flush=True)
+        workers.append(_spawn(WORKER_SRC, wenv, "WORKER_READY"))
+    worker = workers[0]
     try:
         base = f"http://127.0.0.1:{http_port}"
         # registration wait
@@ -192,7 +202,8 @@ def main():
         with httpx.Client(base_url=base, timeout=30) as c:
             for _ in range(200):
                 r = c.get("/health/workers")
-                if r.status_code == 200 and r.json()["counts"]["total"] >= 1:
+                if r.status_code == 200 and \
+                        r.json()["counts"]["total"] >= args.workers:
                     break
                 time.sleep(0.1)
         print("workers registered; warmup...", file=sys.stderr, flush=True)
@@ -217,13 +228,14 @@ def main():
             if ttfts else None,
             "model": args.model,
             "device": args.device,
+            "workers": args.workers,
         }
         print(json.dumps(out))
     finally:
-        for p in (worker, server):
+        for p in workers + [server]:
             if p.poll() is None:
                 p.send_signal(signal.SIGKILL)
-        for p in (worker, server):
+        for p in workers + [server]:
             p.wait(timeout=10)
 
 
