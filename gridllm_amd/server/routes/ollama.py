@@ -165,7 +165,8 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                        "template", "raw", "keep_alive", "context")
                       if k in body},
         )
-        if body.get("system"):
+        if body.get("system") and not body.get("raw"):
+            # raw=true bypasses all prompt shaping (Ollama semantics)
             req.prompt = f"{body['system']}\n\n{req.prompt}"
 
         def chunk_json(c):
