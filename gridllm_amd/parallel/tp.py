@@ -64,6 +64,14 @@ def init_tp_from_env(backend: str = "") -> TPContext:
     if not dist.is_initialized():
         if not backend:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
-        dist.init_process_group(backend=backend)
+        # bounded collectives make TP-group failure ATOMIC (SURVEY.md §7
+        # hard-part 5): if any rank dies, the others' collectives time out
+        # and abort the whole process -> the registry removes the worker
+        # and in-flight jobs orphan-requeue, instead of hanging forever.
+        # (RCCL honors this via the process-group watchdog.)
+        from datetime import timedelta
+        timeout_s = int(os.environ.get("TP_COLLECTIVE_TIMEOUT_S", "120"))
+        dist.init_process_group(backend=backend,
+                                timeout=timedelta(seconds=timeout_s))
     return TPContext(rank=dist.get_rank(), size=dist.get_world_size(),
                      group=None)
