@@ -86,6 +86,7 @@ def main() -> None:
     class _R:  # minimal row shims for the engine's batched sampler
         def __init__(self):
             self.state = SamplerState.create(opts, eng.device)
+            self.constraint = None
     rows = [_R() for _ in range(batch)]
     tokens = [int(torch.randint(0, vocab, (1,)).item()) for _ in range(batch)]
 
