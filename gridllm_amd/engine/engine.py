@@ -191,8 +191,9 @@ class InferenceEngine:
         """Batched sampling; vectorized for the common uniform cases."""
         states = [r.state for r in rows]
         for i, r in enumerate(rows):
-            if r.constraint is not None:
-                self._apply_constraint_mask(logits[i], r.constraint)
+            c = getattr(r, "constraint", None)   # tolerate row shims
+            if c is not None:
+                self._apply_constraint_mask(logits[i], c)
         if all(s.greedy for s in states):
             return logits.argmax(dim=-1).tolist()
         from gridllm_amd.engine.sampler import sample_batch_vectorized
