@@ -27,6 +27,11 @@ class SamplerState:
     device: torch.device
     # token frequency for penalties (lazy, vocab-sized)
     counts: Optional[torch.Tensor] = None
+    # counter-based RNG stream for the BATCHED sampler: per-row noise is a
+    # pure function of (noise_seed, draws), so the whole batch draws with
+    # ONE tensor expression instead of a torch.rand launch per row
+    noise_seed: int = 0
+    draws: int = 0
 
     @classmethod
     def create(cls, opts: SamplingOptions, device: torch.device) -> "SamplerState":
@@ -34,7 +39,10 @@ class SamplerState:
         if opts.seed is not None:
             gen = torch.Generator(device=device)
             gen.manual_seed(int(opts.seed))
-        return cls(opts=opts, gen=gen, device=device)
+            noise_seed = int(opts.seed)
+        else:
+            noise_seed = int(torch.randint(0, 2**62, (1,)).item())
+        return cls(opts=opts, gen=gen, device=device, noise_seed=noise_seed)
 
     @property
     def greedy(self) -> bool:
@@ -42,6 +50,7 @@ class SamplerState:
         return t is not None and t <= 0.0
 
     def observe(self, token_id: int, vocab: int) -> None:
+        self.draws += 1
         o = self.opts
         if (o.repeat_penalty or o.presence_penalty or o.frequency_penalty):
             if self.counts is None:
@@ -148,10 +157,21 @@ def sample_batch_vectorized(logits: torch.Tensor,
         cnt = ((cum - sp) < tps).sum(1, keepdim=True).clamp(min=1)
         thr = torch.maximum(thr, sp.gather(1, cnt - 1))
         x = torch.where(probs >= thr, x, torch.full_like(x, float("-inf")))
-    noise = torch.zeros_like(x)
-    for i, s in enumerate(states):
-        if not s.greedy:
-            u = torch.rand(v, generator=s.gen, device=dev)
-            inner = (-torch.log(u.clamp_min(1e-20))).clamp_min(1e-20)
-            noise[i] = -torch.log(inner)
-    return (x + noise).argmax(-1).tolist()
+    # batched Gumbel noise from a counter-based hash (splitmix64-style):
+    # deterministic per (seed, draw index), zero python-per-row launches
+    keys = torch.tensor(
+        [0 if s.greedy else
+         ((s.noise_seed * 6364136223846793005 +
+           (s.draws + 1) * 1442695040888963407) & (2**63 - 1))
+         for s in states], device=dev, dtype=torch.int64).unsqueeze(1)
+    idx = torch.arange(v, device=dev, dtype=torch.int64).unsqueeze(0)
+    z = (idx ^ keys) * -7046029254386353131        # 0x9E3779B97F4A7C15
+    z = z ^ (z >> 29)
+    z = z * -4658895280553007687                   # 0xBF58476D1CE4E5B9
+    z = z ^ (z >> 32)
+    u = ((z >> 38) & 0xFFFFFF).to(torch.float32) / float(1 << 24)
+    u = u.clamp(1e-6, 1.0 - 1e-6)
+    noise = -torch.log(-torch.log(u))
+    greedy_mask = torch.tensor([1.0 if not s.greedy else 0.0
+                                for s in states], device=dev).unsqueeze(1)
+    return (x + noise * greedy_mask).argmax(-1).tolist()
