@@ -25,8 +25,10 @@ class SamplerState:
     opts: SamplingOptions
     gen: Optional[torch.Generator]
     device: torch.device
-    # token frequency for penalties (lazy, vocab-sized)
-    counts: Optional[torch.Tensor] = None
+    # token -> count for penalties (SPARSE: a generation sees few unique
+    # tokens; dense [vocab] tensors per row cost memory and per-row GPU
+    # launches at serving batch sizes)
+    counts: Optional[dict] = None
     # counter-based RNG stream for the BATCHED sampler: per-row noise is a
     # pure function of (noise_seed, draws), so the whole batch draws with
     # ONE tensor expression instead of a torch.rand launch per row
@@ -54,24 +56,28 @@ class SamplerState:
         o = self.opts
         if (o.repeat_penalty or o.presence_penalty or o.frequency_penalty):
             if self.counts is None:
-                self.counts = torch.zeros(vocab, dtype=torch.float32,
-                                          device=self.device)
-            self.counts[token_id] += 1
+                self.counts = {}
+            self.counts[token_id] = self.counts.get(token_id, 0) + 1
 
 
 def sample(logits: torch.Tensor, state: SamplerState) -> int:
     """Pick one token id from logits [vocab] (f32)."""
     o = state.opts
-    if state.counts is not None:
-        seen = state.counts > 0
+    if state.counts:
+        toks = torch.tensor(list(state.counts), dtype=torch.long,
+                            device=logits.device)
+        cnts = torch.tensor([float(c) for c in state.counts.values()],
+                            device=logits.device)
+        vals = logits[toks]
         if o.repeat_penalty and o.repeat_penalty != 1.0:
-            pos = logits > 0
-            logits = torch.where(seen & pos, logits / o.repeat_penalty, logits)
-            logits = torch.where(seen & ~pos, logits * o.repeat_penalty, logits)
+            vals = torch.where(vals > 0, vals / o.repeat_penalty,
+                               vals * o.repeat_penalty)
         if o.presence_penalty:
-            logits = logits - o.presence_penalty * seen.float()
+            vals = vals - o.presence_penalty
         if o.frequency_penalty:
-            logits = logits - o.frequency_penalty * state.counts
+            vals = vals - o.frequency_penalty * cnts
+        logits = logits.clone()
+        logits[toks] = vals
 
     temp = o.temperature if o.temperature is not None else 0.8
     if temp <= 0.0:
@@ -113,23 +119,41 @@ def sample_batch_vectorized(logits: torch.Tensor,
     dev = logits.device
     b, v = logits.shape
     # penalties on RAW logits (before temperature), matching sample().
-    # NOTE: mutates `logits` rows in place — callers pass per-step decode
-    # outputs that are fully rewritten before their next use.
+    # ONE global sparse scatter for the whole batch: (row, token) pairs
+    # gathered from every penalized row's seen-token dict — no per-row
+    # GPU launches. Mutates `logits` rows in place (callers pass per-step
+    # decode outputs that are fully rewritten before their next use).
+    pi: list[int] = []
+    pt: list[int] = []
+    pc: list[float] = []
+    prp: list[float] = []
+    ppp: list[float] = []
+    pfp: list[float] = []
     for i, s in enumerate(states):
-        if s.counts is None:
+        if not s.counts:
             continue
         o = s.opts
-        seen = s.counts > 0
-        xi = logits[i]
-        if o.repeat_penalty and o.repeat_penalty != 1.0:
-            pos = xi > 0
-            xi = torch.where(seen & pos, xi / o.repeat_penalty, xi)
-            xi = torch.where(seen & ~pos, xi * o.repeat_penalty, xi)
-        if o.presence_penalty:
-            xi = xi - o.presence_penalty * seen.float()
-        if o.frequency_penalty:
-            xi = xi - o.frequency_penalty * s.counts
-        logits[i] = xi
+        rp = o.repeat_penalty if o.repeat_penalty else 1.0
+        pp = o.presence_penalty or 0.0
+        fp = o.frequency_penalty or 0.0
+        for t_, c_ in s.counts.items():
+            pi.append(i)
+            pt.append(t_)
+            pc.append(float(c_))
+            prp.append(rp)
+            ppp.append(pp)
+            pfp.append(fp)
+    if pi:
+        ii = torch.tensor(pi, dtype=torch.long, device=dev)
+        tt = torch.tensor(pt, dtype=torch.long, device=dev)
+        cc = torch.tensor(pc, device=dev)
+        rp_t = torch.tensor(prp, device=dev)
+        pp_t = torch.tensor(ppp, device=dev)
+        fp_t = torch.tensor(pfp, device=dev)
+        vals = logits[ii, tt]
+        vals = torch.where(vals > 0, vals / rp_t, vals * rp_t)
+        vals = vals - pp_t - fp_t * cc
+        logits[ii, tt] = vals
     temps = torch.tensor(
         [max(s.opts.temperature if s.opts.temperature is not None else 0.8,
              1e-4) for s in states], device=dev).unsqueeze(1)
