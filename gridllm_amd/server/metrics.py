@@ -33,6 +33,11 @@ class _Metrics:
             "gridllm_ttft_seconds", "Time to first streamed token",
             buckets=(.005, .01, .025, .05, .1, .25, .5, 1, 2.5, 5, 10, 30),
             registry=self.reg)
+        self.queue_wait = Histogram(
+            "gridllm_queue_wait_seconds", "Enqueue-to-assignment latency "
+            "(the reference's floor was its 1 s dispatch tick)",
+            buckets=(.001, .005, .01, .05, .1, .5, 1, 5, 30),
+            registry=self.reg)
         self.queue_depth = Gauge("gridllm_queue_depth", "Queued jobs",
                                  registry=self.reg)
         self.active_jobs = Gauge("gridllm_active_jobs", "Jobs executing",
@@ -59,6 +64,11 @@ def init_metrics(scheduler, worker_registry) -> None:
 def observe_ttft(seconds: float) -> None:
     if _M is not None:
         _M.ttft.observe(seconds)
+
+
+def observe_queue_wait(seconds: float) -> None:
+    if _M is not None:
+        _M.queue_wait.observe(seconds)
 
 
 def add_tokens(n) -> None:

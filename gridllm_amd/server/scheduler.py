@@ -230,6 +230,8 @@ class JobScheduler:
         await self.bus.publish(Channels.worker_job(worker.id),
                                {"type": "job_assignment", "job": req.to_dict()})
         self._arm_timeout(a)
+        from gridllm_amd.server import metrics
+        metrics.observe_queue_wait(max(0, now_ms() - req.created_at) / 1000.0)
         log.job("job assigned", job_id=req.id, worker_id=worker.id,
                 model=req.model, priority=req.priority)
         return True

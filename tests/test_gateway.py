@@ -291,6 +291,7 @@ def test_metrics_endpoint(client):
     assert "gridllm_jobs_completed_total" in body
     assert "gridllm_workers" in body
     assert "gridllm_ttft_seconds" in body
+    assert "gridllm_queue_wait_seconds" in body
 
 
 def test_version_show_ps(client):
