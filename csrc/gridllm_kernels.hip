@@ -1218,7 +1218,10 @@ static int launch_attn_decode(hipStream_t st, bf16_t* out, const bf16_t* q,
                               float* ws, int B, int Hk, int G, int Smax,
                               int nsplit, float scale) {
     dim3 grid((unsigned)(B * Hk * nsplit));
-    const bool onewave = (B * Hk * nsplit) >= 2048;   // grid fills the chip
+    // one-wave variant only on the unsplit path (the auto-nsplit rule
+    // yields nsplit==1 whenever B*Hk >= 512, so this is the common case;
+    // explicit nsplit>1 calls keep the 4-wave cooperative kernel)
+    const bool onewave = nsplit == 1 && (B * Hk) >= 2048;
     switch (G) {   // compile-time G keeps softmax state in registers
 #define CASE_G(GG) \
         case GG: if (onewave) k_attn_decode1<D, GG><<<grid, dim3(64), 0, st>>>( \
