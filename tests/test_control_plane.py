@@ -341,3 +341,29 @@ async def test_mixed_fleet_model_aware_routing():
     assert set(w_big.jobs_seen) <= {r.id for r in reqs if r.model == "llama3:70b"}
     assert not (set(w_small.jobs_seen) & set(w_big.jobs_seen))
     await teardown(reg, sched, w_small, w_big, w_both)
+
+
+@pytest.mark.asyncio
+async def test_registry_model_index_and_alias_counts():
+    """get_all_models aggregates every advertised name (incl. aliases)
+    with worker counts; availability respects capacity and PAUSED."""
+    core = MemoryBusCore()
+    bus, reg, sched = await make_stack(core)
+    w1 = FakeWorker(core, "w1", models=("m1", "m1:latest"))
+    w2 = FakeWorker(core, "w2", models=("m1", "m2"), concurrency=2)
+    await w1.start()
+    await w2.start()
+    await asyncio.sleep(0.1)
+    models = reg.get_all_models()
+    assert models["m1"] == 2
+    assert models["m1:latest"] == 1
+    assert models["m2"] == 1
+    assert len(reg.get_available_workers_by_model("m1")) == 2
+    # a paused worker leaves the schedulable pool but stays registered
+    reg.workers["w1"].status = "paused"
+    assert len(reg.get_available_workers_by_model("m1")) == 1
+    assert len(reg.get_workers_by_model("m1")) == 2
+    # capacity exhaustion removes from the available set
+    reg.workers["w2"].current_jobs = 2
+    assert not reg.get_available_workers_by_model("m2")
+    await teardown(reg, sched, w1, w2)
