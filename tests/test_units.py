@@ -1,0 +1,71 @@
+"""Direct unit tests for leaf modules (logging, telemetry, chat template,
+model registry API)."""
+
+import json
+import logging
+
+import pytest
+
+from gridllm_amd.engine import chat as chat_mod
+from gridllm_amd.models.registry import get_spec, list_specs, names_for
+from gridllm_amd.utils.logging import JsonFormatter, get_logger
+from gridllm_amd.worker import telemetry
+
+
+def test_json_formatter_structured_and_unserializable():
+    fmt = JsonFormatter()
+    rec = logging.LogRecord("gridllm.x", logging.INFO, __file__, 1,
+                            "hello %s", ("world",), None)
+    rec.gridllm = {"type": "job", "job_id": "j1"}
+    out = json.loads(fmt.format(rec))
+    assert out["msg"] == "hello world"
+    assert out["type"] == "job" and out["job_id"] == "j1"
+    # unserializable extra falls back to repr instead of raising
+    rec2 = logging.LogRecord("gridllm.x", logging.INFO, __file__, 1,
+                             "x", (), None)
+    rec2.gridllm = {"type": "worker", "obj": object()}
+    out2 = json.loads(fmt.format(rec2))
+    assert "extra" in out2 or "obj" in out2
+
+
+def test_tagged_logger_helpers(caplog):
+    log = get_logger("unit")
+    with caplog.at_level(logging.INFO, logger="gridllm.unit"):
+        log.job("did a thing", job_id="abc")
+    assert any("did a thing" in r.message for r in caplog.records)
+
+
+def test_telemetry_cpu_fallback():
+    stats = telemetry.gpu_stats(0)
+    d = stats.to_dict()
+    assert "name" in d
+    assert telemetry.performance_tier(stats) in ("high", "medium", "low")
+
+
+def test_chat_template_roles_and_multimodal():
+    txt = chat_mod.render_chat([
+        {"role": "system", "content": "be terse"},
+        {"role": "user", "content": [
+            {"type": "text", "text": "hi"},
+            {"type": "image_url", "image_url": {"url": "x"}}]},
+        {"role": "assistant", "content": "ok"},
+        {"role": "user", "content": "more"},
+    ])
+    assert txt.startswith("<|begin_of_text|>")
+    assert "<|start_header_id|>system<|end_header_id|>\n\nbe terse" in txt
+    assert "hi" in txt and "image_url" not in txt
+    assert txt.endswith("<|start_header_id|>assistant<|end_header_id|>\n\n")
+
+
+def test_registry_api():
+    spec = get_spec("llama3.2")              # alias resolves
+    assert spec.name == "llama3.2:3b"
+    assert "llama3.2:3b" in names_for("llama3.2:latest")
+    assert "llama3.2:3b" in list_specs()
+    with pytest.raises(KeyError):
+        get_spec("not-a-model")
+    # max_seq override returns a copy, not a mutated global
+    s1 = get_spec("tiny", 32)
+    s2 = get_spec("tiny")
+    assert s1.max_seq_len == 32 and s2.max_seq_len != 32 or \
+        s2.max_seq_len == 256
