@@ -89,21 +89,25 @@ def attn_decode(q: torch.Tensor, kcache: torch.Tensor, vcache: torch.Tensor,
     bsz, hq, d = q.shape
     hk = kcache.shape[1]
     g = hq // hk
-    out = torch.empty_like(q)
-    for b in range(bsz):
-        n = int(seqlen[b])
-        if n == 0:
-            out[b].zero_()
-            continue
-        bc = int(slot_map[b]) if slot_map is not None else b
-        kk = kcache[bc, :, :n].float()                  # [Hk,n,D]
-        vv = vcache[bc, :, :n].float()
-        qq = q[b].float().view(hk, g, d)                # [Hk,G,D]
-        scores = torch.einsum("hgd,hnd->hgn", qq, kk) * scale
-        p = torch.softmax(scores, dim=-1)
-        o = torch.einsum("hgn,hnd->hgd", p, vv)
-        out[b] = o.reshape(hq, d).to(q.dtype)
-    return out
+    idx = slot_map.long() if slot_map is not None else \
+        torch.arange(bsz, device=q.device)
+    n = seqlen.long()
+    nmax = int(n.max()) if bsz else 0
+    if nmax == 0:
+        return torch.zeros_like(q)
+    kk = kcache[idx, :, :nmax].float()                  # [B,Hk,nmax,D]
+    vv = vcache[idx, :, :nmax].float()
+    qq = q.float().view(bsz, hk, g, d)
+    scores = torch.einsum("bhgd,bhnd->bhgn", qq, kk) * scale
+    mask = torch.arange(nmax, device=q.device).view(1, 1, 1, nmax) >= \
+        n.view(bsz, 1, 1, 1)
+    scores = scores.masked_fill(mask, float("-inf"))
+    p = torch.softmax(scores, dim=-1)
+    o = torch.einsum("bhgn,bhnd->bhgd", p, vv)
+    empty = n == 0                                      # all -inf -> nan
+    if bool(empty.any()):
+        o[empty] = 0
+    return o.reshape(bsz, hq, d).to(q.dtype)
 
 
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
