@@ -196,8 +196,12 @@ class InferenceEngine:
                 self._apply_constraint_mask(logits[i], c)
         if all(s.greedy for s in states):
             return logits.argmax(dim=-1).tolist()
-        from gridllm_amd.engine.sampler import sample_batch_vectorized
-        return sample_batch_vectorized(logits, states)
+        from gridllm_amd.engine.sampler import (BatchPlan,
+                                                sample_batch_vectorized)
+        plan = getattr(self, "_sampler_plan", None)
+        if plan is None:
+            plan = self._sampler_plan = BatchPlan()
+        return sample_batch_vectorized(logits, states, plan)
 
     def _apply_constraint_mask(self, row_logits: torch.Tensor,
                                constraint) -> None:

@@ -38,6 +38,21 @@ from gridllm_amd.worker import telemetry
 log = get_logger("worker")
 
 
+class _Stream:
+    """One streaming job's token buffer, drained by the agent's single
+    shared flusher task."""
+    __slots__ = ("job_id", "buf", "lk", "done_ev", "sent_any", "fin")
+
+    def __init__(self, job_id: str, buf: list, lk: threading.Lock,
+                 done_ev: threading.Event):
+        self.job_id = job_id
+        self.buf = buf
+        self.lk = lk
+        self.done_ev = done_ev
+        self.sent_any = False
+        self.fin = asyncio.Event()     # set after the final chunk publishes
+
+
 class WorkerAgent:
     def __init__(self, bus: Bus, engine: InferenceEngine,
                  cfg: Optional[WorkerConfig] = None,
@@ -65,6 +80,9 @@ class WorkerAgent:
                 self._by_model.setdefault(name, (eng, b))
         self.current_jobs = 0
         self._cancel_events: dict[str, threading.Event] = {}
+        self._streams: dict[str, "_Stream"] = {}
+        self._flush_task: Optional[asyncio.Task] = None
+        self._flush_wake = asyncio.Event()
         self._tasks: list[asyncio.Task] = []
         self._job_tasks: set[asyncio.Task] = set()
         self._last_status: tuple = ()
@@ -262,10 +280,9 @@ class WorkerAgent:
             if done:
                 done_ev.set()
 
-        pump_task = None
+        stream = None
         if req.stream:
-            pump_task = loop.create_task(
-                self._pump_stream(req.id, buf, lk, done_ev))
+            stream = self._register_stream(req.id, buf, lk, done_ev)
         try:
             pair = self._by_model.get(req.model)
             if pair is None:
@@ -328,10 +345,14 @@ class WorkerAgent:
                 if req.request_type == "chat":
                     payload["message"] = {"role": "assistant",
                                           "content": result.text}
-            if pump_task is not None:
+            if stream is not None:
                 done_ev.set()                  # flush remaining and stop
-                await pump_task
-                pump_task = None
+                try:                           # flusher sets fin after the
+                    await asyncio.wait_for(    # done-chunk publish
+                        stream.fin.wait(), timeout=10.0)
+                except asyncio.TimeoutError:
+                    self._streams.pop(req.id, None)
+                stream = None
             await self.bus.publish(Channels.JOB_COMPLETED, {
                 "jobId": req.id, "workerId": self.worker_id})
             await self.bus.publish(Channels.job_result(req.id), payload)
@@ -347,8 +368,8 @@ class WorkerAgent:
             except Exception:  # noqa: BLE001
                 pass
         finally:
-            if pump_task is not None:
-                pump_task.cancel()
+            if stream is not None:             # error path: drop the stream
+                self._streams.pop(req.id, None)
             self._cancel_events.pop(req.id, None)
             self.current_jobs = max(0, self.current_jobs - 1)
             try:
@@ -356,33 +377,52 @@ class WorkerAgent:
             except Exception:  # noqa: BLE001
                 pass
 
-    async def _pump_stream(self, job_id: str, buf: list,
-                           lk: threading.Lock,
-                           done_ev: threading.Event) -> None:
+    def _register_stream(self, job_id: str, buf: list, lk: threading.Lock,
+                         done_ev: threading.Event) -> _Stream:
+        s = _Stream(job_id, buf, lk, done_ev)
+        self._streams[job_id] = s
+        self._flush_wake.set()         # break out of a long flush sleep
+        if self._flush_task is None or self._flush_task.done():
+            self._flush_task = asyncio.get_running_loop().create_task(
+                self._flush_loop())
+        return s
+
+    async def _flush_loop(self) -> None:
         """Publish token chunks to job:stream:<id> (reference :613-626).
-        Polls the job's token buffer every stream_flush_ms and publishes
-        whatever accumulated as ONE chunk (wire-compatible: "response"
-        carries a few tokens). Poll interval shrinks while the buffer is
-        empty and the job just started, so TTFT pays at most a few ms."""
-        ch = Channels.job_stream(job_id)
+        ONE shared task drains EVERY active stream's token buffer each
+        cycle and publishes whatever accumulated as one chunk per job
+        (wire-compatible: "response" carries a few tokens). A per-job
+        polling task — the previous design — cost ~10k timer wakeups/s
+        at 192 concurrent streams (measured: 1.1M TimerHandle comparisons
+        per 14 s in the worker loop thread); one shared timer removes
+        that churn. Cycle shrinks to 4 ms while any stream still owes its
+        first chunk, so TTFT pays a few ms at most."""
         flush_s = max(0.002, self.cfg.stream_flush_ms / 1000.0)
-        sent_any = False
-        while True:
-            await asyncio.sleep(flush_s if sent_any else 0.005)
-            with lk:
-                pieces = buf[:]
-                buf.clear()
-            # done only once the buffer is fully drained (done_ev is
-            # set AFTER the engine's final piece lands in buf)
-            done = done_ev.is_set() and len(buf) == 0
-            if not pieces and not done:
-                continue
-            sent_any = True
-            await self.bus.publish(ch, {
-                "jobId": job_id, "workerId": self.worker_id,
-                "chunk": {"id": job_id, "response": "".join(pieces),
-                          "done": done},
-                "timestamp": now_ms(),
-            })
-            if done:
-                break
+        while self._streams:
+            fast = any(not s.sent_any for s in self._streams.values())
+            self._flush_wake.clear()
+            try:
+                await asyncio.wait_for(self._flush_wake.wait(),
+                                       0.004 if fast else flush_s)
+            except asyncio.TimeoutError:
+                pass
+            for s in list(self._streams.values()):
+                with s.lk:
+                    pieces = s.buf[:]
+                    s.buf.clear()
+                # done only once the buffer is fully drained (done_ev is
+                # set AFTER the engine's final piece lands in buf)
+                done = s.done_ev.is_set() and len(s.buf) == 0
+                if not pieces and not done:
+                    continue
+                s.sent_any = True
+                await self.bus.publish(Channels.job_stream(s.job_id), {
+                    "jobId": s.job_id, "workerId": self.worker_id,
+                    "chunk": {"id": s.job_id,
+                              "response": "".join(pieces), "done": done},
+                    "timestamp": now_ms(),
+                })
+                if done:
+                    self._streams.pop(s.job_id, None)
+                    s.fin.set()
+        self._flush_task = None

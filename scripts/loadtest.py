@@ -61,6 +61,29 @@ from gridllm_amd.worker.agent import WorkerAgent
 async def main():
     conc = int(os.environ["LT_CONC"])
     wid = os.environ.get("LT_WID", "lt-worker")
+    profs = []
+    if os.environ.get("LT_PROFILE"):
+        # profile BOTH worker threads: asyncio loop (this thread) and the
+        # batcher decode thread; dump on SIGTERM
+        import cProfile, pstats, signal
+        from gridllm_amd.engine.batcher import Batcher
+        pr_loop = cProfile.Profile(); pr_loop.enable()
+        profs.append(("loop", pr_loop))
+        orig = Batcher._loop_continuous
+        def prof_batcher(self):
+            pr = cProfile.Profile(); pr.enable()
+            profs.append(("batcher", pr))
+            orig(self)
+        Batcher._loop_continuous = prof_batcher
+        def dump(*a):
+            for tag, pr in profs:
+                pr.disable()
+                st = pstats.Stats(pr); st.sort_stats("tottime")
+                with open(f"gpurun_out/lt_worker_{tag}.prof.txt", "w") as f:
+                    st.stream = f
+                    st.print_stats(30)
+            os._exit(0)
+        signal.signal(signal.SIGTERM, dump)
     wcfg = WorkerConfig(worker_id=wid,
                         bus_port=int(os.environ["BUS_PORT"]),
                         max_concurrent_tasks=conc)
@@ -71,6 +94,26 @@ async def main():
                         max_batch_size=conc,
                         use_hipgraph=os.environ["LT_DEVICE"].startswith("cuda"))
     engine = InferenceEngine(ecfg)
+    if os.environ.get("LT_FAKE"):
+        # stub the model: measures the serving GLUE ceiling (bus, pump,
+        # gateway, GIL) with near-zero compute per token
+        import torch
+        m, V = engine.model, engine.spec.vocab
+        eos = engine.tokenizer.eos_id
+        fake = torch.randn(conc + 1, V)
+        fake[:, eos] = -100.0
+
+        def fake_prefill_batch(token_lists, slots, return_hidden=False):
+            for t, s in zip(token_lists, slots):
+                m.cache.seq_lens[s] = len(t)
+            return fake[:len(slots)].clone()   # sampler may scatter in place
+
+        def fake_decode(tokens, slot_map):
+            m.cache.seq_lens[slot_map.long()] += 1
+            return fake[:tokens.shape[0]].clone()
+
+        m.forward_prefill_batch = fake_prefill_batch
+        m.forward_decode = fake_decode
     bus = TcpBus("127.0.0.1", wcfg.bus_port)
     await bus.connect()
     agent = WorkerAgent(bus, engine, wcfg)
@@ -232,6 +275,10 @@ def main():
         }
         print(json.dumps(out))
     finally:
+        for p in workers + [server]:
+            if p.poll() is None:
+                p.send_signal(signal.SIGTERM)
+        time.sleep(1.0 if os.environ.get("LT_PROFILE") else 0)
         for p in workers + [server]:
             if p.poll() is None:
                 p.send_signal(signal.SIGKILL)
