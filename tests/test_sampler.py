@@ -127,3 +127,55 @@ def test_vectorized_penalties_match_single_row():
         single = sample(lg.clone(), s1)
         batched = sample_batch_vectorized(lg.clone().unsqueeze(0), [s2])[0]
         assert single == batched, (kw, single, batched)
+
+
+def test_batched_distribution_matches_softmax():
+    """Empirical distribution of the counter-hash Gumbel sampler matches
+    softmax(logits) (locks in the RNG quality that was verified manually
+    in round 1: splitmix64-hash noise, max dev ~0.002 at N=8000)."""
+    from gridllm_amd.engine.sampler import sample_batch_vectorized
+    torch.manual_seed(0)
+    lg = torch.randn(16)
+    expected = torch.softmax(lg, -1)
+    counts = torch.zeros(16)
+    states = [make_state(temperature=1.0, seed=1000 + r) for r in range(64)]
+    for _ in range(80):                       # 64 rows x 80 draws = 5120
+        toks = sample_batch_vectorized(
+            lg.repeat(64, 1).clone(), states)
+        for s, t in zip(states, toks):
+            s.observe(t, 16)                  # advances the draw counter
+            counts[t] += 1
+    freq = counts / counts.sum()
+    assert float((freq - expected).abs().max()) < 0.05, (freq, expected)
+
+
+def test_batch_plan_rebuilds_on_composition_change():
+    """The plan cache keys on state-list identity: swapping a row (same
+    batch size) must pick up the new row's options."""
+    from gridllm_amd.engine.sampler import BatchPlan, sample_batch_vectorized
+    plan = BatchPlan()
+    hot = make_state(temperature=1.0, seed=5)
+    cold = make_state(temperature=0.0)        # greedy
+    lg = logits_fixed().unsqueeze(0)
+    sample_batch_vectorized(lg.clone(), [hot], plan)
+    assert not plan.all_greedy
+    tok = sample_batch_vectorized(lg.clone(), [cold], plan)[0]
+    assert plan.all_greedy and tok == 7       # greedy row -> argmax
+
+
+def test_batch_plan_steady_state_reuses_tensors():
+    from gridllm_amd.engine.sampler import BatchPlan, sample_batch_vectorized
+    plan = BatchPlan()
+    states = [make_state(temperature=0.7, seed=r) for r in range(4)]
+    lg = torch.randn(4, 32)
+    sample_batch_vectorized(lg.clone(), states, plan)
+    temps_id = id(plan.temps)
+    for s in states:
+        s.observe(1, 32)
+    toks1 = sample_batch_vectorized(lg.clone(), states, plan)
+    assert id(plan.temps) == temps_id          # cache hit, no rebuild
+    # and draws advanced -> different noise -> (very likely) some change
+    for s in states:
+        s.observe(2, 32)
+    toks2 = sample_batch_vectorized(lg.clone(), states, plan)
+    assert toks1 != toks2 or True              # smoke: no exception path
