@@ -1,0 +1,183 @@
+#!/usr/bin/env python3
+"""Fault-injection soak: a live cluster (server + N workers) under
+continuous traffic while random faults fire — SIGKILL+respawn a worker,
+pause/resume via the control channel. Asserts NO request is lost: every
+submitted request must complete (orphan requeue + retries make worker
+death invisible to the client, SURVEY.md §3.5 / §5.3 — the reference had
+the recovery path but no injection tooling to exercise it).
+
+Usage: python scripts/chaos.py [--duration 30] [--workers 2]
+         [--concurrency 8] [--fault-every 4.0]
+Exit 0 = no lost requests and final cluster healthy.
+"""
+
+import argparse
+import asyncio
+import json
+import os
+import random
+import signal
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+import loadtest  # noqa: E402  (scripts/ sibling: _spawn + process sources)
+
+
+async def drive(base: str, args, stats: dict, stop: asyncio.Event):
+    import aiohttp
+    sem = asyncio.Semaphore(args.concurrency)
+    conn = aiohttp.TCPConnector(limit=args.concurrency + 8)
+    async with aiohttp.ClientSession(base_url=base, connector=conn) as sess:
+
+        async def one(i):
+            async with sem:
+                stream = i % 2 == 0
+                stats["attempted"] += 1
+                try:
+                    if stream:
+                        async with sess.post("/api/generate", json={
+                                "model": "tiny", "stream": True,
+                                "prompt": f"chaos {i} " + "lorem " * 8,
+                                "options": {"seed": i, "num_predict": 24}},
+                                timeout=aiohttp.ClientTimeout(total=90)) as r:
+                            done = False
+                            async for raw in r.content:
+                                if raw.strip() and \
+                                        json.loads(raw).get("done"):
+                                    done = True
+                            if not done:
+                                raise RuntimeError("stream ended w/o done")
+                    else:
+                        async with sess.post("/api/generate", json={
+                                "model": "tiny", "stream": False,
+                                "prompt": f"chaos {i} " + "lorem " * 8,
+                                "options": {"seed": i, "num_predict": 24}},
+                                timeout=aiohttp.ClientTimeout(total=90)) as r:
+                            body = await r.json()
+                            if r.status != 200 or "response" not in body:
+                                raise RuntimeError(f"bad response {r.status}")
+                    stats["completed"] += 1
+                except Exception as e:  # noqa: BLE001
+                    stats["lost"] += 1
+                    stats["errors"].append(f"req {i}: {e}")
+
+        i = 0
+        tasks = set()
+        while not stop.is_set():
+            t = asyncio.ensure_future(one(i))
+            tasks.add(t)
+            t.add_done_callback(tasks.discard)
+            i += 1
+            await asyncio.sleep(0.02)
+        if tasks:
+            await asyncio.wait(tasks, timeout=120)
+
+
+async def inject_faults(args, env, workers: list, stats: dict,
+                        stop: asyncio.Event):
+    """Random faults until `stop`: kill+respawn a worker, pause/resume."""
+    from gridllm_amd.bus.tcp import TcpBus
+    from gridllm_amd.types import Channels
+    bus = TcpBus("127.0.0.1", int(env["BUS_PORT"]))
+    await bus.connect()
+    while not stop.is_set():
+        await asyncio.sleep(args.fault_every * (0.5 + random.random()))
+        if stop.is_set():
+            break
+        kind = random.choice(["kill", "pause"])
+        idx = random.randrange(len(workers))
+        wid = f"chaos-worker-{idx}"
+        if kind == "kill":
+            proc = workers[idx]
+            if proc.poll() is None:
+                proc.send_signal(signal.SIGKILL)
+                proc.wait(timeout=10)
+            stats["faults"].append(f"SIGKILL {wid}")
+            wenv = dict(env)
+            wenv["LT_WID"] = wid
+            workers[idx] = await asyncio.to_thread(
+                loadtest._spawn, loadtest.WORKER_SRC, wenv, "WORKER_READY")
+            stats["faults"].append(f"respawn {wid}")
+        else:
+            await bus.publish(Channels.worker_control(wid),
+                              {"type": "pause"})
+            stats["faults"].append(f"pause {wid}")
+            await asyncio.sleep(1.5)
+            await bus.publish(Channels.worker_control(wid),
+                              {"type": "resume"})
+            stats["faults"].append(f"resume {wid}")
+    await bus.close()
+
+
+async def main_async(args, env, base, workers):
+    stats = {"attempted": 0, "completed": 0, "lost": 0,
+             "errors": [], "faults": []}
+    stop = asyncio.Event()
+
+    async def timer():
+        await asyncio.sleep(args.duration)
+        stop.set()
+
+    await asyncio.gather(timer(),
+                         drive(base, args, stats, stop),
+                         inject_faults(args, env, workers, stats, stop))
+    return stats
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--duration", type=float, default=30.0)
+    ap.add_argument("--workers", type=int, default=2)
+    ap.add_argument("--concurrency", type=int, default=8)
+    ap.add_argument("--fault-every", type=float, default=4.0)
+    args = ap.parse_args()
+
+    bus_port, http_port = loadtest._free_port(), loadtest._free_port()
+    env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port),
+           "LT_CONC": "16", "LT_MODEL": "tiny", "LT_DEVICE": "cpu",
+           "LT_DTYPE": "fp32", "LT_MAXSEQ": "256"}
+    server = loadtest._spawn(loadtest.SERVER_SRC, env, "SERVER_READY")
+    workers = []
+    for i in range(args.workers):
+        wenv = dict(env)
+        wenv["LT_WID"] = f"chaos-worker-{i}"
+        workers.append(loadtest._spawn(loadtest.WORKER_SRC, wenv,
+                                       "WORKER_READY"))
+    base = f"http://127.0.0.1:{http_port}"
+    try:
+        import httpx
+        with httpx.Client(base_url=base, timeout=30) as c:
+            for _ in range(200):
+                r = c.get("/health/workers")
+                if r.status_code == 200 and \
+                        r.json()["counts"]["total"] >= args.workers:
+                    break
+                time.sleep(0.1)
+        stats = asyncio.run(main_async(args, env, base, workers))
+        out = {"attempted": stats["attempted"],
+               "completed": stats["completed"],
+               "lost": stats["lost"],
+               "faults": len(stats["faults"]),
+               "fault_log": stats["faults"],
+               "ok": stats["lost"] == 0 and stats["attempted"] > 0}
+        print(json.dumps(out))
+        if stats["errors"]:
+            for e in stats["errors"][:10]:
+                print("ERR:", e, file=sys.stderr)
+        sys.exit(0 if out["ok"] else 1)
+    finally:
+        for p in workers + [server]:
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in workers + [server]:
+            try:
+                p.wait(timeout=10)
+            except Exception:  # noqa: BLE001
+                pass
+
+
+if __name__ == "__main__":
+    main()
