@@ -1204,6 +1204,72 @@ static void launch_gemm_skinny(hipStream_t st, bf16_t* out, float* ws,
 }
 
 // ===========================================================================
+// Fused sampling: Gumbel-argmax token pick for the no-filter path
+// (temperature + optional seed). The torch eager chain is ~15 separate
+// [B,V] launches per decode step (int64 splitmix hash, two logs, argmax)
+// that re-read the logits several times; this kernel reads them ONCE.
+// RNG contract matches gridllm_amd/engine/sampler.py exactly: noise is a
+// pure function of (key, candidate index) via the same splitmix64
+// constants with torch's ARITHMETIC int64 shifts (emulated on unsigned),
+// u = ((z>>38)&0xFFFFFF)/2^24 clamped to [1e-6, 1-1e-6], g=-log(-log u).
+// key==0 marks a greedy row: no noise (sampler.py greedy_mask semantics).
+// Ties pick the lowest index (torch.argmax first-max rule). One workgroup
+// per row; lanes stride the vocab, then wave-shuffle + LDS argmax.
+// ===========================================================================
+__global__ __launch_bounds__(256, 4) void k_sample_gumbel(
+        int* __restrict__ out, const float* __restrict__ logits,
+        const long long* __restrict__ keys,
+        const float* __restrict__ temps, int v) {
+    const int row = blockIdx.x;
+    const float* x = logits + (long)row * (long)v;
+    const unsigned long long key = (unsigned long long)keys[row];
+    const float t = temps[row];
+    const float uhi = (float)(1.0 - 1e-6);   // double-rounded like torch
+    float best = -INFINITY;
+    int besti = 2147483647;
+    for (int i = threadIdx.x; i < v; i += 256) {
+        float val = x[i] / t;
+        if (key != 0ull) {
+            unsigned long long z =
+                ((unsigned long long)i ^ key) * 0x9E3779B97F4A7C15ull;
+            z ^= (unsigned long long)((long long)z >> 29);  // arithmetic >>
+            z *= 0xBF58476D1CE4E5B9ull;
+            z ^= (unsigned long long)((long long)z >> 32);
+            float u = (float)((z >> 38) & 0xFFFFFFull) *
+                      (1.0f / 16777216.0f);
+            u = fminf(fmaxf(u, 1e-6f), uhi);
+            val += -logf(-logf(u));
+        }
+        if (val > best || (val == best && i < besti)) {
+            best = val;
+            besti = i;
+        }
+    }
+    #pragma unroll
+    for (int off = 32; off; off >>= 1) {
+        float ob = __shfl_down(best, off);
+        int oi = __shfl_down(besti, off);
+        if (ob > best || (ob == best && oi < besti)) {
+            best = ob;
+            besti = oi;
+        }
+    }
+    __shared__ float sb[4];
+    __shared__ int si[4];
+    const int wave = threadIdx.x >> 6;
+    if ((threadIdx.x & 63) == 0) { sb[wave] = best; si[wave] = besti; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int w = 1; w < 4; ++w)
+            if (sb[w] > best || (sb[w] == best && si[w] < besti)) {
+                best = sb[w];
+                besti = si[w];
+            }
+        out[row] = besti == 2147483647 ? 0 : besti;  // all -inf -> 0
+    }
+}
+
+// ===========================================================================
 // Host-callable entry points (extern "C", explicit stream — ctypes-bound).
 // All return hipError_t as int (0 == success); callers must check.
 // ===========================================================================
@@ -1416,6 +1482,16 @@ int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
             (bf16_t*)out, (const float*)ws, mn, splitk);
         LAUNCH_CHECK();
     }
+    return 0;
+}
+
+int gl_sample_gumbel(void* stream, void* out, const void* logits,
+                     const void* keys, const void* temps, int b, int v) {
+    if (b <= 0) return 0;
+    k_sample_gumbel<<<dim3((unsigned)b), dim3(256), 0, (hipStream_t)stream>>>(
+        (int*)out, (const float*)logits, (const long long*)keys,
+        (const float*)temps, v);
+    LAUNCH_CHECK();
     return 0;
 }
 

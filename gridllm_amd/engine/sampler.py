@@ -12,6 +12,7 @@ greedy is a pure argmax. Per-request RNG via a seeded torch.Generator.
 
 from __future__ import annotations
 
+import os
 from dataclasses import dataclass
 from typing import Optional
 
@@ -244,6 +245,17 @@ def sample_batch_vectorized(logits: torch.Tensor,
             vals = torch.where(vals > 0, vals / rp_t, vals * rp_t)
             vals = vals - pp_t - fp_t * cc
             logits[ii, tt] = vals
+    if (logits.is_cuda and not plan.need_filter
+            and os.environ.get("GRIDLLM_FUSED_SAMPLER") == "1"):
+        # EXPERIMENTAL (round-2 validation pending): one HIP launch
+        # replacing the ~15-op eager chain below; same RNG contract
+        # (csrc k_sample_gumbel docstring). Greedy rows ride along with
+        # key==0. Off by default until measured on hardware.
+        from gridllm_amd.ops import hip_ops
+        return hip_ops.sample_gumbel(
+            logits.contiguous(),
+            plan.keys(states, dev).view(-1).contiguous(),
+            plan.temps.view(-1).contiguous()).tolist()
     x = logits / plan.temps
     if plan.need_filter:
         probs = torch.softmax(x, -1)

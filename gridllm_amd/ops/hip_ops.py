@@ -58,6 +58,7 @@ def _load() -> ctypes.CDLL:
                                     c_i, c_i, c_i, c_i, c_i, c_i, c_f]
     lib.gl_attn_prefill.restype = c_i
     lib.gl_swiglu_il.argtypes = [c_vp, c_vp, c_vp, c_l, c_i]
+    lib.gl_sample_gumbel.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_i, c_i]
     lib.gl_swiglu_il.restype = c_i
     lib.gl_device_sync.argtypes = []
     lib.gl_device_sync.restype = c_i
@@ -323,6 +324,22 @@ def swiglu_il(gu: torch.Tensor, out: Optional[torch.Tensor] = None
         out = torch.empty(*gu.shape[:-1], f, dtype=gu.dtype, device=gu.device)
     _check(lib.gl_swiglu_il(_stream(), _ptr(out), _ptr(gu),
                             ctypes.c_long(n), f), "swiglu_il")
+    return out
+
+
+def sample_gumbel(logits: torch.Tensor, keys: torch.Tensor,
+                  temps: torch.Tensor) -> torch.Tensor:
+    """Fused Gumbel-argmax token pick (no-filter sampling path):
+    out[b] = argmax_v logits[b,v]/temps[b] + gumbel(splitmix64(v ^ keys[b])),
+    key==0 rows greedy. logits [B,V] f32, keys [B] i64, temps [B] f32 ->
+    [B] i32. One launch replacing the eager ~15-op torch chain
+    (engine/sampler.py::sample_batch_vectorized)."""
+    lib = _load()
+    b, v = logits.shape
+    out = torch.empty(b, dtype=torch.int32, device=logits.device)
+    _check(lib.gl_sample_gumbel(_stream(), _ptr(out), _ptr(logits),
+                                _ptr(keys), _ptr(temps), b, v),
+           "sample_gumbel")
     return out
 
 

@@ -5,6 +5,7 @@ on the same data; tolerances sized for bf16 I/O (~3 decimal digits).
 """
 
 import math
+import os
 
 import pytest
 import torch
@@ -302,3 +303,37 @@ def test_gemm_proj_wide_gpu(dev, m, n, k):
     o1 = hip_ops.gemm_proj_wide(x, w, splitk=1)
     o4 = hip_ops.gemm_proj_wide(x, w, splitk=4)
     assert (o1.float() - o4.float()).abs().max().item() < 0.05
+
+
+@pytest.mark.skipif(os.environ.get("GRIDLLM_FUSED_SAMPLER") != "1",
+                    reason="experimental fused sampler (round-2 "
+                           "validation); set GRIDLLM_FUSED_SAMPLER=1")
+def test_sample_gumbel_matches_eager(dev):
+    """Fused Gumbel-argmax vs the torch eager chain in sampler.py —
+    identical splitmix64 stream, so tokens must match except ulp-level
+    log() near-ties (allow 0.5% of rows)."""
+    from gridllm_amd.engine.sampler import BatchPlan, sample_batch_vectorized
+    from gridllm_amd.engine.sampler import SamplerState
+    from gridllm_amd.ops import hip_ops
+    from gridllm_amd.types import SamplingOptions
+    torch.manual_seed(5)
+    b, v = 512, 4096
+    logits = torch.randn(b, v, device=dev) * 3
+    states = [SamplerState.create(
+        SamplingOptions.from_dict(
+            {"temperature": 0.3 + (r % 7) * 0.2, "seed": r}
+            if r % 5 else {"temperature": 0.0}),   # mix in greedy rows
+        torch.device(dev)) for r in range(b)]
+    plan = BatchPlan()
+    plan.refresh(states, logits.device, v)
+    fused = hip_ops.sample_gumbel(
+        logits.contiguous(),
+        plan.keys(states, logits.device).view(-1).contiguous(),
+        plan.temps.view(-1).contiguous()).tolist()
+    os.environ["GRIDLLM_FUSED_SAMPLER"] = "0"
+    try:
+        eager = sample_batch_vectorized(logits.clone(), states, BatchPlan())
+    finally:
+        os.environ["GRIDLLM_FUSED_SAMPLER"] = "1"
+    mismatch = sum(1 for a, b2 in zip(fused, eager) if a != b2)
+    assert mismatch <= b * 0.005, (mismatch, b)
