@@ -27,6 +27,7 @@ from typing import Optional
 from fastapi import APIRouter, HTTPException, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
+from gridllm_amd.server import tools as tools_mod
 from gridllm_amd.server.registry import WorkerRegistry
 from gridllm_amd.server.scheduler import JobScheduler
 from gridllm_amd.types import InferenceRequest
@@ -206,6 +207,11 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         opts = _clean_options(body.get("options"))
         if body.get("format") == "json":
             opts["format"] = "json"
+        # tools: render into a system instruction; parse the JSON reply
+        # back into message.tool_calls (arguments as OBJECT per Ollama)
+        fns = tools_mod.extract_functions(body.get("tools"))
+        if fns:
+            messages = [tools_mod.system_message(fns)] + list(messages)
         req = InferenceRequest(
             model=model, messages=messages, request_type="chat",
             stream=stream, timeout_ms=300_000, options=opts)
@@ -228,10 +234,16 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         res = await scheduler.submit_and_wait(req)
         if res.get("error"):
             raise HTTPException(status_code=500, detail=res["error"])
+        msg = res.get("message") or \
+            {"role": "assistant", "content": res.get("response", "")}
+        if fns and msg.get("content"):
+            calls = tools_mod.parse_calls(msg["content"])
+            if calls:
+                msg = {"role": "assistant", "content": "",
+                       "tool_calls": tools_mod.to_ollama(calls)}
         return JSONResponse({
             "model": model, "created_at": _now_iso(),
-            "message": res.get("message") or
-            {"role": "assistant", "content": res.get("response", "")},
+            "message": msg,
             "done": True, "done_reason": res.get("done_reason", "stop"),
             **_timings(res)})
 

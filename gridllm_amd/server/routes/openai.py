@@ -17,6 +17,7 @@ import uuid
 from fastapi import APIRouter, HTTPException, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
+from gridllm_amd.server import tools as tools_mod
 from gridllm_amd.server.registry import WorkerRegistry
 from gridllm_amd.server.scheduler import JobScheduler
 from gridllm_amd.types import InferenceRequest
@@ -64,6 +65,19 @@ def _convert_messages(messages: list) -> list:
     for m in messages:
         role = m.get("role", "user")
         content = m.get("content", "")
+        if content is None:
+            content = ""
+        # assistant tool-call history: re-serialize the calls so the
+        # conversation context survives the round trip
+        if role == "assistant" and m.get("tool_calls"):
+            try:
+                calls = [{"name": (c.get("function") or {}).get("name"),
+                          "arguments": (c.get("function") or {})
+                          .get("arguments")} for c in m["tool_calls"]]
+                content = (content + "\n" if content else "") + \
+                    json.dumps(calls if len(calls) > 1 else calls[0])
+            except Exception:  # noqa: BLE001
+                pass
         images = []
         if isinstance(content, list):
             texts = []
@@ -190,15 +204,28 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         stream = bool(body.get("stream", False))
         cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         created = int(time.time())
-        metadata = {}
+        options = _map_options(body)
         rf = body.get("response_format")
-        if isinstance(rf, dict):
-            metadata["format"] = rf.get("type")
+        if isinstance(rf, dict) and rf.get("type") in ("json_object",
+                                                       "json_schema"):
+            # engine reads format from OPTIONS (SamplingOptions.format);
+            # json_schema degrades to json-value constraint
+            options["format"] = "json"
+        conv = _convert_messages(messages)
+        # tools (reference forwarded them to Ollama's template and
+        # converted returned tool_calls, openai.ts:246-304,651-654):
+        # render specs into a system instruction; force format=json
+        # (byte-level automaton) when the call is required
+        fns = tools_mod.extract_functions(body.get("tools"))
+        tool_choice = body.get("tool_choice")
+        if fns and tool_choice != "none":
+            conv = [tools_mod.system_message(fns, tool_choice)] + conv
+            if tools_mod.forced_call(tool_choice):
+                options["format"] = "json"
         req = InferenceRequest(model=model,
-                               messages=_convert_messages(messages),
+                               messages=conv,
                                request_type="chat", stream=stream,
-                               timeout_ms=300_000, options=_map_options(body),
-                               metadata=metadata)
+                               timeout_ms=300_000, options=options)
 
         def make_chunk(c):
             return {"id": cid, "object": "chat.completion.chunk",
@@ -221,11 +248,18 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
             raise HTTPException(status_code=500, detail=res["error"])
         msg = res.get("message") or {"role": "assistant",
                                      "content": res.get("response", "")}
+        finish = _finish_reason(res)
+        if fns and tool_choice != "none" and msg.get("content"):
+            calls = tools_mod.parse_calls(msg["content"])
+            if calls:
+                msg = {"role": "assistant", "content": None,
+                       "tool_calls": tools_mod.to_openai(calls)}
+                finish = "tool_calls"
         return JSONResponse({
             "id": cid, "object": "chat.completion", "created": created,
             "model": model,
             "choices": [{"index": 0, "message": msg, "logprobs": None,
-                         "finish_reason": _finish_reason(res)}],
+                         "finish_reason": finish}],
             "usage": _usage(res)})
 
     # --------------------------------------------------------------- models

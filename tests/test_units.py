@@ -69,3 +69,37 @@ def test_registry_api():
     s2 = get_spec("tiny")
     assert s1.max_seq_len == 32 and s2.max_seq_len != 32 or \
         s2.max_seq_len == 256
+
+
+def test_tools_module_parse_and_shapes():
+    from gridllm_amd.server import tools as T
+    fns = T.extract_functions([
+        {"type": "function", "function": {
+            "name": "get_weather",
+            "parameters": {"type": "object",
+                           "properties": {"city": {"type": "string"}}}}},
+        {"type": "retrieval"},                       # non-function: dropped
+        "garbage",
+    ])
+    assert [f["name"] for f in fns] == ["get_weather"]
+    sysmsg = T.system_message(fns, {"function": {"name": "get_weather"}})
+    assert sysmsg["role"] == "system" and "get_weather" in sysmsg["content"]
+    assert T.forced_call("required")
+    assert T.forced_call({"function": {"name": "x"}})
+    assert not T.forced_call("auto") and not T.forced_call(None)
+
+    calls = T.parse_calls('{"name": "get_weather", "arguments": '
+                          '{"city": "SF"}}')
+    assert calls == [{"name": "get_weather", "arguments": {"city": "SF"}}]
+    assert T.parse_calls('[{"name": "a", "arguments": {}}, '
+                         '{"name": "b", "arguments": {"x": 1}}]')
+    assert T.parse_calls("just a normal sentence") is None
+    assert T.parse_calls('{"no_name": 1}') is None
+    assert T.parse_calls('{"name": "f", "arguments": "notadict"}') is None
+
+    oa = T.to_openai(calls)
+    assert oa[0]["type"] == "function"
+    assert oa[0]["id"].startswith("call_")
+    assert json.loads(oa[0]["function"]["arguments"]) == {"city": "SF"}
+    ol = T.to_ollama(calls)
+    assert ol[0]["function"]["arguments"] == {"city": "SF"}   # object
