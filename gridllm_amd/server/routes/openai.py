@@ -109,7 +109,11 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                                 detail=f"model '{model}' not found")
 
     def _sse(scheduler: JobScheduler, req: InferenceRequest, make_chunk,
-             make_final):
+             make_final, buffered_finals=None):
+        """buffered_finals: when set (tool-call streams), token deltas are
+        withheld and the callback maps the FINAL result to the list of SSE
+        payloads — a tool call can only be recognized on the complete
+        text, matching OpenAI's tool_call-delta framing."""
         async def gen():
             q: asyncio.Queue = asyncio.Queue()
 
@@ -128,11 +132,16 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                 while True:
                     kind, item = await q.get()
                     if kind == "chunk":
-                        if item.get("response"):
+                        if item.get("response") and buffered_finals is None:
                             yield f"data: {json.dumps(make_chunk(item))}\n\n"
                     elif kind == "final":
                         if not item.get("error"):
-                            yield f"data: {json.dumps(make_final(item))}\n\n"
+                            if buffered_finals is not None:
+                                for payload in buffered_finals(item):
+                                    yield f"data: {json.dumps(payload)}\n\n"
+                            else:
+                                yield \
+                                    f"data: {json.dumps(make_final(item))}\n\n"
                         yield "data: [DONE]\n\n"
                         break
                     else:
@@ -242,7 +251,33 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                     "usage": _usage(res)}
 
         if stream:
-            return _sse(scheduler, req, make_chunk, make_final)
+            finals = None
+            if fns and tool_choice != "none":
+                def finals(res):           # noqa: F811 — buffered stream
+                    text = res.get("response", "") or \
+                        (res.get("message") or {}).get("content", "")
+                    calls = tools_mod.parse_calls(text) if text else None
+                    base = {"id": cid, "object": "chat.completion.chunk",
+                            "created": created, "model": model}
+                    if calls:
+                        deltas = [{**base, "choices": [{
+                            "index": 0, "finish_reason": None,
+                            "delta": {"role": "assistant",
+                                      "tool_calls": [{"index": i, **c}]}}]}
+                            for i, c in
+                            enumerate(tools_mod.to_openai(calls))]
+                        return deltas + [{**base, "choices": [{
+                            "index": 0, "delta": {},
+                            "finish_reason": "tool_calls"}],
+                            "usage": _usage(res)}]
+                    out = []
+                    if text:
+                        out.append({**base, "choices": [{
+                            "index": 0, "finish_reason": None,
+                            "delta": {"content": text}}]})
+                    return out + [make_final(res)]
+            return _sse(scheduler, req, make_chunk, make_final,
+                        buffered_finals=finals)
         res = await scheduler.submit_and_wait(req)
         if res.get("error"):
             raise HTTPException(status_code=500, detail=res["error"])

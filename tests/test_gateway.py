@@ -445,3 +445,25 @@ def test_ollama_chat_with_tools(client):
                           dict)                        # Ollama: object
     else:
         assert isinstance(msg["content"], str)
+
+
+def test_openai_chat_tools_streaming_buffered(client):
+    """Tool-call streams withhold token deltas and emit either tool_call
+    deltas or one buffered content delta + final (can only be decided on
+    the complete text)."""
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny", "stream": True, "tools": TOOLS,
+        "tool_choice": "required",
+        "messages": [{"role": "user", "content": "call the tool"}],
+        "max_tokens": 24, "temperature": 1.0, "seed": 7})
+    assert r.status_code == 200, r.text
+    events = [json.loads(ln[len("data: "):])
+              for ln in r.text.splitlines()
+              if ln.startswith("data: ") and ln != "data: [DONE]"]
+    assert r.text.strip().endswith("data: [DONE]")
+    assert events, r.text
+    last = events[-1]["choices"][0]
+    assert last["finish_reason"] in ("tool_calls", "stop", "length")
+    if last["finish_reason"] == "tool_calls":
+        assert any(e["choices"][0]["delta"].get("tool_calls")
+                   for e in events[:-1])
