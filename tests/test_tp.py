@@ -144,3 +144,58 @@ def test_tp_serving_wave_matches_single():
     assert tokens_tp == ref.token_ids
     emb_ref = eng.embed(["vec me"])[0][:4]
     assert all(abs(a - b) < 1e-3 for a, b in zip(emb_tp, emb_ref))
+
+
+def _tp_cancel_worker(rank: int, world: int, port: int, q):
+    try:
+        dist.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{port}",
+            rank=rank, world_size=world)
+        import threading
+        from gridllm_amd.config import EngineConfig
+        from gridllm_amd.engine.engine import InferenceEngine
+        from gridllm_amd.parallel.tp import TPContext
+        from gridllm_amd.worker.tp_runner import TPBatcher, follower_loop
+        tp = TPContext(rank=rank, size=world, group=None)
+        eng = InferenceEngine(EngineConfig(
+            model="tiny", device="cpu", dtype="fp32", max_seq_len=128,
+            max_batch_size=4, use_hipgraph=False, seed=7), tp=tp)
+        if rank == 0:
+            batcher = TPBatcher(eng, tp)
+            cancel = threading.Event()
+            # cancel as soon as the first token streams out — deterministic
+            # mid-generation cancel regardless of host speed; the decision
+            # must reach every rank through the engine's cancel broadcast
+            fut = batcher.submit_generate(
+                "long generation", {"temperature": 0.0, "num_predict": 100},
+                on_token=lambda piece, done: cancel.set(),
+                cancel=cancel)
+            res = fut.result(timeout=120)
+            batcher.shutdown()
+            q.put(("ok", res.cancelled, len(res.token_ids)))
+        else:
+            follower_loop(eng)
+        dist.destroy_process_group()
+    except Exception:  # noqa: BLE001
+        import traceback
+        if rank == 0:
+            q.put(("err", traceback.format_exc(), None))
+
+
+def test_tp_cancellation_is_synchronized():
+    """Cancel mid-generation under TP: rank 0 observes the client cancel
+    and BROADCASTS the decision so all ranks retire the row on the same
+    step (engine._sync_cancel) — a rank-local check would desync the
+    collectives."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_tp_cancel_worker, args=(r, 2, PORT + 2, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    status, cancelled, ntok = q.get()
+    for p in procs:
+        p.join(timeout=180)
+    assert status == "ok", f"TP cancel failed:\n{cancelled}"
+    assert cancelled is True
+    assert ntok < 100, ntok        # stopped early
