@@ -223,3 +223,63 @@ class JsonByteConstraint:
             self.mode = AFTER
         else:
             raise ValueError(f"illegal delimiter {b!r}")
+
+    # -------------------------------------------------- token-level helpers
+
+    def accepts(self, bs: bytes) -> bool:
+        """Would pushing this byte string keep the document legal?
+        State is saved/restored — the automaton does NOT advance. Used for
+        token-level constrained decoding with multi-byte tokenizers."""
+        saved = (list(self.stack), self.mode, self._lit, self._key)
+        try:
+            for b in bs:
+                if b not in self.allowed()[0]:
+                    return False
+                self.push(b)
+            return True
+        finally:
+            self.stack = list(saved[0])
+            self.mode, self._lit, self._key = saved[1], saved[2], saved[3]
+
+    def push_bytes(self, bs: bytes) -> None:
+        for b in bs:
+            self.push(b)
+
+
+def mask_logits_token_level(row_logits, constraint, tokenizer,
+                            eos_id: int) -> None:
+    """format="json" enforcement for MULTI-BYTE tokenizers (HF BPE): a
+    token is legal iff its raw bytes keep the automaton legal. Checking
+    the whole vocab per step is wasteful, so candidates are checked
+    lazily in descending-logit order (top-256 then top-4096 — a legal
+    token outside the window could never win the sample anyway once the
+    window's tokens are unmasked); if nothing in the window is legal the
+    single-byte fallback tokens of the byte-level alphabet guarantee
+    progress. In-place; one host sync per constrained row per step."""
+    import torch
+    allowed_bytes, eos_ok = constraint.allowed()
+    v = row_logits.shape[0]
+    legal: list[int] = []
+    for k in (256, 4096):
+        ids = row_logits.topk(min(k, v)).indices.tolist()
+        for tid in ids:
+            if tid == eos_id:
+                continue
+            bs = tokenizer.token_bytes(tid)
+            if bs and constraint.accepts(bs):
+                legal.append(tid)
+        if legal or k >= v:
+            break
+    if not legal:
+        for b in allowed_bytes:
+            tid = tokenizer.byte_token_id(b)
+            if tid is not None:
+                legal.append(tid)
+    if eos_ok:
+        legal.append(eos_id)
+    keep = torch.full_like(row_logits, float("-inf"))
+    if legal:
+        ix = torch.tensor(sorted(set(legal)), dtype=torch.long,
+                          device=row_logits.device)
+        keep[ix] = row_logits[ix]
+    row_logits.copy_(keep)

@@ -207,7 +207,14 @@ class InferenceEngine:
                                constraint) -> None:
         """Mask row logits to the automaton's legal next bytes (+EOS when a
         complete value has been emitted). Byte tokenizer: ids 0..255 are
-        raw bytes. In-place; no host sync."""
+        raw bytes. In-place; no host sync. HF tokenizers dispatch to
+        token-level lazy masking (constraints.mask_logits_token_level)."""
+        if hasattr(self.tokenizer, "token_bytes"):
+            from gridllm_amd.engine.constraints import \
+                mask_logits_token_level
+            mask_logits_token_level(row_logits, constraint, self.tokenizer,
+                                    self.tokenizer.eos_id)
+            return
         allowed, eos_ok = constraint.allowed()
         idx = list(allowed)
         if eos_ok:
@@ -349,9 +356,12 @@ class InferenceEngine:
                     max_new = min(max_new,
                                   self.model.max_seq - len(ids) - 1)
                 constraint = None
-                if opts.format == "json" and self.tokenizer.vocab_size and \
-                        getattr(self.tokenizer, "BOS", None) is not None:
-                    # byte-level tokenizer -> exact byte-grammar masking
+                if opts.format == "json" and (
+                        getattr(self.tokenizer, "BOS", None) is not None
+                        or hasattr(self.tokenizer, "token_bytes")):
+                    # byte tokenizer -> exact byte-grammar masking;
+                    # HF tokenizer -> token-level lazy masking over the
+                    # same byte automaton (constraints.py)
                     from gridllm_amd.engine.constraints import \
                         JsonByteConstraint
                     constraint = JsonByteConstraint()
@@ -435,9 +445,14 @@ class InferenceEngine:
         res.eval_count += 1
         row.last_token = tok
         row.state.observe(tok, self.spec.vocab)
-        if row.constraint is not None and 0 <= tok < 256:
+        if row.constraint is not None:
             try:
-                row.constraint.push(tok)
+                if hasattr(self.tokenizer, "token_bytes"):
+                    bs = self.tokenizer.token_bytes(tok)
+                    if bs:
+                        row.constraint.push_bytes(bs)
+                elif 0 <= tok < 256:
+                    row.constraint.push(tok)
             except ValueError:
                 res.done_reason = "stop"   # automaton refused (shouldn't
                 return False               # happen with masking in place)

@@ -9,7 +9,7 @@ no network). If a HuggingFace tokenizer.json path is supplied, the
 from __future__ import annotations
 
 import os
-from typing import List
+from typing import List, Optional
 
 
 class ByteTokenizer:
@@ -63,6 +63,21 @@ class ByteStreamDecoder:
         return out
 
 
+def _byte_level_decoder() -> dict:
+    """Invert the GPT-2/byte-level-BPE printable-alphabet mapping:
+    token-string char -> original raw byte."""
+    bs = list(range(ord("!"), ord("~") + 1)) + \
+        list(range(0xA1, 0xAC + 1)) + list(range(0xAE, 0xFF + 1))
+    cs = bs[:]
+    n = 0
+    for b in range(256):
+        if b not in bs:
+            bs.append(b)
+            cs.append(256 + n)
+            n += 1
+    return {chr(c): b for b, c in zip(bs, cs)}
+
+
 class HFTokenizer:
     def __init__(self, path: str):
         from tokenizers import Tokenizer  # offline wheel
@@ -74,6 +89,55 @@ class HFTokenizer:
             if tid is not None:
                 self.eos_id = tid
                 break
+        self._tb: Optional[list] = None      # id -> raw bytes (lazy)
+        self._byte_tok: Optional[dict] = None
+
+    def _build_token_bytes(self) -> None:
+        dec = _byte_level_decoder()
+        special: set = set()
+        try:
+            special = {i for i, t in
+                       self._tok.get_added_tokens_decoder().items()
+                       if t.special}
+        except Exception:  # noqa: BLE001 — older tokenizers versions
+            pass
+        tb: list = []
+        for i in range(self.vocab_size):
+            s = self._tok.id_to_token(i)
+            if s is None or i in special:
+                tb.append(None)
+                continue
+            try:
+                tb.append(bytes(dec[ch] for ch in s))
+            except KeyError:
+                # special token or non-byte-level alphabet: approximate
+                # with the decoded text (empty -> not representable)
+                t = self._tok.decode([i])
+                tb.append(t.encode("utf-8") if t else None)
+        self._tb = tb
+        self._byte_tok = {}
+        for i, bs in enumerate(tb):
+            if bs is not None and len(bs) == 1 and bs[0] not in \
+                    self._byte_tok:
+                self._byte_tok[bs[0]] = i
+
+    def token_bytes(self, tid: int):
+        """Raw bytes the token contributes to the output stream (None for
+        specials) — token-level grammar constraints check these against
+        the byte automaton (engine/constraints.py)."""
+        if self._tb is None:
+            self._build_token_bytes()
+        if 0 <= tid < len(self._tb):
+            return self._tb[tid]
+        return None
+
+    def byte_token_id(self, b: int):
+        """Token id of the single-byte fallback token for byte b (byte-
+        level BPE vocabs contain all 256) — the guaranteed-progress path
+        when no high-probability token is grammar-legal."""
+        if self._byte_tok is None:
+            self._build_token_bytes()
+        return self._byte_tok.get(b)
 
     def encode(self, text: str, bos: bool = True) -> List[int]:
         return self._tok.encode(text).ids
