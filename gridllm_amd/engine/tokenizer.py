@@ -150,25 +150,29 @@ class HFTokenizer:
 
 
 class HFStreamDecoder:
+    """Streams decoded text token-by-token, holding back spans whose tail
+    is an incomplete UTF-8 sequence. Decodes only the UNFLUSHED tail ids
+    (not the whole history — a full re-decode per token is O(n^2) over a
+    generation), emitting and resetting at each clean boundary."""
+
     def __init__(self, tok) -> None:
         self._tok = tok
-        self._ids: List[int] = []
-        self._emitted = 0
+        self._tail: List[int] = []
 
     def push(self, token_id: int) -> str:
-        self._ids.append(token_id)
-        text = self._tok.decode(self._ids)
+        self._tail.append(token_id)
+        text = self._tok.decode(self._tail)
         if text.endswith("�"):
-            return ""
-        out = text[self._emitted:]
-        self._emitted = len(text)
-        return out
+            return ""                   # mid-codepoint: hold the span
+        self._tail.clear()
+        return text
 
     def flush(self) -> str:
-        text = self._tok.decode(self._ids)
-        out = text[self._emitted:]
-        self._emitted = len(text)
-        return out
+        if not self._tail:
+            return ""
+        text = self._tok.decode(self._tail)
+        self._tail.clear()
+        return text
 
 
 def load_tokenizer(weights_path: str = "", vocab_size: int = 512):

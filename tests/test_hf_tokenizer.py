@@ -95,3 +95,18 @@ def test_engine_with_hf_tokenizer_enforces_json(hf_tok):
     if res.done_reason != "length":
         json.loads(res.text.encode("utf-8", errors="replace")
                    .decode("utf-8", errors="replace"))
+
+
+def test_hf_stream_decoder_matches_full_decode(hf_tok):
+    """Span-wise streaming decode (O(tail) per token, not O(history))
+    reassembles exactly the full decode for byte-level BPE — including
+    code points split across tokens."""
+    for txt in ['{"key": "vérité ✓"}', "héllo wörld ünïcode",
+                "plain ascii text here", "日本語 mixed ascii"]:
+        ids = hf_tok.encode(txt)
+        dec = hf_tok.decode_stream()
+        out = []
+        for i in ids:
+            out.append(dec.push(i))
+        out.append(dec.flush())
+        assert "".join(out) == hf_tok.decode(ids), txt
