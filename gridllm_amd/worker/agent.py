@@ -416,12 +416,16 @@ class WorkerAgent:
                 if not pieces and not done:
                     continue
                 s.sent_any = True
-                await self.bus.publish(Channels.job_stream(s.job_id), {
-                    "jobId": s.job_id, "workerId": self.worker_id,
-                    "chunk": {"id": s.job_id,
-                              "response": "".join(pieces), "done": done},
-                    "timestamp": now_ms(),
-                })
+                try:
+                    await self.bus.publish(Channels.job_stream(s.job_id), {
+                        "jobId": s.job_id, "workerId": self.worker_id,
+                        "chunk": {"id": s.job_id,
+                                  "response": "".join(pieces), "done": done},
+                        "timestamp": now_ms(),
+                    })
+                except Exception:  # noqa: BLE001 — bus hiccup: chunks are
+                    pass           # fire-and-forget; the job result path
+                                   # (and orphan recovery) owns delivery
                 if done:
                     self._streams.pop(s.job_id, None)
                     s.fin.set()
