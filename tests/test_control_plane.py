@@ -61,6 +61,14 @@ class FakeWorker:
             await asyncio.sleep(0.05)
 
     async def _on_job(self, _ch, payload):
+        if payload.get("type") == "job_cancellation":
+            # honor cancellation like the real worker (the hung job's
+            # slot frees, so the scheduler can reassign)
+            self.current = max(0, self.current - 1)
+            await self.bus.publish(Channels.WORKER_HEARTBEAT, {
+                "workerId": self.wid, "status": "online",
+                "currentJobs": self.current})
+            return
         if payload.get("type") != "job_assignment":
             return
         job = payload["job"]
@@ -367,3 +375,35 @@ async def test_registry_model_index_and_alias_counts():
     reg.workers["w2"].current_jobs = 2
     assert not reg.get_available_workers_by_model("m2")
     await teardown(reg, sched, w1, w2)
+
+
+@pytest.mark.asyncio
+async def test_job_timeout_cancels_and_fails():
+    """A job exceeding its timeout gets a job_cancellation publish to the
+    worker and the awaiter receives a timeout error (reference scheduled
+    the timeout but the worker ignored the cancellation,
+    JobScheduler.ts:416-419 / WorkerClientService.ts:483-491 — here the
+    worker honors it; this test checks the scheduler side with a hung
+    stub)."""
+    core = MemoryBusCore()
+    cfg = ServerConfig(worker_heartbeat_timeout_ms=10_000,
+                       worker_cleanup_interval_ms=5_000,
+                       job_timeout_ms=300, job_retry_attempts=1,
+                       job_retry_delay_ms=50)
+    bus, reg, sched = await make_stack(core, cfg)
+    w = FakeWorker(core, "w-hang", behavior="hang")
+    cancellations = []
+
+    async def on_ctrl(_ch, payload):
+        if payload.get("type") == "job_cancellation":
+            cancellations.append(payload.get("jobId"))
+    await w.start()
+    await w.bus.subscribe(Channels.worker_job("w-hang"), on_ctrl)
+    await asyncio.sleep(0.1)
+    req = InferenceRequest(model="m1", prompt="x", timeout_ms=300)
+    res = await sched.submit_and_wait(req)
+    assert res.get("error"), res              # timed out, not completed
+    assert "timeout" in res["error"].lower() or "timed" in res["error"].lower()
+    await asyncio.sleep(0.1)
+    assert req.id in cancellations            # worker was told to abort
+    await teardown(reg, sched, w)
