@@ -223,8 +223,14 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                     "done": False}
 
         def final_json(res):
+            msg = {"role": "assistant", "content": ""}
+            if fns:
+                # Ollama emits tool_calls on the final chunk of a stream
+                calls = tools_mod.parse_calls(res.get("response", "") or "")
+                if calls:
+                    msg["tool_calls"] = tools_mod.to_ollama(calls)
             return {"model": model, "created_at": _now_iso(),
-                    "message": {"role": "assistant", "content": ""},
+                    "message": msg,
                     "done": True,
                     "done_reason": res.get("done_reason", "stop"),
                     **_timings(res)}
