@@ -356,15 +356,21 @@ class InferenceEngine:
                     max_new = min(max_new,
                                   self.model.max_seq - len(ids) - 1)
                 constraint = None
-                if opts.format == "json" and (
-                        getattr(self.tokenizer, "BOS", None) is not None
+                if (getattr(self.tokenizer, "BOS", None) is not None
                         or hasattr(self.tokenizer, "token_bytes")):
                     # byte tokenizer -> exact byte-grammar masking;
                     # HF tokenizer -> token-level lazy masking over the
-                    # same byte automaton (constraints.py)
-                    from gridllm_amd.engine.constraints import \
-                        JsonByteConstraint
-                    constraint = JsonByteConstraint()
+                    # same byte automaton (constraints.py). A dict format
+                    # is a JSON schema (Ollama contract) -> the
+                    # schema-directed automaton (schema_constraints.py)
+                    if opts.format == "json":
+                        from gridllm_amd.engine.constraints import \
+                            JsonByteConstraint
+                        constraint = JsonByteConstraint()
+                    elif isinstance(opts.format, dict):
+                        from gridllm_amd.engine.schema_constraints import \
+                            JsonSchemaByteConstraint
+                        constraint = JsonSchemaByteConstraint(opts.format)
                 row = _Row(slot=slot,
                            state=SamplerState.create(opts, self.device),
                            decoder=self.tokenizer.decode_stream(),

@@ -1,0 +1,411 @@
+"""JSON-Schema-constrained decoding: a schema-directed byte automaton.
+
+Ollama enforces `format: {<json schema>}` through llama.cpp grammars; the
+reference forwarded the object untouched (server/src/routes/ollama.ts:45,
+openai.ts:637-642). Here the engine enforces a practical schema subset
+directly, byte-by-byte, stacked on the plain JSON automaton
+(constraints.py): masking keeps every sampled byte inside the schema, so
+a completed generation is guaranteed schema-valid by construction.
+
+Supported subset:
+  type: object / array / string / number / integer / boolean / null
+  object: properties, required, additionalProperties (False restricts
+          keys to `properties` and lets masking FORCE the required keys;
+          a dict supplies the schema for unknown keys)
+  array:  items, minItems, maxItems
+  enum / const: any JSON values, matched by their compact serialization
+Anything else (anyOf, $ref, patterns, numeric bounds, type lists…)
+degrades that subtree to the plain syntactic-JSON constraint.
+
+Output is COMPACT JSON: insignificant whitespace is not offered to the
+sampler (standard for grammar-constrained decoding; any JSON consumer
+accepts it).
+
+The interface matches JsonByteConstraint — allowed() -> (frozenset bytes,
+eos_ok), push(byte), accepts(bytes), push_bytes — so both masking paths
+(byte-exact and HF token-level, engine.py::_apply_constraint_mask) work
+unchanged.
+"""
+
+from __future__ import annotations
+
+import json
+from typing import Optional
+
+from gridllm_amd.engine import constraints as C
+
+_STR_MODES = (C.STRING, C.STR_ESC, C.STR_U1, C.STR_U2, C.STR_U3, C.STR_U4)
+_VALUE_START = frozenset(b'{["-0123456789tfn')
+_NUM_START = frozenset(b"-0123456789")
+
+
+def _dumps(v) -> bytes:
+    return json.dumps(v, separators=(",", ":"), ensure_ascii=False) \
+        .encode("utf-8")
+
+
+class _Gen:
+    """Subtree constrained only syntactically (nested plain automaton),
+    with optional first-byte restriction and banned bytes (used for
+    string / number / integer / fallback)."""
+
+    def __init__(self, first: Optional[frozenset] = None,
+                 banned: frozenset = frozenset()):
+        self.inner = C.JsonByteConstraint()
+        self.started = False
+        self.first = first
+        self.banned = banned
+        self.done = False                 # never self-closes; popped lazily
+
+    def allowed(self) -> set:
+        a = set(self.inner.allowed()[0])
+        if self.inner.mode not in _STR_MODES:
+            a -= C._WS                    # compact output
+            a -= self.banned
+        if not self.started and self.first is not None:
+            a &= self.first
+        return a
+
+    def complete(self) -> bool:
+        return self.started and self.inner.allowed()[1]
+
+    def push(self, ctx, b: int):
+        self.started = True
+        self.inner.push(b)
+        return None
+
+    def child_done(self, f) -> None:  # pragma: no cover - leaf frame
+        pass
+
+
+class _Lit:
+    """Byte-trie over candidate serializations (enum/const, true/false,
+    null, restricted object keys)."""
+
+    def __init__(self, options: list, values: Optional[list] = None):
+        self.opts = list(options)
+        self.vals = list(values) if values is not None else [None] * len(
+            self.opts)
+        self.pos = 0
+        self.matched = None               # value of a fully-matched option
+        self.done = False
+
+    def allowed(self) -> set:
+        return {o[self.pos] for o in self.opts if self.pos < len(o)}
+
+    def complete(self) -> bool:
+        return self.done or self.matched is not None
+
+    def push(self, ctx, b: int):
+        keep = [(o, v) for o, v in zip(self.opts, self.vals)
+                if self.pos < len(o) and o[self.pos] == b]
+        self.opts = [o for o, _ in keep]
+        self.vals = [v for _, v in keep]
+        self.pos += 1
+        self.matched = None
+        for o, v in keep:
+            if len(o) == self.pos:
+                self.matched = v if v is not None else o
+        if all(len(o) == self.pos for o in self.opts):
+            self.done = True
+        return None
+
+    def child_done(self, f) -> None:  # pragma: no cover - leaf frame
+        pass
+
+
+class _KeyAny:
+    """An arbitrary object key, entered AFTER the opening quote; records
+    the key text so the object can pick the right value schema."""
+
+    def __init__(self):
+        self.inner = C.JsonByteConstraint()
+        self.inner.push(0x22)             # prime: inside a string
+        self.raw = bytearray()            # content + closing quote
+        self.done = False
+        self.matched = None               # decoded key on completion
+
+    def allowed(self) -> set:
+        return set(self.inner.allowed()[0]) if not self.done else set()
+
+    def complete(self) -> bool:
+        return self.done
+
+    def push(self, ctx, b: int):
+        self.inner.push(b)
+        self.raw.append(b)
+        if self.inner.allowed()[1]:       # string closed -> complete value
+            self.done = True
+            try:
+                self.matched = json.loads(
+                    b'"' + bytes(self.raw), strict=False)
+            except Exception:  # noqa: BLE001 — non-UTF8 content
+                self.matched = bytes(self.raw[:-1]).decode(
+                    "utf-8", errors="replace")
+        return None
+
+    def child_done(self, f) -> None:  # pragma: no cover - leaf frame
+        pass
+
+
+class _Obj:
+    OPEN, KEY_OR_CLOSE, KEY, COLON, AFTER = range(5)
+
+    def __init__(self, schema: dict):
+        self.props = schema.get("properties") or {}
+        self.required = set(schema.get("required") or [])
+        self.required &= set(self.props) | self.required  # as given
+        addl = schema.get("additionalProperties", True)
+        self.addl = addl
+        self.seen: set = set()
+        self.state = self.OPEN
+        self.pending_key = None
+        self.expect_key = False
+        self.done = False
+
+    def _open_keys(self) -> list:
+        return [k for k in self.props if k not in self.seen]
+
+    def _may_close(self) -> bool:
+        return not (self.required - self.seen)
+
+    def allowed(self) -> set:
+        if self.done:
+            return set()
+        if self.state == self.OPEN:
+            return {0x7B}
+        if self.state == self.KEY_OR_CLOSE:
+            a = set()
+            if self._may_close():
+                a.add(0x7D)
+            if self._open_keys() or self.addl is not False:
+                a.add(0x22)
+            return a
+        if self.state == self.KEY:
+            return {0x22}                 # after ',' a key is mandatory
+        if self.state == self.COLON:
+            return {0x3A}
+        if self.state == self.AFTER:
+            a = set()
+            if self._may_close():
+                a.add(0x7D)
+            if self._open_keys() or self.addl is not False:
+                a.add(0x2C)
+            return a
+        return set()
+
+    def complete(self) -> bool:
+        return self.done
+
+    def push(self, ctx, b: int):
+        if self.state == self.OPEN and b == 0x7B:
+            self.state = self.KEY_OR_CLOSE
+            return None
+        if self.state in (self.KEY_OR_CLOSE, self.AFTER) and b == 0x7D:
+            self.done = True
+            return None
+        if self.state == self.AFTER and b == 0x2C:
+            self.state = self.KEY
+            return None
+        if self.state in (self.KEY_OR_CLOSE, self.KEY) and b == 0x22:
+            self.expect_key = True
+            self.state = self.COLON
+            if self.addl is False:
+                opts = [_dumps(k)[1:] for k in self._open_keys()]
+                return _Lit(opts, self._open_keys())
+            return _KeyAny()
+        if self.state == self.COLON and b == 0x3A:
+            key = self.pending_key
+            if key in self.props:
+                sub = self.props[key]
+            elif isinstance(self.addl, dict):
+                sub = self.addl
+            else:
+                sub = True                # generic value
+            self.state = self.AFTER
+            return make_frame(sub)
+        raise ValueError(f"object: illegal byte {b!r} in state {self.state}")
+
+    def child_done(self, f) -> None:
+        if self.expect_key:
+            self.expect_key = False
+            self.pending_key = f.matched
+            self.seen.add(f.matched)
+
+
+class _Arr:
+    OPEN, FIRST, AFTER = range(3)
+
+    def __init__(self, schema: dict):
+        self.items = schema.get("items", True)
+        self.min_items = int(schema.get("minItems") or 0)
+        mx = schema.get("maxItems")
+        self.max_items = int(mx) if mx is not None else None
+        self.count = 0
+        self.state = self.OPEN
+        self.done = False
+
+    def allowed(self) -> set:
+        if self.done:
+            return set()
+        if self.state == self.OPEN:
+            return {0x5B}
+        if self.state == self.FIRST:
+            a = set(_first_bytes(self.items)) \
+                if self.max_items is None or self.max_items > 0 else set()
+            if self.min_items <= 0:
+                a.add(0x5D)
+            return a
+        if self.state == self.AFTER:
+            a = set()
+            if self.count >= self.min_items:
+                a.add(0x5D)
+            if self.max_items is None or self.count < self.max_items:
+                a.add(0x2C)
+            return a
+        return set()
+
+    def complete(self) -> bool:
+        return self.done
+
+    def push(self, ctx, b: int):
+        if self.state == self.OPEN and b == 0x5B:
+            self.state = self.FIRST
+            return None
+        if self.state == self.FIRST and b == 0x5D and self.min_items <= 0:
+            self.done = True
+            return None
+        if self.state == self.AFTER and b == 0x5D and \
+                self.count >= self.min_items:
+            self.done = True
+            return None
+        if self.state == self.FIRST:
+            self.count += 1
+            self.state = self.AFTER
+            return (make_frame(self.items), b)      # replay b into child
+        if self.state == self.AFTER and b == 0x2C:
+            self.count += 1
+            return make_frame(self.items)
+        raise ValueError(f"array: illegal byte {b!r} in state {self.state}")
+
+    def child_done(self, f) -> None:
+        pass
+
+
+def _first_bytes(schema) -> frozenset:
+    """Bytes a value of this schema may start with (compact form)."""
+    if not isinstance(schema, dict) or not schema:
+        return _VALUE_START
+    if "enum" in schema or "const" in schema:
+        vals = schema.get("enum", [schema.get("const")])
+        return frozenset(_dumps(v)[0] for v in vals if _dumps(v))
+    t = schema.get("type")
+    if isinstance(t, list):
+        out = frozenset()
+        for ti in t:
+            out |= _first_bytes({**schema, "type": ti})
+        return out
+    if t == "object" or (t is None and "properties" in schema):
+        return frozenset(b"{")
+    if t == "array":
+        return frozenset(b"[")
+    if t == "string":
+        return frozenset(b'"')
+    if t in ("number", "integer"):
+        return _NUM_START
+    if t == "boolean":
+        return frozenset(b"tf")
+    if t == "null":
+        return frozenset(b"n")
+    return _VALUE_START
+
+
+def make_frame(schema):
+    """Schema node -> automaton frame. Unsupported constructs degrade to
+    the plain syntactic constraint for that subtree."""
+    if not isinstance(schema, dict) or not schema:
+        return _Gen()
+    if "enum" in schema or "const" in schema:
+        vals = schema.get("enum", [schema.get("const")])
+        return _Lit([_dumps(v) for v in vals], vals)
+    t = schema.get("type")
+    if isinstance(t, list):
+        return _Gen(first=_first_bytes(schema))
+    if t == "object" or (t is None and "properties" in schema):
+        return _Obj(schema)
+    if t == "array":
+        return _Arr(schema)
+    if t == "string":
+        return _Gen(first=frozenset(b'"'))
+    if t == "integer":
+        return _Gen(first=_NUM_START, banned=frozenset(b".eE"))
+    if t == "number":
+        return _Gen(first=_NUM_START)
+    if t == "boolean":
+        return _Lit([b"true", b"false"], [True, False])
+    if t == "null":
+        return _Lit([b"null"], [None])
+    return _Gen()
+
+
+class JsonSchemaByteConstraint:
+    """Drop-in sibling of JsonByteConstraint, directed by a JSON schema."""
+
+    def __init__(self, schema):
+        self.stack = [make_frame(schema)]
+
+    def allowed(self) -> tuple:
+        out: set = set()
+        for f in reversed(self.stack):
+            out |= f.allowed()
+            if not f.complete():
+                break
+        eos_ok = all(f.complete() for f in self.stack)
+        return frozenset(out), eos_ok
+
+    def push(self, b: int) -> None:
+        while True:
+            top = self.stack[-1]
+            if b in top.allowed():
+                res = top.push(self, b)
+                if res is not None:
+                    if isinstance(res, tuple):    # (child, replay byte)
+                        child, rb = res
+                        self.stack.append(child)
+                        b = rb
+                        continue
+                    self.stack.append(res)
+                break
+            if top.complete() and len(self.stack) > 1:
+                f = self.stack.pop()
+                self.stack[-1].child_done(f)
+                continue
+            raise ValueError(f"schema: illegal byte {b!r}")
+        while len(self.stack) > 1 and self.stack[-1].done:
+            f = self.stack.pop()
+            self.stack[-1].child_done(f)
+
+    # same token-level helpers as JsonByteConstraint -------------------
+
+    def accepts(self, bs: bytes) -> bool:
+        saved = self._snapshot()
+        try:
+            for b in bs:
+                if b not in self.allowed()[0]:
+                    return False
+                self.push(b)
+            return True
+        finally:
+            self._restore(saved)
+
+    def push_bytes(self, bs: bytes) -> None:
+        for b in bs:
+            self.push(b)
+
+    # frames hold mutable state; deep-snapshot via pickle-free copy
+    def _snapshot(self):
+        import copy
+        return copy.deepcopy(self.stack)
+
+    def _restore(self, saved) -> None:
+        self.stack = saved

@@ -154,8 +154,9 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
 
         _validate_model(registry, model)
         opts = _clean_options(body.get("options"))
-        if body.get("format") == "json":
-            opts["format"] = "json"    # engine enforces via a byte-level
+        fmt = body.get("format")
+        if fmt == "json" or isinstance(fmt, dict):
+            opts["format"] = fmt       # engine enforces via a byte-level
                                        # JSON automaton (engine/constraints)
         req = InferenceRequest(
             model=model, prompt=prompt, request_type="generate",
@@ -205,8 +206,9 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         _validate_model(registry, model)
         stream = bool(body.get("stream", True))
         opts = _clean_options(body.get("options"))
-        if body.get("format") == "json":
-            opts["format"] = "json"
+        fmt = body.get("format")
+        if fmt == "json" or isinstance(fmt, dict):
+            opts["format"] = fmt       # dict = JSON schema (enforced)
         # tools: render into a system instruction; parse the JSON reply
         # back into message.tool_calls (arguments as OBJECT per Ollama)
         fns = tools_mod.extract_functions(body.get("tools"))

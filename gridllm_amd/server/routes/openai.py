@@ -215,11 +215,14 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         created = int(time.time())
         options = _map_options(body)
         rf = body.get("response_format")
-        if isinstance(rf, dict) and rf.get("type") in ("json_object",
-                                                       "json_schema"):
-            # engine reads format from OPTIONS (SamplingOptions.format);
-            # json_schema degrades to json-value constraint
+        if isinstance(rf, dict) and rf.get("type") == "json_object":
             options["format"] = "json"
+        elif isinstance(rf, dict) and rf.get("type") == "json_schema":
+            # the engine ENFORCES the schema (schema-directed byte
+            # automaton, engine/schema_constraints.py)
+            schema = (rf.get("json_schema") or {}).get("schema")
+            options["format"] = schema if isinstance(schema, dict) \
+                else "json"
         conv = _convert_messages(messages)
         # tools (reference forwarded them to Ollama's template and
         # converted returned tool_calls, openai.ts:246-304,651-654):
@@ -230,7 +233,9 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         if fns and tool_choice != "none":
             conv = [tools_mod.system_message(fns, tool_choice)] + conv
             if tools_mod.forced_call(tool_choice):
-                options["format"] = "json"
+                # schema-enforced call: name from the allowed set and
+                # arguments valid against the function's parameters
+                options["format"] = tools_mod.call_schema(fns, tool_choice)
         req = InferenceRequest(model=model,
                                messages=conv,
                                request_type="chat", stream=stream,

@@ -64,7 +64,8 @@ class SamplingOptions:
     repeat_penalty: Optional[float] = None
     presence_penalty: Optional[float] = None
     frequency_penalty: Optional[float] = None
-    format: Optional[str] = None   # "json" -> grammar-constrained decoding
+    format: object = None          # "json" or a JSON-schema dict ->
+                                   # grammar-constrained decoding
 
     @classmethod
     def from_dict(cls, d: Optional[dict]) -> "SamplingOptions":
