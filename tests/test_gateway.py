@@ -468,3 +468,64 @@ def test_openai_chat_tools_streaming_buffered(client):
     if last["finish_reason"] == "tool_calls":
         assert any(e["choices"][0]["delta"].get("tool_calls")
                    for e in events[:-1])
+
+
+def test_generate_schema_format_enforced(client):
+    """format as a JSON-schema dict (Ollama contract) is ENFORCED — the
+    reference forwarded the object to Ollama (ollama.ts:45)."""
+    schema = {"type": "object",
+              "properties": {"ok": {"type": "boolean"}},
+              "required": ["ok"], "additionalProperties": False}
+    r = client.post("/api/generate", json={
+        "model": "tiny", "prompt": "emit", "stream": False,
+        "format": schema,
+        "options": {"temperature": 1.0, "seed": 6, "num_predict": 64}})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    if body["done_reason"] != "length":
+        doc = json.loads(body["response"])
+        assert set(doc) == {"ok"} and isinstance(doc["ok"], bool)
+
+
+def test_openai_json_schema_response_format(client):
+    schema = {"type": "object",
+              "properties": {"mode": {"enum": ["x", "y"]}},
+              "required": ["mode"], "additionalProperties": False}
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny", "stream": False,
+        "response_format": {"type": "json_schema",
+                            "json_schema": {"name": "m", "schema": schema}},
+        "messages": [{"role": "user", "content": "go"}],
+        "max_tokens": 48, "temperature": 1.0, "seed": 2})
+    assert r.status_code == 200, r.text
+    ch = r.json()["choices"][0]
+    if ch["finish_reason"] != "length":
+        assert json.loads(ch["message"]["content"]) in (
+            {"mode": "x"}, {"mode": "y"})
+
+
+def test_forced_tool_call_is_schema_valid(client):
+    """tool_choice=required + parameters schema: the call is enforced by
+    the schema automaton, so every completed reply converts to
+    tool_calls with arguments matching the function parameters."""
+    tools = [{"type": "function", "function": {
+        "name": "get_weather",
+        "parameters": {"type": "object",
+                       "properties": {"city": {"type": "string"}},
+                       "required": ["city"],
+                       "additionalProperties": False}}}]
+    for seed in (1, 2, 3):
+        r = client.post("/v1/chat/completions", json={
+            "model": "tiny", "stream": False, "tools": tools,
+            "tool_choice": "required",
+            "messages": [{"role": "user", "content": "weather in SF"}],
+            "max_tokens": 96, "temperature": 1.0, "seed": seed})
+        assert r.status_code == 200, r.text
+        ch = r.json()["choices"][0]
+        if ch["finish_reason"] == "length":
+            continue
+        assert ch["finish_reason"] == "tool_calls", ch
+        fn = ch["message"]["tool_calls"][0]["function"]
+        assert fn["name"] == "get_weather"
+        args = json.loads(fn["arguments"])
+        assert set(args) == {"city"} and isinstance(args["city"], str)
