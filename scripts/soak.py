@@ -138,6 +138,21 @@ async def run_mix(base: str, args) -> dict:
                     if not msg.get("tool_calls") and \
                             ch["finish_reason"] != "length":
                         json.loads(msg["content"])     # forced json
+            elif kind == "schema":
+                sch = {"type": "object",
+                       "properties": {"ok": {"type": "boolean"},
+                                      "n": {"type": "integer"}},
+                       "required": ["ok"], "additionalProperties": False}
+                async with sess.post("/api/generate", json={
+                        "model": "tiny", "stream": False, "format": sch,
+                        "prompt": "emit", "options":
+                        {"seed": i, "temperature": 1.0,
+                         "num_predict": 64}}) as r:
+                    b = await r.json()
+                    if b["done_reason"] != "length":
+                        doc = json.loads(b["response"])
+                        assert isinstance(doc.get("ok"), bool) and \
+                            set(doc) <= {"ok", "n"}, doc
             elif kind == "openai":
                 async with sess.post("/v1/completions", json={
                         "model": "tiny", "prompt": f"c{i}",
@@ -149,7 +164,8 @@ async def run_mix(base: str, args) -> dict:
             stats["failed"].append(f"{kind}#{i}: {e!r}")
 
     KINDS = ["generate", "stream", "chat", "embed", "json", "penalties",
-             "seeded_pair", "context", "cancel", "tools", "openai"]
+             "seeded_pair", "context", "cancel", "tools", "openai",
+             "schema"]
     conn = aiohttp.TCPConnector(limit=args.concurrency + 8)
     async with aiohttp.ClientSession(
             base_url=base, connector=conn,
