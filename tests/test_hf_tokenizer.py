@@ -110,3 +110,23 @@ def test_hf_stream_decoder_matches_full_decode(hf_tok):
             out.append(dec.push(i))
         out.append(dec.flush())
         assert "".join(out) == hf_tok.decode(ids), txt
+
+
+def test_engine_hf_tokenizer_with_schema(hf_tok):
+    """Schema constraint through the token-level mask path."""
+    from gridllm_amd.config import EngineConfig
+    from gridllm_amd.engine import InferenceEngine
+    from gridllm_amd.models.registry import get_spec
+    if get_spec("tiny").vocab < hf_tok.vocab_size:
+        pytest.skip("tiny vocab smaller than HF tokenizer")
+    eng = InferenceEngine(EngineConfig(
+        model="tiny", device="cpu", dtype="fp32", max_seq_len=256,
+        max_batch_size=2, use_hipgraph=False, seed=3))
+    eng.tokenizer = hf_tok
+    schema = {"type": "object",
+              "properties": {"mode": {"enum": ["a", "b"]}},
+              "required": ["mode"], "additionalProperties": False}
+    res = eng.generate("x", {"temperature": 1.0, "seed": 9,
+                             "format": schema, "num_predict": 60})
+    if res.done_reason != "length":
+        assert json.loads(res.text) in ({"mode": "a"}, {"mode": "b"})
