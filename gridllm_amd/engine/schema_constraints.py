@@ -152,9 +152,12 @@ class _Obj:
     OPEN, KEY_OR_CLOSE, KEY, COLON, AFTER = range(5)
 
     def __init__(self, schema: dict):
-        self.props = schema.get("properties") or {}
+        self.props = dict(schema.get("properties") or {})
         self.required = set(schema.get("required") or [])
-        self.required &= set(self.props) | self.required  # as given
+        for k in self.required:
+            # required key without a property schema: give it a generic
+            # one so the automaton can still offer (and force) it
+            self.props.setdefault(k, True)
         addl = schema.get("additionalProperties", True)
         self.addl = addl
         self.seen: set = set()
