@@ -115,3 +115,62 @@ def test_request_wire_roundtrip(rtype, pri, prompt, stream, seed):
     assert back.stream == stream
     assert back.options == req.options
     assert back.metadata == req.metadata
+
+
+SCHEMA_P = {"type": "object",
+            "properties": {"a": {"type": "integer"},
+                           "b": {"type": "array",
+                                 "items": {"type": "string"},
+                                 "maxItems": 3},
+                           "c": {"enum": ["x", "y", True]},
+                           "d": {"type": "number"}},
+            "required": ["a"], "additionalProperties": False}
+
+
+@settings(max_examples=120, deadline=None)
+@given(st.integers(min_value=-10**9, max_value=10**9),
+       st.lists(st.text(max_size=6), max_size=3),
+       st.sampled_from(["x", "y", True, None]),
+       st.floats(allow_nan=False, allow_infinity=False, width=32),
+       st.booleans(), st.booleans(), st.booleans())
+def test_schema_automaton_accepts_conforming_docs(a, b, c, d,
+                                                  use_b, use_c, use_d):
+    """Any document conforming to the schema must byte-walk through the
+    schema automaton to a complete state."""
+    from gridllm_amd.engine.schema_constraints import JsonSchemaByteConstraint
+    doc = {"a": a}
+    if use_b:
+        doc["b"] = b
+    if use_c and c is not None:
+        doc["c"] = c
+    if use_d:
+        doc["d"] = d
+    data = json.dumps(doc, separators=(",", ":"), ensure_ascii=False).encode()
+    con = JsonSchemaByteConstraint(SCHEMA_P)
+    for by in data:
+        allowed, _ = con.allowed()
+        assert by in allowed, (doc, data, chr(by))
+        con.push(by)
+    assert con.allowed()[1], doc
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.text(max_size=6), st.floats(allow_nan=False, allow_infinity=False))
+def test_schema_automaton_rejects_wrong_types(s, f):
+    """'a' must be an integer: a string or fractional float there is
+    refused at some byte."""
+    from gridllm_amd.engine.schema_constraints import JsonSchemaByteConstraint
+    for bad in (s, f):
+        if isinstance(bad, float) and bad == int(bad) and \
+                "e" not in repr(bad) and "E" not in repr(bad):
+            continue                      # 3.0 serializes with '.', ok to test
+        data = json.dumps({"a": bad}, separators=(",", ":"),
+                          ensure_ascii=False).encode()
+        con = JsonSchemaByteConstraint(SCHEMA_P)
+        ok = True
+        for by in data:
+            if by not in con.allowed()[0]:
+                ok = False
+                break
+            con.push(by)
+        assert not (ok and con.allowed()[1]), (bad, data)
