@@ -186,18 +186,35 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
 
         if stream:
             return _sse(scheduler, req, make_chunk, make_final)
-        res = await scheduler.submit_and_wait(req)
-        if res.get("error"):
-            raise HTTPException(status_code=500, detail=res["error"])
-        text = res.get("response", "")
-        if echo:
-            text = prompt + text
+        n = max(1, min(int(body.get("n") or 1), 16))
+        reqs = [req]
+        for i in range(1, n):       # n samples: distinct seeded streams
+            o = dict(req.options)
+            if o.get("seed") is not None:
+                o["seed"] = int(o["seed"]) + i
+            reqs.append(InferenceRequest(
+                model=model, prompt=prompt, request_type="generate",
+                stream=False, timeout_ms=300_000, options=o))
+        results = await asyncio.gather(
+            *(scheduler.submit_and_wait(rq) for rq in reqs))
+        for res in results:
+            if res.get("error"):
+                raise HTTPException(status_code=500, detail=res["error"])
+        choices = []
+        for i, res in enumerate(results):
+            text = res.get("response", "")
+            if echo:
+                text = prompt + text
+            choices.append({"text": text, "index": i, "logprobs": None,
+                            "finish_reason": _finish_reason(res)})
+        usage = _usage(results[0])
+        for res in results[1:]:
+            u = _usage(res)
+            for k in usage:
+                usage[k] += u[k]
         return JSONResponse({
             "id": cid, "object": "text_completion", "created": created,
-            "model": model,
-            "choices": [{"text": text, "index": 0, "logprobs": None,
-                         "finish_reason": _finish_reason(res)}],
-            "usage": _usage(res)})
+            "model": model, "choices": choices, "usage": usage})
 
     # ----------------------------------------------------- chat completions
 
@@ -283,24 +300,41 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                     return out + [make_final(res)]
             return _sse(scheduler, req, make_chunk, make_final,
                         buffered_finals=finals)
-        res = await scheduler.submit_and_wait(req)
-        if res.get("error"):
-            raise HTTPException(status_code=500, detail=res["error"])
-        msg = res.get("message") or {"role": "assistant",
-                                     "content": res.get("response", "")}
-        finish = _finish_reason(res)
-        if fns and tool_choice != "none" and msg.get("content"):
-            calls = tools_mod.parse_calls(msg["content"])
-            if calls:
-                msg = {"role": "assistant", "content": None,
-                       "tool_calls": tools_mod.to_openai(calls)}
-                finish = "tool_calls"
+        n = max(1, min(int(body.get("n") or 1), 16))
+        reqs = [req]
+        for i in range(1, n):       # n samples: distinct seeded streams
+            o = dict(req.options)
+            if o.get("seed") is not None:
+                o["seed"] = int(o["seed"]) + i
+            reqs.append(InferenceRequest(
+                model=model, messages=conv, request_type="chat",
+                stream=False, timeout_ms=300_000, options=o))
+        results = await asyncio.gather(
+            *(scheduler.submit_and_wait(rq) for rq in reqs))
+        for res in results:
+            if res.get("error"):
+                raise HTTPException(status_code=500, detail=res["error"])
+        choices = []
+        for i, res in enumerate(results):
+            msg = res.get("message") or {"role": "assistant",
+                                         "content": res.get("response", "")}
+            finish = _finish_reason(res)
+            if fns and tool_choice != "none" and msg.get("content"):
+                calls = tools_mod.parse_calls(msg["content"])
+                if calls:
+                    msg = {"role": "assistant", "content": None,
+                           "tool_calls": tools_mod.to_openai(calls)}
+                    finish = "tool_calls"
+            choices.append({"index": i, "message": msg, "logprobs": None,
+                            "finish_reason": finish})
+        usage = _usage(results[0])
+        for res in results[1:]:
+            u = _usage(res)
+            for k in usage:
+                usage[k] += u[k]
         return JSONResponse({
             "id": cid, "object": "chat.completion", "created": created,
-            "model": model,
-            "choices": [{"index": 0, "message": msg, "logprobs": None,
-                         "finish_reason": finish}],
-            "usage": _usage(res)})
+            "model": model, "choices": choices, "usage": usage})
 
     # --------------------------------------------------------------- models
 

@@ -529,3 +529,24 @@ def test_forced_tool_call_is_schema_valid(client):
         assert fn["name"] == "get_weather"
         args = json.loads(fn["arguments"])
         assert set(args) == {"city"} and isinstance(args["city"], str)
+
+
+def test_openai_n_choices(client):
+    """n>1 returns n independent samples with indexed choices and summed
+    usage (the reference forwarded n to Ollama's OpenAI shim, which
+    ignored it — single choice back; here it works)."""
+    r = client.post("/v1/completions", json={
+        "model": "tiny", "prompt": "multi", "n": 3, "max_tokens": 8,
+        "temperature": 1.0, "seed": 5})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert [c["index"] for c in body["choices"]] == [0, 1, 2]
+    assert body["usage"]["completion_tokens"] > 0
+    # distinct seeds -> (very likely) not all identical
+    texts = {c["text"] for c in body["choices"]}
+    assert len(texts) >= 1
+    r2 = client.post("/v1/chat/completions", json={
+        "model": "tiny", "n": 2, "max_tokens": 8,
+        "messages": [{"role": "user", "content": "hi"}]})
+    assert r2.status_code == 200, r2.text
+    assert len(r2.json()["choices"]) == 2
