@@ -14,8 +14,10 @@ Supported subset:
           a dict supplies the schema for unknown keys)
   array:  items, minItems, maxItems
   enum / const: any JSON values, matched by their compact serialization
-Anything else (anyOf, $ref, patterns, numeric bounds, type lists…)
-degrades that subtree to the plain syntactic-JSON constraint.
+  anyOf / oneOf: parallel branch automata (a byte is legal if any live
+          branch accepts it)
+Anything else ($ref, patterns, numeric bounds, type lists…) degrades
+that subtree to the plain syntactic-JSON constraint.
 
 Output is COMPACT JSON: insignificant whitespace is not offered to the
 sampler (standard for grammar-constrained decoding; any JSON consumer
@@ -295,10 +297,49 @@ class _Arr:
         pass
 
 
+class _Union:
+    """anyOf / oneOf: parallel branch automata over the same byte stream.
+    A byte is legal if ANY live branch accepts it; branches that refuse a
+    consumed byte are dropped; the value completes when any surviving
+    branch is complete."""
+
+    def __init__(self, branches: list):
+        self.branches = [JsonSchemaByteConstraint(s) for s in branches]
+        self.done = False
+
+    def allowed(self) -> set:
+        out: set = set()
+        for br in self.branches:
+            out |= br.allowed()[0]
+        return out
+
+    def complete(self) -> bool:
+        return any(br.allowed()[1] for br in self.branches)
+
+    def push(self, ctx, b: int):
+        keep = []
+        for br in self.branches:
+            if b in br.allowed()[0]:
+                br.push(b)
+                keep.append(br)
+        if not keep:
+            raise ValueError(f"union: illegal byte {b!r}")
+        self.branches = keep
+        return None
+
+    def child_done(self, f) -> None:  # pragma: no cover - leaf frame
+        pass
+
+
 def _first_bytes(schema) -> frozenset:
     """Bytes a value of this schema may start with (compact form)."""
     if not isinstance(schema, dict) or not schema:
         return _VALUE_START
+    if "anyOf" in schema or "oneOf" in schema:
+        out = frozenset()
+        for s in schema.get("anyOf") or schema.get("oneOf") or []:
+            out |= _first_bytes(s)
+        return out or _VALUE_START
     if "enum" in schema or "const" in schema:
         vals = schema.get("enum", [schema.get("const")])
         return frozenset(_dumps(v)[0] for v in vals if _dumps(v))
@@ -327,6 +368,13 @@ def make_frame(schema):
     """Schema node -> automaton frame. Unsupported constructs degrade to
     the plain syntactic constraint for that subtree."""
     if not isinstance(schema, dict) or not schema:
+        return _Gen()
+    if "anyOf" in schema or "oneOf" in schema:
+        branches = [s for s in (schema.get("anyOf")
+                                or schema.get("oneOf") or [])
+                    if isinstance(s, (dict, bool))]
+        if branches:
+            return _Union(branches)
         return _Gen()
     if "enum" in schema or "const" in schema:
         vals = schema.get("enum", [schema.get("const")])

@@ -157,3 +157,36 @@ def test_engine_schema_many_seeds(seed):
                              "format": schema, "num_predict": 40})
     if res.done_reason != "length":
         assert json.loads(res.text) in ({"mode": "a"}, {"mode": "b"})
+
+
+def test_anyof_branches():
+    sch = {"anyOf": [{"type": "integer"},
+                     {"type": "object",
+                      "properties": {"k": {"type": "string"}},
+                      "required": ["k"], "additionalProperties": False}]}
+    assert _accepts_doc(sch, -7)
+    assert _accepts_doc(sch, {"k": "v"})
+    assert _rejects_bytes(sch, b'"plain string"')
+    assert _rejects_bytes(sch, b'{"other":1}')
+    assert _rejects_bytes(sch, b"1.5")            # not integer, not object
+    # nested inside an object property
+    outer = {"type": "object",
+             "properties": {"v": sch}, "required": ["v"],
+             "additionalProperties": False}
+    assert _accepts_doc(outer, {"v": 3})
+    assert _accepts_doc(outer, {"v": {"k": "x"}})
+    assert _rejects_bytes(outer, b'{"v":true}')
+
+
+def test_oneof_ambiguous_prefix():
+    """Two object branches sharing a prefix disambiguate on the first
+    distinguishing byte."""
+    sch = {"oneOf": [
+        {"type": "object", "properties": {"alpha": {"type": "integer"}},
+         "required": ["alpha"], "additionalProperties": False},
+        {"type": "object", "properties": {"alt": {"type": "boolean"}},
+         "required": ["alt"], "additionalProperties": False}]}
+    assert _accepts_doc(sch, {"alpha": 1})
+    assert _accepts_doc(sch, {"alt": True})
+    assert _rejects_bytes(sch, b'{"alpha":true}')  # branch-1 type
+    assert _rejects_bytes(sch, b'{"alx"')          # neither key
