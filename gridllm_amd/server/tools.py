@@ -92,25 +92,28 @@ def to_ollama(calls: list[dict]) -> list[dict]:
 
 
 def call_schema(fns: list[dict], tool_choice=None) -> dict:
-    """JSON schema a FORCED tool call must satisfy: the tool-call shape
-    with `name` restricted to the allowed functions and `arguments`
-    validated against that function's `parameters`. With one candidate
-    the arguments schema applies exactly; with several, arguments are
-    any object (per-name dispatch needs oneOf, outside the enforced
-    subset — the gateway still parses and validates shape on return)."""
+    """JSON schema a FORCED tool call must satisfy: anyOf over the
+    allowed functions, each branch pinning `name` to that function and
+    validating `arguments` against ITS `parameters` — per-name dispatch,
+    enforced by the schema automaton's parallel branches
+    (engine/schema_constraints.py::_Union)."""
     names = [f["name"] for f in fns]
     if isinstance(tool_choice, dict):
         forced = (tool_choice.get("function") or {}).get("name")
         if forced in names:
             names = [forced]
-    cands = [f for f in fns if f["name"] in names]
-    if len(cands) == 1:
-        args = cands[0].get("parameters")
+    branches = []
+    for f in fns:
+        if f["name"] not in names:
+            continue
+        args = f.get("parameters")
         if not isinstance(args, dict) or not args:
             args = {"type": "object"}
-    else:
-        args = {"type": "object"}
-    return {"type": "object",
-            "properties": {"name": {"enum": names}, "arguments": args},
-            "required": ["name", "arguments"],
-            "additionalProperties": False}
+        branches.append({"type": "object",
+                         "properties": {"name": {"const": f["name"]},
+                                        "arguments": args},
+                         "required": ["name", "arguments"],
+                         "additionalProperties": False})
+    if len(branches) == 1:
+        return branches[0]
+    return {"anyOf": branches}

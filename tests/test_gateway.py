@@ -550,3 +550,40 @@ def test_openai_n_choices(client):
         "messages": [{"role": "user", "content": "hi"}]})
     assert r2.status_code == 200, r2.text
     assert len(r2.json()["choices"]) == 2
+
+
+def test_forced_call_two_tools_per_name_dispatch(client):
+    """With TWO tools and tool_choice=required, each completed call's
+    arguments must match the schema of the function it NAMED (anyOf
+    branch dispatch)."""
+    tools = [
+        {"type": "function", "function": {
+            "name": "add", "parameters": {
+                "type": "object",
+                "properties": {"x": {"type": "integer"},
+                               "y": {"type": "integer"}},
+                "required": ["x", "y"], "additionalProperties": False}}},
+        {"type": "function", "function": {
+            "name": "greet", "parameters": {
+                "type": "object",
+                "properties": {"who": {"type": "string"}},
+                "required": ["who"], "additionalProperties": False}}},
+    ]
+    for seed in (1, 2, 3, 4):
+        r = client.post("/v1/chat/completions", json={
+            "model": "tiny", "stream": False, "tools": tools,
+            "tool_choice": "required",
+            "messages": [{"role": "user", "content": "do something"}],
+            "max_tokens": 120, "temperature": 1.0, "seed": seed})
+        assert r.status_code == 200, r.text
+        ch = r.json()["choices"][0]
+        if ch["finish_reason"] == "length":
+            continue
+        fn = ch["message"]["tool_calls"][0]["function"]
+        args = json.loads(fn["arguments"])
+        if fn["name"] == "add":
+            assert set(args) == {"x", "y"} and \
+                all(isinstance(v, int) for v in args.values())
+        else:
+            assert fn["name"] == "greet"
+            assert set(args) == {"who"} and isinstance(args["who"], str)
