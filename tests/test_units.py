@@ -103,3 +103,18 @@ def test_tools_module_parse_and_shapes():
     assert json.loads(oa[0]["function"]["arguments"]) == {"city": "SF"}
     ol = T.to_ollama(calls)
     assert ol[0]["function"]["arguments"] == {"city": "SF"}   # object
+
+
+def test_call_schema_shapes():
+    from gridllm_amd.server import tools as T
+    fns = [{"name": "a", "parameters": {"type": "object",
+                                        "properties": {"x": {"type": "integer"}},
+                                        "required": ["x"]}},
+           {"name": "b"}]
+    one = T.call_schema(fns, {"function": {"name": "a"}})
+    assert one["properties"]["name"]["const"] == "a"
+    assert one["properties"]["arguments"]["required"] == ["x"]
+    both = T.call_schema(fns)
+    assert "anyOf" in both and len(both["anyOf"]) == 2
+    names = {br["properties"]["name"]["const"] for br in both["anyOf"]}
+    assert names == {"a", "b"}
