@@ -16,8 +16,10 @@ Supported subset:
   enum / const: any JSON values, matched by their compact serialization
   anyOf / oneOf: parallel branch automata (a byte is legal if any live
           branch accepts it)
-Anything else ($ref, patterns, numeric bounds, type lists…) degrades
-that subtree to the plain syntactic-JSON constraint.
+  $ref:   local ('#/'-rooted) pointers, incl. RECURSIVE schemas (frames
+          build lazily as the document descends)
+Anything else (patterns, numeric bounds, type lists…) degrades that
+subtree to the plain syntactic-JSON constraint.
 
 Output is COMPACT JSON: insignificant whitespace is not offered to the
 sampler (standard for grammar-constrained decoding; any JSON consumer
@@ -39,6 +41,30 @@ from gridllm_amd.engine import constraints as C
 _STR_MODES = (C.STRING, C.STR_ESC, C.STR_U1, C.STR_U2, C.STR_U3, C.STR_U4)
 _VALUE_START = frozenset(b'{["-0123456789tfn')
 _NUM_START = frozenset(b"-0123456789")
+
+
+def _deref(schema, root):
+    """Resolve local $ref chains ('#/$defs/x', '#/definitions/x', any
+    '#/'-rooted pointer). Lazy frame construction makes RECURSIVE schemas
+    safe: a frame for a $ref subtree is only built when the document
+    actually descends into it."""
+    hops = 0
+    while isinstance(schema, dict) and isinstance(schema.get("$ref"), str) \
+            and schema["$ref"].startswith("#") and root is not None:
+        hops += 1
+        if hops > 32:
+            return True                   # cyclic ref chain: degrade
+        node = root
+        for part in schema["$ref"].lstrip("#/").split("/"):
+            if not part:
+                continue
+            part = part.replace("~1", "/").replace("~0", "~")
+            if isinstance(node, dict) and part in node:
+                node = node[part]
+            else:
+                return True               # unresolvable: degrade
+        schema = node
+    return schema
 
 
 def _dumps(v) -> bytes:
@@ -153,7 +179,8 @@ class _KeyAny:
 class _Obj:
     OPEN, KEY_OR_CLOSE, KEY, COLON, AFTER = range(5)
 
-    def __init__(self, schema: dict):
+    def __init__(self, schema: dict, root=None):
+        self.root = root
         self.props = dict(schema.get("properties") or {})
         self.required = set(schema.get("required") or [])
         for k in self.required:
@@ -228,7 +255,7 @@ class _Obj:
             else:
                 sub = True                # generic value
             self.state = self.AFTER
-            return make_frame(sub)
+            return make_frame(sub, self.root)
         raise ValueError(f"object: illegal byte {b!r} in state {self.state}")
 
     def child_done(self, f) -> None:
@@ -241,7 +268,8 @@ class _Obj:
 class _Arr:
     OPEN, FIRST, AFTER = range(3)
 
-    def __init__(self, schema: dict):
+    def __init__(self, schema: dict, root=None):
+        self.root = root
         self.items = schema.get("items", True)
         self.min_items = int(schema.get("minItems") or 0)
         mx = schema.get("maxItems")
@@ -256,7 +284,7 @@ class _Arr:
         if self.state == self.OPEN:
             return {0x5B}
         if self.state == self.FIRST:
-            a = set(_first_bytes(self.items)) \
+            a = set(_first_bytes(self.items, self.root)) \
                 if self.max_items is None or self.max_items > 0 else set()
             if self.min_items <= 0:
                 a.add(0x5D)
@@ -287,10 +315,10 @@ class _Arr:
         if self.state == self.FIRST:
             self.count += 1
             self.state = self.AFTER
-            return (make_frame(self.items), b)      # replay b into child
+            return (make_frame(self.items, self.root), b)  # replay b
         if self.state == self.AFTER and b == 0x2C:
             self.count += 1
-            return make_frame(self.items)
+            return make_frame(self.items, self.root)
         raise ValueError(f"array: illegal byte {b!r} in state {self.state}")
 
     def child_done(self, f) -> None:
@@ -303,8 +331,9 @@ class _Union:
     consumed byte are dropped; the value completes when any surviving
     branch is complete."""
 
-    def __init__(self, branches: list):
-        self.branches = [JsonSchemaByteConstraint(s) for s in branches]
+    def __init__(self, branches: list, root=None):
+        self.branches = [JsonSchemaByteConstraint(s, root)
+                         for s in branches]
         self.done = False
 
     def allowed(self) -> set:
@@ -331,14 +360,15 @@ class _Union:
         pass
 
 
-def _first_bytes(schema) -> frozenset:
+def _first_bytes(schema, root=None) -> frozenset:
     """Bytes a value of this schema may start with (compact form)."""
+    schema = _deref(schema, root)
     if not isinstance(schema, dict) or not schema:
         return _VALUE_START
     if "anyOf" in schema or "oneOf" in schema:
         out = frozenset()
         for s in schema.get("anyOf") or schema.get("oneOf") or []:
-            out |= _first_bytes(s)
+            out |= _first_bytes(s, root)
         return out or _VALUE_START
     if "enum" in schema or "const" in schema:
         vals = schema.get("enum", [schema.get("const")])
@@ -347,7 +377,7 @@ def _first_bytes(schema) -> frozenset:
     if isinstance(t, list):
         out = frozenset()
         for ti in t:
-            out |= _first_bytes({**schema, "type": ti})
+            out |= _first_bytes({**schema, "type": ti}, root)
         return out
     if t == "object" or (t is None and "properties" in schema):
         return frozenset(b"{")
@@ -364,9 +394,10 @@ def _first_bytes(schema) -> frozenset:
     return _VALUE_START
 
 
-def make_frame(schema):
+def make_frame(schema, root=None):
     """Schema node -> automaton frame. Unsupported constructs degrade to
     the plain syntactic constraint for that subtree."""
+    schema = _deref(schema, root)
     if not isinstance(schema, dict) or not schema:
         return _Gen()
     if "anyOf" in schema or "oneOf" in schema:
@@ -374,7 +405,7 @@ def make_frame(schema):
                                 or schema.get("oneOf") or [])
                     if isinstance(s, (dict, bool))]
         if branches:
-            return _Union(branches)
+            return _Union(branches, root)
         return _Gen()
     if "enum" in schema or "const" in schema:
         vals = schema.get("enum", [schema.get("const")])
@@ -383,9 +414,9 @@ def make_frame(schema):
     if isinstance(t, list):
         return _Gen(first=_first_bytes(schema))
     if t == "object" or (t is None and "properties" in schema):
-        return _Obj(schema)
+        return _Obj(schema, root)
     if t == "array":
-        return _Arr(schema)
+        return _Arr(schema, root)
     if t == "string":
         return _Gen(first=frozenset(b'"'))
     if t == "integer":
@@ -402,8 +433,10 @@ def make_frame(schema):
 class JsonSchemaByteConstraint:
     """Drop-in sibling of JsonByteConstraint, directed by a JSON schema."""
 
-    def __init__(self, schema):
-        self.stack = [make_frame(schema)]
+    def __init__(self, schema, root=None):
+        if root is None and isinstance(schema, dict):
+            root = schema
+        self.stack = [make_frame(schema, root)]
 
     def allowed(self) -> tuple:
         out: set = set()

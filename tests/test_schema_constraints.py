@@ -190,3 +190,28 @@ def test_oneof_ambiguous_prefix():
     assert _accepts_doc(sch, {"alt": True})
     assert _rejects_bytes(sch, b'{"alpha":true}')  # branch-1 type
     assert _rejects_bytes(sch, b'{"alx"')          # neither key
+
+
+def test_ref_and_recursive_schema():
+    """Local $refs resolve; RECURSIVE schemas work because frames build
+    lazily as the document descends (a tree of nodes)."""
+    sch = {"$defs": {"node": {
+               "type": "object",
+               "properties": {
+                   "v": {"type": "integer"},
+                   "kids": {"type": "array",
+                            "items": {"$ref": "#/$defs/node"},
+                            "maxItems": 2}},
+               "required": ["v"], "additionalProperties": False}},
+           "$ref": "#/$defs/node"}
+    assert _accepts_doc(sch, {"v": 1})
+    assert _accepts_doc(sch, {"v": 1, "kids": [
+        {"v": 2}, {"v": 3, "kids": [{"v": 4}]}]})
+    assert _rejects_bytes(sch, b'{"v":"str"}')
+    assert _rejects_bytes(sch, b'{"v":1,"kids":[{"nope":2}]}')
+    assert _rejects_bytes(sch, b'{"v":1,"kids":[{"v":1},{"v":2},{"v":3}]}')
+
+
+def test_unresolvable_ref_degrades():
+    sch = {"$ref": "#/nothing/here"}
+    assert _accepts_doc(sch, {"any": ["thing", 1]})   # generic JSON
