@@ -251,3 +251,69 @@ def test_chaos_sigkill_during_streaming(cluster):
         if victim.poll() is None:
             victim.send_signal(signal.SIGKILL)
             victim.wait(timeout=10)
+
+
+def test_module_entrypoints_e2e():
+    """The DOCUMENTED deployment commands (`python -m gridllm_amd.server`,
+    `python -m gridllm_amd.worker`, README quickstart) boot, register,
+    and serve a request."""
+    import httpx
+
+    def free():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    hp, bp, wp = free(), free(), free()
+    env = {**os.environ, "PORT": str(hp), "BUS_PORT": str(bp),
+           "PYTHONPATH": REPO}
+    srv = subprocess.Popen([sys.executable, "-m", "gridllm_amd.server"],
+                           env=env, stdout=subprocess.PIPE,
+                           stderr=subprocess.STDOUT, text=True)
+    wenv = {**env, "ENGINE_MODEL": "tiny", "ENGINE_DEVICE": "cpu",
+            "ENGINE_DTYPE": "fp32", "WORKER_PORT": str(wp),
+            "BUS_HOST": "127.0.0.1", "WORKER_ID": "mod-entry-w"}
+    wrk = None
+    try:
+        for _ in range(100):            # server HTTP up
+            time.sleep(0.2)
+            try:
+                if httpx.get(f"http://127.0.0.1:{hp}/health",
+                             timeout=2).status_code == 200:
+                    break
+            except Exception:  # noqa: BLE001
+                pass
+        else:
+            raise AssertionError("server module entry did not come up")
+        wrk = subprocess.Popen([sys.executable, "-m", "gridllm_amd.worker"],
+                               env=wenv, stdout=subprocess.PIPE,
+                               stderr=subprocess.STDOUT, text=True)
+        for _ in range(150):            # worker registered
+            time.sleep(0.2)
+            try:
+                r = httpx.get(f"http://127.0.0.1:{hp}/health/workers",
+                              timeout=2)
+                if r.status_code == 200 and \
+                        r.json()["counts"].get("total", 0) >= 1:
+                    break
+            except Exception:  # noqa: BLE001
+                pass
+        else:
+            raise AssertionError("worker module entry never registered")
+        g = httpx.post(f"http://127.0.0.1:{hp}/api/generate",
+                       json={"model": "tiny", "prompt": "hi",
+                             "stream": False,
+                             "options": {"num_predict": 4}}, timeout=60)
+        assert g.status_code == 200 and g.json()["eval_count"] == 4, g.text
+    finally:
+        for p in (wrk, srv):
+            if p is not None and p.poll() is None:
+                p.send_signal(signal.SIGTERM)
+        for p in (wrk, srv):
+            if p is not None:
+                try:
+                    p.wait(timeout=10)
+                except Exception:  # noqa: BLE001
+                    p.kill()
