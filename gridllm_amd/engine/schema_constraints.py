@@ -13,6 +13,7 @@ Supported subset:
           keys to `properties` and lets masking FORCE the required keys;
           a dict supplies the schema for unknown keys)
   array:  items, minItems, maxItems
+  string: minLength / maxLength (code-point counted)
   enum / const: any JSON values, matched by their compact serialization
   anyOf / oneOf: parallel branch automata (a byte is legal if any live
           branch accepts it)
@@ -171,6 +172,88 @@ class _KeyAny:
                 self.matched = bytes(self.raw[:-1]).decode(
                     "utf-8", errors="replace")
         return None
+
+    def child_done(self, f) -> None:  # pragma: no cover - leaf frame
+        pass
+
+
+_STR_CONTENT = frozenset(b for b in range(0x20, 0x100)
+                         if b not in (0x22, 0x5C))
+_CONT = frozenset(range(0x80, 0xC0))          # UTF-8 continuation bytes
+
+
+class _Str:
+    """String with minLength/maxLength: counts CODE POINTS (UTF-8 lead
+    bytes and escapes count 1, continuation bytes 0; a lead byte's
+    continuations are then mandatory, so a capped string cannot be padded
+    with stray continuation bytes)."""
+
+    BODY, ESC, U1, U2, U3, U4 = range(6)
+
+    def __init__(self, schema: dict):
+        self.min_l = int(schema.get("minLength") or 0)
+        mx = schema.get("maxLength")
+        self.max_l = int(mx) if mx is not None else None
+        self.n = 0
+        self.rem = 0                   # continuation bytes still owed
+        self.state = None              # None = before opening quote
+        self.done = False
+
+    def allowed(self) -> set:
+        if self.done:
+            return set()
+        if self.state is None:
+            return {0x22}
+        if self.state == self.ESC:
+            return frozenset(b'"\\/bfnrtu')
+        if self.state in (self.U1, self.U2, self.U3, self.U4):
+            return frozenset(b"0123456789abcdefABCDEF")
+        # BODY
+        if self.rem:
+            return set(_CONT)
+        a: set = set()
+        if self.max_l is None or self.n < self.max_l:
+            a |= _STR_CONTENT - _CONT  # new chars: ASCII + lead bytes
+            a.add(0x5C)
+        if self.n >= self.min_l:
+            a.add(0x22)
+        return a
+
+    def complete(self) -> bool:
+        return self.done
+
+    def push(self, ctx, b: int):
+        if self.state is None and b == 0x22:
+            self.state = self.BODY
+            return None
+        if self.state == self.BODY:
+            if self.rem:
+                self.rem -= 1
+                return None
+            if b == 0x22:
+                self.done = True
+            elif b == 0x5C:
+                self.state = self.ESC
+                self.n += 1
+            else:
+                self.n += 1
+                if 0xC2 <= b <= 0xDF:
+                    self.rem = 1
+                elif 0xE0 <= b <= 0xEF:
+                    self.rem = 2
+                elif 0xF0 <= b <= 0xF4:
+                    self.rem = 3
+            return None
+        if self.state == self.ESC:
+            self.state = self.U1 if b == 0x75 else self.BODY
+            return None
+        if self.state in (self.U1, self.U2, self.U3):
+            self.state += 1
+            return None
+        if self.state == self.U4:
+            self.state = self.BODY
+            return None
+        raise ValueError(f"string: illegal byte {b!r}")
 
     def child_done(self, f) -> None:  # pragma: no cover - leaf frame
         pass
@@ -418,6 +501,8 @@ def make_frame(schema, root=None):
     if t == "array":
         return _Arr(schema, root)
     if t == "string":
+        if schema.get("minLength") or schema.get("maxLength") is not None:
+            return _Str(schema)
         return _Gen(first=frozenset(b'"'))
     if t == "integer":
         return _Gen(first=_NUM_START, banned=frozenset(b".eE"))

@@ -215,3 +215,36 @@ def test_ref_and_recursive_schema():
 def test_unresolvable_ref_degrades():
     sch = {"$ref": "#/nothing/here"}
     assert _accepts_doc(sch, {"any": ["thing", 1]})   # generic JSON
+
+
+def test_string_length_bounds():
+    sch = {"type": "string", "minLength": 2, "maxLength": 4}
+    assert _accepts_doc(sch, "ab")
+    assert _accepts_doc(sch, "abcd")
+    assert _accepts_doc(sch, "héé")              # code points, not bytes
+    assert _accepts_doc(sch, 'a"b'[:2])
+    assert _rejects_bytes(sch, b'"a"')            # too short
+    assert _rejects_bytes(sch, b'"abcde"')        # too long
+    assert _rejects_bytes(sch, b'"\xc3\xa9\xc3\xa9\xc3\xa9\xc3\xa9\xc3\xa9"')
+    assert _accepts_doc(sch, "a\\b"[:3])          # escape counts 1
+    # inside an object, and forced by masking: a random walk respects it
+    obj = {"type": "object",
+           "properties": {"s": {"type": "string", "minLength": 1,
+                                "maxLength": 3}},
+           "required": ["s"], "additionalProperties": False}
+    rng = random.Random(3)
+    for _ in range(40):
+        c = JsonSchemaByteConstraint(obj)
+        out = bytearray()
+        for _ in range(4000):
+            a, eos = c.allowed()
+            if eos and (not a or rng.random() < 0.3):
+                break
+            assert a, bytes(out)
+            b = rng.choice(sorted(a))
+            c.push(b)
+            out.append(b)
+        if not c.allowed()[1]:
+            continue
+        doc = json.loads(out.decode("utf-8", "replace"))
+        assert 1 <= len(doc["s"]) <= 3, doc
