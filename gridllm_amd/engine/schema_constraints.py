@@ -177,16 +177,22 @@ class _KeyAny:
         pass
 
 
-_STR_CONTENT = frozenset(b for b in range(0x20, 0x100)
-                         if b not in (0x22, 0x5C))
-_CONT = frozenset(range(0x80, 0xC0))          # UTF-8 continuation bytes
+_ASCII_CONTENT = frozenset(b for b in range(0x20, 0x80)
+                           if b not in (0x22, 0x5C))
+# valid UTF-8 lead bytes (no overlongs via C0/C1, nothing past F4)
+_LEADS = frozenset(range(0xC2, 0xF5))
+_CONT_DEFAULT = (0x80, 0xBF)
+# first-continuation range per lead (excludes overlongs, surrogates,
+# and code points beyond U+10FFFF)
+_FIRST_CONT = {0xE0: (0xA0, 0xBF), 0xED: (0x80, 0x9F),
+               0xF0: (0x90, 0xBF), 0xF4: (0x80, 0x8F)}
 
 
 class _Str:
-    """String with minLength/maxLength: counts CODE POINTS (UTF-8 lead
-    bytes and escapes count 1, continuation bytes 0; a lead byte's
-    continuations are then mandatory, so a capped string cannot be padded
-    with stray continuation bytes)."""
+    """String with minLength/maxLength: counts CODE POINTS. Content is
+    restricted to VALID UTF-8 (lead-dependent continuation ranges — no
+    overlongs/surrogates), so the automaton's count equals the decoded
+    string's length; escapes count 1."""
 
     BODY, ESC, U1, U2, U3, U4 = range(6)
 
@@ -196,6 +202,7 @@ class _Str:
         self.max_l = int(mx) if mx is not None else None
         self.n = 0
         self.rem = 0                   # continuation bytes still owed
+        self.crange = _CONT_DEFAULT    # range for the NEXT continuation
         self.state = None              # None = before opening quote
         self.done = False
 
@@ -210,10 +217,11 @@ class _Str:
             return frozenset(b"0123456789abcdefABCDEF")
         # BODY
         if self.rem:
-            return set(_CONT)
+            lo, hi = self.crange
+            return set(range(lo, hi + 1))
         a: set = set()
         if self.max_l is None or self.n < self.max_l:
-            a |= _STR_CONTENT - _CONT  # new chars: ASCII + lead bytes
+            a |= _ASCII_CONTENT | _LEADS
             a.add(0x5C)
         if self.n >= self.min_l:
             a.add(0x22)
@@ -229,6 +237,7 @@ class _Str:
         if self.state == self.BODY:
             if self.rem:
                 self.rem -= 1
+                self.crange = _CONT_DEFAULT
                 return None
             if b == 0x22:
                 self.done = True
@@ -243,6 +252,7 @@ class _Str:
                     self.rem = 2
                 elif 0xF0 <= b <= 0xF4:
                     self.rem = 3
+                self.crange = _FIRST_CONT.get(b, _CONT_DEFAULT)
             return None
         if self.state == self.ESC:
             self.state = self.U1 if b == 0x75 else self.BODY
