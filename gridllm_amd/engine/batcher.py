@@ -158,6 +158,9 @@ class Batcher:
                     self.engine.step_rows([row for _, row in live])
                 except Exception as e:  # noqa: BLE001
                     log.logger.exception("decode step failed")
+                    # retire the rows or their KV slots leak permanently
+                    # and admission wedges
+                    self.engine.abort_rows([row for _, row in live])
                     for p, _ in live:
                         if not p.future.done():
                             p.future.set_exception(e)

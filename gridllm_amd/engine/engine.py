@@ -297,8 +297,12 @@ class InferenceEngine:
             [(prompt, opts_d, on_tokens[i], cancels[i], extra_stops)
              for i, (prompt, opts_d) in enumerate(requests)])
         active = [r for r in rows if not r.done]
-        while active:
-            active = self._step_rows(active)
+        try:
+            while active:
+                active = self._step_rows(active)
+        except Exception:
+            self.abort_rows(active, locked=True)   # no slot leak on failure
+            raise
         return [r.result for r in rows]
 
     # -------------------------------------------------- row-stepping API
@@ -385,13 +389,19 @@ class InferenceEngine:
             for sl in slots:          # release what this batch allocated
                 self.model.cache.free(sl)
             raise
-        if len(rows) == 1:
-            logits = self.model.forward_prefill(token_lists[0],
-                                                slots[0]).unsqueeze(0)
-        else:
-            logits = self.model.forward_prefill_batch(token_lists, slots)
-        self._sync_cancel(rows)
-        toks = self._sample_batch(logits, rows)           # syncs
+        try:
+            if len(rows) == 1:
+                logits = self.model.forward_prefill(token_lists[0],
+                                                    slots[0]).unsqueeze(0)
+            else:
+                logits = self.model.forward_prefill_batch(token_lists, slots)
+            self._sync_cancel(rows)
+            toks = self._sample_batch(logits, rows)           # syncs
+        except Exception:
+            # prefill/first-sample failure: release every slot this batch
+            # allocated or capacity leaks permanently
+            self.abort_rows(rows, locked=True)
+            raise
         t1 = time.monotonic_ns()
         per = (t1 - t0) // max(1, len(rows))
         for row, tok in zip(rows, toks):
@@ -400,6 +410,18 @@ class InferenceEngine:
             if not (self._emit(row, tok) and row.max_new > 1):
                 self._retire(row)
         return rows
+
+    def abort_rows(self, rows: Sequence["_Row"], locked: bool = False) -> None:
+        """Free the KV slots of rows that have not retired (step/prefill
+        failure cleanup). Safe to call with already-retired rows: _retire
+        sets done and frees exactly once."""
+        if not locked:
+            with self._lock:
+                return self.abort_rows(rows, locked=True)
+        for r in rows:
+            if not r.done:
+                r.done = True
+                self.model.cache.free(r.slot)
 
     def _step_rows(self, active: list["_Row"]) -> list["_Row"]:
         """One batched decode step; returns the rows still running."""
