@@ -196,6 +196,10 @@ class _Str:
 
     BODY, ESC, U1, U2, U3, U4 = range(6)
 
+    _HEX = frozenset(b"0123456789abcdefABCDEF")
+    _LOW2 = frozenset(b"cdefCDEF")       # 2nd hex digit of DC00-DFFF
+    _HIGH2 = frozenset(b"89abAB")        # 2nd hex digit of D800-DBFF
+
     def __init__(self, schema: dict):
         self.min_l = int(schema.get("minLength") or 0)
         mx = schema.get("maxLength")
@@ -205,6 +209,13 @@ class _Str:
         self.crange = _CONT_DEFAULT    # range for the NEXT continuation
         self.state = None              # None = before opening quote
         self.done = False
+        # surrogate-pair escapes: a \uD800-\uDBFF escape counts ONE code
+        # point together with the \uDC00-\uDFFF escape that MUST follow
+        # (round-1 advisor finding: pairs were counted as 2 and a lone
+        # high surrogate could be force-closed into invalid text)
+        self.pending_high = False      # inside a pair, low half owed
+        self.hex0 = 0                  # first hex digit of current escape
+        self.is_high = False           # current escape is a high surrogate
 
     def allowed(self) -> set:
         if self.done:
@@ -212,10 +223,25 @@ class _Str:
         if self.state is None:
             return {0x22}
         if self.state == self.ESC:
+            if self.pending_high:
+                return {0x75}          # only \uDC00-\uDFFF may follow
             return frozenset(b'"\\/bfnrtu')
-        if self.state in (self.U1, self.U2, self.U3, self.U4):
-            return frozenset(b"0123456789abcdefABCDEF")
+        if self.state == self.U1:
+            if self.pending_high:
+                return frozenset(b"dD")
+            return self._HEX
+        if self.state == self.U2:
+            if self.hex0 in b"dD":
+                if self.pending_high:
+                    return self._LOW2          # low half of the pair
+                # D800-DBFF starts a pair (allowed); lone DC00-DFFF banned
+                return self._HEX - self._LOW2
+            return self._HEX
+        if self.state in (self.U3, self.U4):
+            return self._HEX
         # BODY
+        if self.pending_high:
+            return {0x5C}              # the pair's low half is mandatory
         if self.rem:
             lo, hi = self.crange
             return set(range(lo, hi + 1))
@@ -243,7 +269,8 @@ class _Str:
                 self.done = True
             elif b == 0x5C:
                 self.state = self.ESC
-                self.n += 1
+                if not self.pending_high:   # pair counts once, at its start
+                    self.n += 1
             else:
                 self.n += 1
                 if 0xC2 <= b <= 0xDF:
@@ -257,10 +284,25 @@ class _Str:
         if self.state == self.ESC:
             self.state = self.U1 if b == 0x75 else self.BODY
             return None
-        if self.state in (self.U1, self.U2, self.U3):
-            self.state += 1
+        if self.state == self.U1:
+            self.hex0 = b
+            self.state = self.U2
+            return None
+        if self.state == self.U2:
+            if self.hex0 in b"dD" and b in self._HIGH2:
+                self.is_high = True     # this escape opens a pair
+            else:
+                self.is_high = False
+            self.state = self.U3
+            return None
+        if self.state == self.U3:
+            self.state = self.U4
             return None
         if self.state == self.U4:
+            if self.pending_high:
+                self.pending_high = False   # pair completed
+            elif getattr(self, "is_high", False):
+                self.pending_high = True    # low half now owed
             self.state = self.BODY
             return None
         raise ValueError(f"string: illegal byte {b!r}")

@@ -53,6 +53,16 @@ def _usage(res: dict) -> dict:
     return {"prompt_tokens": p, "completion_tokens": c, "total_tokens": p + c}
 
 
+def _usage_n(results: list) -> dict:
+    """Usage across n samples of ONE prompt: the shared prompt counts once;
+    completion tokens sum (OpenAI semantics — round-1 advisor finding)."""
+    p = int((results[0].get("timings") or {}).get("prompt_eval_count", 0))
+    c = sum(int((r.get("timings") or {}).get("eval_count", 0))
+            for r in results)
+    return {"prompt_tokens": p, "completion_tokens": c,
+            "total_tokens": p + c}
+
+
 def _finish_reason(res: dict) -> str:
     return {"length": "length", "cancelled": "stop"}.get(
         res.get("done_reason", "stop"), "stop")
@@ -150,6 +160,14 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                         break
             finally:
                 task.cancel()
+                if not task.done():
+                    # client disconnected mid-stream: abort the job on the
+                    # worker too, or it keeps generating to completion
+                    # (round-1 advisor finding)
+                    try:
+                        await scheduler.cancel_job(req.id)
+                    except Exception:  # noqa: BLE001
+                        pass
 
         return StreamingResponse(gen(), media_type="text/event-stream")
 
@@ -207,14 +225,10 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                 text = prompt + text
             choices.append({"text": text, "index": i, "logprobs": None,
                             "finish_reason": _finish_reason(res)})
-        usage = _usage(results[0])
-        for res in results[1:]:
-            u = _usage(res)
-            for k in usage:
-                usage[k] += u[k]
         return JSONResponse({
             "id": cid, "object": "text_completion", "created": created,
-            "model": model, "choices": choices, "usage": usage})
+            "model": model, "choices": choices,
+            "usage": _usage_n(results)})
 
     # ----------------------------------------------------- chat completions
 
@@ -327,14 +341,10 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                     finish = "tool_calls"
             choices.append({"index": i, "message": msg, "logprobs": None,
                             "finish_reason": finish})
-        usage = _usage(results[0])
-        for res in results[1:]:
-            u = _usage(res)
-            for k in usage:
-                usage[k] += u[k]
         return JSONResponse({
             "id": cid, "object": "chat.completion", "created": created,
-            "model": model, "choices": choices, "usage": usage})
+            "model": model, "choices": choices,
+            "usage": _usage_n(results)})
 
     # --------------------------------------------------------------- models
 

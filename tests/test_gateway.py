@@ -542,6 +542,11 @@ def test_openai_n_choices(client):
     body = r.json()
     assert [c["index"] for c in body["choices"]] == [0, 1, 2]
     assert body["usage"]["completion_tokens"] > 0
+    # OpenAI semantics: the shared prompt is counted ONCE, completions sum
+    u = body["usage"]
+    assert u["completion_tokens"] >= 3          # 3 samples, >=1 token each
+    assert u["prompt_tokens"] < 3 * len("multi") + 3
+    assert u["total_tokens"] == u["prompt_tokens"] + u["completion_tokens"]
     # distinct seeds -> (very likely) not all identical
     texts = {c["text"] for c in body["choices"]}
     assert len(texts) >= 1

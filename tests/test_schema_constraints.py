@@ -248,3 +248,43 @@ def test_string_length_bounds():
             continue
         doc = json.loads(out.decode("utf-8", "replace"))
         assert 1 <= len(doc["s"]) <= 3, doc
+
+
+def test_surrogate_pair_escapes():
+    """\\uXXXX surrogate-pair escapes count ONE code point; lone
+    surrogates are unreachable (round-1 advisor finding)."""
+    sch = {"type": "string", "maxLength": 1}
+    # an astral char written as a surrogate-pair escape fits maxLength 1
+    pair = b'"\\ud83d\\ude00"'
+    c = JsonSchemaByteConstraint(sch)
+    for b in pair:
+        assert b in c.allowed()[0], (bytes([b]), pair)
+        c.push(b)
+    assert c.allowed()[1]
+    # after a high-surrogate escape the ONLY continuation is \\uDC00-DFFF:
+    # no closing quote, no raw bytes, and the json.dumps ensure_ascii
+    # encoding of an astral char is accepted end-to-end
+    c = JsonSchemaByteConstraint(sch)
+    for b in b'"\\ud800':
+        c.push(b)
+    a, eos = c.allowed()
+    assert a == {0x5C} and not eos
+    c.push(0x5C)
+    assert c.allowed()[0] == {0x75}
+    c.push(0x75)
+    assert c.allowed()[0] == frozenset(b"dD")
+    c.push(ord("d"))
+    assert c.allowed()[0] == frozenset(b"cdefCDEF")
+    # a lone low surrogate can never start: second hex digit c-f banned
+    c = JsonSchemaByteConstraint(sch)
+    for b in b'"\\ud':
+        c.push(b)
+    assert not (c.allowed()[0] & set(b"cdefCDEF"))
+    # and the pair still respects maxLength: a second char is too long
+    sch2 = {"type": "string", "maxLength": 2, "minLength": 2}
+    ok = json.dumps("\U0001F600a", ensure_ascii=True).encode()
+    c = JsonSchemaByteConstraint(sch2)
+    for b in ok:
+        assert b in c.allowed()[0], (bytes([b]), ok)
+        c.push(b)
+    assert c.allowed()[1]
