@@ -35,6 +35,7 @@ class _Pending:
     cancel: Optional[threading.Event] = None
     future: Future = field(default_factory=Future)
     extra_stops: Optional[list[str]] = None
+    think: bool = False
 
 
 class Batcher:
@@ -76,9 +77,9 @@ class Batcher:
         return p.future
 
     def submit_chat(self, messages: list, options: dict,
-                    on_token=None, cancel=None) -> Future:
+                    on_token=None, cancel=None, think: bool = False) -> Future:
         p = _Pending(kind="chat", messages=messages, options=options,
-                     on_token=on_token, cancel=cancel)
+                     on_token=on_token, cancel=cancel, think=think)
         self.q.put(p)
         return p.future
 
@@ -92,7 +93,7 @@ class Batcher:
     def _prep(self, p: _Pending) -> tuple[str, dict]:
         """Render the prompt + merge stop sequences for a generate/chat row."""
         if p.kind == "chat":
-            prompt = chat_mod.render_chat(p.messages or [])
+            prompt = chat_mod.render_chat(p.messages or [], think=p.think)
             opts = dict(p.options)
             opts["stop"] = list(opts.get("stop") or []) + \
                 list(chat_mod.CHAT_STOPS)
@@ -220,7 +221,8 @@ class Batcher:
         stops_all: Optional[list[str]] = None
         for p in gens:
             if p.kind == "chat":
-                prompt = chat_mod.render_chat(p.messages or [])
+                prompt = chat_mod.render_chat(p.messages or [],
+                                              think=p.think)
                 opts = dict(p.options)
                 opts["stop"] = list(opts.get("stop") or []) + \
                     list(chat_mod.CHAT_STOPS)

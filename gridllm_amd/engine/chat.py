@@ -11,7 +11,7 @@ from __future__ import annotations
 from typing import Iterable
 
 
-def render_chat(messages: Iterable[dict]) -> str:
+def render_chat(messages: Iterable[dict], think: bool = False) -> str:
     parts = ["<|begin_of_text|>"]
     for m in messages:
         role = m.get("role", "user")
@@ -21,7 +21,58 @@ def render_chat(messages: Iterable[dict]) -> str:
                                 if isinstance(p, dict) and p.get("type") == "text")
         parts.append(f"<|start_header_id|>{role}<|end_header_id|>\n\n{content}<|eot_id|>")
     parts.append("<|start_header_id|>assistant<|end_header_id|>\n\n")
+    if think:
+        # thinking mode (Ollama `think: true`, reference forwards it at
+        # server/src/routes/ollama.ts:216-239): generation starts inside a
+        # think block; output before THINK_CLOSE is the `thinking` field
+        parts.append(THINK_OPEN + "\n")
     return "".join(parts)
 
 
 CHAT_STOPS = ["<|eot_id|>", "<|start_header_id|>"]
+THINK_OPEN = "<think>"
+THINK_CLOSE = "</think>"
+
+
+class ThinkSplitter:
+    """Incremental splitter for thinking-mode output: text before the
+    first THINK_CLOSE is `thinking`, the rest is `content`. Streaming-safe
+    (holds back a partial close tag across piece boundaries)."""
+
+    def __init__(self) -> None:
+        self.in_think = True
+        self._tail = ""
+
+    def push(self, piece: str) -> tuple[str, str]:
+        """Returns (thinking_delta, content_delta) ready to emit."""
+        if not self.in_think:
+            return "", piece
+        buf = self._tail + piece
+        idx = buf.find(THINK_CLOSE)
+        if idx >= 0:
+            self.in_think = False
+            self._tail = ""
+            content = buf[idx + len(THINK_CLOSE):].lstrip("\n")
+            return buf[:idx], content
+        # hold back the longest buf suffix that could begin THINK_CLOSE
+        keep = 0
+        for k in range(min(len(buf), len(THINK_CLOSE) - 1), 0, -1):
+            if THINK_CLOSE.startswith(buf[-k:]):
+                keep = k
+                break
+        self._tail = buf[len(buf) - keep:] if keep else ""
+        return (buf[:len(buf) - keep] if keep else buf), ""
+
+    def flush(self) -> tuple[str, str]:
+        t, self._tail = self._tail, ""
+        if self.in_think:
+            return t, ""
+        return "", t
+
+
+def split_think(text: str) -> tuple[str, str]:
+    """One-shot split of a completed thinking-mode generation."""
+    idx = text.find(THINK_CLOSE)
+    if idx < 0:
+        return text, ""
+    return text[:idx], text[idx + len(THINK_CLOSE):].lstrip("\n")

@@ -110,7 +110,7 @@ def _stream_ndjson(scheduler: JobScheduler, req: InferenceRequest,
             while True:
                 kind, item = await q.get()
                 if kind == "chunk":
-                    if item.get("response"):
+                    if item.get("response") or item.get("thinking"):
                         yield json.dumps(to_chunk(item)) + "\n"
                     if item.get("done"):
                         continue  # final payload carries the done record
@@ -172,8 +172,11 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
             req.prompt = f"{body['system']}\n\n{req.prompt}"
 
         def chunk_json(c):
-            return {"model": model, "created_at": _now_iso(),
-                    "response": c.get("response", ""), "done": False}
+            out = {"model": model, "created_at": _now_iso(),
+                   "response": c.get("response", ""), "done": False}
+            if c.get("thinking"):
+                out["thinking"] = c["thinking"]
+            return out
 
         def final_json(res):
             return {"model": model, "created_at": _now_iso(), "response": "",
@@ -186,12 +189,15 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         res = await scheduler.submit_and_wait(req)
         if res.get("error"):
             raise HTTPException(status_code=500, detail=res["error"])
-        return JSONResponse({
+        out = {
             "model": model, "created_at": _now_iso(),
             "response": res.get("response", ""), "done": True,
             "done_reason": res.get("done_reason", "stop"),
             "context": res.get("context") or [],
-            **_timings(res)})
+            **_timings(res)}
+        if res.get("thinking") is not None:
+            out["thinking"] = res["thinking"]
+        return JSONResponse(out)
 
     # ----------------------------------------------------------------- chat
 
@@ -216,16 +222,21 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
             messages = [tools_mod.system_message(fns)] + list(messages)
         req = InferenceRequest(
             model=model, messages=messages, request_type="chat",
-            stream=stream, timeout_ms=300_000, options=opts)
+            stream=stream, timeout_ms=300_000, options=opts,
+            metadata={"think": True} if body.get("think") else None)
 
         def chunk_json(c):
+            msg = {"role": "assistant", "content": c.get("response", "")}
+            if c.get("thinking"):
+                msg["thinking"] = c["thinking"]
             return {"model": model, "created_at": _now_iso(),
-                    "message": {"role": "assistant",
-                                "content": c.get("response", "")},
+                    "message": msg,
                     "done": False}
 
         def final_json(res):
             msg = {"role": "assistant", "content": ""}
+            if res.get("thinking") is not None:
+                msg["thinking"] = ""
             if fns:
                 # Ollama emits tool_calls on the final chunk of a stream
                 calls = tools_mod.parse_calls(res.get("response", "") or "")
