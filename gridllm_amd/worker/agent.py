@@ -88,6 +88,9 @@ class WorkerAgent:
         self._tasks: list[asyncio.Task] = []
         self._job_tasks: set[asyncio.Task] = set()
         self._last_status: tuple = ()
+        self.stream_drops = 0          # chunk publishes swallowed (at-most-
+                                       # once streaming; result path owns
+                                       # delivery) — surfaced in telemetry
         self._started = False
         self.paused = False
         self.draining = False
@@ -141,7 +144,7 @@ class WorkerAgent:
             max_concurrent_jobs=self.cfg.max_concurrent_tasks,
             current_jobs=self.current_jobs,
             performance_tier=telemetry.performance_tier(stats),
-            gpu=stats.to_dict(),
+            gpu={**stats.to_dict(), "stream_drops": self.stream_drops},
             tp_degree=self.engine.tp.size,
         )
 
@@ -454,8 +457,10 @@ class WorkerAgent:
                         "timestamp": now_ms(),
                     })
                 except Exception:  # noqa: BLE001 — bus hiccup: chunks are
-                    pass           # fire-and-forget; the job result path
-                                   # (and orphan recovery) owns delivery
+                    self.stream_drops += 1   # fire-and-forget; the result
+                                   # path (and orphan recovery) owns
+                                   # delivery; drops are COUNTED so a
+                                   # lossy stream is observable (/metrics)
                 if done:
                     self._streams.pop(s.job_id, None)
                     s.fin.set()
