@@ -124,8 +124,8 @@ class BatchPlan:
     function of (seed, draw index), recomputed vectorized each step."""
 
     __slots__ = ("states", "temps", "greedy_mask", "all_greedy",
-                 "need_filter", "kidx", "mps", "tps", "seeds", "greedy_np",
-                 "any_penalty")
+                 "need_filter", "kidx", "mps", "tps", "any_penalty",
+                 "seeds_dev", "draws_dev", "nongreedy_dev", "_draws0")
 
     def __init__(self):
         self.states: list = []
@@ -139,46 +139,61 @@ class BatchPlan:
             self.any_penalty = any(s.counts for s in states)
             return
         self.states = list(states)
-        self.temps = torch.tensor(
-            [max(s.opts.temperature if s.opts.temperature is not None
-                 else 0.8, 1e-4) for s in states],
-            device=dev).unsqueeze(1)
-        g = [s.greedy for s in states]
-        self.greedy_np = np.array(g, dtype=bool)
-        self.all_greedy = all(g)
-        self.greedy_mask = torch.tensor(
-            [0.0 if x else 1.0 for x in g], device=dev).unsqueeze(1)
-        self.seeds = np.array([s.noise_seed for s in states],
-                              dtype=np.uint64)
+        b = len(states)
+        # ONE packed H2D per composition change instead of ~8 per-field
+        # tensor() calls; the per-step key math runs fully ON DEVICE
+        # (draws_dev += 1 after each draw) — the per-step host->device key
+        # upload measured 1.7 ms/step of GPU-sync stalls at c192 (round-2
+        # GPU profile, lt_worker_batcher.prof)
+        host = np.zeros((6, b), dtype=np.float64)
+        g = np.zeros(b, dtype=bool)
+        seeds = np.zeros(b, dtype=np.int64)
+        draws = np.zeros(b, dtype=np.int64)
+        for i, s in enumerate(states):
+            o = s.opts
+            host[0, i] = max(o.temperature if o.temperature is not None
+                             else 0.8, 1e-4)
+            g[i] = s.greedy
+            host[1, i] = 0.0 if s.greedy else 1.0
+            host[2, i] = (int(o.top_k) - 1
+                          if o.top_k and 0 < o.top_k < v else v - 1)
+            host[3, i] = (o.min_p if o.min_p and 0 < o.min_p < 1 else 0.0)
+            host[4, i] = (o.top_p if o.top_p and 0 < o.top_p < 1 else 1.0)
+            # mask to 63 bits: the device-side key math runs in int64
+            seeds[i] = s.noise_seed & ((1 << 63) - 1)
+            draws[i] = s.draws
+        self.all_greedy = bool(g.all())
+        packed = torch.from_numpy(host).to(dev, non_blocking=False)
+        self.temps = packed[0].float().unsqueeze(1)
+        self.greedy_mask = packed[1].float().unsqueeze(1)
         self.need_filter = any(
             (s.opts.top_k and 0 < s.opts.top_k < v)
             or (s.opts.top_p and 0.0 < s.opts.top_p < 1.0)
             or (s.opts.min_p and 0.0 < s.opts.min_p < 1.0)
             for s in states)
         if self.need_filter:
-            self.kidx = torch.tensor(
-                [int(s.opts.top_k) - 1
-                 if s.opts.top_k and 0 < s.opts.top_k < v else v - 1
-                 for s in states], device=dev).unsqueeze(1)
-            self.mps = torch.tensor(
-                [s.opts.min_p if s.opts.min_p and 0 < s.opts.min_p < 1
-                 else 0.0 for s in states], device=dev).unsqueeze(1)
-            self.tps = torch.tensor(
-                [s.opts.top_p if s.opts.top_p and 0 < s.opts.top_p < 1
-                 else 1.0 for s in states], device=dev).unsqueeze(1)
+            self.kidx = packed[2].long().unsqueeze(1)
+            self.mps = packed[3].float().unsqueeze(1)
+            self.tps = packed[4].float().unsqueeze(1)
+        ints = torch.from_numpy(np.stack([seeds, draws])).to(dev)
+        self.seeds_dev = ints[0]
+        self.draws_dev = ints[1]
+        self._draws0 = draws.copy()          # host mirror for validation
+        self.nongreedy_dev = packed[1].to(torch.int64)
         self.any_penalty = any(s.counts for s in states)
 
     def keys(self, states: list["SamplerState"], dev) -> torch.Tensor:
-        """Per-row splitmix keys for this step's draw — vectorized numpy
-        uint64 (wraps mod 2^64 exactly like the masked python-int math)."""
-        import numpy as np
-        draws = np.fromiter((s.draws for s in states), dtype=np.uint64,
-                            count=len(states))
-        k = (self.seeds * np.uint64(6364136223846793005) +
-             (draws + np.uint64(1)) * np.uint64(1442695040888963407)) \
-            & np.uint64((1 << 63) - 1)
-        k[self.greedy_np] = 0
-        return torch.from_numpy(k.astype(np.int64)).to(dev).unsqueeze(1)
+        """Per-row splitmix keys for this step's draw — computed ON DEVICE
+        from the cached (seed, draw-counter) tensors; int64 wrap-around
+        matches the masked python-int math bit-exactly (unit-tested).
+        Advances the device draw counter; SamplerState.observe advances
+        the host mirror so a composition-change rebuild stays in sync."""
+        k = (self.seeds_dev * 6364136223846793005 +
+             (self.draws_dev + 1) * 1442695040888963407) \
+            & ((1 << 63) - 1)
+        k = k * self.nongreedy_dev           # greedy rows draw key 0
+        self.draws_dev += 1
+        return k.unsqueeze(1)
 
 
 _ARANGE_CACHE: dict = {}
@@ -246,11 +261,12 @@ def sample_batch_vectorized(logits: torch.Tensor,
             vals = vals - pp_t - fp_t * cc
             logits[ii, tt] = vals
     if (logits.is_cuda and not plan.need_filter
-            and os.environ.get("GRIDLLM_FUSED_SAMPLER") == "1"):
-        # EXPERIMENTAL (round-2 validation pending): one HIP launch
+            and os.environ.get("GRIDLLM_FUSED_SAMPLER", "1") != "0"):
+        # DEFAULT ON (hardware-validated round 2: test green on MI355X,
+        # serving +6% — gpurun_out/r02_lt_c192_fused.log): one HIP launch
         # replacing the ~15-op eager chain below; same RNG contract
         # (csrc k_sample_gumbel docstring). Greedy rows ride along with
-        # key==0. Off by default until measured on hardware.
+        # key==0. GRIDLLM_FUSED_SAMPLER=0 restores the eager chain.
         from gridllm_amd.ops import hip_ops
         return hip_ops.sample_gumbel(
             logits.contiguous(),

@@ -54,6 +54,10 @@ class JobScheduler:
             "job_completed": [], "job_failed": [], "job_orphaned": []}
         self.stats = {"completed": 0, "failed": 0, "orphaned": 0,
                       "timeouts": 0, "retries": 0}
+        # direct token-stream plane (bus/stream.py): when set, streaming
+        # jobs carry this server's address so workers send chunks straight
+        # here instead of double-hopping through the bus
+        self.stream_server = None
 
     def on(self, event: str, cb: Callable) -> None:
         self._listeners[event].append(cb)
@@ -70,12 +74,42 @@ class JobScheduler:
         await self.bus.subscribe(Channels.JOB_COMPLETED, self._h_completed)
         await self.bus.subscribe(Channels.JOB_FAILED, self._h_failed)
         await self.bus.subscribe(Channels.JOB_TIMEOUT, self._h_timeout)
+        # remote-submit plane: detached gateway replicas enqueue/cancel via
+        # bus messages instead of in-process calls (server/remote.py); the
+        # per-request result/stream channels work unchanged for them
+        await self.bus.subscribe(Channels.JOB_SUBMIT, self._h_submit)
+        await self.bus.subscribe(Channels.JOB_CANCEL, self._h_cancel)
         self.registry.on("worker_removed", self._on_worker_removed)
         self.registry.on("worker_registered", lambda _info: self.kick())
         await self._load_existing()
         loop = asyncio.get_running_loop()
         self._tasks.append(loop.create_task(self._dispatch_loop()))
         self._tasks.append(loop.create_task(self._sweep_loop()))
+        self._tasks.append(loop.create_task(self._stats_loop()))
+
+    async def _h_submit(self, _ch: str, payload: dict) -> None:
+        job = payload.get("job")
+        if job:
+            await self.add_job(InferenceRequest.from_dict(job))
+
+    async def _h_cancel(self, _ch: str, payload: dict) -> None:
+        jid = payload.get("jobId")
+        if jid:
+            await self.cancel_job(jid)
+
+    async def _stats_loop(self) -> None:
+        """Publish a cluster-stats snapshot for detached gateways (they
+        serve /health, /inference/queue and metrics gauges from it)."""
+        while True:
+            try:
+                await self.bus.kv_set(Keys.CLUSTER_STATS, json.dumps({
+                    "queue": self.queue_stats(),
+                    "workers": self.registry.counts(),
+                    "ts": now_ms(),
+                }))
+            except Exception:  # noqa: BLE001
+                pass
+            await asyncio.sleep(1.0)
 
     async def shutdown(self) -> None:
         for t in self._tasks + list(self._timeout_tasks.values()) + \

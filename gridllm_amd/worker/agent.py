@@ -42,10 +42,10 @@ class _Stream:
     """One streaming job's token buffer, drained by the agent's single
     shared flusher task."""
     __slots__ = ("job_id", "buf", "lk", "done_ev", "sent_any", "fin",
-                 "splitter")
+                 "splitter", "addr")
 
     def __init__(self, job_id: str, buf: list, lk: threading.Lock,
-                 done_ev: threading.Event, splitter=None):
+                 done_ev: threading.Event, splitter=None, addr=None):
         self.job_id = job_id
         self.buf = buf
         self.lk = lk
@@ -53,6 +53,7 @@ class _Stream:
         self.sent_any = False
         self.fin = asyncio.Event()     # set after the final chunk publishes
         self.splitter = splitter       # think mode: thinking/content split
+        self.addr = addr               # direct gateway addr (bus/stream.py)
 
 
 class WorkerAgent:
@@ -83,6 +84,8 @@ class WorkerAgent:
         self.current_jobs = 0
         self._cancel_events: dict[str, threading.Event] = {}
         self._streams: dict[str, "_Stream"] = {}
+        from gridllm_amd.bus.stream import StreamSink
+        self._sink = StreamSink()      # direct worker->gateway chunk plane
         self._flush_task: Optional[asyncio.Task] = None
         self._flush_wake = asyncio.Event()
         self._tasks: list[asyncio.Task] = []
@@ -128,6 +131,7 @@ class WorkerAgent:
         self.batcher.shutdown()
         for b in self._extra_batchers:
             b.shutdown()
+        await self._sink.close()
 
     def _info(self) -> WorkerInfo:
         stats = telemetry.gpu_stats(self.cfg.device)
@@ -294,7 +298,8 @@ class WorkerAgent:
             from gridllm_amd.engine import chat as chat_mod
             stream = self._register_stream(
                 req.id, buf, lk, done_ev,
-                splitter=chat_mod.ThinkSplitter() if think else None)
+                splitter=chat_mod.ThinkSplitter() if think else None,
+                addr=(req.metadata or {}).get("streamAddr"))
         try:
             pair = self._by_model.get(req.model)
             if pair is None:
@@ -402,8 +407,9 @@ class WorkerAgent:
                 pass
 
     def _register_stream(self, job_id: str, buf: list, lk: threading.Lock,
-                         done_ev: threading.Event, splitter=None) -> _Stream:
-        s = _Stream(job_id, buf, lk, done_ev, splitter=splitter)
+                         done_ev: threading.Event, splitter=None,
+                         addr=None) -> _Stream:
+        s = _Stream(job_id, buf, lk, done_ev, splitter=splitter, addr=addr)
         self._streams[job_id] = s
         self._flush_wake.set()         # break out of a long flush sleep
         if self._flush_task is None or self._flush_task.done():
@@ -430,6 +436,12 @@ class WorkerAgent:
                                        0.004 if fast else flush_s)
             except asyncio.TimeoutError:
                 pass
+            ts = now_ms()
+            # direct-path chunks batch into ONE frame per gateway per
+            # cycle (bus/stream.py); addr-less streams fall back to the
+            # bus, one publish each (reference wire behavior)
+            batches: dict[str, list] = {}
+            finished: list[_Stream] = []
             for s in list(self._streams.values()):
                 with s.lk:
                     pieces = s.buf[:]
@@ -450,18 +462,28 @@ class WorkerAgent:
                     chunk["response"] = co
                     if th:
                         chunk["thinking"] = th
-                try:
-                    await self.bus.publish(Channels.job_stream(s.job_id), {
-                        "jobId": s.job_id, "workerId": self.worker_id,
-                        "chunk": chunk,
-                        "timestamp": now_ms(),
-                    })
-                except Exception:  # noqa: BLE001 — bus hiccup: chunks are
-                    self.stream_drops += 1   # fire-and-forget; the result
-                                   # path (and orphan recovery) owns
-                                   # delivery; drops are COUNTED so a
-                                   # lossy stream is observable (/metrics)
+                payload = {"jobId": s.job_id, "workerId": self.worker_id,
+                           "chunk": chunk, "timestamp": ts}
+                if s.addr:
+                    batches.setdefault(s.addr, []).append(
+                        (Channels.job_stream(s.job_id), payload))
+                else:
+                    try:
+                        await self.bus.publish(
+                            Channels.job_stream(s.job_id), payload)
+                    except Exception:  # noqa: BLE001 — bus hiccup: chunks
+                        self.stream_drops += 1   # are fire-and-forget; the
+                                       # result path (and orphan recovery)
+                                       # owns delivery; drops are COUNTED
+                                       # (/metrics)
                 if done:
-                    self._streams.pop(s.job_id, None)
-                    s.fin.set()
+                    finished.append(s)
+            for addr, items in batches.items():
+                try:
+                    await self._sink.send(addr, items)
+                except Exception:  # noqa: BLE001 — dead gateway: at-most-
+                    self.stream_drops += len(items)   # once, counted
+            for s in finished:
+                self._streams.pop(s.job_id, None)
+                s.fin.set()
         self._flush_task = None
