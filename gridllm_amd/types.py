@@ -257,6 +257,11 @@ class Channels:
     JOB_COMPLETED = "job:completed"
     JOB_FAILED = "job:failed"
     JOB_TIMEOUT = "job:timeout"
+    # remote-submit plane (no reference analog: the reference's routes
+    # called the scheduler in-process; detached gateway replicas submit
+    # over the bus — server/remote.py)
+    JOB_SUBMIT = "job:submit"
+    JOB_CANCEL = "job:cancel"
 
     @staticmethod
     def worker_job(worker_id: str) -> str:
@@ -285,6 +290,8 @@ class Keys:
     WORKERS = "workers"          # hash worker_id -> WorkerInfo json
     ACTIVE_JOBS = "active_jobs"  # hash job_id -> JobAssignment json
     JOB_QUEUE = "job_queue"      # json array of queued InferenceRequest
+    CLUSTER_STATS = "cluster:stats"  # scheduler stats snapshot (1 s) for
+                                     # detached gateways
 
     @staticmethod
     def heartbeat(worker_id: str) -> str:
