@@ -1039,12 +1039,16 @@ k_gemm_combine(bf16_t* __restrict__ out, const float* __restrict__ ws,
 
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
+// Shared body: one 128-q-row tile of one q-head for one sequence.
+// q0 = first q row of this tile; vt/plds = caller's LDS buffers.
 template <int D>
-__global__ void __launch_bounds__(256, 2)
-k_attn_prefill(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
-               const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
-               int T, int Hq, int G, int Hk, int Smax, int slot,
-               int s_total, int causal_off, float scale) {
+__device__ __forceinline__ void
+attn_prefill_tile(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
+                  const bf16_t* __restrict__ kc,
+                  const bf16_t* __restrict__ vc,
+                  int T, int Hq, int G, int Hk, int Smax, int slot,
+                  int s_total, int causal_off, float scale, int q0,
+                  bf16_t* vt, bf16_t* plds_all) {
     constexpr int NS = D / 16;     // 16-wide d slices (QK^T fragments)
     constexpr int NT = D / 32;     // 32-wide d tiles (output / V^T)
     constexpr int VST = 40;        // padded k-stride of V^T tile (16B-mult)
@@ -1055,11 +1059,7 @@ k_attn_prefill(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     const int ln = lane & 31;
     const int hq = blockIdx.y;
     const int hk = hq / G;
-    const int q0 = blockIdx.x * 128;
     const int qw = q0 + wid * 32;
-
-    __shared__ __attribute__((aligned(16))) bf16_t vt[D * VST];
-    __shared__ __attribute__((aligned(16))) bf16_t plds[4][2 * 64 * 8];
 
     const bf16_t* kbase = kc + ((long)slot * Hk + hk) * (long)Smax * D;
     const bf16_t* vbase = vc + ((long)slot * Hk + hk) * (long)Smax * D;
@@ -1142,7 +1142,7 @@ k_attn_prefill(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         // (kpos, q=ln) lands where slice sidx = kpos>>4, lane
         // lr = ln + 32*((kpos>>3)&1), word i = kpos&7 will read it; regs
         // pair (r, r+1) = (kpos, kpos+1) -> one packed b32 write.
-        bf16_t* ps = plds[wid];
+        bf16_t* ps = plds_all + wid * (2 * 64 * 8);
 #pragma unroll
         for (int r = 0; r < 16; r += 2) {
             const int kpos = (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -1192,6 +1192,46 @@ k_attn_prefill(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
             }
         }
     }
+}
+
+template <int D>
+__global__ void __launch_bounds__(256, 2)
+k_attn_prefill(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
+               const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
+               int T, int Hq, int G, int Hk, int Smax, int slot,
+               int s_total, int causal_off, float scale) {
+    constexpr int VST = 40;
+    __shared__ __attribute__((aligned(16))) bf16_t vt[D * VST];
+    __shared__ __attribute__((aligned(16))) bf16_t plds[4][2 * 64 * 8];
+    attn_prefill_tile<D>(out, q, kc, vc, T, Hq, G, Hk, Smax, slot, s_total,
+                         causal_off, scale, blockIdx.x * 128, vt,
+                         &plds[0][0]);
+}
+
+// ===========================================================================
+// Varlen BATCHED prefill: one launch covers MANY fresh sequences.
+//   meta[tile] = {q_off, len, slot, q0}: tile `blockIdx.x` works on rows
+//   [q0, q0+128) of the sequence packed at q_off in q/out, against KV slot
+//   `slot` with s_total = len, causal_off = 0.
+// Motivation (round-2 serving profile, lt_worker_batcher.prof): admission
+// batches ran attention per-sequence per-layer from Python — 52k host
+// launches, 36 ms per 32-prompt admission. One meta-driven launch per
+// layer removes the Python loop and fills the chip (ntiles * Hq blocks).
+// ===========================================================================
+template <int D>
+__global__ void __launch_bounds__(256, 2)
+k_attn_prefill_batch(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
+                     const bf16_t* __restrict__ kc,
+                     const bf16_t* __restrict__ vc,
+                     const int4* __restrict__ meta, int Hq, int G, int Hk,
+                     int Smax, float scale) {
+    constexpr int VST = 40;
+    __shared__ __attribute__((aligned(16))) bf16_t vt[D * VST];
+    __shared__ __attribute__((aligned(16))) bf16_t plds[4][2 * 64 * 8];
+    const int4 m = meta[blockIdx.x];
+    attn_prefill_tile<D>(out + (long)m.x * Hq * D, q + (long)m.x * Hq * D,
+                         kc, vc, m.y, Hq, G, Hk, Smax, m.z, m.y,
+                         /*causal_off=*/0, scale, m.w, vt, &plds[0][0]);
 }
 
 template <int MT>
@@ -1523,6 +1563,28 @@ int gl_attn_prefill(void* stream, void* out, const void* q, const void* kc,
             (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)kc,
             (const bf16_t*)vc, T, Hq, G, Hk, Smax, slot, s_total,
             causal_off, scale);
+    else
+        return -111;
+    LAUNCH_CHECK();
+    return 0;
+}
+
+int gl_attn_prefill_batch(void* stream, void* out, const void* q,
+                          const void* kc, const void* vc, const void* meta,
+                          int ntiles, int Hq, int Hk, int D, int Smax,
+                          float scale) {
+    hipStream_t st = (hipStream_t)stream;
+    if (Hq % Hk) return -110;
+    const int G = Hq / Hk;
+    dim3 grid((unsigned)ntiles, (unsigned)Hq), blk(256);
+    if (D == 128)
+        k_attn_prefill_batch<128><<<grid, blk, 0, st>>>(
+            (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)kc,
+            (const bf16_t*)vc, (const int4*)meta, Hq, G, Hk, Smax, scale);
+    else if (D == 64)
+        k_attn_prefill_batch<64><<<grid, blk, 0, st>>>(
+            (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)kc,
+            (const bf16_t*)vc, (const int4*)meta, Hq, G, Hk, Smax, scale);
     else
         return -111;
     LAUNCH_CHECK();

@@ -351,18 +351,32 @@ class LlamaModel:
         logits [k, vocab] f32; with return_hidden, (logits, mean-pooled
         final hidden [k, H] f32) — the embedding path."""
         s = self.spec
+        import numpy as np
         lens = [int(t.shape[0]) for t in token_lists]
         bounds = [0]
         for ln in lens:
             bounds.append(bounds[-1] + ln)
         tokens = torch.cat(token_lists)
-        pos = torch.cat([torch.arange(ln, dtype=torch.int32,
-                                      device=self.device) for ln in lens])
-        slot_t = torch.cat([
-            torch.full((ln,), sl * self.cache.max_seq, dtype=torch.int32,
-                       device=self.device)
-            + torch.arange(ln, dtype=torch.int32, device=self.device)
-            for ln, sl in zip(lens, slots)])
+        # position/slot indices built host-side: TWO H2D copies instead of
+        # ~3k tiny device aranges per admission batch
+        pos_np = np.concatenate(
+            [np.arange(ln, dtype=np.int32) for ln in lens])
+        slot_np = np.concatenate(
+            [sl * self.cache.max_seq + np.arange(ln, dtype=np.int32)
+             for ln, sl in zip(lens, slots)])
+        idx = torch.from_numpy(np.stack([pos_np, slot_np])).to(self.device)
+        pos, slot_t = idx[0], idx[1]
+        # varlen batched attention (one meta-driven kernel launch per layer
+        # instead of a per-sequence Python loop — the round-2 profile's
+        # 36 ms-per-admission hot spot); meta rows: {q_off, len, slot, q0}
+        meta = None
+        if (tokens.is_cuda and self.dtype == torch.bfloat16
+                and s.head_dim in (64, 128)):
+            rows = [(bounds[i], ln, sl, q0)
+                    for i, (ln, sl) in enumerate(zip(lens, slots))
+                    for q0 in range(0, ln, 128)]
+            meta = torch.from_numpy(
+                np.asarray(rows, dtype=np.int32)).to(self.device)
         x = self.embed[tokens]
         resid = x
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
@@ -371,11 +385,15 @@ class LlamaModel:
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
                             self.cos, self.sin, pos, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)
-            attn = torch.empty_like(q)
-            for i, sl in enumerate(slots):
-                attn[bounds[i]:bounds[i + 1]] = ops.attn_prefill_cache(
-                    q[bounds[i]:bounds[i + 1]], self.cache.k[li],
-                    self.cache.v[li], sl, lens[i], 0, self.scale)
+            if meta is not None:
+                attn = ops.attn_prefill_cache_batch(
+                    q, self.cache.k[li], self.cache.v[li], meta, self.scale)
+            else:
+                attn = torch.empty_like(q)
+                for i, sl in enumerate(slots):
+                    attn[bounds[i]:bounds[i + 1]] = ops.attn_prefill_cache(
+                        q[bounds[i]:bounds[i + 1]], self.cache.k[li],
+                        self.cache.v[li], sl, lens[i], 0, self.scale)
             attn_out = _linear(attn.reshape(bounds[-1], -1), lw.wo)
             self.tp.all_reduce(attn_out)
             normed, resid = ops.rmsnorm(attn_out, lw.mlp_norm, s.norm_eps,

@@ -113,6 +113,16 @@ def attn_prefill_cache(q: torch.Tensor, kcache: torch.Tensor,
                                   causal_offset)[0]
 
 
+def attn_prefill_cache_batch(q: torch.Tensor, kcache: torch.Tensor,
+                             vcache: torch.Tensor, meta: torch.Tensor,
+                             scale: float) -> torch.Tensor:
+    """Varlen batched GQA prefill (GPU bf16 only — callers gate on the
+    same condition LlamaModel.forward_prefill_batch uses; the CPU path
+    keeps the per-sequence loop)."""
+    from gridllm_amd.ops import hip_ops
+    return hip_ops.attn_prefill_cache_batch(q, kcache, vcache, meta, scale)
+
+
 def _attn_prefill_chunked(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                           scale: float, causal_offset: int = 0,
                           chunk: int = 512) -> torch.Tensor:

@@ -57,6 +57,9 @@ def _load() -> ctypes.CDLL:
     lib.gl_attn_prefill.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_i, c_i,
                                     c_i, c_i, c_i, c_i, c_i, c_i, c_f]
     lib.gl_attn_prefill.restype = c_i
+    lib.gl_attn_prefill_batch.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
+                                          c_i, c_i, c_i, c_i, c_i, c_f]
+    lib.gl_attn_prefill_batch.restype = c_i
     lib.gl_swiglu_il.argtypes = [c_vp, c_vp, c_vp, c_l, c_i]
     lib.gl_sample_gumbel.argtypes = [c_vp, c_vp, c_vp, c_vp, c_vp, c_i, c_i]
     lib.gl_swiglu_il.restype = c_i
@@ -215,6 +218,33 @@ def attn_prefill_cache(q: torch.Tensor, kcache: torch.Tensor,
                                _ptr(vcache), t, hq, hk, d, smax, slot,
                                s_total, causal_offset,
                                ctypes.c_float(scale)), "attn_prefill")
+    return out
+
+
+def attn_prefill_cache_batch(q: torch.Tensor, kcache: torch.Tensor,
+                             vcache: torch.Tensor, meta: torch.Tensor,
+                             scale: float,
+                             out: Optional[torch.Tensor] = None
+                             ) -> torch.Tensor:
+    """Varlen batched MFMA GQA prefill: many fresh sequences packed in q
+    ([sum(T_i), Hq, D]), one launch per layer. meta: [ntiles, 4] int32 on
+    device, rows {q_off, len, slot, q0} (built once per admission batch by
+    LlamaModel.forward_prefill_batch and reused across layers)."""
+    lib = _load()
+    hq, d = q.shape[1], q.shape[2]
+    hk, smax = kcache.shape[1], kcache.shape[2]
+    ntiles = meta.shape[0]
+    if out is None:
+        out = torch.empty_like(q)
+    if _VALIDATE:
+        _check_range("meta.slot", meta[:, 2], kcache.shape[0])
+        _check_range("meta.len", meta[:, 1] - 1, smax)
+        _check_range("meta.q_off", meta[:, 0], q.shape[0])
+    _check(lib.gl_attn_prefill_batch(_stream(), _ptr(out), _ptr(q),
+                                     _ptr(kcache), _ptr(vcache), _ptr(meta),
+                                     ntiles, hq, hk, d, smax,
+                                     ctypes.c_float(scale)),
+           "attn_prefill_batch")
     return out
 
 
