@@ -82,6 +82,50 @@ class BusServer:
                 pass
         self.conns.clear()
 
+    # -- durability (standalone bus: python -m gridllm_amd.bus) ------------
+    # The reference's control-plane state survives server death because it
+    # lives in an external Redis with AOF (docker-compose.yml:22; reloaded
+    # at JobScheduler.ts:82-126). The standalone bus process provides the
+    # same guarantee: periodic atomic snapshots of kv+hashes, restored on
+    # boot. TTL keys keep their REMAINING ttl across restart.
+
+    def snapshot_state(self) -> dict:
+        now = time.monotonic()
+        kv = {}
+        for k, (v, exp) in self.kv.items():
+            if exp and exp < now:
+                continue
+            kv[k] = [v, int((exp - now) * 1000) if exp else 0]
+        return {"kv": kv, "hashes": {k: dict(h)
+                                     for k, h in self.hashes.items()}}
+
+    def restore_state(self, state: dict) -> None:
+        now = time.monotonic()
+        for k, (v, ttl_ms) in (state.get("kv") or {}).items():
+            self.kv[k] = (v, now + ttl_ms / 1000.0 if ttl_ms else 0.0)
+        for k, h in (state.get("hashes") or {}).items():
+            self.hashes[k].update(h)
+
+    def save_snapshot(self, path: str) -> None:
+        import os
+        import msgpack as _mp
+        blob = _mp.packb(self.snapshot_state(), use_bin_type=True)
+        tmp = path + ".tmp"
+        with open(tmp, "wb") as f:
+            f.write(blob)
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(tmp, path)          # atomic: crash never truncates
+
+    def load_snapshot(self, path: str) -> bool:
+        import os
+        import msgpack as _mp
+        if not os.path.exists(path):
+            return False
+        with open(path, "rb") as f:
+            self.restore_state(_mp.unpackb(f.read(), raw=False))
+        return True
+
     async def _on_conn(self, reader: asyncio.StreamReader,
                        writer: asyncio.StreamWriter) -> None:
         conn = _Conn(self, reader, writer)

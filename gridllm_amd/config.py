@@ -62,6 +62,9 @@ class ServerConfig:
     log_level: str = "info"
     rate_limit_max_requests: int = 0            # 0 = unlimited (ref: built
     rate_limit_window_ms: int = 900_000         # but unmounted; we mount it)
+    bus_external: bool = False                  # connect to a standalone bus
+                                                # (python -m gridllm_amd.bus)
+                                                # instead of embedding one
 
     @classmethod
     def from_env(cls) -> "ServerConfig":
@@ -79,6 +82,7 @@ class ServerConfig:
             log_level=_env_str("LOG_LEVEL", "info", ["debug", "info", "warn", "error"]),
             rate_limit_max_requests=_env_int("RATE_LIMIT_MAX_REQUESTS", 0, 0),
             rate_limit_window_ms=_env_int("RATE_LIMIT_WINDOW_MS", 900_000, 1000),
+            bus_external=_env_int("BUS_EXTERNAL", 0, 0, 1) == 1,
         )
 
 

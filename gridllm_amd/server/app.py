@@ -78,7 +78,11 @@ def create_app(scheduler: JobScheduler, registry: WorkerRegistry,
 class GridLLMServer:
     def __init__(self, cfg: Optional[ServerConfig] = None):
         self.cfg = cfg or ServerConfig.from_env()
-        self.bus_server = BusServer(self.cfg.bus_host, self.cfg.bus_port)
+        # external bus (python -m gridllm_amd.bus, durable across server
+        # death — the reference's external-Redis guarantee,
+        # docker-compose.yml:12-30) vs embedded (single-process default)
+        self.bus_server = None if self.cfg.bus_external else \
+            BusServer(self.cfg.bus_host, self.cfg.bus_port)
         self.bus = TcpBus(self.cfg.bus_host, self.cfg.bus_port)
         self.registry = WorkerRegistry(self.bus, self.cfg)
         self.scheduler = JobScheduler(self.bus, self.registry, self.cfg)
@@ -86,22 +90,33 @@ class GridLLMServer:
         self._status_task: Optional[asyncio.Task] = None
 
     async def start(self) -> None:
-        await self.bus_server.start()
+        if self.bus_server is not None:
+            await self.bus_server.start()
         await self.bus.connect()
+        # direct worker->gateway chunk plane for this process's own
+        # streaming requests (bus/stream.py)
+        from gridllm_amd.bus.stream import StreamServer
+        self.stream_server = StreamServer("127.0.0.1", 0)
+        await self.stream_server.start()
+        self.scheduler.stream_server = self.stream_server
         await self.registry.initialize()
         await self.scheduler.initialize()
         self._status_task = asyncio.get_running_loop().create_task(
             self._status_loop())
         log.worker("GridLLM-AMD server started", port=self.cfg.port,
-                   bus_port=self.cfg.bus_port)
+                   bus_port=self.cfg.bus_port,
+                   bus="external" if self.bus_server is None else "embedded")
 
     async def shutdown(self) -> None:
         if self._status_task:
             self._status_task.cancel()
         await self.scheduler.shutdown()
         await self.registry.shutdown()
+        if getattr(self, "stream_server", None) is not None:
+            await self.stream_server.stop()
         await self.bus.close()
-        await self.bus_server.stop()
+        if self.bus_server is not None:
+            await self.bus_server.stop()
         log.worker("server shut down")
 
     async def _status_loop(self) -> None:

@@ -423,7 +423,14 @@ class JobScheduler:
 
         rch, sch = Channels.job_result(req.id), Channels.job_stream(req.id)
         await self.bus.subscribe(rch, _on_result)
-        await self.bus.subscribe(sch, _on_stream)
+        direct = self.stream_server is not None
+        if direct:
+            # direct worker->gateway chunk plane (bus/stream.py): chunks
+            # skip the bus broker entirely
+            req.metadata["streamAddr"] = self.stream_server.addr
+            self.stream_server.subscribe(sch, _on_stream)
+        else:
+            await self.bus.subscribe(sch, _on_stream)
         try:
             await self.add_job(req)
             budget = (req.timeout_ms
@@ -432,7 +439,10 @@ class JobScheduler:
             return await asyncio.wait_for(fut, timeout=budget + 30.0)
         finally:
             await self.bus.unsubscribe(rch, _on_result)
-            await self.bus.unsubscribe(sch, _on_stream)
+            if direct:
+                self.stream_server.unsubscribe(sch, _on_stream)
+            else:
+                await self.bus.unsubscribe(sch, _on_stream)
 
     async def cancel_job(self, job_id: str) -> bool:
         """Cancel queued or active job (reference: :874-908)."""
