@@ -1,81 +1,365 @@
 #!/usr/bin/env python3
-"""Flagship serving benchmark: llama3.2-3B bf16 batched streaming decode.
+"""Flagship serving benchmark — the BASELINE.json headline metric.
 
-Measures the BASELINE.json headline metric — output tokens/sec for the whole
-node — on synthetic prompts with random-init weights (no network for real
-checkpoints). One rank per GPU (data-parallel request sharding, the
-reference's only parallelism — SURVEY.md §2.4); each rank runs the full
-engine decode path: hipGraph-replayed forward (HIP kernels + hipBLASLt
-GEMMs) + batched sampling, exactly as the serving loop does.
+Default mode times the FULL serving path: HTTP /api/generate streaming
+through gateway -> scheduler -> bus -> worker -> engine -> direct stream
+back (the reference contract at server/src/routes/ollama.ts:161-319).
+One rank per GPU hosts a worker (engine pinned to cuda:LOCAL_RANK); rank 0
+additionally spawns the scheduler server, N gateway replicas and N async
+load generators. A "step" is one wave of `--concurrency` streaming
+requests per GPU, each generating `--num-predict` tokens; K steps are
+timed between barriers and the whole-node tokens/s printed as ONE JSON
+line.
 
-Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--batch B]
+`--mode engine` preserves the raw engine-decode benchmark (hipGraph decode
++ batched sampling, no serving glue) for kernel-path comparisons.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--mode M]
 Launched for N>1 via torch.distributed.run (reads RANK/WORLD_SIZE env).
 """
 
 from __future__ import annotations
 
 import argparse
+import asyncio
 import json
 import os
 import statistics
+import subprocess
+import sys
 import time
 
 import torch
 
+REPO = os.path.dirname(os.path.abspath(__file__))
 
-def main() -> None:
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=32)
-    ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--batch", type=int, default=1024, help="sequences per GPU")
-    ap.add_argument("--model", default="llama3.2:3b")
-    ap.add_argument("--prompt-len", type=int, default=128)
-    ap.add_argument("--max-seq", type=int, default=1024)
-    ap.add_argument("--device", default="")
-    args = ap.parse_args()
+SERVER_SRC = r"""
+import asyncio, os
+import uvicorn
+from gridllm_amd.config import ServerConfig
+from gridllm_amd.server.app import GridLLMServer
 
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+cfg = ServerConfig(host="127.0.0.1", port=int(os.environ["HTTP_PORT"]),
+                   bus_port=int(os.environ["BUS_PORT"]),
+                   max_concurrent_jobs_per_worker=4096)
+server = GridLLMServer(cfg)
+app = server.app
+
+@app.on_event("startup")
+async def _s():
+    await server.start()
+    print("SERVER_READY", flush=True)
+
+uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error")
+"""
+
+LOADGEN_SRC = r"""
+import asyncio, json, os, statistics, sys, time
+import aiohttp
+
+BASE = os.environ["LG_BASE"]
+CONC = int(os.environ["LG_CONC"])
+NPRED = int(os.environ["LG_NPREDICT"])
+MODEL = os.environ["LG_MODEL"]
+PROMPT = "benchmark prompt " + "lorem ipsum dolor " * 8
+
+async def wave(sess, seed0):
+    sem = asyncio.Semaphore(CONC)
+    tokens = [0]
+    ttfts = []
+    errors = [0]
+
+    async def one(i):
+        async with sem:
+            t0 = time.monotonic()
+            first = None
+            try:
+                async with sess.post("/api/generate", json={
+                        "model": MODEL,
+                        "prompt": f"req {seed0+i}: " + PROMPT,
+                        "stream": True,
+                        "options": {"temperature": 0.8, "seed": seed0 + i,
+                                    "num_predict": NPRED}},
+                        timeout=aiohttp.ClientTimeout(total=600)) as resp:
+                    async for raw in resp.content:
+                        if not raw.strip():
+                            continue
+                        c = json.loads(raw)
+                        if first is None and c.get("response"):
+                            first = time.monotonic() - t0
+                        if c.get("done"):
+                            tokens[0] += int(c.get("eval_count", 0))
+            except Exception:
+                errors[0] += 1
+                return
+            if first is not None:
+                ttfts.append(first * 1000.0)
+
+    await asyncio.gather(*(one(i) for i in range(CONC)))
+    p50 = statistics.median(ttfts) if ttfts else 0.0
+    return tokens[0], p50, errors[0]
+
+async def main():
+    conn = aiohttp.TCPConnector(limit=CONC + 16)
+    seed = int(os.environ.get("LG_SEED", "0")) * 1_000_000
+    async with aiohttp.ClientSession(base_url=BASE, connector=conn) as sess:
+        print("LOADGEN_READY", flush=True)
+        loop = asyncio.get_running_loop()
+        while True:
+            line = await loop.run_in_executor(None, sys.stdin.readline)
+            if not line or line.strip() == "QUIT":
+                return
+            toks, p50, errs = await wave(sess, seed)
+            seed += CONC
+            print(f"DONE {toks} {p50:.2f} {errs}", flush=True)
+
+asyncio.run(main())
+"""
+
+
+def _spawn(argv, env, ready_tag, timeout=420):
+    proc = subprocess.Popen(argv, env={**os.environ, **env,
+                                       "PYTHONPATH": REPO},
+                            stdin=subprocess.PIPE, stdout=subprocess.PIPE,
+                            stderr=subprocess.STDOUT, text=True, cwd=REPO)
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        line = proc.stdout.readline()
+        if ready_tag in line:
+            return proc
+        if proc.poll() is not None:
+            raise RuntimeError(f"{ready_tag} process died:\n"
+                               f"{proc.stdout.read()[-4000:]}")
+    proc.kill()
+    raise TimeoutError(f"no {ready_tag} within {timeout}s")
+
+
+def _worker_main(args, device, bus_port, worker_id):
+    """Run engine + worker agent in THIS rank process (asyncio loop);
+    returns (engine, loop_thread, ready_event)."""
+    import threading
+
+    from gridllm_amd.bus.tcp import TcpBus
+    from gridllm_amd.config import EngineConfig, WorkerConfig
+    from gridllm_amd.engine import InferenceEngine
+    from gridllm_amd.worker.agent import WorkerAgent
+
+    on_gpu = device.startswith("cuda")
+    ecfg = EngineConfig(
+        model=args.model if on_gpu else "tiny",
+        device=device, dtype="bf16" if on_gpu else "fp32",
+        max_seq_len=args.max_seq,
+        max_batch_size=args.concurrency + 8,
+        use_hipgraph=on_gpu)
+    engine = InferenceEngine(ecfg)
+    wcfg = WorkerConfig(worker_id=worker_id, bus_port=bus_port,
+                        max_concurrent_tasks=args.concurrency + 8,
+                        stream_flush_ms=args.flush_ms)
+    ready = threading.Event()
+
+    def run():
+        async def amain():
+            bus = TcpBus("127.0.0.1", bus_port)
+            for _ in range(120):
+                try:
+                    await bus.connect()
+                    break
+                except OSError:
+                    await asyncio.sleep(0.5)
+            agent = WorkerAgent(bus, engine, wcfg)
+            await agent.start()
+            ready.set()
+            await asyncio.Event().wait()
+
+        asyncio.run(amain())
+
+    t = threading.Thread(target=run, daemon=True, name="bench-worker")
+    t.start()
+    return engine, t, ready
+
+
+def serving_mode(args, rank, world, local_rank, dist, device, on_gpu):
+    import httpx
+
     n_gpus = world if world > 1 else args.gpus
+    master_port = int(os.environ.get("MASTER_PORT", "29500"))
+    bus_port = master_port + 101
+    http_port = master_port + 102
+    gw_port0 = master_port + 110
 
-    on_gpu = torch.cuda.is_available()
-    device = args.device or (f"cuda:{local_rank}" if on_gpu else "cpu")
-    if on_gpu:
-        torch.cuda.set_device(device)
+    procs = []
+    try:
+        if rank == 0:
+            env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port)}
+            procs.append(_spawn([sys.executable, "-u", "-c", SERVER_SRC],
+                                env, "SERVER_READY"))
+        if dist is not None:
+            dist.barrier()          # bus up before workers join
 
-    dist = None
-    if world > 1:
-        import torch.distributed as tdist
-        dist = tdist
-        backend = "nccl" if on_gpu else "gloo"
-        dist.init_process_group(backend=backend)
+        _engine, _t, ready = _worker_main(
+            args, device, bus_port, f"bench-w{rank}")
+        if not ready.wait(timeout=600):
+            raise TimeoutError("worker never started")
 
+        gateways = []
+        loadgens = []
+        if rank == 0:
+            # every worker registered?
+            with httpx.Client(timeout=10) as c:
+                for _ in range(600):
+                    try:
+                        r = c.get(f"http://127.0.0.1:{http_port}"
+                                  "/health/workers")
+                        if r.status_code == 200 and \
+                                r.json()["counts"]["total"] >= world:
+                            break
+                    except Exception:  # noqa: BLE001
+                        pass
+                    time.sleep(0.25)
+                else:
+                    raise TimeoutError("workers never all registered")
+            # one gateway replica + one load generator per GPU
+            for i in range(n_gpus):
+                gp = gw_port0 + i
+                gateways.append(_spawn(
+                    [sys.executable, "-m", "gridllm_amd.gateway",
+                     "--port", str(gp), "--host", "127.0.0.1",
+                     "--bus", f"127.0.0.1:{bus_port}"],
+                    {}, "GATEWAY_READY"))
+                loadgens.append(_spawn(
+                    [sys.executable, "-u", "-c", LOADGEN_SRC],
+                    {"LG_BASE": f"http://127.0.0.1:{gp}",
+                     "LG_CONC": str(args.concurrency),
+                     "LG_NPREDICT": str(args.num_predict),
+                     "LG_MODEL": args.model if on_gpu else "tiny",
+                     "LG_SEED": str(i)},
+                    "LOADGEN_READY"))
+            procs.extend(gateways)
+            procs.extend(loadgens)
+
+        def run_step():
+            """One wave on every load generator; returns (tokens, ttft)."""
+            for lg in loadgens:
+                lg.stdin.write("STEP\n")
+                lg.stdin.flush()
+            toks = 0
+            ttfts = []
+            errs = 0
+            for lg in loadgens:
+                line = lg.stdout.readline()
+                while line and not line.startswith("DONE"):
+                    line = lg.stdout.readline()
+                if not line:
+                    raise RuntimeError("load generator died")
+                _, t, p50, e = line.split()
+                toks += int(t)
+                ttfts.append(float(p50))
+                errs += int(e)
+            if errs:
+                print(f"WARNING: {errs} request errors", file=sys.stderr)
+            return toks, statistics.median(ttfts) if ttfts else 0.0
+
+        def barrier_sync():
+            if on_gpu:
+                torch.cuda.synchronize()
+            if dist is not None:
+                dist.barrier()
+            if on_gpu:
+                torch.cuda.synchronize()
+
+        # warmup (captures hipGraphs, warms gateways/conn pools)
+        for _ in range(args.warmup):
+            if rank == 0:
+                run_step()
+        barrier_sync()
+        t0 = time.monotonic()
+        total_tokens = 0
+        ttft_p50s = []
+        for _ in range(args.steps):
+            if rank == 0:
+                tok, p50 = run_step()
+                total_tokens += tok
+                ttft_p50s.append(p50)
+        barrier_sync()
+        t1 = time.monotonic()
+
+        elapsed = t1 - t0
+        if dist is not None:
+            e = torch.tensor([elapsed], dtype=torch.float64)
+            if on_gpu:
+                e = e.to(device)
+            dist.all_reduce(e, op=dist.ReduceOp.MAX)
+            elapsed = float(e.item())
+
+        if rank == 0:
+            toks_per_sec = total_tokens / elapsed
+            out = {
+                "metric": "output tokens/sec (node), llama3.2 "
+                          "/api/generate streaming",
+                "value": round(toks_per_sec, 1),
+                "unit": "tokens/s",
+                "n_gpus": n_gpus,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(elapsed * 1000.0 / args.steps, 3),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "bf16" if on_gpu else "fp32",
+                "data": "synthetic",
+                "config": {
+                    "model": args.model if on_gpu else "tiny",
+                    "global_batch": args.concurrency * n_gpus,
+                    "seq_len": args.prompt_tokens,
+                    "num_predict": args.num_predict,
+                    "parallelism": f"dp{n_gpus}",
+                    "mode": "serving",
+                    "gateways": n_gpus,
+                    "ttft_p50_ms": round(statistics.median(ttft_p50s), 2)
+                    if ttft_p50s else None,
+                    "tokens_timed": total_tokens,
+                },
+            }
+            print(json.dumps(out))
+            for lg in loadgens:
+                try:
+                    lg.stdin.write("QUIT\n")
+                    lg.stdin.flush()
+                except Exception:  # noqa: BLE001
+                    pass
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.kill()
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except Exception:  # noqa: BLE001
+                pass
+
+
+def engine_mode(args, rank, world, local_rank, dist, device, on_gpu):
     from gridllm_amd.config import EngineConfig
     from gridllm_amd.engine import InferenceEngine
     from gridllm_amd.engine.sampler import SamplerState
     from gridllm_amd.types import SamplingOptions
 
+    n_gpus = world if world > 1 else args.gpus
     model = args.model if on_gpu else "tiny"
     batch = args.batch if on_gpu else 4
-    # +1 slot so the TTFT measurement can allocate while the decode batch
-    # holds `batch` slots
     cfg = EngineConfig(model=model, device=device,
                        dtype="bf16" if on_gpu else "fp32",
                        max_seq_len=args.max_seq, max_batch_size=batch + 1,
                        use_hipgraph=on_gpu)
     eng = InferenceEngine(cfg)
 
-    # --- synthetic prompts: prefill `batch` slots with prompt_len random
-    # tokens (batched prefill: chunks of 64 sequences per pass)
     torch.manual_seed(1000 + rank)
     slots = [eng.model.cache.alloc() for _ in range(batch)]
     vocab = eng.spec.vocab
     for i in range(0, batch, 64):
         chunk = slots[i:i + 64]
-        toks = [torch.randint(0, vocab, (args.prompt_len,), device=eng.device)
-                for _ in chunk]
+        toks = [torch.randint(0, vocab, (args.prompt_tokens,),
+                              device=eng.device) for _ in chunk]
         if len(chunk) > 1:
             eng.model.forward_prefill_batch(toks, chunk)
         else:
@@ -88,13 +372,13 @@ def main() -> None:
             self.state = SamplerState.create(opts, eng.device)
             self.constraint = None
     rows = [_R() for _ in range(batch)]
-    tokens = [int(torch.randint(0, vocab, (1,)).item()) for _ in range(batch)]
+    tokens = [int(torch.randint(0, vocab, (1,)).item())
+              for _ in range(batch)]
 
-    def step() -> list[int]:
+    def step():
         logits = eng._decode_step(tokens, slots)
         return eng._sample_batch(logits, rows)
 
-    # --- warmup (also captures the hipGraph)
     for _ in range(args.warmup):
         tokens = step()
 
@@ -121,15 +405,12 @@ def main() -> None:
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e.item())
 
-    ms_per_step = elapsed * 1000.0 / args.steps
-    total_tokens = args.steps * batch * n_gpus
-    toks_per_sec = total_tokens / elapsed
-
-    # --- p50 TTFT (prefill + first token), measured outside the timed region
+    # p50 TTFT (prefill + first token), outside the timed region
     ttfts = []
     for _ in range(5):
         s = eng.model.cache.alloc()
-        ids = torch.randint(0, vocab, (args.prompt_len,), device=eng.device)
+        ids = torch.randint(0, vocab, (args.prompt_tokens,),
+                            device=eng.device)
         if on_gpu:
             torch.cuda.synchronize()
         tt = time.monotonic()
@@ -137,17 +418,17 @@ def main() -> None:
         _ = int(logits.argmax().item())
         ttfts.append((time.monotonic() - tt) * 1000.0)
         eng.model.cache.free(s)
-    ttft_p50 = statistics.median(ttfts)
 
     if rank == 0:
+        total_tokens = args.steps * batch * n_gpus
         out = {
             "metric": "output tokens/sec (node), llama3.2 decode",
-            "value": round(toks_per_sec, 1),
+            "value": round(total_tokens / elapsed, 1),
             "unit": "tokens/s",
             "n_gpus": n_gpus,
             "steps": args.steps,
             "warmup": args.warmup,
-            "ms_per_step": round(ms_per_step, 3),
+            "ms_per_step": round(elapsed * 1000.0 / args.steps, 3),
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
@@ -156,13 +437,67 @@ def main() -> None:
             "config": {
                 "model": model,
                 "global_batch": batch * n_gpus,
-                "seq_len": args.prompt_len,
+                "seq_len": args.prompt_tokens,
                 "max_seq": args.max_seq,
                 "parallelism": f"dp{n_gpus}",
-                "ttft_p50_ms": round(ttft_p50, 2),
+                "mode": "engine",
+                "ttft_p50_ms": round(statistics.median(ttfts), 2),
             },
         }
         print(json.dumps(out))
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--mode", choices=["serving", "engine"],
+                    default="serving")
+    ap.add_argument("--model", default="llama3.2:3b")
+    ap.add_argument("--concurrency", type=int, default=192,
+                    help="serving: concurrent streams per GPU")
+    ap.add_argument("--num-predict", type=int, default=64,
+                    help="serving: tokens generated per request")
+    ap.add_argument("--flush-ms", type=int, default=25,
+                    help="serving: worker stream flush interval")
+    ap.add_argument("--batch", type=int, default=1024,
+                    help="engine: sequences per GPU")
+    ap.add_argument("--prompt-tokens", type=int, default=128)
+    ap.add_argument("--prompt-len", type=int, dest="prompt_tokens",
+                    help=argparse.SUPPRESS)   # back-compat alias
+    ap.add_argument("--max-seq", type=int, default=1024)
+    ap.add_argument("--device", default="")
+    args = ap.parse_args()
+
+    # keep stdout clean for the one result JSON line (worker/engine logs
+    # go to stderr-level warn only)
+    os.environ.setdefault("LOG_LEVEL", "warn")
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    on_gpu = torch.cuda.is_available()
+    device = args.device or (f"cuda:{local_rank}" if on_gpu else "cpu")
+    if on_gpu:
+        torch.cuda.set_device(device)
+    if not on_gpu:
+        # CPU smoke defaults: tiny waves so the no-GPU path finishes fast
+        args.concurrency = min(args.concurrency, 8)
+        args.num_predict = min(args.num_predict, 8)
+        args.prompt_tokens = min(args.prompt_tokens, 16)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        dist.init_process_group(backend="nccl" if on_gpu else "gloo")
+
+    if args.mode == "serving":
+        serving_mode(args, rank, world, local_rank, dist, device, on_gpu)
+    else:
+        engine_mode(args, rank, world, local_rank, dist, device, on_gpu)
 
     if dist is not None:
         dist.destroy_process_group()
