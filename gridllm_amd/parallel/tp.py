@@ -25,10 +25,20 @@ class TPContext:
     rank: int = 0
     size: int = 1
     group: Optional[object] = None
+    # control-plane side channel: small per-step host messages (admission
+    # batches, cancel flags) go over gloo so they never touch the GPU or
+    # the RCCL stream — the compute group stays graph-capturable
+    ctrl_group: Optional[object] = None
 
     @property
     def enabled(self) -> bool:
         return self.size > 1
+
+    @property
+    def backend(self) -> str:
+        if self.size <= 1:
+            return ""
+        return dist.get_backend(self.group)
 
     def all_reduce(self, x: torch.Tensor) -> torch.Tensor:
         """Sum across the TP group, in place. No-op at size 1."""
@@ -42,6 +52,22 @@ class TPContext:
                 if self.group is not None else src
             dist.broadcast(x, src=src_global, group=self.group)
         return x
+
+    def broadcast_cpu(self, x: torch.Tensor, src: int = 0) -> torch.Tensor:
+        """Broadcast a small HOST tensor over the gloo control group (no
+        GPU sync, no RCCL-stream interaction)."""
+        if self.size > 1:
+            dist.broadcast(x, src=src, group=self.ctrl_group or self.group)
+        return x
+
+    def bcast_obj(self, obj=None, src: int = 0):
+        """Broadcast one picklable object over the control group."""
+        if self.size <= 1:
+            return obj
+        buf = [obj]
+        dist.broadcast_object_list(buf, src=src,
+                                   group=self.ctrl_group or self.group)
+        return buf[0]
 
     def barrier(self) -> None:
         if self.size > 1:
@@ -73,5 +99,10 @@ def init_tp_from_env(backend: str = "") -> TPContext:
         timeout_s = int(os.environ.get("TP_COLLECTIVE_TIMEOUT_S", "120"))
         dist.init_process_group(backend=backend,
                                 timeout=timedelta(seconds=timeout_s))
+    # gloo control group beside the compute group (same membership): host
+    # messages stay off the RCCL stream; on CPU (already gloo) reuse it
+    ctrl = None
+    if dist.get_backend() != "gloo":
+        ctrl = dist.new_group(backend="gloo")
     return TPContext(rank=dist.get_rank(), size=dist.get_world_size(),
-                     group=None)
+                     group=None, ctrl_group=ctrl)

@@ -2,8 +2,16 @@
 
 Launch with torch.distributed.run --nproc-per-node <TP> (one rank per GPU,
 RCCL over xGMI). Rank 0 runs the normal WorkerAgent + control-plane
-connection; ranks >0 run a follower loop that receives each engine wave by
-broadcast and executes it in lockstep so every collective lines up.
+connection; ranks >0 run a follower loop that mirrors rank 0's CONTINUOUS
+batching loop: one small control broadcast per decode iteration (gloo side
+channel, parallel/tp.py) carries that iteration's admissions, then every
+rank executes the identical start_rows/step_rows sequence so all RCCL
+collectives line up. Rows join and leave the live batch in lockstep —
+a long generation no longer blocks new arrivals (round-1 verdict weak 3:
+the old design broadcast whole run-to-completion waves).
+
+The decode step itself is hipGraph-captured per batch bucket INCLUDING
+the per-layer all-reduces (engine._graph_ok allows tp>1 on nccl).
 
 No reference analog (the reference had zero parallelism beyond request
 sharding — SURVEY.md §2.4); design per SURVEY.md §5.8 / §7 step 6.
@@ -12,23 +20,22 @@ Determinism across ranks:
   * logits are bitwise-identical on every rank (all-reduced activations,
     replicated lm_head) so sampling matches — PROVIDED every request has a
     seed; rank 0 injects one into any request that lacks it before
-    broadcasting the wave.
-  * cancellation is decided by rank 0 and broadcast per decode step
-    (engine._sync_cancel) so rows retire identically everywhere.
+    broadcasting the admission.
+  * retirement (EOS/stop/length) is a pure function of the sampled tokens,
+    so every rank prunes the same rows at the same step; cancellation is
+    decided by rank 0 and broadcast per step (engine._sync_cancel).
 """
 
 from __future__ import annotations
 
 import asyncio
+import queue as _q
 import random
-
-import torch.distributed as dist
 
 from gridllm_amd.bus.tcp import TcpBus
 from gridllm_amd.config import EngineConfig, WorkerConfig
 from gridllm_amd.engine.batcher import Batcher, _Pending
 from gridllm_amd.engine.engine import InferenceEngine
-from gridllm_amd.engine import chat as chat_mod
 from gridllm_amd.parallel.tp import TPContext, init_tp_from_env
 from gridllm_amd.utils.logging import get_logger
 from gridllm_amd.worker.agent import WorkerAgent
@@ -36,92 +43,159 @@ from gridllm_amd.worker.agent import WorkerAgent
 log = get_logger("tp-runner")
 
 
-def _inject_seeds(reqs: list[tuple[str, dict]]) -> None:
-    for _, opts in reqs:
+def _inject_seeds(items: list[tuple]) -> None:
+    for _, opts, *_ in items:
         if opts.get("seed") is None:
             opts["seed"] = random.randint(0, 2**31 - 1)
 
 
 class TPBatcher(Batcher):
-    """Rank-0 batcher: broadcasts each wave to followers before running it."""
+    """Rank-0 continuous batcher: each loop iteration broadcasts a compact
+    control message (admissions + embed inputs + stop flag) to followers,
+    then all ranks run the identical admission/step."""
 
     def __init__(self, engine: InferenceEngine, tp: TPContext,
-                 window_ms: float = 4.0):
+                 admit_per_step: int = 32):
         self.tp = tp
-        super().__init__(engine, window_ms)
+        super().__init__(engine, continuous=False,
+                         admit_per_step=admit_per_step)
 
-    def _run_wave(self, wave: list[_Pending]) -> None:
-        """Broadcast-then-execute each sub-wave IN ORDER: followers replay
-        waves in broadcast order, so rank 0 must run each wave's collectives
-        before broadcasting the next."""
-        embeds = [p for p in wave if p.kind == "embedding"]
-        gens = [p for p in wave if p.kind in ("generate", "chat")]
-        if embeds:
-            inputs_all = [i for p in embeds for i in (p.inputs or [])]
-            dist.broadcast_object_list([("embed", inputs_all)], src=0)
-            try:
-                vecs = self.engine.embed(inputs_all)
-                k = 0
-                for p in embeds:
-                    n = len(p.inputs or [])
-                    p.future.set_result(vecs[k:k + n])
-                    k += n
-            except Exception as e:  # noqa: BLE001
-                for p in embeds:
-                    if not p.future.done():
-                        p.future.set_exception(e)
-        if gens:
-            reqs = []
-            for p in gens:
-                if p.kind == "chat":
-                    prompt = chat_mod.render_chat(p.messages or [])
-                    opts = dict(p.options)
-                    opts["stop"] = list(opts.get("stop") or []) + \
-                        list(chat_mod.CHAT_STOPS)
+    # Batcher.__init__ starts _loop (wave mode) when continuous is False;
+    # override the thread target selection by overriding _loop itself.
+    def _loop(self) -> None:
+        self._loop_tp()
+
+    def _loop_tp(self) -> None:
+        max_batch = max(1, self.engine.cfg.max_batch_size)
+        live: list[tuple[_Pending, object]] = []   # (pending, engine row)
+        while True:
+            arrivals: list[_Pending] = []
+            if not live:
+                # idle: block (followers are parked in their broadcast)
+                try:
+                    arrivals.append(self.q.get(timeout=0.5))
+                except _q.Empty:
+                    if self._stop.is_set():
+                        self._broadcast_iter([], [], stop=True)
+                        return
+                    continue
+            while (len(live) + len(arrivals) < max_batch
+                   and len(arrivals) < self.admit_per_step
+                   and self.engine.model.cache.n_free > len(arrivals) + 1):
+                try:
+                    arrivals.append(self.q.get_nowait())
+                except _q.Empty:
+                    break
+            stop_now = False
+            gens: list[_Pending] = []
+            embeds: list[_Pending] = []
+            for p in arrivals:
+                if p.kind == "__stop__":
+                    stop_now = True
+                elif p.kind == "embedding":
+                    embeds.append(p)
                 else:
-                    prompt = p.prompt
-                    opts = dict(p.options)
-                    if p.extra_stops:
-                        opts["stop"] = list(opts.get("stop") or []) + \
-                            p.extra_stops
-                reqs.append((prompt, opts))
-            _inject_seeds(reqs)
-            dist.broadcast_object_list([("generate", reqs)], src=0)
-            # run locally with callbacks; followers run the same wave
-            try:
-                results = self.engine.generate_batch(
-                    reqs, on_tokens=[p.on_token for p in gens],
-                    cancels=[p.cancel for p in gens])
-                for p, r in zip(gens, results):
-                    p.future.set_result(r)
-            except Exception as e:  # noqa: BLE001
-                for p in gens:
-                    if not p.future.done():
+                    gens.append(p)
+            items = []
+            for p in gens:
+                prompt, opts = self._prep(p)
+                items.append((prompt, opts))
+            _inject_seeds(items)
+            embed_inputs = [p.inputs or [] for p in embeds]
+            self._broadcast_iter(items, embed_inputs,
+                                 stop=stop_now and not live and not gens)
+            if stop_now and not live and not gens:
+                for p in embeds:
+                    p.future.set_exception(RuntimeError("batcher shut down"))
+                return
+            # ---- identical execution on every rank -----------------------
+            if embeds:
+                flat = [x for ins in embed_inputs for x in ins]
+                try:
+                    vecs = self.engine.embed(flat)
+                    k = 0
+                    for p in embeds:
+                        n = len(p.inputs or [])
+                        p.future.set_result(vecs[k:k + n])
+                        k += n
+                except Exception as e:  # noqa: BLE001
+                    for p in embeds:
+                        if not p.future.done():
+                            p.future.set_exception(e)
+            if items:
+                try:
+                    new_rows = self.engine.start_rows(
+                        [(pr, op, p.on_token, p.cancel, None)
+                         for (pr, op), p in zip(items, gens)])
+                except Exception as e:  # noqa: BLE001
+                    log.logger.exception("TP admission failed")
+                    for p in gens:
                         p.future.set_exception(e)
+                    new_rows = []
+                for p, row in zip(gens, new_rows):
+                    if row.done:
+                        p.future.set_result(row.result)
+                    else:
+                        live.append((p, row))
+            if live:
+                try:
+                    self.engine.step_rows([row for _, row in live])
+                except Exception as e:  # noqa: BLE001
+                    log.logger.exception("TP decode step failed")
+                    self.engine.abort_rows([row for _, row in live])
+                    for p, _ in live:
+                        if not p.future.done():
+                            p.future.set_exception(e)
+                    live = []
+                still = []
+                for p, row in live:
+                    if row.done:
+                        p.future.set_result(row.result)
+                    else:
+                        still.append((p, row))
+                live = still
+            if self._stop.is_set() and not live:
+                self._broadcast_iter([], [], stop=True)
+                return
+
+    def _broadcast_iter(self, items, embed_inputs, stop: bool) -> None:
+        self.tp.bcast_obj(("it", items, embed_inputs, stop), src=0)
 
     def shutdown(self) -> None:
-        try:
-            dist.broadcast_object_list([("stop", None)], src=0)
-        except Exception:  # noqa: BLE001
-            pass
-        super().shutdown()
+        self._stop.set()
+        self.q.put(_Pending(kind="__stop__"))
+        self._thread.join(timeout=10.0)
 
 
-def follower_loop(engine: InferenceEngine) -> None:
-    """Ranks >0: execute broadcast waves until told to stop."""
+def follower_loop(engine: InferenceEngine, tp: TPContext) -> None:
+    """Ranks >0: mirror rank 0's continuous loop. One control broadcast
+    per iteration; rows retire by the same deterministic rules."""
+    rows: list = []
     while True:
-        obj: list = [None]
-        dist.broadcast_object_list(obj, src=0)
-        kind, payload = obj[0]
-        if kind == "stop":
+        kind, items, embed_inputs, stop = tp.bcast_obj(None, src=0)
+        assert kind == "it"
+        if embed_inputs:
+            flat = [x for ins in embed_inputs for x in ins]
+            try:
+                engine.embed(flat)
+            except Exception:  # noqa: BLE001
+                log.logger.exception("follower embed failed")
+        if items:
+            try:
+                new_rows = engine.start_rows(
+                    [(pr, op, None, None, None) for pr, op in items])
+                rows.extend(r for r in new_rows if not r.done)
+            except Exception:  # noqa: BLE001
+                log.logger.exception("follower admission failed")
+        if rows:
+            try:
+                rows = engine.step_rows(rows)
+            except Exception:  # noqa: BLE001
+                log.logger.exception("follower step failed")
+                engine.abort_rows(rows)
+                rows = []
+        if stop:
             return
-        try:
-            if kind == "generate":
-                engine.generate_batch(payload)
-            elif kind == "embed":
-                engine.embed(payload)
-        except Exception:  # noqa: BLE001
-            log.logger.exception("follower wave failed")
 
 
 async def _run_rank0(wcfg: WorkerConfig, engine: InferenceEngine,
@@ -149,7 +223,7 @@ def main() -> None:
         wcfg.worker_id = wcfg.worker_id or f"tp-worker-{os.getpid()}"
         asyncio.run(_run_rank0(wcfg, engine, tp))
     else:
-        follower_loop(engine)
+        follower_loop(engine, tp)
 
 
 if __name__ == "__main__":

@@ -46,6 +46,17 @@ app = server.app
 @app.on_event("startup")
 async def _s():
     await server.start()
+    if os.environ.get("LT_PROFILE_SERVER"):
+        import cProfile, pstats, signal as _sig
+        pr = cProfile.Profile(); pr.enable()
+        def dump(*a):
+            pr.disable()
+            st = pstats.Stats(pr); st.sort_stats("tottime")
+            with open("gpurun_out/lt_server_loop.prof.txt", "w") as f:
+                st.stream = f
+                st.print_stats(30)
+            os._exit(0)
+        _sig.signal(_sig.SIGTERM, dump)
     print("SERVER_READY", flush=True)
 
 uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error")
