@@ -82,6 +82,26 @@ def tune_shape(m: int, n: int, k: int, device) -> tuple[str, int]:
             if t < t_best * 0.97:
                 t_best = t
                 best = ("proj", sk)
+    if hip_ops.gemm_proj_supported(m, n, k) and n % 128 == 0:
+        # m-tiled decomposition (grid.y m-tiles, W slice re-read from XCD
+        # L2): candidate mtiles keep >=64 rows per tile; splitk composes
+        # when the grid is still thin
+        for mt in (2, 4, 8):
+            if m // mt < 64 and not (mt == 2 and m > 64):
+                continue
+            if (m + mt - 1) // mt > 256:
+                continue
+            rule = max(1, min(k // 256,
+                              256 // max(1, (n // 128) * mt)))
+            for sk in sorted({1, rule}):
+                try:
+                    t = _time_fn(lambda w: hip_ops.gemm_proj(
+                        x, w, splitk=sk, mtiles=mt), ws)
+                except RuntimeError:
+                    continue
+                if t < t_best * 0.97:
+                    t_best = t
+                    best = ("projm", sk, mt)
     if hip_ops.gemm_proj_wide_supported(m, n, k):
         rule = max(1, min(k // 256, 256 // max(1, n // 256)))
         for sk in sorted({1, rule, rule * 2}):
@@ -126,6 +146,17 @@ def tune_gateup(m: int, n: int, k: int, device) -> tuple[str, int]:
             if t < t_best * 0.97:
                 t_best = t
                 best = ("proj", sk)
+        for mt in (2, 4, 8):        # m-tiled + fused epilogue
+            if m // mt < 64 and not (mt == 2 and m > 64):
+                continue
+            try:
+                t = _time_fn(lambda w: hip_ops.gemm_proj(
+                    x, w, splitk=1, fuse_swiglu=True, mtiles=mt), ws)
+            except RuntimeError:
+                continue
+            if t < t_best * 0.97:
+                t_best = t
+                best = ("projm", 1, mt)
     del ws
     _CHOICE_GU[key] = best
     return best
@@ -137,6 +168,9 @@ def linear_gateup(x: torch.Tensor, w_il: torch.Tensor) -> torch.Tensor:
     c = _CHOICE_GU.get((x.shape[0], w_il.shape[0], w_il.shape[1]))
     if c is not None and c[0] == "proj":
         return hip_ops.gemm_proj(x, w_il, splitk=c[1], fuse_swiglu=True)
+    if c is not None and c[0] == "projm":
+        return hip_ops.gemm_proj(x, w_il, splitk=c[1], fuse_swiglu=True,
+                                 mtiles=c[2])
     return hip_ops.swiglu_il(F.linear(x, w_il))
 
 
@@ -153,6 +187,8 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
                 return hip_ops.gemm_skinny(x, w, splitk=c[1])
             if c[0] == "proj":
                 return hip_ops.gemm_proj(x, w, splitk=c[1])
+            if c[0] == "projm":
+                return hip_ops.gemm_proj(x, w, splitk=c[1], mtiles=c[2])
             if c[0] == "projw":
                 return hip_ops.gemm_proj_wide(x, w, splitk=c[1])
     return F.linear(x, w)

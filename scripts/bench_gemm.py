@@ -56,6 +56,29 @@ def main():
                     t = bench_cold(lambda w: hip_ops.gemm_proj(x, w, splitk=sk), ws)
                     print(f"    {name} proj splitk={sk:2d}: {t:7.1f}us "
                           f"({gb/t*1e6/1e3:5.2f} TB/s, grid={(N//64)*sk})")
+                # m-tiled decomposition sweep (W slice re-read from XCD L2)
+                best_mt = None
+                for mt in (2, 4, 8):
+                    if M // mt < 64 and not (mt == 2 and M > 64):
+                        continue
+                    if N % 128:
+                        continue
+                    for sk in (1, 2, 3, 4):
+                        if sk > max(1, K // 256):
+                            continue
+                        err = (hip_ops.gemm_proj(x, w, splitk=sk, mtiles=mt)
+                               .float() - ref).abs().max().item()
+                        t = bench_cold(lambda w: hip_ops.gemm_proj(
+                            x, w, splitk=sk, mtiles=mt), ws)
+                        ok = "OK" if err < 0.02 * scale + 1e-6 else \
+                            f"FAIL({err:.3f})"
+                        print(f"    {name} projm mt={mt} sk={sk}: "
+                              f"{t:7.1f}us ({gb/t*1e6/1e3:5.2f} TB/s, "
+                              f"grid={(N//128)*sk}x{mt}) {ok}")
+                        if best_mt is None or t < best_mt[0]:
+                            best_mt = (t, mt, sk)
+                if best_mt:
+                    tot["projm"] = tot.get("projm", 0.0) + best_mt[0]
         line = f"{name:8s} K={K:6d} N={N:6d} W={gb*1000:7.1f}MB "
         for kname, (t, err) in res.items():
             ok = "OK" if err < 0.02 * scale + 1e-6 else f"FAIL({err:.3f})"

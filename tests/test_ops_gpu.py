@@ -399,3 +399,46 @@ def test_prefill_batch_engine_parity(dev):
             dk = _mx(m1.cache.k[li][seq_slots[i], :, :n].cpu(),
                      m1.cache.k[li][bat_slots[i], :, :n].cpu())
             assert dk < 1e-4, (li, i, dk)
+
+
+@pytest.mark.parametrize("m,n,k,mt,sk", [
+    (256, 5120, 3072, 2, 1),   # qkv m-tiled
+    (256, 5120, 3072, 2, 3),   # + splitk compose
+    (256, 16384, 3072, 4, 1),  # gateup, 64-row tiles (4-wave path)
+    (192, 3072, 8192, 2, 4),   # down, ragged M
+    (96, 3072, 3072, 2, 1),    # rpt<=64 with partial last tile
+])
+def test_gemm_proj_mtiled(dev, m, n, k, mt, sk):
+    """M-tiled projection decomposition vs fp32 reference."""
+    torch.manual_seed(m + mt)
+    from gridllm_amd.ops import hip_ops
+    x = (torch.randn(m, k, device=dev) * 0.1).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) * 0.1).to(torch.bfloat16)
+    ref = x.float() @ w.float().t()
+    out = hip_ops.gemm_proj(x, w, splitk=sk, mtiles=mt)
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.02 * ref.abs().max().item() + 1e-6, err
+
+
+def test_gemm_proj_mtiled_fused_swiglu(dev):
+    """M-tiled + fused SwiGLU epilogue on an interleaved gate/up weight."""
+    torch.manual_seed(9)
+    from gridllm_amd.ops import hip_ops
+    m, f, k = 256, 8192, 3072
+    x = (torch.randn(m, k, device=dev) * 0.1).to(torch.bfloat16)
+    wg = (torch.randn(f, k, device=dev) * 0.1).to(torch.bfloat16)
+    wu = (torch.randn(f, k, device=dev) * 0.1).to(torch.bfloat16)
+    # interleave rows into 128-row groups [64 gate | 64 up]
+    g = torch.arange(f)
+    perm = torch.empty(2 * f, dtype=torch.int64)
+    perm[(g // 64) * 128 + g % 64] = g
+    perm[(g // 64) * 128 + g % 64 + 64] = f + g
+    w_il = torch.cat([wg, wu])[perm.to(dev)].contiguous()
+    gate = (x.float() @ wg.float().t())
+    up = (x.float() @ wu.float().t())
+    ref = gate * torch.sigmoid(gate) * up
+    for mt in (2, 4):
+        out = hip_ops.gemm_proj(x, w_il, splitk=1, fuse_swiglu=True,
+                                mtiles=mt)
+        err = (out.float() - ref).abs().max().item()
+        assert err < 0.03 * ref.abs().max().item() + 1e-5, (mt, err)
