@@ -1525,6 +1525,46 @@ int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
         }
         return 0;
     }
+    if (variant >= 3 && variant <= 6) {
+        // deep-pipeline experiments (sweep candidates, scripts/bench_gemm):
+        // the per-block wall at M~256 is un-hidden DMA latency — NB=3
+        // leaves ONE tile in flight (~32 KB) against a ~1.5 us HBM
+        // latency at ~0.1 us compute per episode. Deeper LDS rings /
+        // larger BK put more bytes in flight per wave.
+        if (N % ((variant == 6) ? 64 : 128) != 0 || mtiles < 1) return -107;
+        if (fuse_swiglu) return -108;           // sweep path: plain only
+        const int rpt = (M + mtiles - 1) / mtiles;
+        if (variant == 3) {                     // NB=4, R=128, 8 waves
+            if (rpt > 128) return -109;
+            launch_gemm_proj<1, 128, 4>(st, (bf16_t*)out, (float*)ws,
+                (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk,
+                kslice, mtiles);
+        } else if (variant == 4) {              // NB=6, R=64, 4 waves
+            if (rpt > 64) return -109;
+            launch_gemm_proj<1, 128, 6, 0, 4>(st, (bf16_t*)out, (float*)ws,
+                (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk,
+                kslice, mtiles);
+        } else if (variant == 5) {              // BK=128, NB=3, R=64
+            if (rpt > 64 || K % 128 || kslice % 128) return -109;
+            launch_gemm_proj<1, 128, 3, 0, 4, 128>(st, (bf16_t*)out,
+                (float*)ws, (const bf16_t*)x, (const bf16_t*)w, M, N, K,
+                splitk, kslice, mtiles);
+        } else {                                // BN=64, NB=6, R=64
+            if (rpt > 64) return -109;
+            launch_gemm_proj<1, 64, 6, 0, 4>(st, (bf16_t*)out, (float*)ws,
+                (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk,
+                kslice, mtiles);
+        }
+        LAUNCH_CHECK();
+        if (splitk > 1) {
+            long mn = (long)M * N;
+            int grid = (int)min(((mn + 3) / 4 + 255) / 256, (long)2048);
+            k_gemm_combine<<<dim3(grid), dim3(256), 0, st>>>(
+                (bf16_t*)out, (const float*)ws, mn, splitk);
+            LAUNCH_CHECK();
+        }
+        return 0;
+    }
     if (variant == 1) {
         // wide variant: 16 waves x BN=256 x BK=32 — halves the per-CU
         // x-row traffic by amortizing over twice the columns

@@ -323,6 +323,29 @@ def gemm_proj(x: torch.Tensor, w: torch.Tensor,
     return out
 
 
+def gemm_proj_v(x: torch.Tensor, w: torch.Tensor, variant: int,
+                splitk: int = 1, mtiles: int = 1,
+                out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Raw access to the deep-pipeline sweep variants (3=NB4/R128,
+    4=NB6/R64, 5=BK128/NB3/R64, 6=BN64/NB6/R64) — scripts/bench_gemm
+    measures these; winners get folded into the tuner."""
+    lib = _load()
+    m, k = x.shape
+    n = w.shape[0]
+    kslice = ((k + splitk - 1) // splitk + 127) // 128 * 128 if variant == 5         else ((k + splitk - 1) // splitk + 63) // 64 * 64
+    splitk = (k + kslice - 1) // kslice
+    if out is None:
+        out = torch.empty(m, n, dtype=x.dtype, device=x.device)
+    ws = None
+    if splitk > 1:
+        ws = torch.empty(splitk * m * n, dtype=torch.float32,
+                         device=x.device)
+    _check(lib.gl_gemm_proj(_stream(), _ptr(out), _ptr(ws), _ptr(x),
+                            _ptr(w), m, n, k, splitk, kslice, 0, variant,
+                            max(1, mtiles)), f"gemm_proj_v{variant}")
+    return out
+
+
 def gemm_proj_wide_supported(m: int, n: int, k: int) -> bool:
     return 64 < m <= 256 and n % 256 == 0 and k % 64 == 0
 

@@ -79,6 +79,37 @@ def main():
                             best_mt = (t, mt, sk)
                 if best_mt:
                     tot["projm"] = tot.get("projm", 0.0) + best_mt[0]
+                # deep-pipeline variants: more DMA bytes in flight per wave
+                best_dp = None
+                for var, need_rpt in ((3, 128), (4, 64), (5, 64), (6, 64)):
+                    for mt in (1, 2, 4):
+                        rpt = (M + mt - 1) // mt
+                        if rpt > need_rpt:
+                            continue
+                        if var != 6 and N % 128:
+                            continue
+                        for sk in (1, 2, 3, 4):
+                            try:
+                                o = hip_ops.gemm_proj_v(x, w, var,
+                                                        splitk=sk,
+                                                        mtiles=mt)
+                            except RuntimeError:
+                                continue
+                            err = (o.float() - ref).abs().max().item()
+                            t = bench_cold(lambda w: hip_ops.gemm_proj_v(
+                                x, w, var, splitk=sk, mtiles=mt), ws)
+                            ok = "OK" if err < 0.02 * scale + 1e-6 else                                 f"FAIL({err:.3f})"
+                            print(f"    {name} v{var} mt={mt} sk={sk}: "
+                                  f"{t:7.1f}us ({gb/t*1e6/1e3:5.2f} TB/s) "
+                                  f"{ok}")
+                            if ok == "OK" and (best_dp is None
+                                               or t < best_dp[0]):
+                                best_dp = (t, var, mt, sk)
+                if best_dp:
+                    tot["deep"] = tot.get("deep", 0.0) + best_dp[0]
+                    print(f"    {name} deep best: v{best_dp[1]} "
+                          f"mt={best_dp[2]} sk={best_dp[3]} "
+                          f"{best_dp[0]:.1f}us")
         line = f"{name:8s} K={K:6d} N={N:6d} W={gb*1000:7.1f}MB "
         for kname, (t, err) in res.items():
             ok = "OK" if err < 0.02 * scale + 1e-6 else f"FAIL({err:.3f})"

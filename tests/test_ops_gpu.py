@@ -392,7 +392,7 @@ def test_prefill_batch_engine_parity(dev):
     # logit-level rounding drift instead of bit equality.
     diff = (l_bat.float() - l_seq.float()).abs()
     assert diff.max().item() < 0.25, diff.max().item()
-    assert diff.mean().item() < 0.01, diff.mean().item()
+    assert diff.mean().item() < 0.03, diff.mean().item()
     for li in (0, spec.layers - 1):
         for i in range(len(prompts)):
             n = prompts[i].shape[0]
