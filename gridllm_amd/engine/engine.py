@@ -112,6 +112,8 @@ class InferenceEngine:
         self._scratch_slot = self.cfg.max_batch_size  # last slot
         self.model.cache.reserve(self._scratch_slot)
         self._graphs: dict[int, tuple] = {}
+        from gridllm_amd.engine.sampler import SlotSampler
+        self.sampler = SlotSampler(self.cfg.max_batch_size + 1, self.device)
         if self.is_gpu and self.cfg.autotune:
             from gridllm_amd.ops.autotune import tune_model_shapes
             buckets = [b for b in _BUCKETS if b <= self.cfg.max_batch_size]
@@ -176,8 +178,12 @@ class InferenceEngine:
         log.performance("hipGraph captured", bucket=bucket)
         return self._graphs[bucket]
 
-    def _decode_step(self, token_ids: list[int], slot_ids: list[int]) -> torch.Tensor:
-        """Run one decode step for rows (token, slot). Returns logits [B, V]."""
+    def _decode_step(self, token_ids: list[int], slot_ids: list[int],
+                     slots_t: "Optional[torch.Tensor]" = None
+                     ) -> torch.Tensor:
+        """Run one decode step for rows (token, slot). Returns logits [B, V].
+        slots_t: optional prebuilt device int32 slot tensor (callers that
+        also sample by slot share ONE upload per step)."""
         b = len(token_ids)
         if self._graph_ok():
             bucket = self._bucket(b)
@@ -200,19 +206,26 @@ class InferenceEngine:
                     sbuf[b:] = self._scratch_slot
                     # keep the scratch slot from overflowing across replays
                     self.model.cache.seq_lens[self._scratch_slot] = 0
-                sbuf[:b] = torch.as_tensor(slot_ids, dtype=torch.int32,
-                                           device=self.device)
+                if slots_t is not None:
+                    sbuf[:b] = slots_t
+                else:
+                    sbuf[:b] = torch.as_tensor(slot_ids, dtype=torch.int32,
+                                               device=self.device)
                 g.replay()
                 return out[:b]
         tokens = torch.as_tensor(token_ids, dtype=torch.int64, device=self.device)
-        slots = torch.as_tensor(slot_ids, dtype=torch.int32, device=self.device)
+        slots = slots_t if slots_t is not None else torch.as_tensor(
+            slot_ids, dtype=torch.int32, device=self.device)
         return self.model.forward_decode(tokens, slots)
 
     # ------------------------------------------------------------- sampling
 
     def _sample_batch(self, logits: torch.Tensor,
-                      rows: Sequence[_Row]) -> list[int]:
-        """Batched sampling; vectorized for the common uniform cases."""
+                      rows: Sequence[_Row],
+                      slots_t: "Optional[torch.Tensor]" = None) -> list[int]:
+        """Batched sampling; vectorized for the common uniform cases.
+        With slots_t (device int64 slot ids) the per-slot device-resident
+        sampler runs — zero per-step H2D (see sampler.SlotSampler)."""
         states = [r.state for r in rows]
         for i, r in enumerate(rows):
             c = getattr(r, "constraint", None)   # tolerate row shims
@@ -220,6 +233,8 @@ class InferenceEngine:
                 self._apply_constraint_mask(logits[i], c)
         if all(s.greedy for s in states):
             return logits.argmax(dim=-1).tolist()
+        if slots_t is not None:
+            return self.sampler.sample(logits, rows, slots_t)
         from gridllm_amd.engine.sampler import (BatchPlan,
                                                 sample_batch_vectorized)
         plan = getattr(self, "_sampler_plan", None)
@@ -426,7 +441,10 @@ class InferenceEngine:
             else:
                 logits = self.model.forward_prefill_batch(token_lists, slots)
             self._sync_cancel(rows)
-            toks = self._sample_batch(logits, rows)           # syncs
+            self.sampler.admit(rows, self.spec.vocab)
+            slots_t = torch.as_tensor(slots, dtype=torch.int64,
+                                      device=self.device)
+            toks = self._sample_batch(logits, rows, slots_t)  # syncs
         except Exception:
             # prefill/first-sample failure: release every slot this batch
             # allocated or capacity leaks permanently
@@ -461,8 +479,13 @@ class InferenceEngine:
                     r.cur_len = self.model.shift_slot(r.slot)
         token_ids = [r.last_token for r in active]
         slot_ids = [r.slot for r in active]
-        logits = self._decode_step(token_ids, slot_ids)
-        next_ids = self._sample_batch(logits, active)
+        # one slot upload per step, shared by decode (int32 view) and the
+        # per-slot sampler (int64)
+        slots64 = torch.as_tensor(slot_ids, dtype=torch.int64,
+                                  device=self.device)
+        logits = self._decode_step(token_ids, slot_ids,
+                                   slots_t=slots64.to(torch.int32))
+        next_ids = self._sample_batch(logits, active, slots64)
         self._sync_cancel(active)
         still: list[_Row] = []
         for r, tok in zip(active, next_ids):

@@ -405,10 +405,15 @@ class LlamaModel:
                          if li + 1 < len(self.layers) else self.final_norm)
             normed, resid = ops.rmsnorm(mlp_out, next_norm, s.norm_eps,
                                         residual=resid)
-        for ln, sl in zip(lens, slots):
-            self.cache.seq_lens[sl] = ln
-        lasts = torch.tensor([b - 1 for b in bounds[1:]], dtype=torch.long,
-                             device=self.device)
+        # ONE scatter instead of k per-sequence device writes (each is a
+        # serialized H2D launch — 15 ms of host time per admission batch
+        # in the round-2 serving profile)
+        import numpy as _np
+        host = _np.asarray([[sl for sl in slots], lens,
+                            [b - 1 for b in bounds[1:]]], dtype=_np.int64)
+        dev3 = torch.from_numpy(host).to(self.device)
+        self.cache.seq_lens[dev3[0]] = dev3[1].to(torch.int32)
+        lasts = dev3[2]
         logits = _linear(normed[lasts], self.lm_head).float()
         if return_hidden:
             hidden = torch.stack([

@@ -101,7 +101,7 @@ async def main():
     if slow > 0:
         orig = engine._decode_step
         import time as _t
-        def slow_step(tokens, slots):
+        def slow_step(tokens, slots, **kw):
             _t.sleep(slow)
             return orig(tokens, slots)
         engine._decode_step = slow_step

@@ -47,7 +47,7 @@ async def main():
     if slow > 0:  # make decode slow so chaos test can kill mid-job
         orig = engine._decode_step
         import time as _t
-        def slow_step(tokens, slots):
+        def slow_step(tokens, slots, **kw):
             _t.sleep(slow)
             return orig(tokens, slots)
         engine._decode_step = slow_step
