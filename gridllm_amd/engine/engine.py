@@ -496,6 +496,127 @@ class InferenceEngine:
                 self._retire(r)
         return still
 
+    # ------------------------------------------------ pipelined stepping
+    # The synchronous loop serializes: GPU step -> host sync -> per-token
+    # bookkeeping -> next step; at serving batch sizes the host segment is
+    # 30-50% of the step. The pipelined path submits step N with step
+    # N-1's sampled ids STILL ON DEVICE (a gather feeds the next graph
+    # replay), then processes N-1's ids on the host while N runs. Costs:
+    # retires land one step late (one wasted decode step per finishing
+    # row); rows with grammar constraints, sampling penalties, or at the
+    # context boundary take the synchronous path; TP stays synchronous
+    # (the follower mirror steps in lockstep).
+
+    def pipeline_ok(self, rows: Sequence["_Row"]) -> bool:
+        if self.tp.enabled:
+            return False
+        for r in rows:
+            if r.constraint is not None or r.state.has_penalty:
+                return False
+            if r.cur_len + 2 > self.model.max_seq:
+                return False
+        return True
+
+    def _pinned(self, n: int) -> torch.Tensor:
+        """Reusable pinned host buffers (ring of 2: one in flight, one
+        being collected)."""
+        cache = getattr(self, "_pin_ring", None)
+        if cache is None:
+            cache = self._pin_ring = {}
+        ring = cache.setdefault(n, [None, None, 0])
+        i = ring[2] = (ring[2] + 1) % 2
+        if ring[i] is None:
+            ring[i] = torch.empty(n, dtype=torch.int64, pin_memory=True)
+        return ring[i]
+
+    def submit_step(self, rows: list["_Row"], prev=None):
+        """Launch one decode step WITHOUT waiting for its result.
+        Token sources: rows present in `prev` gather their token from
+        prev's on-device ids; others (new admissions, sync-stepped rows)
+        use their host-known last_token. Returns an inflight handle for
+        collect_step. Caller ensures pipeline_ok(rows)."""
+        with self._lock:
+            import numpy as np
+            b = len(rows)
+            if self.cfg.context_shift:
+                for r in rows:
+                    if r.cur_len >= self.model.max_seq:
+                        r.cur_len = self.model.shift_slot(r.slot)
+            src = np.full(b, -1, dtype=np.int64)
+            host_tok = np.zeros(b, dtype=np.int64)
+            slots_np = np.zeros(b, dtype=np.int64)
+            pos = prev["pos"] if prev is not None else {}
+            for i, r in enumerate(rows):
+                j = pos.get(id(r), -1)
+                src[i] = j
+                host_tok[i] = r.last_token
+                slots_np[i] = r.slot
+            up = torch.from_numpy(np.stack([src, host_tok, slots_np])) \
+                .to(self.device)
+            if prev is not None and (up[0] >= 0).any():
+                tokens_dev = torch.where(
+                    up[0] >= 0,
+                    prev["ids_dev"][up[0].clamp(min=0)], up[1])
+            else:
+                tokens_dev = up[1]
+            slots64 = up[2]
+            slots32 = slots64.to(torch.int32)
+            if self._graph_ok():
+                bucket = self._bucket(b)
+                try:
+                    g, tbuf, sbuf, out = self._ensure_graph(bucket)
+                except Exception:  # noqa: BLE001
+                    log.logger.exception(
+                        "hipGraph capture failed — eager decode")
+                    self.cfg.use_hipgraph = False
+                    g = None
+                if g is not None:
+                    tbuf[:b] = tokens_dev
+                    if b < bucket:
+                        tbuf[b:] = 0
+                        sbuf[b:] = self._scratch_slot
+                        self.model.cache.seq_lens[self._scratch_slot] = 0
+                    sbuf[:b] = slots32
+                    g.replay()
+                    logits = out[:b]
+                else:
+                    logits = self.model.forward_decode(tokens_dev, slots32)
+            else:
+                logits = self.model.forward_decode(tokens_dev, slots32)
+            ids_dev = self.sampler.sample_ids(logits, rows, slots64)
+            if self.is_gpu:
+                ids_host = self._pinned(b)
+                ids_host[:b].copy_(ids_dev, non_blocking=True)
+                ev = torch.cuda.Event()
+                ev.record()
+            else:
+                ids_host, ev = ids_dev, None
+            for r in rows:
+                r.cur_len += 1          # the step appended to the KV slot
+            free_after = getattr(self, "_deferred_free", [])
+            self._deferred_free = []
+            return {"rows": rows, "ids_dev": ids_dev, "ids_host": ids_host,
+                    "ev": ev, "pos": {id(r): i for i, r in enumerate(rows)},
+                    "free_after": free_after}
+
+    def collect_step(self, inflight) -> None:
+        """Wait for an inflight step's sampled ids and run the per-token
+        bookkeeping (emit/stop/EOS/retire). Slots retired here stay
+        allocated until the NEXT collect (they were already inside the
+        following in-flight step)."""
+        with self._lock:
+            ev = inflight["ev"]
+            if ev is not None:
+                ev.synchronize()
+            ids = inflight["ids_host"][:len(inflight["rows"])].tolist()
+            for r, tok in zip(inflight["rows"], ids):
+                if r.done:
+                    continue            # retired while this step flew
+                if not self._emit(r, int(tok)):
+                    self._retire(r, defer_free=True)
+            for sl in inflight["free_after"]:
+                self.model.cache.free(sl)
+
     def _sync_cancel(self, rows: Sequence[_Row]) -> None:
         """Under TP every rank must retire the same rows at the same step, or
         the next collective deadlocks (SURVEY.md §7 hard-part 5). Rank 0 owns

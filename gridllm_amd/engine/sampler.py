@@ -59,6 +59,12 @@ class SamplerState:
                     or (o.top_p and 0.0 < o.top_p < 1.0)
                     or (o.min_p and 0.0 < o.min_p < 1.0))
 
+    @property
+    def has_penalty(self) -> bool:
+        o = self.opts
+        return bool(o.repeat_penalty or o.presence_penalty
+                    or o.frequency_penalty)
+
     def observe(self, token_id: int, vocab: int) -> None:
         self.draws += 1
         o = self.opts
@@ -416,9 +422,17 @@ class SlotSampler:
         states = [r.state for r in rows]
         if any(s.counts for s in states):
             _apply_penalties(logits, states)
-        all_greedy = all(s.greedy for s in states)
-        if all_greedy:
-            return logits.argmax(dim=-1).tolist()
+        return self.sample_ids(logits, rows, slot_ids).tolist()
+
+    def sample_ids(self, logits: torch.Tensor, rows,
+                   slot_ids: torch.Tensor) -> torch.Tensor:
+        """Device-tensor sampling (no host sync): the pipelined decode
+        path keeps the sampled ids ON DEVICE and feeds them into the next
+        step. Penalty application (host-sparse) is the caller's concern;
+        rows with penalties take the synchronous path instead."""
+        states = [r.state for r in rows]
+        if all(s.greedy for s in states):
+            return logits.argmax(dim=-1)
         need_filter = any(s.need_filter for s in states)
         temps = self.temps[slot_ids].unsqueeze(1)
         if (logits.is_cuda and not need_filter
@@ -426,7 +440,7 @@ class SlotSampler:
             from gridllm_amd.ops import hip_ops
             return hip_ops.sample_gumbel(
                 logits.contiguous(), self._keys(slot_ids).contiguous(),
-                temps.view(-1).contiguous()).tolist()
+                temps.view(-1).contiguous())
         x = logits / temps
         v = logits.shape[1]
         if need_filter:
@@ -451,4 +465,4 @@ class SlotSampler:
         u = u.clamp(1e-6, 1.0 - 1e-6)
         noise = -torch.log(-torch.log(u))
         gm = self.nongreedy[slot_ids].unsqueeze(1).float()
-        return (x + noise * gm).argmax(-1).tolist()
+        return (x + noise * gm).argmax(-1)

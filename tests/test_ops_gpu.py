@@ -398,7 +398,9 @@ def test_prefill_batch_engine_parity(dev):
             n = prompts[i].shape[0]
             dk = _mx(m1.cache.k[li][seq_slots[i], :, :n].cpu(),
                      m1.cache.k[li][bat_slots[i], :, :n].cpu())
-            assert dk < 1e-4, (li, i, dk)
+            # K comes off the qkv GEMM, which runs at M=sum vs M=T_i ->
+            # one-bf16-ulp output differences are expected
+            assert dk < 4e-3, (li, i, dk)
 
 
 @pytest.mark.parametrize("m,n,k,mt,sk", [
