@@ -106,20 +106,56 @@ class Batcher:
 
     def _loop_continuous(self) -> None:
         max_batch = max(1, self.engine.cfg.max_batch_size)
+        eng = self.engine
         live: list[tuple[_Pending, object]] = []   # (pending, engine row)
+        inflight = None                            # pipelined step handle
+
+        def prune() -> None:
+            nonlocal live
+            still = []
+            for p, row in live:
+                if row.done:
+                    if not p.future.done():
+                        p.future.set_result(row.result)
+                else:
+                    still.append((p, row))
+            live = still
+
+        def fail_live(e: Exception) -> None:
+            nonlocal live
+            eng.abort_rows([row for _, row in live])
+            eng.flush_deferred()
+            for p, _ in live:
+                if not p.future.done():
+                    p.future.set_exception(e)
+            live = []
+
+        def drain_inflight() -> None:
+            nonlocal inflight
+            if inflight is not None:
+                try:
+                    eng.collect_step(inflight)
+                except Exception as e:  # noqa: BLE001
+                    log.logger.exception("pipelined collect failed")
+                    fail_live(e)
+                inflight = None
+                eng.flush_deferred()
+                prune()
+
         while True:
             # -- admit: block only when idle, otherwise drain without waiting
             arrivals: list[_Pending] = []
-            if not live:
+            if not live and inflight is None:
                 try:
                     arrivals.append(self.q.get(timeout=0.5))
                 except queue.Empty:
                     if self._stop.is_set():
                         break
                     continue
+            # one slot spare: a pipelined retire frees one step late
             while (len(live) + len(arrivals) < max_batch
                    and len(arrivals) < self.admit_per_step
-                   and self.engine.model.cache.n_free > len(arrivals)):
+                   and eng.model.cache.n_free > len(arrivals) + 1):
                 try:
                     arrivals.append(self.q.get_nowait())
                 except queue.Empty:
@@ -130,8 +166,9 @@ class Batcher:
                 if p.kind == "__stop__":
                     stop_now = True
                 elif p.kind == "embedding":
+                    drain_inflight()   # embeds share slots: settle state
                     try:
-                        p.future.set_result(self.engine.embed(p.inputs or []))
+                        p.future.set_result(eng.embed(p.inputs or []))
                     except Exception as e:  # noqa: BLE001
                         p.future.set_exception(e)
                 else:
@@ -142,7 +179,7 @@ class Batcher:
                     prompt, opts = self._prep(p)
                     items.append((prompt, opts, p.on_token, p.cancel, None))
                 try:
-                    new_rows = self.engine.start_rows(items)
+                    new_rows = eng.start_rows(items)
                 except Exception as e:  # noqa: BLE001
                     log.logger.exception("batched admission failed")
                     for p in gens:
@@ -155,25 +192,41 @@ class Batcher:
                         live.append((p, row))
             # -- one decode step for everything in flight
             if live:
-                try:
-                    self.engine.step_rows([row for _, row in live])
-                except Exception as e:  # noqa: BLE001
-                    log.logger.exception("decode step failed")
-                    # retire the rows or their KV slots leak permanently
-                    # and admission wedges
-                    self.engine.abort_rows([row for _, row in live])
-                    for p, _ in live:
-                        if not p.future.done():
-                            p.future.set_exception(e)
-                    live = []
-                still: list[tuple[_Pending, object]] = []
-                for p, row in live:
-                    if row.done:
-                        p.future.set_result(row.result)
-                    else:
-                        still.append((p, row))
-                live = still
-            if stop_now or (self._stop.is_set() and not live):
+                rows = [row for _, row in live]
+                if eng.pipeline_ok(rows):
+                    # pipelined: submit step N, then process step N-1's
+                    # tokens on the host while N runs on the GPU
+                    try:
+                        nxt = eng.submit_step(rows, prev=inflight)
+                    except Exception as e:  # noqa: BLE001
+                        log.logger.exception("pipelined submit failed")
+                        drain_inflight()
+                        fail_live(e)
+                        nxt = None
+                    if inflight is not None:
+                        try:
+                            eng.collect_step(inflight)
+                        except Exception as e:  # noqa: BLE001
+                            log.logger.exception("pipelined collect failed")
+                            fail_live(e)
+                            nxt = None
+                    inflight = nxt
+                    prune()
+                else:
+                    drain_inflight()
+                    try:
+                        eng.step_rows([row for _, row in live])
+                    except Exception as e:  # noqa: BLE001
+                        log.logger.exception("decode step failed")
+                        # retire the rows or their KV slots leak
+                        # permanently and admission wedges
+                        fail_live(e)
+                    prune()
+            elif inflight is not None:
+                drain_inflight()
+            if stop_now or (self._stop.is_set() and not live
+                            and inflight is None):
+                drain_inflight()
                 for p, _ in live:
                     if not p.future.done():
                         p.future.set_exception(

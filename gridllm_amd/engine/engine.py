@@ -617,6 +617,14 @@ class InferenceEngine:
             for sl in inflight["free_after"]:
                 self.model.cache.free(sl)
 
+    def flush_deferred(self) -> None:
+        """Free retire-deferred slots once NO step is in flight (pipeline
+        drain / fallback to the synchronous path)."""
+        with self._lock:
+            for sl in getattr(self, "_deferred_free", []):
+                self.model.cache.free(sl)
+            self._deferred_free = []
+
     def _sync_cancel(self, rows: Sequence[_Row]) -> None:
         """Under TP every rank must retire the same rows at the same step, or
         the next collective deadlocks (SURVEY.md §7 hard-part 5). Rank 0 owns
@@ -696,7 +704,7 @@ class InferenceEngine:
             return False
         return True
 
-    def _retire(self, row: _Row) -> None:
+    def _retire(self, row: _Row, defer_free: bool = False) -> None:
         tail = row.decoder.flush() if hasattr(row.decoder, "flush") else ""
         final = row.text_tail + tail
         if final and not row.result.cancelled:
@@ -710,7 +718,14 @@ class InferenceEngine:
         row.done = True
         if row.on_token:
             row.on_token("", True)
-        self.model.cache.free(row.slot)
+        if defer_free:
+            # pipelined retire: the slot is still referenced by the step
+            # currently in flight — free only after that step collects
+            if not hasattr(self, "_deferred_free"):
+                self._deferred_free = []
+            self._deferred_free.append(row.slot)
+        else:
+            self.model.cache.free(row.slot)
 
     # ------------------------------------------------------------- metadata
 
