@@ -225,6 +225,9 @@ def main():
     ap.add_argument("--max-seq", type=int, default=1024)
     ap.add_argument("--workers", type=int, default=1,
                     help="DP workers; worker i pins cuda:i")
+    ap.add_argument("--same-device", action="store_true",
+                    help="all DP workers share --device (single-GPU DP "
+                         "validation, BASELINE config 3 on one lease)")
     args = ap.parse_args()
 
     import torch
@@ -244,7 +247,7 @@ def main():
     for i in range(args.workers):
         wenv = dict(env)
         wenv["LT_WID"] = f"lt-worker-{i}"
-        if args.device.startswith("cuda"):
+        if args.device.startswith("cuda") and not args.same_device:
             wenv["LT_DEVICE"] = f"cuda:{i}"
         print(f"spawning worker {i}...", file=sys.stderr, flush=True)
         workers.append(_spawn(WORKER_SRC, wenv, "WORKER_READY"))
