@@ -60,10 +60,14 @@ _reg(ModelSpec("llama3:70b", hidden=8192, layers=80, heads=64, kv_heads=8,
                head_dim=128, intermediate=28672, vocab=128256,
                tie_embeddings=False),
      "llama3.1:70b", "llama3-70b")
-# tiny config for CPU tests and CI
+# tiny configs for CPU tests and CI (tiny2: a SECOND architecture so
+# mixed-fleet routing tests have two distinct models on CPU)
 _reg(ModelSpec("tiny", hidden=256, layers=2, heads=4, kv_heads=2, head_dim=64,
                intermediate=512, vocab=512, max_seq_len=256),
      "tiny:latest")
+_reg(ModelSpec("tiny2", hidden=128, layers=2, heads=2, kv_heads=2,
+               head_dim=64, intermediate=256, vocab=512, max_seq_len=256),
+     "tiny2:latest")
 
 
 def get_spec(name: str, max_seq_len: int = 0) -> ModelSpec:
