@@ -332,7 +332,11 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     const int Hq = Hk * G;
     for (int i = threadIdx.x; i < G * D; i += 256) {
         int g = i / D, d = i % D;
-        q_lds[i] = bf2f(q[((long)b * Hq + hk * (long)G + g) * D + d]) * scale;
+        // log2-domain softmax: fold log2(e) into the q prescale so every
+        // exp becomes a native v_exp_f32 exp2 (identical values:
+        // exp2((x-m)*log2e) == exp(x-m))
+        q_lds[i] = bf2f(q[((long)b * Hq + hk * (long)G + g) * D + d])
+                   * (scale * 1.4426950408889634f);
     }
     __syncthreads();
 
@@ -389,11 +393,13 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
             float bmax = wave_reduce_max(sv);
             float mn = fmaxf(m[g], bmax);
             // guard: all-invalid block -> bmax=-inf, mn may stay -inf
-            float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
-            float p = (sv == -INFINITY) ? 0.f : __expf(sv - mn);
+            float alpha = (m[g] == -INFINITY) ? 0.f : exp2f(m[g] - mn);
+            float p = (sv == -INFINITY) ? 0.f : exp2f(sv - mn);
             l[g] = l[g] * alpha + wave_reduce_sum(p);
+            if (mn != m[g]) {          // defer-rescale: skip when max held
 #pragma unroll
-            for (int i = 0; i < 8; ++i) facc[g][i] *= alpha;
+                for (int i = 0; i < 8; ++i) facc[g][i] *= alpha;
+            }
             m[g] = mn;
             red[wid][g * 64 + lane] = p;      // per-wave scratch: p for 64 keys
         }
@@ -438,7 +444,7 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     for (int g = 0; g < G; ++g) {
         float M = fmaxf(fmaxf(wml[0][g][0], wml[1][g][0]),
                         fmaxf(wml[2][g][0], wml[3][g][0]));
-        float f = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - M);
+        float f = (m[g] == -INFINITY) ? 0.f : exp2f(m[g] - M);
         if (lane < DV)
 #pragma unroll
             for (int i = 0; i < 8; ++i)
@@ -504,7 +510,7 @@ k_attn_decode1(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     const int Hq = Hk * G;
     for (int i = lane; i < G * D; i += 64)
         q_lds[i] = bf2f(q[((long)b * Hq + hk * (long)G + i / D) * D
-                          + i % D]) * scale;
+                          + i % D]) * (scale * 1.4426950408889634f);
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     float m[G], l[G];
@@ -552,11 +558,13 @@ k_attn_decode1(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
             float sv = valid ? s[g] : -INFINITY;
             float bmax = wave_reduce_max(sv);
             float mn = fmaxf(m[g], bmax);
-            float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
-            float p = (sv == -INFINITY) ? 0.f : __expf(sv - mn);
+            float alpha = (m[g] == -INFINITY) ? 0.f : exp2f(m[g] - mn);
+            float p = (sv == -INFINITY) ? 0.f : exp2f(sv - mn);
             l[g] = l[g] * alpha + wave_reduce_sum(p);
+            if (mn != m[g]) {          // defer-rescale: skip when max held
 #pragma unroll
-            for (int i = 0; i < 8; ++i) facc[g][i] *= alpha;
+                for (int i = 0; i < 8; ++i) facc[g][i] *= alpha;
+            }
             m[g] = mn;
             p_lds[g * 64 + lane] = p;
         }
@@ -625,7 +633,7 @@ k_attn_combine(bf16_t* __restrict__ out, const float* __restrict__ ws,
     for (int s = 0; s < nsplit; ++s) {
         const float* p = base + s * (long)(D + 2);
         float ms = p[D];
-        float f = (ms == -INFINITY) ? 0.f : __expf(ms - M);
+        float f = (ms == -INFINITY) ? 0.f : exp2f(ms - M);
         num += f * p[d];
         den += f * p[D + 1];
     }
@@ -1124,20 +1132,21 @@ attn_prefill_tile(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
 
         // ---- mask + online softmax (lane owns one q-row) ---------------
         float sv[16], tmax = -INFINITY;
+        const float scl2 = scale * 1.4426950408889634f;   // log2 domain
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
             const int kg = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
             const bool ok = (kg < s_total) && (kg <= q_abs);
-            sv[r] = ok ? s_acc[r] * scale : -INFINITY;
+            sv[r] = ok ? s_acc[r] * scl2 : -INFINITY;
             tmax = fmaxf(tmax, sv[r]);
         }
         tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
         const float mn = fmaxf(m, tmax);
-        const float alpha = (m == -INFINITY) ? 0.f : __expf(m - mn);
+        const float alpha = (m == -INFINITY) ? 0.f : exp2f(m - mn);
         float p[16], psum = 0.f;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-            p[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mn);
+            p[r] = (sv[r] == -INFINITY) ? 0.f : exp2f(sv[r] - mn);
             psum += p[r];
         }
         psum += __shfl_xor(psum, 32, WAVE);
