@@ -369,6 +369,7 @@ class SlotSampler:
 
     def __init__(self, n_slots: int, device):
         self.device = device
+        self._up = None
         self.temps = torch.full((n_slots,), 1.0, device=device)
         self.nongreedy = torch.zeros(n_slots, dtype=torch.int64,
                                      device=device)
@@ -397,8 +398,11 @@ class SlotSampler:
             ints[0, i] = r.slot
             ints[1, i] = s.noise_seed & ((1 << 63) - 1)
             ints[2, i] = s.draws
-        t = torch.from_numpy(host).to(self.device)
-        it = torch.from_numpy(ints).to(self.device)
+        if self._up is None:
+            from gridllm_amd.utils.pinned import PinnedUploader
+            self._up = PinnedUploader(self.device)
+        t = self._up.upload(host)
+        it = self._up.upload(ints)
         sl = it[0]
         self.temps[sl] = t[0].float()
         self.nongreedy[sl] = t[1].long()

@@ -120,6 +120,8 @@ class LlamaModel:
         self.cache = KVCache(spec, kv_slots, self.max_seq, self.hkv_l,
                              self.device, self.dtype if self.dtype != torch.float32
                              else torch.float32)
+        from gridllm_amd.utils.pinned import PinnedUploader
+        self.up = PinnedUploader(self.device)   # stall-free H2D staging
 
     # -- weights -----------------------------------------------------------
 
@@ -367,7 +369,7 @@ class LlamaModel:
         slot_np = np.concatenate(
             [sl * self.cache.max_seq + np.arange(ln, dtype=np.int32)
              for ln, sl in zip(lens, slots)])
-        idx = torch.from_numpy(np.stack([pos_np, slot_np])).to(self.device)
+        idx = self.up.upload(np.stack([pos_np, slot_np]))
         pos, slot_t = idx[0], idx[1]
         # varlen batched attention (one meta-driven kernel launch per layer
         # instead of a per-sequence Python loop — the round-2 profile's
@@ -378,8 +380,7 @@ class LlamaModel:
             rows = [(bounds[i], ln, sl, q0)
                     for i, (ln, sl) in enumerate(zip(lens, slots))
                     for q0 in range(0, ln, 128)]
-            meta = torch.from_numpy(
-                np.asarray(rows, dtype=np.int32)).to(self.device)
+            meta = self.up.upload(np.asarray(rows, dtype=np.int32))
         x = self.embed[tokens]
         resid = x
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
@@ -414,7 +415,7 @@ class LlamaModel:
         import numpy as _np
         host = _np.asarray([[sl for sl in slots], lens,
                             [b - 1 for b in bounds[1:]]], dtype=_np.int64)
-        dev3 = torch.from_numpy(host).to(self.device)
+        dev3 = self.up.upload(host)
         self.cache.seq_lens[dev3[0]] = dev3[1].to(torch.int32)
         # fresh slots start unshifted (alloc no longer touches the device)
         self.cache.rope_off.index_fill_(0, dev3[0], 0)
