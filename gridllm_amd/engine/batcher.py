@@ -47,7 +47,7 @@ class Batcher:
     whole waves and runs each to completion."""
 
     def __init__(self, engine: InferenceEngine, window_ms: float = 4.0,
-                 continuous: bool = True, admit_per_step: int = 32):
+                 continuous: bool = True, admit_per_step: int = 64):
         self.engine = engine
         self.window_s = window_ms / 1000.0
         self.continuous = continuous and engine.tp.size == 1
