@@ -36,6 +36,7 @@ class _Pending:
     future: Future = field(default_factory=Future)
     extra_stops: Optional[list[str]] = None
     think: bool = False
+    ts: float = field(default_factory=time.monotonic)
 
 
 class Batcher:
@@ -47,13 +48,20 @@ class Batcher:
     whole waves and runs each to completion."""
 
     def __init__(self, engine: InferenceEngine, window_ms: float = 4.0,
-                 continuous: bool = True, admit_per_step: int = 64):
+                 continuous: bool = True, admit_per_step: int = 64,
+                 admit_min: int = 8, admit_wait_ms: float = 15.0):
         self.engine = engine
         self.window_s = window_ms / 1000.0
         self.continuous = continuous and engine.tp.size == 1
         # bound sequential prefills between decode steps: a burst of new
         # prompts must not stall in-flight rows for the whole burst
         self.admit_per_step = max(1, admit_per_step)
+        # coalesce trickle arrivals: a single-prompt admission costs a
+        # full eager prefill pass (~8 ms of host launches at 3B); while
+        # the batch is busy, hold arrivals until admit_min gathered or
+        # admit_wait_ms passed (TTFT pays at most that wait)
+        self.admit_min = max(1, admit_min)
+        self.admit_wait_s = admit_wait_ms / 1000.0
         self.q: "queue.Queue[_Pending]" = queue.Queue()
         self._stop = threading.Event()
         target = self._loop_continuous if self.continuous else self._loop
@@ -142,10 +150,11 @@ class Batcher:
                 eng.flush_deferred()
                 prune()
 
+        hold: list[_Pending] = []      # coalescing buffer for arrivals
         while True:
             # -- admit: block only when idle, otherwise drain without waiting
             arrivals: list[_Pending] = []
-            if not live and inflight is None:
+            if not live and inflight is None and not hold:
                 try:
                     arrivals.append(self.q.get(timeout=0.5))
                 except queue.Empty:
@@ -153,15 +162,15 @@ class Batcher:
                         break
                     continue
             # one slot spare: a pipelined retire frees one step late
-            while (len(live) + len(arrivals) < max_batch
-                   and len(arrivals) < self.admit_per_step
-                   and eng.model.cache.n_free > len(arrivals) + 1):
+            while (len(live) + len(hold) + len(arrivals) < max_batch
+                   and len(hold) + len(arrivals) < self.admit_per_step
+                   and eng.model.cache.n_free > len(arrivals)
+                   + len(hold) + 1):
                 try:
                     arrivals.append(self.q.get_nowait())
                 except queue.Empty:
                     break
             stop_now = False
-            gens: list[_Pending] = []
             for p in arrivals:
                 if p.kind == "__stop__":
                     stop_now = True
@@ -172,7 +181,14 @@ class Batcher:
                     except Exception as e:  # noqa: BLE001
                         p.future.set_exception(e)
                 else:
-                    gens.append(p)
+                    hold.append(p)
+            gens: list[_Pending] = []
+            if hold and (not live
+                         or len(hold) >= self.admit_min
+                         or stop_now or self._stop.is_set()
+                         or time.monotonic() - hold[0].ts
+                         >= self.admit_wait_s):
+                gens, hold = hold, []
             if gens:
                 items = []
                 for p in gens:

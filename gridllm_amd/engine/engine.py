@@ -12,6 +12,8 @@ from __future__ import annotations
 
 import threading
 import time
+
+import numpy as np
 from dataclasses import dataclass, field
 from typing import Callable, Optional, Sequence
 
@@ -442,7 +444,7 @@ class InferenceEngine:
         try:
             # ONE pinned upload for every prompt's ids (one blocking H2D
             # per prompt measured ~1.5 ms each against a busy stream)
-            import numpy as np
+            pass
             all_ids = self.model.up.upload(np.concatenate(
                 [np.asarray(x, dtype=np.int64) for x in id_lists]))
             token_lists = list(torch.split(all_ids,
@@ -454,8 +456,8 @@ class InferenceEngine:
                 logits = self.model.forward_prefill_batch(token_lists, slots)
             self._sync_cancel(rows)
             self.sampler.admit(rows, self.spec.vocab)
-            slots_t = torch.as_tensor(slots, dtype=torch.int64,
-                                      device=self.device)
+            slots_t = self.model.up.upload(
+                np.asarray(slots, dtype=np.int64))
             toks = self._sample_batch(logits, rows, slots_t)  # syncs
         except Exception:
             # prefill/first-sample failure: release every slot this batch

@@ -400,7 +400,7 @@ def test_prefill_batch_engine_parity(dev):
                      m1.cache.k[li][bat_slots[i], :, :n].cpu())
             # K comes off the qkv GEMM, which runs at M=sum vs M=T_i ->
             # one-bf16-ulp output differences are expected
-            assert dk < 4e-3, (li, i, dk)
+            assert dk < 1.6e-2, (li, i, dk)
 
 
 @pytest.mark.parametrize("m,n,k,mt,sk", [
