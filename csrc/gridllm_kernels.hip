@@ -371,12 +371,17 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
 #pragma unroll
         for (int g = 0; g < G; ++g) s[g] = 0.f;
         if (valid) {
+            // burst-load the whole K row first (PMC: 78% SQ_WAIT —
+            // same memory-level-parallelism lever as k_attn_decode1)
             const bf16_t* krow = kbase + (long)j * D;
+            uint4 kvv[D / 8];
+#pragma unroll
+            for (int c = 0; c < D / 8; ++c)
+                kvv[c] = *(const uint4*)(krow + c * 8);
 #pragma unroll
             for (int dd = 0; dd < D; dd += 8) {
-                uint4 kv = *(const uint4*)(krow + dd);
                 float kf[8];
-                unpack8(kv, kf);
+                unpack8(kvv[dd / 8], kf);
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
                     float a = 0.f;
@@ -405,12 +410,25 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         }
         // wave-local LDS visibility (no cross-wave traffic here)
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        // accumulate P·V: RPI rows per iteration, one b128 load per lane
+        // accumulate P·V: RPI rows per iteration, one b128 load per
+        // lane, 4-deep load pipeline (SQ_WAIT lever)
         const int jmax = min(64, s1 - kb);
-        for (int jb = 0; jb < jmax; jb += RPI) {
-            const int j2 = jb + rg;
-            const uint4 vv = *(const uint4*)(
+        const int iters = (jmax + RPI - 1) / RPI;
+        uint4 vbuf[4];
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+            const int j2 = k * RPI + rg;
+            vbuf[k] = *(const uint4*)(
                 vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
+        }
+        for (int it = 0; it < iters; ++it) {
+            const uint4 vv = vbuf[it & 3];
+            if (it + 4 < iters) {
+                const int j2 = (it + 4) * RPI + rg;
+                vbuf[it & 3] = *(const uint4*)(
+                    vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
+            }
+            const int j2 = it * RPI + rg;
             float vf[8];
             unpack8(vv, vf);
 #pragma unroll
