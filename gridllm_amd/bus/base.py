@@ -59,7 +59,8 @@ class Bus(ABC):
     async def kv_del(self, key: str) -> None: ...
 
     @abstractmethod
-    async def hset(self, key: str, field: str, value: str) -> None: ...
+    async def hset(self, key: str, field: str, value: str,
+               ack: bool = True) -> None: ...
 
     @abstractmethod
     async def hget(self, key: str, field: str) -> Optional[str]: ...

@@ -101,7 +101,8 @@ class MemoryBus(Bus):
     async def kv_del(self, key: str) -> None:
         self.core.kv.pop(key, None)
 
-    async def hset(self, key: str, field: str, value: str) -> None:
+    async def hset(self, key: str, field: str, value: str,
+                   ack: bool = True) -> None:
         self.core.hashes[key][field] = value
 
     async def hget(self, key: str, field: str) -> Optional[str]:

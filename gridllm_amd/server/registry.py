@@ -145,7 +145,10 @@ class WorkerRegistry:
         await self._persist(info)
 
     async def _persist(self, info: WorkerInfo) -> None:
-        await self.bus.hset(Keys.WORKERS, info.id, json.dumps(info.to_dict()))
+        # fire-and-forget: runs on every job-count flip (2x per job) —
+        # an awaited RTT here serializes the assignment chain
+        await self.bus.hset(Keys.WORKERS, info.id,
+                            json.dumps(info.to_dict()), ack=False)
 
     # --------------------------------------------------------------- sweeps
 
