@@ -266,7 +266,7 @@ class TcpBus(Bus):
             await self._call("identify", worker_id=wid)
         for ch in list(self._handlers):
             if self._handlers[ch]:
-                await self._call("subscribe", channel=ch)
+                await self._send_nf("subscribe", channel=ch)
 
     async def close(self) -> None:
         self.closed = True
@@ -345,11 +345,22 @@ class TcpBus(Bus):
                                   "payload": payload}))
         await self._writer.drain()
 
+    async def _send_nf(self, op: str, **kw) -> None:
+        if self._writer is None or self.closed:
+            raise ConnectionError("bus not connected")
+        self._writer.write(_pack({"op": op, **kw}))
+        await self._writer.drain()
+
     async def subscribe(self, channel: str, handler: Handler) -> None:
+        """No ack round trip: frames on one connection are processed in
+        order by the bus, so anything published AFTER this call (on this
+        connection) is observed by the subscription — which is the only
+        ordering the submit-then-await pattern needs. The awaited ack
+        cost 2 RTTs per request at the gateway (subscribe result+stream)."""
         first = not self._handlers[channel]
         self._handlers[channel].append(handler)
         if first:
-            await self._call("subscribe", channel=channel)
+            await self._send_nf("subscribe", channel=channel)
 
     async def unsubscribe(self, channel: str,
                           handler: Optional[Handler] = None) -> None:
@@ -360,7 +371,7 @@ class TcpBus(Bus):
                 h for h in self._handlers[channel] if h is not handler]
         if not self._handlers[channel]:
             try:
-                await self._call("unsubscribe", channel=channel)
+                await self._send_nf("unsubscribe", channel=channel)
             except ConnectionError:
                 pass
 
