@@ -371,17 +371,12 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
 #pragma unroll
         for (int g = 0; g < G; ++g) s[g] = 0.f;
         if (valid) {
-            // burst-load the whole K row first (PMC: 78% SQ_WAIT —
-            // same memory-level-parallelism lever as k_attn_decode1)
             const bf16_t* krow = kbase + (long)j * D;
-            uint4 kvv[D / 8];
-#pragma unroll
-            for (int c = 0; c < D / 8; ++c)
-                kvv[c] = *(const uint4*)(krow + c * 8);
 #pragma unroll
             for (int dd = 0; dd < D; dd += 8) {
+                uint4 kv = *(const uint4*)(krow + dd);
                 float kf[8];
-                unpack8(kvv[dd / 8], kf);
+                unpack8(kv, kf);
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
                     float a = 0.f;
@@ -410,25 +405,12 @@ k_attn_decode(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         }
         // wave-local LDS visibility (no cross-wave traffic here)
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        // accumulate P·V: RPI rows per iteration, one b128 load per
-        // lane, 4-deep load pipeline (SQ_WAIT lever)
+        // accumulate P·V: RPI rows per iteration, one b128 load per lane
         const int jmax = min(64, s1 - kb);
-        const int iters = (jmax + RPI - 1) / RPI;
-        uint4 vbuf[4];
-#pragma unroll
-        for (int k = 0; k < 4; ++k) {
-            const int j2 = k * RPI + rg;
-            vbuf[k] = *(const uint4*)(
+        for (int jb = 0; jb < jmax; jb += RPI) {
+            const int j2 = jb + rg;
+            const uint4 vv = *(const uint4*)(
                 vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
-        }
-        for (int it = 0; it < iters; ++it) {
-            const uint4 vv = vbuf[it & 3];
-            if (it + 4 < iters) {
-                const int j2 = (it + 4) * RPI + rg;
-                vbuf[it & 3] = *(const uint4*)(
-                    vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
-            }
-            const int j2 = it * RPI + rg;
             float vf[8];
             unpack8(vv, vf);
 #pragma unroll
@@ -555,19 +537,12 @@ k_attn_decode1(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
 #pragma unroll
         for (int g = 0; g < G; ++g) s[g] = 0.f;
         if (valid) {
-            // burst-load the WHOLE K row before any use: PMC showed 63%
-            // SQ_WAIT at 8 blocks/CU — interleaved load-use chains expose
-            // one load latency per 16 B; a 16-wide burst exposes one per
-            // row (memory-level parallelism is the only lever here)
             const bf16_t* krow = kbase + (long)j * D;
-            uint4 kvv[D / 8];
-#pragma unroll
-            for (int c = 0; c < D / 8; ++c)
-                kvv[c] = *(const uint4*)(krow + c * 8);
 #pragma unroll
             for (int dd = 0; dd < D; dd += 8) {
+                uint4 kv = *(const uint4*)(krow + dd);
                 float kf[8];
-                unpack8(kvv[dd / 8], kf);
+                unpack8(kv, kf);
 #pragma unroll
                 for (int g = 0; g < G; ++g) {
                     float a = 0.f;
@@ -596,25 +571,12 @@ k_attn_decode1(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         // P.V: RPI rows per b128 iteration; p read from same-wave LDS
         // (an LDS hop, not shfl: it bounds how far the scheduler can
-        // interleave the K and V phases — shfl spilled 1.5 KB here).
-        // 4-deep software pipeline on the V loads (same SQ_WAIT lever).
+        // interleave the K and V phases — shfl spilled 1.5 KB here)
         const int jmax = min(64, s1 - kb);
-        const int iters = (jmax + RPI - 1) / RPI;
-        uint4 vbuf[4];
-#pragma unroll
-        for (int k = 0; k < 4; ++k) {
-            const int j2 = k * RPI + rg;
-            vbuf[k] = *(const uint4*)(
+        for (int jb = 0; jb < jmax; jb += RPI) {
+            const int j2 = jb + rg;
+            const uint4 vv = *(const uint4*)(
                 vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
-        }
-        for (int it = 0; it < iters; ++it) {
-            const uint4 vv = vbuf[it & 3];
-            if (it + 4 < iters) {
-                const int j2 = (it + 4) * RPI + rg;
-                vbuf[it & 3] = *(const uint4*)(
-                    vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
-            }
-            const int j2 = it * RPI + rg;
             float vf[8];
             unpack8(vv, vf);
 #pragma unroll
