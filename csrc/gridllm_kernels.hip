@@ -1358,130 +1358,6 @@ __global__ __launch_bounds__(256, 4) void k_sample_gumbel(
 #define LAUNCH_CHECK() do { hipError_t e = hipGetLastError(); \
     if (e != hipSuccess) return (int)e; } while (0)
 
-// ===========================================================================
-// Head-parallel one-wave decode: one wave per (b, q-HEAD) — G times the
-// blocks of k_attn_decode1, each doing 1/G of its VALU work. PMC showed
-// the one-wave kernel 63% SQ_WAIT at 2 waves/SIMD (grid-limited): the
-// fix consistent with the counters is MORE independent waves, not deeper
-// per-wave load queues (that variant measured 40% slower — r02_attn_pmc).
-// Block id layout g*(B*Hk*nsplit) + inner keeps a (b,hk)'s G waves at ids
-// ≡ same value mod 8 whenever B*Hk*nsplit % 8 == 0, so they land on the
-// SAME XCD (round-robin dispatch) and the 2nd..G-th waves read K/V from
-// that XCD's L2 instead of HBM.
-// ===========================================================================
-template <int D>
-__global__ void __launch_bounds__(64, 4)
-k_attn_decode1h(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
-                const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
-                const int* __restrict__ seqlen,
-                const int* __restrict__ slot_map,
-                float* __restrict__ ws,
-                int B, int Hk, int G, int Smax, int nsplit, float scale) {
-    const int inner = blockIdx.x % (B * Hk * nsplit);
-    const int g = blockIdx.x / (B * Hk * nsplit);
-    const int split = inner % nsplit;
-    const int hk = (inner / nsplit) % Hk;
-    const int b = inner / (nsplit * Hk);
-    const int bc = slot_map ? slot_map[b] : b;
-    const int lane = threadIdx.x;
-
-    __shared__ float q_lds[D];
-    __shared__ float p_lds[64];
-
-    const int seq = seqlen[b];
-    const int chunks = (seq + 63) >> 6;
-    const int per = (chunks + nsplit - 1) / nsplit;
-    const int s0 = (split * per) << 6;
-    const int s1 = min(seq, (split * per + per) << 6);
-
-    const int Hq = Hk * G;
-    for (int i = lane; i < D; i += 64)
-        q_lds[i] = bf2f(q[((long)b * Hq + hk * (long)G + g) * D + i])
-                   * (scale * 1.4426950408889634f);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-
-    float m = -INFINITY, l = 0.f;
-    constexpr int DV = D / 8;
-    constexpr int RPI = 64 / DV;
-    float facc[8];
-#pragma unroll
-    for (int i = 0; i < 8; ++i) facc[i] = 0.f;
-    const int rg = lane / DV;
-    const int d8 = (lane % DV) * 8;
-
-    const bf16_t* kbase = kc + ((long)bc * Hk + hk) * (long)Smax * D;
-    const bf16_t* vbase = vc + ((long)bc * Hk + hk) * (long)Smax * D;
-
-    for (int kb = s0; kb < s1; kb += 64) {
-        const int j = kb + lane;
-        const bool valid = j < s1;
-        float s = 0.f;
-        if (valid) {
-            const bf16_t* krow = kbase + (long)j * D;
-#pragma unroll
-            for (int dd = 0; dd < D; dd += 8) {
-                uint4 kv = *(const uint4*)(krow + dd);
-                float kf[8];
-                unpack8(kv, kf);
-#pragma unroll
-                for (int i = 0; i < 8; ++i)
-                    s += kf[i] * q_lds[dd + i];
-            }
-        }
-        float sv = valid ? s : -INFINITY;
-        float bmax = wave_reduce_max(sv);
-        float mn = fmaxf(m, bmax);
-        float alpha = (m == -INFINITY) ? 0.f : exp2f(m - mn);
-        float p = (sv == -INFINITY) ? 0.f : exp2f(sv - mn);
-        l = l * alpha + wave_reduce_sum(p);
-        if (mn != m) {
-#pragma unroll
-            for (int i = 0; i < 8; ++i) facc[i] *= alpha;
-        }
-        m = mn;
-        p_lds[lane] = p;
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        const int jmax = min(64, s1 - kb);
-        for (int jb = 0; jb < jmax; jb += RPI) {
-            const int j2 = jb + rg;
-            const uint4 vv = *(const uint4*)(
-                vbase + (long)(kb + min(j2, jmax - 1)) * D + d8);
-            float vf[8];
-            unpack8(vv, vf);
-            const float pp = (j2 < jmax) ? p_lds[j2] : 0.f;
-#pragma unroll
-            for (int i = 0; i < 8; ++i)
-                facc[i] = fmaf(pp, vf[i], facc[i]);
-        }
-    }
-
-#pragma unroll
-    for (int i = 0; i < 8; ++i)
-#pragma unroll
-        for (int off = 32; off >= DV; off >>= 1)
-            facc[i] += __shfl_xor(facc[i], off, WAVE);
-
-    if (lane < DV) {
-        const int hq = hk * G + g;
-        if (nsplit == 1) {
-            const float L = l;
-#pragma unroll
-            for (int i = 0; i < 8; ++i)
-                out[((long)b * Hq + hq) * D + d8 + i] =
-                    f2bf(L > 0.f ? facc[i] / L : 0.f);
-        } else {
-            float* wsp = ws + (((long)b * Hq + hq) * nsplit + split)
-                         * (D + 2);
-#pragma unroll
-            for (int i = 0; i < 8; ++i) wsp[d8 + i] = facc[i];
-            if (lane == 0) {
-                wsp[D] = m;
-                wsp[D + 1] = l;
-            }
-        }
-    }
-}
-
 template <int D>
 static int launch_attn_decode(hipStream_t st, bf16_t* out, const bf16_t* q,
                               const bf16_t* kc, const bf16_t* vc,
@@ -1489,21 +1365,10 @@ static int launch_attn_decode(hipStream_t st, bf16_t* out, const bf16_t* q,
                               float* ws, int B, int Hk, int G, int Smax,
                               int nsplit, float scale) {
     dim3 grid((unsigned)(B * Hk * nsplit));
-    // one-wave variants only on the unsplit path (the auto-nsplit rule
+    // one-wave variant only on the unsplit path (the auto-nsplit rule
     // yields nsplit==1 whenever B*Hk >= 512, so this is the common case;
-    // explicit nsplit>1 calls keep the 4-wave cooperative kernel).
-    // Head-parallel (one wave per q-head, G x blocks) when the XCD
-    // co-location condition holds — its grid layout keeps a (b,hk)'s
-    // G waves on one XCD so K/V re-reads stay in that L2.
+    // explicit nsplit>1 calls keep the 4-wave cooperative kernel)
     const bool onewave = nsplit == 1 && (B * Hk) >= 2048;
-    const bool headpar = onewave && G > 1 && ((B * Hk * nsplit) % 8 == 0);
-    if (headpar) {
-        dim3 hgrid((unsigned)(B * Hk * nsplit * G));
-        k_attn_decode1h<D><<<hgrid, dim3(64), 0, st>>>(
-            out, q, kc, vc, seqlen, slot_map, ws, B, Hk, G, Smax, nsplit,
-            scale);
-        return 0;
-    }
     switch (G) {   // compile-time G keeps softmax state in registers
 #define CASE_G(GG) \
         case GG: if (onewave) k_attn_decode1<D, GG><<<grid, dim3(64), 0, st>>>( \
