@@ -1065,8 +1065,7 @@ attn_prefill_tile(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
                   bf16_t* vt, bf16_t* plds_all) {
     constexpr int NS = D / 16;     // 16-wide d slices (QK^T fragments)
     constexpr int NT = D / 32;     // 32-wide d tiles (output / V^T)
-    constexpr int VST = 72;        // padded k-stride of the 64-row V^T
-                                   // tile (144 B = odd multiple of 16 B)
+    constexpr int VST = 40;        // padded k-stride of V^T tile (16B-mult)
 
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -1099,15 +1098,12 @@ attn_prefill_tile(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
     // causal upper bound for the workgroup's LAST q-row
     const int kmax = min(s_total, causal_off + q0 + 128);
 
-    for (int k0 = 0; k0 < kmax; k0 += 64) {
-        // KVBLK=64 (round-2 ladder): one barrier pair, one softmax
-        // rescale and one P hand-off per 64 keys instead of per 32 —
-        // twice the MFMA work between synchronization episodes.
+    for (int k0 = 0; k0 < kmax; k0 += 32) {
         // ---- stage V^T tile (cooperative transpose) --------------------
         __syncthreads();               // previous tile's vt readers done
         {
             constexpr int CPR = D / 8;            // b128 chunks per V row
-            for (int c = threadIdx.x; c < 64 * CPR; c += 256) {
+            for (int c = threadIdx.x; c < 32 * CPR; c += 256) {
                 const int kr = c / CPR;
                 const int d8 = (c % CPR) * 8;
                 uint4 v = *(const uint4*)(
@@ -1119,48 +1115,39 @@ attn_prefill_tile(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         }
         __syncthreads();
 
-        // ---- S^T = K·Q^T for BOTH 32-row halves ------------------------
-        f32x16 s_acc0, s_acc1;
+        // ---- S^T = K·Q^T (K fragments straight from HBM/L2) ------------
+        f32x16 s_acc;
 #pragma unroll
-        for (int r = 0; r < 16; ++r) { s_acc0[r] = 0.f; s_acc1[r] = 0.f; }
+        for (int r = 0; r < 16; ++r) s_acc[r] = 0.f;
         {
-            const bf16_t* krow0 =
+            const bf16_t* krow =
                 kbase + (long)min(k0 + ln, s_total - 1) * D + hi * 8;
-            const bf16_t* krow1 =
-                kbase + (long)min(k0 + 32 + ln, s_total - 1) * D + hi * 8;
 #pragma unroll
             for (int s = 0; s < NS; ++s) {
-                const bf16x8 kf0 = *(const bf16x8*)(krow0 + s * 16);
-                const bf16x8 kf1 = *(const bf16x8*)(krow1 + s * 16);
-                s_acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    kf0, qf[s], s_acc0, 0, 0, 0);
-                s_acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    kf1, qf[s], s_acc1, 0, 0, 0);
+                const bf16x8 kf = *(const bf16x8*)(krow + s * 16);
+                s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    kf, qf[s], s_acc, 0, 0, 0);
             }
         }
 
-        // ---- mask + online softmax over 64 keys (lane owns one q-row) --
-        float sv0[16], sv1[16], tmax = -INFINITY;
+        // ---- mask + online softmax (lane owns one q-row) ---------------
+        float sv[16], tmax = -INFINITY;
         const float scl2 = scale * 1.4426950408889634f;   // log2 domain
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-            const int crow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-            const bool ok0 = (k0 + crow < s_total) && (k0 + crow <= q_abs);
-            const bool ok1 = (k0 + 32 + crow < s_total)
-                             && (k0 + 32 + crow <= q_abs);
-            sv0[r] = ok0 ? s_acc0[r] * scl2 : -INFINITY;
-            sv1[r] = ok1 ? s_acc1[r] * scl2 : -INFINITY;
-            tmax = fmaxf(tmax, fmaxf(sv0[r], sv1[r]));
+            const int kg = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            const bool ok = (kg < s_total) && (kg <= q_abs);
+            sv[r] = ok ? s_acc[r] * scl2 : -INFINITY;
+            tmax = fmaxf(tmax, sv[r]);
         }
         tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
         const float mn = fmaxf(m, tmax);
         const float alpha = (m == -INFINITY) ? 0.f : exp2f(m - mn);
-        float p0[16], p1[16], psum = 0.f;
+        float p[16], psum = 0.f;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-            p0[r] = (sv0[r] == -INFINITY) ? 0.f : exp2f(sv0[r] - mn);
-            p1[r] = (sv1[r] == -INFINITY) ? 0.f : exp2f(sv1[r] - mn);
-            psum += p0[r] + p1[r];
+            p[r] = (sv[r] == -INFINITY) ? 0.f : exp2f(sv[r] - mn);
+            psum += p[r];
         }
         psum += __shfl_xor(psum, 32, WAVE);
         l = l * alpha + psum;
@@ -1170,45 +1157,32 @@ attn_prefill_tile(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
         // (kpos, q=ln) lands where slice sidx = kpos>>4, lane
         // lr = ln + 32*((kpos>>3)&1), word i = kpos&7 will read it; regs
         // pair (r, r+1) = (kpos, kpos+1) -> one packed b32 write.
-        // kpos spans 0..63 now: 4 slices per wave.
-        bf16_t* ps = plds_all + wid * (4 * 64 * 8);
+        bf16_t* ps = plds_all + wid * (2 * 64 * 8);
 #pragma unroll
         for (int r = 0; r < 16; r += 2) {
             const int kpos = (r & 3) + 8 * (r >> 2) + 4 * hi;
-            {
-                const int sidx = kpos >> 4;
-                const int lr = ln + 32 * ((kpos >> 3) & 1);
-                u32 w = (u32)f2bf(p0[r]) | ((u32)f2bf(p0[r + 1]) << 16);
-                *(u32*)(ps + (sidx * 64 + lr) * 8 + (kpos & 7)) = w;
-            }
-            {
-                const int kpos1 = kpos + 32;
-                const int sidx = kpos1 >> 4;
-                const int lr = ln + 32 * ((kpos1 >> 3) & 1);
-                u32 w = (u32)f2bf(p1[r]) | ((u32)f2bf(p1[r + 1]) << 16);
-                *(u32*)(ps + (sidx * 64 + lr) * 8 + (kpos1 & 7)) = w;
-            }
+            const int sidx = kpos >> 4;
+            const int lr = ln + 32 * ((kpos >> 3) & 1);
+            u32 w = (u32)f2bf(p[r]) | ((u32)f2bf(p[r + 1]) << 16);
+            *(u32*)(ps + (sidx * 64 + lr) * 8 + (kpos & 7)) = w;
         }
         // rescale output accumulators while the P writes land
-        // (skip when the running max held: alpha == 1 exactly)
-        if (alpha != 1.f) {
 #pragma unroll
-            for (int t = 0; t < NT; ++t)
+        for (int t = 0; t < NT; ++t)
 #pragma unroll
-                for (int r = 0; r < 16; ++r) acc[t][r] *= alpha;
-        }
+            for (int r = 0; r < 16; ++r) acc[t][r] *= alpha;
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-        // ---- O^T += V^T · P (4 k-slices) -------------------------------
-        bf16x8 pf[4];
+        // ---- O^T += V^T · P --------------------------------------------
+        bf16x8 pf[2];
 #pragma unroll
-        for (int s = 0; s < 4; ++s)
+        for (int s = 0; s < 2; ++s)
             pf[s] = *(const bf16x8*)(ps + (s * 64 + lane) * 8);
 #pragma unroll
         for (int t = 0; t < NT; ++t) {
             const bf16_t* vrow = vt + (t * 32 + ln) * VST + hi * 8;
 #pragma unroll
-            for (int s = 0; s < 4; ++s) {
+            for (int s = 0; s < 2; ++s) {
                 const bf16x8 vf = *(const bf16x8*)(vrow + s * 16);
                 acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     vf, pf[s], acc[t], 0, 0, 0);
@@ -1241,9 +1215,9 @@ k_attn_prefill(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
                const bf16_t* __restrict__ kc, const bf16_t* __restrict__ vc,
                int T, int Hq, int G, int Hk, int Smax, int slot,
                int s_total, int causal_off, float scale) {
-    constexpr int VST = 72;
+    constexpr int VST = 40;
     __shared__ __attribute__((aligned(16))) bf16_t vt[D * VST];
-    __shared__ __attribute__((aligned(16))) bf16_t plds[4][4 * 64 * 8];
+    __shared__ __attribute__((aligned(16))) bf16_t plds[4][2 * 64 * 8];
     attn_prefill_tile<D>(out, q, kc, vc, T, Hq, G, Hk, Smax, slot, s_total,
                          causal_off, scale, blockIdx.x * 128, vt,
                          &plds[0][0]);
@@ -1266,9 +1240,9 @@ k_attn_prefill_batch(bf16_t* __restrict__ out, const bf16_t* __restrict__ q,
                      const bf16_t* __restrict__ vc,
                      const int4* __restrict__ meta, int Hq, int G, int Hk,
                      int Smax, float scale) {
-    constexpr int VST = 72;
+    constexpr int VST = 40;
     __shared__ __attribute__((aligned(16))) bf16_t vt[D * VST];
-    __shared__ __attribute__((aligned(16))) bf16_t plds[4][4 * 64 * 8];
+    __shared__ __attribute__((aligned(16))) bf16_t plds[4][2 * 64 * 8];
     const int4 m = meta[blockIdx.x];
     attn_prefill_tile<D>(out + (long)m.x * Hq * D, q + (long)m.x * Hq * D,
                          kc, vc, m.y, Hq, G, Hk, Smax, m.z, m.y,
