@@ -455,9 +455,9 @@ def main() -> None:
     ap.add_argument("--mode", choices=["serving", "engine"],
                     default="serving")
     ap.add_argument("--model", default="llama3.2:3b")
-    ap.add_argument("--concurrency", type=int, default=768,
+    ap.add_argument("--concurrency", type=int, default=1024,
                     help="serving: concurrent streams per GPU")
-    ap.add_argument("--num-predict", type=int, default=96,
+    ap.add_argument("--num-predict", type=int, default=128,
                     help="serving: tokens generated per request")
     ap.add_argument("--flush-ms", type=int, default=25,
                     help="serving: worker stream flush interval")

@@ -1480,7 +1480,11 @@ int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
                  const void* w, int M, int N, int K, int splitk,
                  int kslice, int fuse_swiglu, int variant, int mtiles) {
     if (K % 64 != 0 || kslice % 64 != 0) return -100;
-    if (M > 512) return -104;
+    // m-tiled variant bounds ROWS PER TILE, not M: serving buckets 768/
+    // 1024 hit hipBLASLt fallbacks measured at 0.7-1.6 TB/s (r02j b768
+    // trace) when this cap shut the in-house kernels out
+    if (M > 512 && variant != 2) return -104;
+    if (M > 2048) return -104;
     hipStream_t st = (hipStream_t)stream;
     if (variant == 2) {
         // m-tiled: grid.y m-tiles of <=128 (or <=64) x rows each, W column

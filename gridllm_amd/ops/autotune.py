@@ -82,14 +82,15 @@ def tune_shape(m: int, n: int, k: int, device) -> tuple[str, int]:
             if t < t_best * 0.97:
                 t_best = t
                 best = ("proj", sk)
-    if hip_ops.gemm_proj_supported(m, n, k) and n % 128 == 0:
+    if 64 < m <= 2048 and n % 128 == 0 and k % 64 == 0:
         # m-tiled decomposition (grid.y m-tiles, W slice re-read from XCD
-        # L2): candidate mtiles keep >=64 rows per tile; splitk composes
-        # when the grid is still thin
-        for mt in (2, 4, 8):
-            if m // mt < 64 and not (mt == 2 and m > 64):
-                continue
-            if (m + mt - 1) // mt > 256:
+        # L2): candidate mtiles keep 64..256 rows per tile; splitk
+        # composes when the grid is still thin. Covers serving buckets
+        # ABOVE the plain-proj 512 cap too (768/1024 fell back to
+        # hipBLASLt at 0.7-1.6 TB/s — r02j b768 trace).
+        for mt in (2, 3, 4, 6, 8, 12, 16):
+            rpt = (m + mt - 1) // mt
+            if rpt > 256 or (rpt < 64 and not (mt == 2 and m > 64)):
                 continue
             rule = max(1, min(k // 256,
                               256 // max(1, (n // 128) * mt)))
@@ -133,7 +134,7 @@ def tune_gateup(m: int, n: int, k: int, device) -> tuple[str, int]:
           for _ in range(ncopies)]
     best = ("blas", 0)
     t_best = _time_fn(lambda w: hip_ops.swiglu_il(F.linear(x, w)), ws)
-    if m <= 256 and n % 128 == 0 and hip_ops.gemm_proj_supported(m, n, k):
+    if m <= 2048 and n % 128 == 0 and k % 64 == 0:
         rule = max(1, min(k // 256, 256 // max(1, n // 128)))
         for sk in sorted({1, rule, rule * 2, max(1, rule // 2)}):
             if sk > max(1, k // 64):
@@ -146,8 +147,9 @@ def tune_gateup(m: int, n: int, k: int, device) -> tuple[str, int]:
             if t < t_best * 0.97:
                 t_best = t
                 best = ("proj", sk)
-        for mt in (2, 4, 8):        # m-tiled + fused epilogue
-            if m // mt < 64 and not (mt == 2 and m > 64):
+        for mt in (2, 3, 4, 6, 8, 12, 16):   # m-tiled + fused epilogue
+            rpt = (m + mt - 1) // mt
+            if rpt > 256 or (rpt < 64 and not (mt == 2 and m > 64)):
                 continue
             try:
                 t = _time_fn(lambda w: hip_ops.gemm_proj(

@@ -409,6 +409,8 @@ def test_prefill_batch_engine_parity(dev):
     (256, 16384, 3072, 4, 1),  # gateup, 64-row tiles (4-wave path)
     (192, 3072, 8192, 2, 4),   # down, ragged M
     (96, 3072, 3072, 2, 1),    # rpt<=64 with partial last tile
+    (768, 5120, 3072, 6, 1),   # serving bucket ABOVE the plain-proj cap
+    (1024, 3072, 8192, 8, 2),  # b1024 down, m-tiled + splitk
 ])
 def test_gemm_proj_mtiled(dev, m, n, k, mt, sk):
     """M-tiled projection decomposition vs fp32 reference."""
