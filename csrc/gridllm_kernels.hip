@@ -1538,13 +1538,14 @@ int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
         }
         return 0;
     }
-    if (variant >= 3 && variant <= 6) {
+    if (variant >= 3 && variant <= 8) {
         // deep-pipeline experiments (sweep candidates, scripts/bench_gemm):
         // the per-block wall at M~256 is un-hidden DMA latency — NB=3
         // leaves ONE tile in flight (~32 KB) against a ~1.5 us HBM
         // latency at ~0.1 us compute per episode. Deeper LDS rings /
         // larger BK put more bytes in flight per wave.
-        if (N % ((variant == 6) ? 64 : 128) != 0 || mtiles < 1) return -107;
+        if (N % ((variant == 6 || variant == 7) ? 64 : 128) != 0
+            || mtiles < 1) return -107;
         if (fuse_swiglu) return -108;           // sweep path: plain only
         const int rpt = (M + mtiles - 1) / mtiles;
         if (variant == 3) {                     // NB=4, R=128, 8 waves
@@ -1562,11 +1563,25 @@ int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
             launch_gemm_proj<1, 128, 3, 0, 4, 128>(st, (bf16_t*)out,
                 (float*)ws, (const bf16_t*)x, (const bf16_t*)w, M, N, K,
                 splitk, kslice, mtiles);
-        } else {                                // BN=64, NB=6, R=64
+        } else if (variant == 6) {              // BN=64, NB=6, R=64
             if (rpt > 64) return -109;
             launch_gemm_proj<1, 64, 6, 0, 4>(st, (bf16_t*)out, (float*)ws,
                 (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk,
                 kslice, mtiles);
+        } else if (variant == 7) {
+            // STREAMING tile: 2 waves, R=32, BN=64, 37 KB LDS -> 4
+            // blocks/CU co-resident (8 waves pulling the weight stream
+            // per CU — the occupancy axis every earlier variant lacked
+            // at 96-160 KB LDS footprints)
+            if (rpt > 32) return -109;
+            launch_gemm_proj<1, 64, 3, 0, 2>(st, (bf16_t*)out, (float*)ws,
+                (const bf16_t*)x, (const bf16_t*)w, M, N, K, splitk,
+                kslice, mtiles);
+        } else {                                // v8: 2 waves, BN=128
+            if (rpt > 32) return -109;
+            launch_gemm_proj<1, 128, 3, 0, 2>(st, (bf16_t*)out,
+                (float*)ws, (const bf16_t*)x, (const bf16_t*)w, M, N, K,
+                splitk, kslice, mtiles);
         }
         LAUNCH_CHECK();
         if (splitk > 1) {

@@ -81,8 +81,9 @@ def main():
                     tot["projm"] = tot.get("projm", 0.0) + best_mt[0]
                 # deep-pipeline variants: more DMA bytes in flight per wave
                 best_dp = None
-                for var, need_rpt in ((3, 128), (4, 64), (5, 64), (6, 64)):
-                    for mt in (1, 2, 4):
+                for var, need_rpt in ((3, 128), (4, 64), (5, 64), (6, 64),
+                                      (7, 32), (8, 32)):
+                    for mt in (1, 2, 4, 6, 8, 12, 24, 32):
                         rpt = (M + mt - 1) // mt
                         if rpt > need_rpt:
                             continue
