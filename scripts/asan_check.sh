@@ -13,6 +13,9 @@ cp gridllm_amd/ops/libgridllm_hip.asan.so gridllm_amd/ops/libgridllm_hip.so.bak_
 cp gridllm_amd/ops/libgridllm_hip.so gridllm_amd/ops/libgridllm_hip.so.orig
 cp gridllm_amd/ops/libgridllm_hip.asan.so gridllm_amd/ops/libgridllm_hip.so
 trap 'mv gridllm_amd/ops/libgridllm_hip.so.orig gridllm_amd/ops/libgridllm_hip.so; rm -f gridllm_amd/ops/libgridllm_hip.so.bak_swap' EXIT
-LD_PRELOAD="$ASAN_RT" ASAN_OPTIONS=detect_leaks=0 \
+# HSA_XNACK=1 + protect_shadow_gap=0: ASan's shadow reservation
+# collides with the HIP runtime's address-space setup otherwise
+LD_PRELOAD="$ASAN_RT" HSA_XNACK=1 \
+  ASAN_OPTIONS=detect_leaks=0:protect_shadow_gap=0 \
   GRIDLLM_VALIDATE=1 python -m pytest tests/test_ops_gpu.py -q -x
 echo "ASAN CHECK OK"

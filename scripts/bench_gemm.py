@@ -46,20 +46,27 @@ def main():
         if hip_ops.gemm_skinny_supported(M, N, K):
             err = (hip_ops.gemm_skinny(x, w).float() - ref).abs().max().item()
             res["skinny"] = (bench_cold(lambda w: hip_ops.gemm_skinny(x, w), ws), err)
-        if hip_ops.gemm_proj_supported(M, N, K):
-            err = (hip_ops.gemm_proj(x, w).float() - ref).abs().max().item()
-            res["proj"] = (bench_cold(lambda w: hip_ops.gemm_proj(x, w), ws), err)
+        projable = (hip_ops.gemm_proj_supported(M, N, K)
+                    or (64 < M <= 2048 and N % 128 == 0 and K % 64 == 0))
+        if projable:
+            if hip_ops.gemm_proj_supported(M, N, K):
+                err = (hip_ops.gemm_proj(x, w).float()
+                       - ref).abs().max().item()
+                res["proj"] = (bench_cold(
+                    lambda w: hip_ops.gemm_proj(x, w), ws), err)
             if sweep:
                 for sk in (1, 2, 3, 4, 6, 8, 12):
-                    if sk > max(1, K // 256):
+                    if sk > max(1, K // 256) \
+                            or not hip_ops.gemm_proj_supported(M, N, K):
                         continue
                     t = bench_cold(lambda w: hip_ops.gemm_proj(x, w, splitk=sk), ws)
                     print(f"    {name} proj splitk={sk:2d}: {t:7.1f}us "
                           f"({gb/t*1e6/1e3:5.2f} TB/s, grid={(N//64)*sk})")
                 # m-tiled decomposition sweep (W slice re-read from XCD L2)
                 best_mt = None
-                for mt in (2, 4, 8):
-                    if M // mt < 64 and not (mt == 2 and M > 64):
+                for mt in (2, 3, 4, 6, 8, 12, 16):
+                    rpt = (M + mt - 1) // mt
+                    if rpt > 256 or (rpt < 64 and not (mt == 2 and M > 64)):
                         continue
                     if N % 128:
                         continue
