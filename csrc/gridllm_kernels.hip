@@ -1483,7 +1483,7 @@ int gl_gemm_proj(void* stream, void* out, void* ws, const void* x,
     // m-tiled variant bounds ROWS PER TILE, not M: serving buckets 768/
     // 1024 hit hipBLASLt fallbacks measured at 0.7-1.6 TB/s (r02j b768
     // trace) when this cap shut the in-house kernels out
-    if (M > 512 && variant != 2) return -104;
+    if (M > 512 && (variant < 2 || variant > 8)) return -104;
     if (M > 2048) return -104;
     hipStream_t st = (hipStream_t)stream;
     if (variant == 2) {
