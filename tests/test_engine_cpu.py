@@ -460,3 +460,32 @@ def test_pipelined_mixed_constraint_falls_back():
         assert by in c.allowed()[0]
         c.push(by)
     assert r2.eval_count == 8
+
+
+def test_pipelined_context_shift_and_penalties():
+    """Rows at the context boundary and rows with penalties must drop to
+    the synchronous path mid-flight and still complete correctly."""
+    from gridllm_amd.engine.batcher import Batcher
+    cfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                       max_seq_len=32, max_batch_size=4,
+                       use_hipgraph=False)
+    eng = InferenceEngine(cfg)
+    b = Batcher(eng)
+    try:
+        # crosses the 32-token KV window -> context shift on the sync path
+        f1 = b.submit_generate("abcdefgh", {"temperature": 0.0,
+                                            "num_predict": 50})
+        # penalties -> never pipelined
+        f2 = b.submit_generate("pqr", {"temperature": 0.9, "seed": 5,
+                                       "repeat_penalty": 1.3,
+                                       "num_predict": 12})
+        r1, r2 = f1.result(timeout=120), f2.result(timeout=120)
+    finally:
+        b.shutdown()
+    assert r1.eval_count == 50
+    assert r2.eval_count == 12
+    # penalty row matches the sync engine exactly
+    ref = InferenceEngine(cfg).generate(
+        "pqr", {"temperature": 0.9, "seed": 5, "repeat_penalty": 1.3,
+                "num_predict": 12})
+    assert r2.token_ids == ref.token_ids
