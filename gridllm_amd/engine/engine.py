@@ -120,6 +120,7 @@ class InferenceEngine:
                                          dtype=torch.int64,
                                          device=self.device)
         self._graphs: dict[int, tuple] = {}
+        self._graph_pool = None
         from gridllm_amd.engine.sampler import SlotSampler
         self.sampler = SlotSampler(self.cfg.max_batch_size + 1, self.device)
         if self.is_gpu and self.cfg.autotune:
@@ -129,6 +130,8 @@ class InferenceEngine:
                 buckets.append(self._bucket(self.cfg.max_batch_size))
             tune_model_shapes(self.model, sorted(set(buckets)))
         self._lock = threading.Lock()
+        if self.is_gpu:
+            self.precapture_graphs()
         log.performance("engine ready", model=spec.name,
                         params=spec.params, device=str(self.device),
                         tp=self.tp.size, load_ms=self.load_duration // 1_000_000)
@@ -179,12 +182,37 @@ class InferenceEngine:
                 self.model.cache.seq_lens[self._scratch_slot] = 0
         torch.cuda.current_stream().wait_stream(s)
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
+        # ONE shared memory pool for every bucket graph: private per-graph
+        # pools totalled 49.7 GB at max_batch 1536 (23 buckets) and OOM'd
+        # the 288 GB card (gpurun_out/r02n_serving_c1536.log). Sharing is
+        # safe under the capture-largest-first discipline (engine init
+        # captures descending; a replay never outgrows the pool's high-
+        # water mark set by the largest capture).
+        if self._graph_pool is None:
+            self._graph_pool = torch.cuda.graph_pool_handle()
+        with torch.cuda.graph(g, pool=self._graph_pool):
             out = self.model.forward_decode(tokens, slots)
         self.model.cache.seq_lens[self._scratch_slot] = 0
         self._graphs[bucket] = (g, tokens, slots, out)
         log.performance("hipGraph captured", bucket=bucket)
         return self._graphs[bucket]
+
+    def precapture_graphs(self) -> None:
+        """Capture every bucket graph <= max_batch_size, LARGEST FIRST,
+        into the shared pool (lazy capture order would let a later,
+        larger bucket outgrow the pool)."""
+        if not self._graph_ok():
+            return
+        try:
+            for b in sorted((x for x in _BUCKETS
+                             if x <= self.cfg.max_batch_size),
+                            reverse=True):
+                self._ensure_graph(b)
+        except Exception:  # noqa: BLE001 — degrade loudly to eager
+            log.logger.exception(
+                "hipGraph precapture failed — running eager decode")
+            self.cfg.use_hipgraph = False
+            self._graphs.clear()
 
     def _decode_step(self, token_ids: list[int], slot_ids: list[int],
                      slots_t: "Optional[torch.Tensor]" = None
