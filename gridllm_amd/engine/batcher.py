@@ -140,6 +140,8 @@ class Batcher:
 
         def drain_inflight() -> None:
             nonlocal inflight
+            eng.flush_admit()
+            prune()
             if inflight is not None:
                 try:
                     eng.collect_step(inflight)
@@ -194,8 +196,25 @@ class Batcher:
                 for p in gens:
                     prompt, opts = self._prep(p)
                     items.append((prompt, opts, p.on_token, p.cancel, None))
+                # async admission (no first-token sync) only while the
+                # pipeline is running and every new row will stay
+                # pipeline-eligible: no grammar constraint, no penalties
+                # (a penalty row forces the whole NEXT step onto the sync
+                # path, which cannot resolve a pending async admission)
+                def _async_ok(op: dict) -> bool:
+                    return not (op.get("format")
+                                or op.get("repeat_penalty")
+                                or op.get("presence_penalty")
+                                or op.get("frequency_penalty"))
+                use_async = (inflight is not None
+                             and not eng.tp.enabled
+                             and all(_async_ok(op) for _, op, *_ in items))
                 try:
-                    new_rows = eng.start_rows(items)
+                    if use_async:
+                        new_rows = eng.start_rows_async(items)
+                    else:
+                        drain_inflight()
+                        new_rows = eng.start_rows(items)
                 except Exception as e:  # noqa: BLE001
                     log.logger.exception("batched admission failed")
                     for p in gens:
@@ -229,15 +248,18 @@ class Batcher:
                     inflight = nxt
                     prune()
                 else:
-                    drain_inflight()
-                    try:
-                        eng.step_rows([row for _, row in live])
-                    except Exception as e:  # noqa: BLE001
-                        log.logger.exception("decode step failed")
-                        # retire the rows or their KV slots leak
-                        # permanently and admission wedges
-                        fail_live(e)
+                    eng.flush_admit()   # a pending async admission must
+                    drain_inflight()    # resolve before any sync step
                     prune()
+                    if live:
+                        try:
+                            eng.step_rows([row for _, row in live])
+                        except Exception as e:  # noqa: BLE001
+                            log.logger.exception("decode step failed")
+                            # retire the rows or their KV slots leak
+                            # permanently and admission wedges
+                            fail_live(e)
+                        prune()
             elif inflight is not None:
                 drain_inflight()
             if stop_now or (self._stop.is_set() and not live

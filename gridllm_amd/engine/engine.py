@@ -406,7 +406,48 @@ class InferenceEngine:
         with self._lock:
             return self._step_rows(rows)
 
+    def start_rows_async(self, items: list[tuple]) -> list["_Row"]:
+        """Pipelined admission: prefill + first-token sampling LAUNCH but
+        never sync — the admission's GPU wait stalled the decode pipeline
+        ~10% at c1024 (round-2 profile: 124 admission syncs x ~20 ms).
+        First tokens are emitted by the NEXT collect_step (the submit
+        that follows attaches this batch). Caller guarantees no
+        constraint/penalty rows and a running pipeline (batcher checks)."""
+        with self._lock:
+            rows, logits, slots_t = self._prefill_rows(items)
+            try:
+                ids_dev = self.sampler.sample_ids(logits, rows, slots_t)
+            except Exception:
+                self.abort_rows(rows, locked=True)
+                raise
+            b = len(rows)
+            if self.is_gpu:
+                ih = self._pinned(b)
+                ih[:b].copy_(ids_dev, non_blocking=True)
+            else:
+                ih = ids_dev
+            self._admit_pending = {
+                "rows": rows, "ids_dev": ids_dev, "ids_host": ih,
+                "pos": {id(r): i for i, r in enumerate(rows)}}
+            return rows
+
     def _start_rows(self, items: list[tuple]) -> list["_Row"]:
+        rows, logits, slots_t = self._prefill_rows(items)
+        try:
+            toks = self._sample_batch(logits, rows, slots_t)  # syncs
+        except Exception:
+            self.abort_rows(rows, locked=True)
+            raise
+        t1 = time.monotonic_ns()
+        per = (t1 - rows[0].t_start_ns) // max(1, len(rows)) if rows else 0
+        for row, tok in zip(rows, toks):
+            row.result.prompt_eval_duration = per
+            row.t_decode_ns = t1
+            if not (self._emit(row, tok) and row.max_new > 1):
+                self._retire(row)
+        return rows
+
+    def _prefill_rows(self, items: list[tuple]):
         t0 = time.monotonic_ns()
         rows: list[_Row] = []
         id_lists: list[list[int]] = []
@@ -472,7 +513,6 @@ class InferenceEngine:
         try:
             # ONE pinned upload for every prompt's ids (one blocking H2D
             # per prompt measured ~1.5 ms each against a busy stream)
-            pass
             all_ids = self.model.up.upload(np.concatenate(
                 [np.asarray(x, dtype=np.int64) for x in id_lists]))
             token_lists = list(torch.split(all_ids,
@@ -486,20 +526,12 @@ class InferenceEngine:
             self.sampler.admit(rows, self.spec.vocab)
             slots_t = self.model.up.upload(
                 np.asarray(slots, dtype=np.int64))
-            toks = self._sample_batch(logits, rows, slots_t)  # syncs
         except Exception:
-            # prefill/first-sample failure: release every slot this batch
-            # allocated or capacity leaks permanently
+            # prefill failure: release every slot this batch allocated
+            # or capacity leaks permanently
             self.abort_rows(rows, locked=True)
             raise
-        t1 = time.monotonic_ns()
-        per = (t1 - t0) // max(1, len(rows))
-        for row, tok in zip(rows, toks):
-            row.result.prompt_eval_duration = per
-            row.t_decode_ns = t1
-            if not (self._emit(row, tok) and row.max_new > 1):
-                self._retire(row)
-        return rows
+        return rows, logits, slots_t
 
     def abort_rows(self, rows: Sequence["_Row"], locked: bool = False) -> None:
         """Free the KV slots of rows that have not retired (step/prefill
@@ -596,7 +628,10 @@ class InferenceEngine:
                 for r in rows:
                     if r.cur_len >= self.model.max_seq:
                         r.cur_len = self.model.shift_slot(r.slot)
-            same = (prev is not None and len(prev["rows"]) == b
+            ap = getattr(self, "_admit_pending", None)
+            self._admit_pending = None
+            same = (ap is None and prev is not None
+                    and len(prev["rows"]) == b
                     and all(a is c for a, c in zip(prev["rows"], rows)))
             if same:
                 # steady state (no admit/retire since the last step): the
@@ -607,12 +642,14 @@ class InferenceEngine:
                 slots32 = prev["slots32"]
             else:
                 src = np.full(b, -1, dtype=np.int64)
+                src2 = np.full(b, -1, dtype=np.int64)
                 host_tok = np.zeros(b, dtype=np.int64)
                 slots_np = np.zeros(b, dtype=np.int64)
                 pos = prev["pos"] if prev is not None else {}
+                apos = ap["pos"] if ap is not None else {}
                 for i, r in enumerate(rows):
-                    j = pos.get(id(r), -1)
-                    src[i] = j
+                    src[i] = pos.get(id(r), -1)
+                    src2[i] = apos.get(id(r), -1)
                     host_tok[i] = r.last_token
                     slots_np[i] = r.slot
                 # pinned staging + non_blocking: a pageable .to(device)
@@ -620,20 +657,26 @@ class InferenceEngine:
                 # on it (~1.5 ms/step measured) — the exact stall the
                 # pipeline exists to remove
                 if self.is_gpu:
-                    stage = self._pinned_stage(3 * b).view(3, b)
-                    stage.numpy()[:] = np.stack([src, host_tok, slots_np])
-                    up = torch.empty(3, b, dtype=torch.int64,
+                    stage = self._pinned_stage(4 * b).view(4, b)
+                    stage.numpy()[:] = np.stack(
+                        [src, host_tok, slots_np, src2])
+                    up = torch.empty(4, b, dtype=torch.int64,
                                      device=self.device)
                     up.copy_(stage, non_blocking=True)
                 else:
                     up = torch.from_numpy(
-                        np.stack([src, host_tok, slots_np]))
+                        np.stack([src, host_tok, slots_np, src2]))
+                tokens_dev = up[1]
                 if prev is not None:
                     tokens_dev = torch.where(
                         up[0] >= 0,
-                        prev["ids_dev"][up[0].clamp(min=0)], up[1])
-                else:
-                    tokens_dev = up[1]
+                        prev["ids_dev"][up[0].clamp(min=0)], tokens_dev)
+                if ap is not None:
+                    # async-admitted rows: first token straight from the
+                    # admission's on-device sample
+                    tokens_dev = torch.where(
+                        up[3] >= 0,
+                        ap["ids_dev"][up[3].clamp(min=0)], tokens_dev)
                 slots64 = up[2]
                 slots32 = slots64.to(torch.int32)
             if self._graph_ok():
@@ -674,7 +717,7 @@ class InferenceEngine:
             return {"rows": rows, "ids_dev": ids_dev, "ids_host": ids_host,
                     "ev": ev, "pos": {id(r): i for i, r in enumerate(rows)},
                     "slots64": slots64, "slots32": slots32,
-                    "free_after": free_after}
+                    "free_after": free_after, "admit": ap}
 
     def collect_step(self, inflight) -> None:
         """Wait for an inflight step's sampled ids and run the per-token
@@ -685,6 +728,22 @@ class InferenceEngine:
             ev = inflight["ev"]
             if ev is not None:
                 ev.synchronize()
+            ap = inflight.get("admit")
+            if ap is not None:
+                # async admission batch: its prefill + first-token sample
+                # ran BEFORE this step on the same stream, so this step's
+                # event covers its pinned copy too
+                ids0 = ap["ids_host"][:len(ap["rows"])].tolist()
+                t1 = time.monotonic_ns()
+                for r, tok in zip(ap["rows"], ids0):
+                    if r.done:
+                        continue
+                    # includes ~one overlapped decode step (no sync point
+                    # exists to time the prefill alone on this path)
+                    r.result.prompt_eval_duration = t1 - r.t_start_ns
+                    r.t_decode_ns = t1
+                    if not (self._emit(r, int(tok)) and r.max_new > 1):
+                        self._retire(r, defer_free=True)
             ids = inflight["ids_host"][:len(inflight["rows"])].tolist()
             for r, tok in zip(inflight["rows"], ids):
                 if r.done:
@@ -693,6 +752,28 @@ class InferenceEngine:
                     self._retire(r, defer_free=True)
             for sl in inflight["free_after"]:
                 self.model.cache.free(sl)
+
+    def flush_admit(self) -> None:
+        """Resolve a pending async admission WITHOUT a following pipelined
+        step (the batcher fell back to the synchronous path, e.g. a live
+        row reached the context boundary): sync the stream and emit the
+        first tokens so the rows have valid last_token state."""
+        with self._lock:
+            ap = getattr(self, "_admit_pending", None)
+            self._admit_pending = None
+            if ap is None:
+                return
+            self._sync()
+            ids0 = ap["ids_host"][:len(ap["rows"])].tolist()
+            t1 = time.monotonic_ns()
+            for r, tok in zip(ap["rows"], ids0):
+                if r.done:
+                    continue
+                r.result.prompt_eval_duration = t1 - r.t_start_ns
+                r.t_decode_ns = t1
+                if not (self._emit(r, int(tok)) and r.max_new > 1):
+                    self._retire(r)
+            return
 
     def flush_deferred(self) -> None:
         """Free retire-deferred slots once NO step is in flight (pipeline

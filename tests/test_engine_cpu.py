@@ -489,3 +489,65 @@ def test_pipelined_context_shift_and_penalties():
         "pqr", {"temperature": 0.9, "seed": 5, "repeat_penalty": 1.3,
                 "num_predict": 12})
     assert r2.token_ids == ref.token_ids
+
+
+def test_async_admission_matches_sync():
+    """A request admitted WHILE the pipeline is mid-flight (the async,
+    no-first-token-sync admission) must produce the same tokens as the
+    sync engine."""
+    from gridllm_amd.engine.batcher import Batcher
+    cfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                       max_seq_len=128, max_batch_size=8,
+                       use_hipgraph=False)
+    eng = InferenceEngine(cfg)
+    b = Batcher(eng)
+    try:
+        started = threading.Event()
+        long_fut = b.submit_generate(
+            "background load", {"temperature": 0.0, "num_predict": 60},
+            on_token=lambda piece, done: piece and started.set())
+        assert started.wait(timeout=30)
+        # admitted mid-pipeline -> start_rows_async path
+        f = b.submit_generate("probe", {"temperature": 0.8, "seed": 21,
+                                        "num_predict": 10})
+        r = f.result(timeout=60)
+        long_fut.result(timeout=60)
+    finally:
+        b.shutdown()
+    ref = InferenceEngine(cfg).generate(
+        "probe", {"temperature": 0.8, "seed": 21, "num_predict": 10})
+    assert r.token_ids == ref.token_ids
+    assert r.prompt_eval_duration > 0
+
+
+def test_async_admission_then_sync_fallback_flushes():
+    """A pending async admission must be resolved (flush_admit) when the
+    batch drops to the sync path (here: a penalty request arrives right
+    after an async-admitted one)."""
+    from gridllm_amd.engine.batcher import Batcher
+    cfg = EngineConfig(model="tiny", device="cpu", dtype="fp32",
+                       max_seq_len=128, max_batch_size=8,
+                       use_hipgraph=False)
+    eng = InferenceEngine(cfg)
+    b = Batcher(eng, admit_wait_ms=1.0, admit_min=1)
+    try:
+        started = threading.Event()
+        long_fut = b.submit_generate(
+            "bg", {"temperature": 0.0, "num_predict": 60},
+            on_token=lambda piece, done: piece and started.set())
+        assert started.wait(timeout=30)
+        f1 = b.submit_generate("asyncrow", {"temperature": 0.0,
+                                            "num_predict": 6})
+        import time as _t
+        _t.sleep(0.05)       # f1 admitted async; now force sync mode
+        f2 = b.submit_generate("penalized", {"temperature": 0.9, "seed": 3,
+                                             "repeat_penalty": 1.4,
+                                             "num_predict": 6})
+        r1, r2 = f1.result(timeout=60), f2.result(timeout=60)
+        long_fut.result(timeout=60)
+    finally:
+        b.shutdown()
+    ref1 = InferenceEngine(cfg).generate("asyncrow", {"temperature": 0.0,
+                                                      "num_predict": 6})
+    assert r1.token_ids == ref1.token_ids
+    assert r2.eval_count == 6
