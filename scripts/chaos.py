@@ -39,7 +39,7 @@ async def drive(base: str, args, stats: dict, stop: asyncio.Event):
                 try:
                     if stream:
                         async with sess.post("/api/generate", json={
-                                "model": "tiny", "stream": True,
+                                "model": MODEL[0], "stream": True,
                                 "prompt": f"chaos {i} " + "lorem " * 8,
                                 "options": {"seed": i, "num_predict": 24}},
                                 timeout=aiohttp.ClientTimeout(total=90)) as r:
@@ -52,7 +52,7 @@ async def drive(base: str, args, stats: dict, stop: asyncio.Event):
                                 raise RuntimeError("stream ended w/o done")
                     else:
                         async with sess.post("/api/generate", json={
-                                "model": "tiny", "stream": False,
+                                "model": MODEL[0], "stream": False,
                                 "prompt": f"chaos {i} " + "lorem " * 8,
                                 "options": {"seed": i, "num_predict": 24}},
                                 timeout=aiohttp.ClientTimeout(total=90)) as r:
@@ -127,6 +127,9 @@ async def main_async(args, env, base, workers):
     return stats
 
 
+MODEL = ["tiny"]
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--duration", type=float, default=30.0)
@@ -136,9 +139,19 @@ def main():
     args = ap.parse_args()
 
     bus_port, http_port = loadtest._free_port(), loadtest._free_port()
+    # CHAOS_DEVICE=cuda:0 runs the fault-injection soak against the REAL
+    # GPU engine (model llama3.2:3b bf16); default stays the CPU tiny
+    # engine for CI
+    dev = os.environ.get("CHAOS_DEVICE", "cpu")
+    on_gpu = dev.startswith("cuda")
+    # both chaos workers must fit one GPU when sharing a device
     env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port),
-           "LT_CONC": "16", "LT_MODEL": "tiny", "LT_DEVICE": "cpu",
-           "LT_DTYPE": "fp32", "LT_MAXSEQ": "256"}
+           "LT_CONC": "32" if on_gpu else "16",
+           "LT_MODEL": "llama3.2:3b" if on_gpu else "tiny",
+           "LT_DEVICE": dev,
+           "LT_DTYPE": "bf16" if on_gpu else "fp32",
+           "LT_MAXSEQ": "512" if on_gpu else "256"}
+    MODEL[0] = env["LT_MODEL"]
     server = loadtest._spawn(loadtest.SERVER_SRC, env, "SERVER_READY")
     workers = []
     for i in range(args.workers):
