@@ -50,7 +50,8 @@ async def _s():
     await server.start()
     print("SERVER_READY", flush=True)
 
-uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error")
+uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error",
+            timeout_keep_alive=120)
 """
 
 LOADGEN_SRC = r"""

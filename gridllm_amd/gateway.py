@@ -92,7 +92,8 @@ def main() -> None:
     async def _down():
         await replica.shutdown()
 
-    uvicorn.run(app, host=cfg.host, port=cfg.port, log_level="warning")
+    uvicorn.run(app, host=cfg.host, port=cfg.port,
+                log_level="warning", timeout_keep_alive=120)
 
 
 if __name__ == "__main__":

@@ -142,7 +142,8 @@ def main() -> None:
     async def _shutdown():
         await server.shutdown()
 
-    uvicorn.run(app, host=cfg.host, port=cfg.port, log_level="warning")
+    uvicorn.run(app, host=cfg.host, port=cfg.port,
+                log_level="warning", timeout_keep_alive=120)
 
 
 if __name__ == "__main__":

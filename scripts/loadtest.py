@@ -59,7 +59,8 @@ async def _s():
         _sig.signal(_sig.SIGTERM, dump)
     print("SERVER_READY", flush=True)
 
-uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error")
+uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error",
+            timeout_keep_alive=120)
 """
 
 WORKER_SRC = r"""
