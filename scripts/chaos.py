@@ -67,10 +67,17 @@ async def drive(base: str, args, stats: dict, stop: asyncio.Event):
         i = 0
         tasks = set()
         while not stop.is_set():
-            t = asyncio.ensure_future(one(i))
-            tasks.add(t)
-            t.add_done_callback(tasks.discard)
-            i += 1
+            # bound outstanding work to ~2x the concurrency window: a
+            # fixed arrival rate on a slow host built an unbounded
+            # semaphore backlog that could not drain inside the post-run
+            # wait and showed up as FALSE "lost" requests (the server
+            # side had completed everything it was given — box repro:
+            # gpurun_out/r02t_chaos_cpu.log vs /inference/queue stats)
+            if len(tasks) < args.concurrency * 2:
+                t = asyncio.ensure_future(one(i))
+                tasks.add(t)
+                t.add_done_callback(tasks.discard)
+                i += 1
             await asyncio.sleep(0.02)
         if tasks:
             await asyncio.wait(tasks, timeout=120)

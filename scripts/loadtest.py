@@ -106,6 +106,21 @@ async def main():
                         max_batch_size=conc,
                         use_hipgraph=os.environ["LT_DEVICE"].startswith("cuda"))
     engine = InferenceEngine(ecfg)
+    if os.environ.get("LT_SLOW"):
+        # artificial per-step delay: reproduces slow-host timing (chaos
+        # wedge hunting) without a GPU
+        import time as _t
+        _slow = float(os.environ["LT_SLOW"])
+        _orig = engine._decode_step
+        def _slow_step(tokens, slots, **kw):
+            _t.sleep(_slow)
+            return _orig(tokens, slots, **kw)
+        engine._decode_step = _slow_step
+        _origsub = engine.submit_step
+        def _slow_submit(rows, prev=None):
+            _t.sleep(_slow)
+            return _origsub(rows, prev)
+        engine.submit_step = _slow_submit
     if os.environ.get("LT_FAKE"):
         # stub the model: measures the serving GLUE ceiling (bus, pump,
         # gateway, GIL) with near-zero compute per token
