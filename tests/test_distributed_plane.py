@@ -317,7 +317,7 @@ def replica_stack():
             time.sleep(0.1)
         else:
             raise TimeoutError("gateway replica never saw the worker")
-    yield {"base": base}
+    yield {"base": base, "bus_port": bus_port}
     for p in procs.values():
         if p.poll() is None:
             p.send_signal(signal.SIGKILL)
@@ -349,3 +349,53 @@ def test_gateway_replica_serves_full_api(replica_stack):
         r = c.get("/inference/queue")
         assert r.status_code == 200
         assert "queue" in r.json()
+
+
+def test_replica_stream_survives_worker_sigkill(replica_stack, request):
+    """SIGKILL the worker mid-stream THROUGH A DETACHED REPLICA: the
+    orphaned job requeues (metadata.streamAddr survives), the surviving
+    worker streams to the replica's OWN stream server, and the original
+    HTTP response completes."""
+    # spawn a second worker so the requeue has somewhere to go
+    bus_port = replica_stack.get("bus_port")
+    if bus_port is None:
+        pytest.skip("fixture lacks bus port")
+    w2 = _spawn([sys.executable, "-u", "-c", WORKER_SRC],
+                {"BUS_PORT": str(bus_port), "WORKER_ID": "w-rep2",
+                 "WORKER_SLOW": "0"}, "WORKER_READY")
+    victim = _spawn([sys.executable, "-u", "-c", WORKER_SRC],
+                    {"BUS_PORT": str(bus_port), "WORKER_ID": "w-victim2",
+                     "WORKER_SLOW": "0.15"}, "WORKER_READY")
+    try:
+        time.sleep(1.0)
+        import concurrent.futures as cf
+
+        def stream_req(i):
+            lines = []
+            with httpx.Client(base_url=replica_stack["base"],
+                              timeout=180) as c:
+                with c.stream("POST", "/api/generate", json={
+                        "model": "tiny", "prompt": f"rs{i}",
+                        "stream": True,
+                        "options": {"temperature": 0.0,
+                                    "num_predict": 40}}) as r:
+                    for ln in r.iter_lines():
+                        if ln:
+                            lines.append(json.loads(ln))
+            return lines
+
+        with cf.ThreadPoolExecutor(6) as ex:
+            futs = [ex.submit(stream_req, i) for i in range(6)]
+            time.sleep(2.0)
+            victim.send_signal(signal.SIGKILL)
+            victim.wait(timeout=10)
+            results = [f.result() for f in futs]
+        for lines in results:
+            assert lines, "stream produced nothing"
+            assert lines[-1].get("done") is True
+            assert not lines[-1].get("error"), lines[-1]
+    finally:
+        for p in (victim, w2):
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+                p.wait(timeout=10)
