@@ -219,14 +219,18 @@ def serving_mode(args, rank, world, local_rank, dist, device, on_gpu):
                     time.sleep(0.25)
                 else:
                     raise TimeoutError("workers never all registered")
-            # one gateway replica + one load generator per GPU
-            for i in range(n_gpus):
+            # one gateway replica + one load generator per GPU (or
+            # --gateways replicas, load spread round-robin)
+            n_gw = args.gateways or n_gpus
+            for i in range(n_gw):
                 gp = gw_port0 + i
                 gateways.append(_spawn(
                     [sys.executable, "-m", "gridllm_amd.gateway",
                      "--port", str(gp), "--host", "127.0.0.1",
                      "--bus", f"127.0.0.1:{bus_port}"],
                     {}, "GATEWAY_READY"))
+            for i in range(n_gpus):
+                gp = gw_port0 + (i % n_gw)
                 loadgens.append(_spawn(
                     [sys.executable, "-u", "-c", LOADGEN_SRC],
                     {"LG_BASE": f"http://127.0.0.1:{gp}",
@@ -462,6 +466,8 @@ def main() -> None:
                     help="serving: tokens generated per request")
     ap.add_argument("--flush-ms", type=int, default=25,
                     help="serving: worker stream flush interval")
+    ap.add_argument("--gateways", type=int, default=0,
+                    help="serving: gateway replicas (0 = one per GPU)")
     ap.add_argument("--batch", type=int, default=1024,
                     help="engine: sequences per GPU")
     ap.add_argument("--prompt-tokens", type=int, default=128)

@@ -48,7 +48,7 @@ class Batcher:
     whole waves and runs each to completion."""
 
     def __init__(self, engine: InferenceEngine, window_ms: float = 4.0,
-                 continuous: bool = True, admit_per_step: int = 64,
+                 continuous: bool = True, admit_per_step: int = 128,
                  admit_min: int = 8, admit_wait_ms: float = 15.0):
         self.engine = engine
         self.window_s = window_ms / 1000.0
