@@ -319,7 +319,7 @@ def serving_mode(args, rank, world, local_rank, dist, device, on_gpu):
                     "num_predict": args.num_predict,
                     "parallelism": f"dp{n_gpus}",
                     "mode": "serving",
-                    "gateways": n_gpus,
+                    "gateways": args.gateways or n_gpus,
                     "ttft_p50_ms": round(statistics.median(ttft_p50s), 2)
                     if ttft_p50s else None,
                     "tokens_timed": total_tokens,
