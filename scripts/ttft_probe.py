@@ -1,0 +1,67 @@
+#!/usr/bin/env python3
+"""Interactive-latency probe: SEQUENTIAL single streaming requests
+through the full serving path — the TTFT a lone user sees (the
+closed-loop loadtest's TTFT is cohort queueing at saturation).
+Prints one JSON line with p50/p95 over --requests solo requests."""
+import json, os, statistics, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+import loadtest
+
+def main():
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--requests", type=int, default=24)
+    ap.add_argument("--num-predict", type=int, default=32)
+    ap.add_argument("--prompt-words", type=int, default=24)
+    args = ap.parse_args()
+    import torch
+    on_gpu = torch.cuda.is_available()
+    bus_port, http_port = loadtest._free_port(), loadtest._free_port()
+    env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port),
+           "LT_CONC": "8",
+           "LT_MODEL": "llama3.2:3b" if on_gpu else "tiny",
+           "LT_DEVICE": "cuda:0" if on_gpu else "cpu",
+           "LT_DTYPE": "bf16" if on_gpu else "fp32",
+           "LT_MAXSEQ": "1024"}
+    server = loadtest._spawn(loadtest.SERVER_SRC, env, "SERVER_READY")
+    worker = loadtest._spawn(loadtest.WORKER_SRC, {**env, "LT_WID": "probe"},
+                             "WORKER_READY")
+    base = f"http://127.0.0.1:{http_port}"
+    import httpx
+    try:
+        with httpx.Client(base_url=base, timeout=60) as c:
+            for _ in range(200):
+                r = c.get("/health/workers")
+                if r.status_code == 200 and r.json()["counts"]["total"] == 1:
+                    break
+                time.sleep(0.1)
+            ttfts, totals = [], []
+            for i in range(args.requests + 4):
+                t0 = time.monotonic()
+                first = None
+                with c.stream("POST", "/api/generate", json={
+                        "model": env["LT_MODEL"],
+                        "prompt": f"probe {i}: " + "lorem " * args.prompt_words,
+                        "stream": True,
+                        "options": {"temperature": 0.0,
+                                    "num_predict": args.num_predict}}) as r:
+                    for ln in r.iter_lines():
+                        if ln and first is None:
+                            first = time.monotonic() - t0
+                if i >= 4:                        # skip warmup/captures
+                    ttfts.append(first * 1000)
+                    totals.append((time.monotonic() - t0) * 1000)
+            print(json.dumps({
+                "metric": "solo-request TTFT, /api/generate streaming",
+                "requests": args.requests,
+                "ttft_p50_ms": round(statistics.median(ttfts), 2),
+                "ttft_p95_ms": round(sorted(ttfts)[int(0.95 * len(ttfts))], 2),
+                "total_p50_ms": round(statistics.median(totals), 2),
+                "num_predict": args.num_predict,
+                "model": env["LT_MODEL"]}))
+    finally:
+        server.kill(); worker.kill()
+
+if __name__ == "__main__":
+    main()
