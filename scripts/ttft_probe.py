@@ -47,7 +47,10 @@ def main():
                 "model": env["LT_MODEL"],
                 "prompt": f"probe {i}: " + "lorem " * args.prompt_words,
                 "stream": True,
-                "options": {"temperature": 0.0,
+                # temperature 0.8: greedy decode on a random-init model
+                # fixates on one token, often an empty-decoding special ->
+                # no text chunks -> no observable TTFT (r02aa logs).
+                "options": {"temperature": 0.8, "seed": i,
                             "num_predict": args.num_predict}})
             req = ("POST /api/generate HTTP/1.1\r\nHost: x\r\n"
                    "Content-Type: application/json\r\n"
