@@ -155,6 +155,14 @@ class HFStreamDecoder:
     (not the whole history — a full re-decode per token is O(n^2) over a
     generation), emitting and resetting at each clean boundary."""
 
+    # Longest hold before emitting with replacement chars. A trained
+    # model emits invalid-UTF-8 spans of at most a few tokens; without a
+    # bound, a pathological stream (e.g. random byte-fragment tokens)
+    # whose tail KEEPS ending mid-codepoint holds the entire generation
+    # until flush() — the stream goes silent and the span grows without
+    # limit (measured: TTFT == total on random-init weights).
+    MAX_HOLD = 16
+
     def __init__(self, tok) -> None:
         self._tok = tok
         self._tail: List[int] = []
@@ -162,7 +170,7 @@ class HFStreamDecoder:
     def push(self, token_id: int) -> str:
         self._tail.append(token_id)
         text = self._tok.decode(self._tail)
-        if text.endswith("�"):
+        if text.endswith("�") and len(self._tail) < self.MAX_HOLD:
             return ""                   # mid-codepoint: hold the span
         self._tail.clear()
         return text

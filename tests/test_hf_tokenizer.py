@@ -130,3 +130,29 @@ def test_engine_hf_tokenizer_with_schema(hf_tok):
                              "format": schema, "num_predict": 60})
     if res.done_reason != "length":
         assert json.loads(res.text) in ({"mode": "a"}, {"mode": "b"})
+
+
+def test_hf_stream_decoder_bounded_holdback(hf_tok):
+    """A token stream whose decode ALWAYS ends mid-codepoint (possible
+    with random/adversarial byte-fragment tokens) must not silence the
+    stream for the whole generation: the decoder emits (with replacement
+    chars) after at most MAX_HOLD held tokens."""
+    from gridllm_amd.engine.tokenizer import HFStreamDecoder
+    # find a token whose bytes END with a dangling UTF-8 lead byte
+    frag = None
+    for i in range(hf_tok.vocab_size):
+        bs = hf_tok.token_bytes(i)
+        if bs and hf_tok._tok.decode([i]).endswith("�"):
+            frag = i
+            break
+    if frag is None:
+        pytest.skip("no dangling-lead token in this vocab")
+    dec = hf_tok.decode_stream()
+    emitted, max_gap, gap = [], 0, 0
+    for _ in range(100):
+        piece = dec.push(frag)
+        gap = gap + 1 if not piece else 0
+        max_gap = max(max_gap, gap)
+        emitted.append(piece)
+    assert any(emitted), "stream stayed silent for 100 tokens"
+    assert max_gap < HFStreamDecoder.MAX_HOLD + 1
