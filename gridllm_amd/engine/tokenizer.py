@@ -12,19 +12,26 @@ import os
 from typing import List, Optional
 
 
+_PSEUDO_CACHE: dict = {}
+
+
 def _pseudo_word(tid: int) -> str:
     """Deterministic pseudo-word for an unassigned vocab id. Real
     tokenizers decode every non-special id to text; mapping the synthetic
     vocab's unassigned ids (258..V) to "" made 99.8% of sampled tokens
     invisible on a random-init model — streaming looked silent and TTFT
-    was unmeasurable (gpurun_out/r02ab traces)."""
-    i = tid - 258
-    n = 3 + (i % 3)
-    out = []
-    for _ in range(n):
-        out.append(chr(97 + i % 26))
-        i //= 26
-    return "".join(out) + " "
+    was unmeasurable (gpurun_out/r02ab traces). Memoized: this runs on
+    the batcher thread once per generated token."""
+    w = _PSEUDO_CACHE.get(tid)
+    if w is None:
+        i = tid - 258
+        n = 3 + (i % 3)
+        out = []
+        for _ in range(n):
+            out.append(chr(97 + i % 26))
+            i //= 26
+        w = _PSEUDO_CACHE[tid] = "".join(out) + " "
+    return w
 
 
 class ByteTokenizer:
