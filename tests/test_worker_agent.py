@@ -278,3 +278,28 @@ async def test_multi_model_worker():
         assert r2.get("response") is not None and not r2.get("error")
     finally:
         await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_streaming_incremental_chunks():
+    """Guards the adaptive flush cadence: a steady stream must flush
+    MID-generation (several chunks), not accumulate until done — the
+    full/fast-cycle gating in _flush_loop skips steady streams on fast
+    cycles, and a bug there would collapse the stream to 1-2 chunks."""
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    chunks = []
+    req = InferenceRequest(model="tiny", prompt="incremental chunks probe",
+                           options={"temperature": 0.8, "seed": 7,
+                                    "num_predict": 40},
+                           stream=True, timeout_ms=30000)
+    res = await asyncio.wait_for(
+        sched.submit_streaming(req, lambda c: chunks.append(c)), 30)
+    await asyncio.sleep(0.1)
+    assert res.get("error") is None
+    text_chunks = [c for c in chunks if c.get("response")]
+    # tiny CPU engine: ~7 ms/token * 40 tokens vs 25 ms cadence -> >=3
+    assert len(text_chunks) >= 3, \
+        f"stream not incremental: {len(text_chunks)} text chunks"
+    assert "".join(c["response"] for c in chunks) == res["response"]
+    await teardown(reg, sched, agent)
