@@ -121,6 +121,9 @@ class InferenceEngine:
                                          device=self.device)
         self._graphs: dict[int, tuple] = {}
         self._graph_pool = None
+        import os as _os
+        self._no_emit_text = \
+            _os.environ.get("GRIDLLM_NO_EMIT_TEXT", "0") == "1"
         from gridllm_amd.engine.sampler import SlotSampler
         self.sampler = SlotSampler(self.cfg.max_batch_size + 1, self.device)
         if self.is_gpu and self.cfg.autotune:
@@ -831,6 +834,14 @@ class InferenceEngine:
         if tok == self.tokenizer.eos_id:
             res.done_reason = "stop"
             return False
+        if self._no_emit_text:
+            # diagnostic mode (GRIDLLM_NO_EMIT_TEXT=1): skip per-token
+            # text decode/stream emission to measure its end-to-end cost;
+            # final text still assembled at retire from token_ids
+            if res.eval_count >= row.max_new:
+                res.done_reason = "length"
+                return False
+            return True
         piece = row.decoder.push(tok)
         if piece:
             row.text_tail += piece
@@ -863,8 +874,12 @@ class InferenceEngine:
         return True
 
     def _retire(self, row: _Row, defer_free: bool = False) -> None:
-        tail = row.decoder.flush() if hasattr(row.decoder, "flush") else ""
-        final = row.text_tail + tail
+        if self._no_emit_text:
+            final = self.tokenizer.decode(row.result.token_ids)
+        else:
+            tail = row.decoder.flush() if hasattr(row.decoder, "flush") \
+                else ""
+            final = row.text_tail + tail
         if final and not row.result.cancelled:
             row.result.text += final
             if row.on_token:
