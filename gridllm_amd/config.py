@@ -102,6 +102,9 @@ class WorkerConfig:
     device: int = 0                             # HIP device index this worker pins
     reconnect_max_attempts: int = 10
     reconnect_base_delay_ms: int = 500
+    tp_wedge_timeout_ms: int = 60_000  # TP group: batcher stuck this long
+                                       # with work in flight -> exit so the
+                                       # launcher fails the group over
 
     @classmethod
     def from_env(cls) -> "WorkerConfig":
@@ -115,6 +118,7 @@ class WorkerConfig:
             stream_flush_ms=_env_int("STREAM_FLUSH_MS", 25, 0),
             max_concurrent_tasks=_env_int("MAX_CONCURRENT_TASKS", 1, 1, 4096),
             device=_env_int("WORKER_DEVICE", 0, 0, 63),
+            tp_wedge_timeout_ms=_env_int("TP_WEDGE_TIMEOUT", 60_000, 1000),
         )
 
 

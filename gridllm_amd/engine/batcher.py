@@ -53,6 +53,13 @@ class Batcher:
         self.engine = engine
         self.window_s = window_ms / 1000.0
         self.continuous = continuous and engine.tp.size == 1
+        # wedge watchdog state: loop threads stamp last_progress each
+        # iteration; busy means rows were in flight at the stamp. The
+        # worker agent's heartbeat turns (busy and stale) into a process
+        # exit under TP — a hung collective must take the whole group
+        # down so the scheduler's orphan recovery can requeue its jobs.
+        self.last_progress = time.monotonic()
+        self.busy = False
         # bound sequential prefills between decode steps: a burst of new
         # prompts must not stall in-flight rows for the whole burst
         self.admit_per_step = max(1, admit_per_step)
@@ -68,6 +75,12 @@ class Batcher:
         self._thread = threading.Thread(target=target, daemon=True,
                                         name="gridllm-batcher")
         self._thread.start()
+
+    def wedged(self, timeout_s: float) -> bool:
+        """True when the loop thread had work in flight at its last stamp
+        and has not completed an iteration within timeout_s."""
+        return (self.busy
+                and time.monotonic() - self.last_progress > timeout_s)
 
     def shutdown(self) -> None:
         self._stop.set()
@@ -154,6 +167,8 @@ class Batcher:
 
         hold: list[_Pending] = []      # coalescing buffer for arrivals
         while True:
+            self.last_progress = time.monotonic()   # wedge watchdog stamp
+            self.busy = bool(live) or inflight is not None
             # -- admit: block only when idle, otherwise drain without waiting
             arrivals: list[_Pending] = []
             if not live and inflight is None and not hold:

@@ -167,6 +167,7 @@ class WorkerAgent:
     async def _heartbeat_loop(self) -> None:
         interval = self.cfg.heartbeat_interval_ms / 1000.0
         while True:
+            self._check_wedge()
             try:
                 await self.bus.publish(Channels.WORKER_HEARTBEAT, {
                     "workerId": self.worker_id,
@@ -180,6 +181,35 @@ class WorkerAgent:
             except Exception:  # noqa: BLE001
                 log.logger.warning("heartbeat publish failed")
             await asyncio.sleep(interval)
+
+    def _check_wedge(self) -> None:
+        """TP group failover must be atomic (SURVEY §7 hard-part 5): a
+        collective that hangs (peer rank dead/stuck) blocks the batcher
+        thread while THIS loop keeps heartbeating, so the scheduler would
+        keep routing jobs to a wedged group forever. When the batcher has
+        had work in flight with no completed iteration for
+        tp_wedge_timeout_ms, exit the process: the launcher (torchrun)
+        tears down the remaining ranks, heartbeats stop, and orphan
+        recovery requeues the jobs on live workers. TP-only — a
+        single-process engine stall is a GPU hang with its own handling."""
+        eng = next(iter(self._by_model.values()))[0] if self._by_model \
+            else None
+        if eng is None or not eng.tp.enabled:
+            return
+        timeout_s = getattr(self.cfg, "tp_wedge_timeout_ms", 60_000) / 1000.0
+        for _, batcher in self._by_model.values():
+            if batcher is not None and hasattr(batcher, "wedged") \
+                    and batcher.wedged(timeout_s):
+                log.logger.error(
+                    "TP batcher wedged >%.0fs with work in flight — "
+                    "exiting so the launcher fails the group over", timeout_s)
+                self._wedge_exit(13)
+                return
+
+    @staticmethod
+    def _wedge_exit(code: int) -> None:   # overridable in tests
+        import os
+        os._exit(code)
 
     async def _resource_loop(self) -> None:
         interval = self.cfg.resource_monitor_interval_ms / 1000.0

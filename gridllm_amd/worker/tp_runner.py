@@ -31,6 +31,7 @@ from __future__ import annotations
 import asyncio
 import queue as _q
 import random
+import time as _time
 
 from gridllm_amd.bus.tcp import TcpBus
 from gridllm_amd.config import EngineConfig, WorkerConfig
@@ -69,6 +70,13 @@ class TPBatcher(Batcher):
         max_batch = max(1, self.engine.cfg.max_batch_size)
         live: list[tuple[_Pending, object]] = []   # (pending, engine row)
         while True:
+            # progress stamp for the TP wedge watchdog (agent heartbeat
+            # checks it): a hung collective blocks THIS thread while the
+            # asyncio loop keeps heartbeating — without the stamp the
+            # scheduler would keep routing jobs to a wedged group
+            # (SURVEY §7 hard-part 5: group failover must be atomic)
+            self.last_progress = _time.monotonic()
+            self.busy = bool(live)
             arrivals: list[_Pending] = []
             if not live:
                 # idle: block (followers are parked in their broadcast)

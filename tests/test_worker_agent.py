@@ -303,3 +303,43 @@ async def test_streaming_incremental_chunks():
         f"stream not incremental: {len(text_chunks)} text chunks"
     assert "".join(c["response"] for c in chunks) == res["response"]
     await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_tp_wedge_watchdog_exits_group():
+    """A TP batcher stuck mid-collective with work in flight must take
+    the worker down (launcher then kills the group; orphan recovery
+    requeues) — otherwise the asyncio loop keeps heartbeating for a
+    wedged group and the scheduler routes jobs into a black hole.
+    Simulated: fake-enabled TP + stale progress stamp."""
+    import time as _t
+    core = MemoryBusCore()
+    reg, sched, agent = await make_world(core)
+    engine, batcher = next(iter(agent._by_model.values()))
+
+    class _FakeTP:
+        enabled = True
+    exits = []
+    real_tp = engine.tp
+    agent._wedge_exit = lambda code: exits.append(code)
+    agent.cfg.tp_wedge_timeout_ms = 1000
+    try:
+        engine.tp = _FakeTP()
+        # healthy: busy but fresh stamp
+        batcher.busy = True
+        batcher.last_progress = _t.monotonic()
+        agent._check_wedge()
+        assert not exits
+        # idle + stale: no exit (nothing in flight, parked collectives ok)
+        batcher.busy = False
+        batcher.last_progress = _t.monotonic() - 999.0
+        agent._check_wedge()
+        assert not exits
+        # busy + stale: wedge -> exit
+        batcher.busy = True
+        agent._check_wedge()
+        assert exits == [13]
+    finally:
+        engine.tp = real_tp
+        batcher.busy = False
+    await teardown(reg, sched, agent)
