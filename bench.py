@@ -6,10 +6,13 @@ through gateway -> scheduler -> bus -> worker -> engine -> direct stream
 back (the reference contract at server/src/routes/ollama.ts:161-319).
 One rank per GPU hosts a worker (engine pinned to cuda:LOCAL_RANK); rank 0
 additionally spawns the scheduler server, N gateway replicas and N async
-load generators. A "step" is one wave of `--concurrency` streaming
-requests per GPU, each generating `--num-predict` tokens; K steps are
-timed between barriers and the whole-node tokens/s printed as ONE JSON
-line.
+load generators. Each load generator keeps a CONTINUOUSLY-REPLENISHED
+closed loop of `--concurrency` streaming requests in flight (a finished
+request is replaced at once); a "step" is the next `--concurrency`
+streamed completions per GPU, i.e. concurrency x num_predict output
+tokens — identical work per step, but without the drained-wave artifact
+that under-measured steady-state capacity. K steps are timed between
+barriers and the whole-node tokens/s printed as ONE JSON line.
 
 `--mode engine` preserves the raw engine-decode benchmark (hipGraph decode
 + batched sampling, no serving glue) for kernel-path comparisons.
@@ -55,6 +58,11 @@ uvicorn.run(app, host="127.0.0.1", port=cfg.port, log_level="error",
 """
 
 LOADGEN_SRC = r"""
+# Continuously-replenished closed loop at concurrency CONC: a finished
+# request is immediately replaced, so the server sees a steady CONC in
+# flight across step boundaries (no inter-wave drain — the drained-wave
+# design under-measured steady-state capacity by ~5%). One "STEP" =
+# the next CONC streamed completions (CONC * NPRED output tokens).
 import asyncio, json, os, statistics, sys, time
 import aiohttp
 
@@ -64,22 +72,27 @@ NPRED = int(os.environ["LG_NPREDICT"])
 MODEL = os.environ["LG_MODEL"]
 PROMPT = "benchmark prompt " + "lorem ipsum dolor " * 8
 
-async def wave(sess, seed0):
+async def main():
+    conn = aiohttp.TCPConnector(limit=CONC + 16)
+    seed0 = int(os.environ.get("LG_SEED", "0")) * 10_000_000
+    completions = []   # (eval_count, ttft_ms|None, errored) per finished req
+    consumed = [0]
+    cond = asyncio.Condition()
     sem = asyncio.Semaphore(CONC)
-    tokens = [0]
-    ttfts = []
-    errors = [0]
 
-    async def one(i):
-        async with sem:
+    async with aiohttp.ClientSession(base_url=BASE, connector=conn) as sess:
+
+        async def one(i):
             t0 = time.monotonic()
             first = None
+            n = 0
+            err = False
             try:
                 async with sess.post("/api/generate", json={
                         "model": MODEL,
-                        "prompt": f"req {seed0+i}: " + PROMPT,
+                        "prompt": f"req {i}: " + PROMPT,
                         "stream": True,
-                        "options": {"temperature": 0.8, "seed": seed0 + i,
+                        "options": {"temperature": 0.8, "seed": i,
                                     "num_predict": NPRED}},
                         timeout=aiohttp.ClientTimeout(total=600)) as resp:
                     async for raw in resp.content:
@@ -89,29 +102,53 @@ async def wave(sess, seed0):
                         if first is None and c.get("response"):
                             first = time.monotonic() - t0
                         if c.get("done"):
-                            tokens[0] += int(c.get("eval_count", 0))
+                            n = int(c.get("eval_count", 0))
             except Exception:
-                errors[0] += 1
-                return
-            if first is not None:
-                ttfts.append(first * 1000.0)
+                err = True
+            async with cond:
+                completions.append(
+                    (n, None if first is None else first * 1000.0, err))
+                cond.notify_all()
+            sem.release()
 
-    await asyncio.gather(*(one(i) for i in range(CONC)))
-    p50 = statistics.median(ttfts) if ttfts else 0.0
-    return tokens[0], p50, errors[0]
+        launched = [0]
 
-async def main():
-    conn = aiohttp.TCPConnector(limit=CONC + 16)
-    seed = int(os.environ.get("LG_SEED", "0")) * 1_000_000
-    async with aiohttp.ClientSession(base_url=BASE, connector=conn) as sess:
+        async def pump():
+            i = seed0
+            while True:
+                await sem.acquire()
+                i += 1
+                launched[0] += 1
+                asyncio.get_running_loop().create_task(one(i))
+
+        pump_task = asyncio.get_running_loop().create_task(pump())
         print("LOADGEN_READY", flush=True)
         loop = asyncio.get_running_loop()
         while True:
             line = await loop.run_in_executor(None, sys.stdin.readline)
             if not line or line.strip() == "QUIT":
+                # drain in-flight requests before exiting: tearing the
+                # in-process worker down mid-generation aborts at exit
+                pump_task.cancel()
+                try:
+                    async with cond:
+                        await asyncio.wait_for(
+                            cond.wait_for(
+                                lambda: len(completions) >= launched[0]),
+                            timeout=120)
+                except asyncio.TimeoutError:
+                    pass
+                print("DRAINED", flush=True)
                 return
-            toks, p50, errs = await wave(sess, seed)
-            seed += CONC
+            target = consumed[0] + CONC
+            async with cond:
+                await cond.wait_for(lambda: len(completions) >= target)
+            batch = completions[consumed[0]:target]
+            consumed[0] = target
+            toks = sum(n for n, _, _ in batch)
+            ttfts = [t for _, t, _ in batch if t is not None]
+            errs = sum(1 for _, _, e in batch if e)
+            p50 = statistics.median(ttfts) if ttfts else 0.0
             print(f"DONE {toks} {p50:.2f} {errs}", flush=True)
 
 asyncio.run(main())
@@ -332,6 +369,17 @@ def serving_mode(args, rank, world, local_rank, dist, device, on_gpu):
                     lg.stdin.flush()
                 except Exception:  # noqa: BLE001
                     pass
+            for lg in loadgens:   # wait for in-flight drain (see loadgen)
+                try:
+                    line = lg.stdout.readline()
+                    while line and "DRAINED" not in line:
+                        line = lg.stdout.readline()
+                except Exception:  # noqa: BLE001
+                    pass
+        # all ranks hold their in-process workers up until rank 0's
+        # loadgens have drained — exiting mid-generation aborts at exit
+        if dist is not None:
+            dist.barrier()
     finally:
         for p in procs:
             if p.poll() is None:
