@@ -12,6 +12,7 @@ from __future__ import annotations
 
 import threading
 import time
+from contextlib import nullcontext as _nullctx
 
 import numpy as np
 from dataclasses import dataclass, field
@@ -123,6 +124,13 @@ class InferenceEngine:
                                          device=self.device)
         self._graphs: dict[int, tuple] = {}
         self._graph_pool = None
+        # admission prefill runs on its own HIP stream: prefill is
+        # MFMA-bound while decode replay is HBM-bound, and at
+        # prompt≈num_predict the workload is ~half prefill FLOPs —
+        # serializing them on one stream left the decode stream idle for
+        # the whole prefill pass (the engine-vs-serving 2x gap)
+        self._pf_stream = torch.cuda.Stream(device=self.device) \
+            if self.is_gpu else None
         import os as _os
         self._no_emit_text = \
             _os.environ.get("GRIDLLM_NO_EMIT_TEXT", "0") == "1"
@@ -171,6 +179,10 @@ class InferenceEngine:
     def _ensure_graph(self, bucket: int):
         if bucket in self._graphs:
             return self._graphs[bucket]
+        if self._pf_stream is not None:
+            # a mid-serving capture must not race an in-flight side-stream
+            # admission prefill (capture invalidates concurrent work)
+            self._pf_stream.synchronize()
         tokens = torch.zeros(bucket, dtype=torch.int64, device=self.device)
         slots = torch.full((bucket,), self._scratch_slot, dtype=torch.int32,
                            device=self.device)
@@ -419,20 +431,34 @@ class InferenceEngine:
         that follows attaches this batch). Caller guarantees no
         constraint/penalty rows and a running pipeline (batcher checks)."""
         with self._lock:
-            rows, logits, slots_t = self._prefill_rows(items)
-            try:
-                ids_dev = self.sampler.sample_ids(logits, rows, slots_t)
-            except Exception:
-                self.abort_rows(rows, locked=True)
-                raise
-            b = len(rows)
-            if self.is_gpu:
-                ih = self._pinned(b)
-                ih[:b].copy_(ids_dev, non_blocking=True)
-            else:
-                ih = ids_dev
+            # Dedicated prefill stream: the whole admission pass (uploads,
+            # prefill kernels, first-token sample, D2H) overlaps the
+            # decode graph replays in flight on the main stream — the
+            # slots are fresh (no step references them until the first
+            # post-admission submit, which waits on ev below). Decode is
+            # HBM-bound and prefill MFMA-bound, so they genuinely share
+            # the chip rather than time-slicing it.
+            pf = self._pf_stream
+            ctx = torch.cuda.stream(pf) if pf is not None else _nullctx()
+            with ctx:
+                rows, logits, slots_t = self._prefill_rows(items)
+                try:
+                    ids_dev = self.sampler.sample_ids(logits, rows, slots_t)
+                except Exception:
+                    self.abort_rows(rows, locked=True)
+                    raise
+                b = len(rows)
+                ev = None
+                if self.is_gpu:
+                    ih = self._pinned(b)
+                    ih[:b].copy_(ids_dev, non_blocking=True)
+                    ev = torch.cuda.Event()
+                    ev.record()
+                else:
+                    ih = ids_dev
             self._admit_pending = {
                 "rows": rows, "ids_dev": ids_dev, "ids_host": ih,
+                "ev": ev,
                 "pos": {id(r): i for i, r in enumerate(rows)}}
             return rows
 
@@ -637,6 +663,13 @@ class InferenceEngine:
                         r.cur_len = self.model.shift_slot(r.slot)
             ap = getattr(self, "_admit_pending", None)
             self._admit_pending = None
+            if ap is not None and ap.get("ev") is not None:
+                # first step that includes the admitted rows: order the
+                # main stream after the side-stream prefill, and tell the
+                # allocator ids_dev is consumed here (torch.where below)
+                cur = torch.cuda.current_stream()
+                cur.wait_event(ap["ev"])
+                ap["ids_dev"].record_stream(cur)
             same = (ap is None and prev is not None
                     and len(prev["rows"]) == b
                     and all(a is c for a, c in zip(prev["rows"], rows)))
