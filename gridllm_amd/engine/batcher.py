@@ -200,11 +200,20 @@ class Batcher:
                 else:
                     hold.append(p)
             gens: list[_Pending] = []
+            # deep batches coalesce admissions harder: each prefill pass
+            # costs ms of host launches on this thread (decode submission
+            # pauses for it), so at live>=256 trade ~4x fewer passes for
+            # ~tens of ms of extra queueing — invisible against the
+            # saturated cohort wait, and the larger varlen prefill is
+            # also more MFMA-efficient
+            deep = len(live) >= 256
+            a_min = 32 if deep else self.admit_min
+            a_wait = (4 * self.admit_wait_s) if deep else self.admit_wait_s
             if hold and (not live
-                         or len(hold) >= self.admit_min
+                         or len(hold) >= a_min
                          or stop_now or self._stop.is_set()
                          or time.monotonic() - hold[0].ts
-                         >= self.admit_wait_s):
+                         >= a_wait):
                 gens, hold = hold, []
             if gens:
                 items = []
