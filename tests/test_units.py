@@ -118,3 +118,40 @@ def test_call_schema_shapes():
     assert "anyOf" in both and len(both["anyOf"]) == 2
     names = {br["properties"]["name"]["const"] for br in both["anyOf"]}
     assert names == {"a", "b"}
+
+
+def test_byte_tokenizer_pseudo_word_decode():
+    """Unassigned synthetic-vocab ids decode to deterministic visible
+    pseudo-words (real tokenizers decode every non-special id to text);
+    BOS/EOS stay invisible; raw-byte runs still decode byte-exactly and
+    mixed streams decode identically via decode() and the stream
+    decoder."""
+    from gridllm_amd.engine.tokenizer import ByteTokenizer, _pseudo_word
+    tok = ByteTokenizer(vocab_size=128256)
+    # determinism + visibility + distinctness
+    assert _pseudo_word(300) == _pseudo_word(300)
+    assert _pseudo_word(300) != _pseudo_word(301)
+    assert _pseudo_word(300).strip()
+    # specials invisible
+    assert tok.decode([tok.BOS, tok.EOS]) == ""
+    # byte-exact roundtrip for real text
+    ids = tok.encode("héllo wörld", bos=False)
+    assert tok.decode(ids) == "héllo wörld"
+    # mixed stream: decode() == concat of stream pushes + flush
+    mixed = [tok.BOS] + list("ab".encode()) + [5000, 99999] + \
+        list("é".encode()) + [tok.EOS, 270]
+    dec = tok.decode_stream()
+    stream = "".join(dec.push(i) for i in mixed) + dec.flush()
+    assert stream == tok.decode(mixed)
+    assert "ab" in stream and _pseudo_word(5000) in stream
+
+
+def test_batcher_deep_coalescing_thresholds():
+    """live>=256 quadruples the admission window and raises admit_min —
+    the constants the serving throughput depends on (r02am)."""
+    import inspect
+    from gridllm_amd.engine import batcher as B
+    src = inspect.getsource(B.Batcher._loop_continuous)
+    assert "len(live) >= 256" in src
+    assert "32 if deep" in src
+    assert "4 * self.admit_wait_s" in src
