@@ -450,7 +450,7 @@ class InferenceEngine:
                 b = len(rows)
                 ev = None
                 if self.is_gpu:
-                    ih = self._pinned(b)
+                    ih = self._pinned_admit(b)
                     ih[:b].copy_(ids_dev, non_blocking=True)
                     ev = torch.cuda.Event()
                     ev.record()
@@ -630,6 +630,21 @@ class InferenceEngine:
         cache = getattr(self, "_pin_ring", None)
         if cache is None:
             cache = self._pin_ring = {}
+        ring = cache.setdefault(n, [None, None, 0])
+        i = ring[2] = (ring[2] + 1) % 2
+        if ring[i] is None:
+            ring[i] = torch.empty(n, dtype=torch.int64, pin_memory=True)
+        return ring[i]
+
+    def _pinned_admit(self, n: int) -> torch.Tensor:
+        """Pinned readback buffers for ASYNC ADMISSIONS — a ring separate
+        from _pinned: an admission interleaved between two same-size steps
+        breaks the step ring's strict alternation, and the next step's
+        enqueued D2H can land in the buffer a collect is still reading
+        (silent token corruption when b_admit == b_step)."""
+        cache = getattr(self, "_pin_admit_ring", None)
+        if cache is None:
+            cache = self._pin_admit_ring = {}
         ring = cache.setdefault(n, [None, None, 0])
         i = ring[2] = (ring[2] + 1) % 2
         if ring[i] is None:
