@@ -192,12 +192,10 @@ class WorkerAgent:
         tears down the remaining ranks, heartbeats stop, and orphan
         recovery requeues the jobs on live workers. TP-only — a
         single-process engine stall is a GPU hang with its own handling."""
-        eng = next(iter(self._by_model.values()))[0] if self._by_model \
-            else None
-        if eng is None or not eng.tp.enabled:
-            return
         timeout_s = getattr(self.cfg, "tp_wedge_timeout_ms", 60_000) / 1000.0
-        for _, batcher in self._by_model.values():
+        for eng, batcher in self._by_model.values():
+            if not eng.tp.enabled:
+                continue
             if batcher is not None and hasattr(batcher, "wedged") \
                     and batcher.wedged(timeout_s):
                 log.logger.error(
