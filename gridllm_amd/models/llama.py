@@ -45,6 +45,7 @@ class LayerWeights:
     wdown: torch.Tensor      # [H, F_l]
     attn_norm: torch.Tensor  # [H]
     mlp_norm: torch.Tensor   # [H]
+    bqkv: object = None      # [(Hq_l + 2*Hkv_l) * D] Qwen2-family q/k/v bias
 
 
 class KVCache:
@@ -167,8 +168,17 @@ class LlamaModel:
             wgateup = torch.cat([wg[f0:f0 + self.f_l],
                                  wu[f0:f0 + self.f_l]]).contiguous()
             wdown = wd_full[:, f0:f0 + self.f_l].contiguous()
+            bqkv = None
+            if s.qkv_bias:
+                bq = self._randn(gen, s.heads * d).flatten()
+                bk = self._randn(gen, s.kv_heads * d).flatten()
+                bv = self._randn(gen, s.kv_heads * d).flatten()
+                bqkv = torch.cat([
+                    bq[q0 * d:(q0 + self.hq_l) * d],
+                    bk[k0 * d:(k0 + self.hkv_l) * d],
+                    bv[k0 * d:(k0 + self.hkv_l) * d]]).contiguous()
             self.layers.append(LayerWeights(
-                wqkv=wqkv, wo=wo, wgateup=wgateup, wdown=wdown,
+                wqkv=wqkv, wo=wo, wgateup=wgateup, wdown=wdown, bqkv=bqkv,
                 attn_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device),
                 mlp_norm=torch.ones(s.hidden, dtype=self.dtype, device=self.device)))
 
@@ -238,6 +248,10 @@ class LlamaModel:
                      p + "mlp.down_proj.weight",
                      p + "input_layernorm.weight",
                      p + "post_attention_layernorm.weight"}
+            if s.qkv_bias:
+                keys |= {p + "self_attn.q_proj.bias",
+                         p + "self_attn.k_proj.bias",
+                         p + "self_attn.v_proj.bias"}
         return keys
 
     def _apply_hf(self, t: dict) -> None:
@@ -271,6 +285,14 @@ class LlamaModel:
             lw.wdown = cv(t[p + "mlp.down_proj.weight"][:, f0:f0 + self.f_l])
             lw.attn_norm = cv(t[p + "input_layernorm.weight"])
             lw.mlp_norm = cv(t[p + "post_attention_layernorm.weight"])
+            if s.qkv_bias:
+                lw.bqkv = cv(torch.cat([
+                    t[p + "self_attn.q_proj.bias"]
+                    [q0 * d:(q0 + self.hq_l) * d],
+                    t[p + "self_attn.k_proj.bias"]
+                    [k0 * d:(k0 + self.hkv_l) * d],
+                    t[p + "self_attn.v_proj.bias"]
+                    [k0 * d:(k0 + self.hkv_l) * d]]))
 
     def param_bytes(self) -> int:
         total = self.embed.numel() + self.final_norm.numel()
@@ -280,6 +302,8 @@ class LlamaModel:
             for t in (lw.wqkv, lw.wo, lw.wgateup, lw.wdown, lw.attn_norm,
                       lw.mlp_norm):
                 total += t.numel()
+            if lw.bqkv is not None:
+                total += lw.bqkv.numel()
         return total * self.embed.element_size()
 
     def shift_slot(self, slot: int) -> int:
@@ -319,6 +343,8 @@ class LlamaModel:
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
             qkv = _linear(normed, lw.wqkv)
+            if lw.bqkv is not None:
+                qkv = qkv + lw.bqkv
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
                             self.cos, self.sin, pos, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)
@@ -386,6 +412,8 @@ class LlamaModel:
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
             qkv = _linear(normed, lw.wqkv)
+            if lw.bqkv is not None:
+                qkv = qkv + lw.bqkv
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
                             self.cos, self.sin, pos, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)
@@ -446,6 +474,8 @@ class LlamaModel:
         normed, _ = ops.rmsnorm(x, self.layers[0].attn_norm, s.norm_eps)
         for li, lw in enumerate(self.layers):
             qkv = _linear(normed, lw.wqkv)
+            if lw.bqkv is not None:
+                qkv = qkv + lw.bqkv
             q = ops.rope_kv(qkv, self.cache.k[li], self.cache.v[li],
                             self.cos, self.sin, pos_rope, slot_t,
                             self.hq_l, self.hkv_l, self.spec.head_dim)

@@ -25,6 +25,7 @@ class ModelSpec:
     norm_eps: float = 1e-5
     max_seq_len: int = 8192
     tie_embeddings: bool = True
+    qkv_bias: bool = False         # Qwen2-family: bias on q/k/v projections
 
     @property
     def params(self) -> int:
@@ -32,6 +33,8 @@ class ModelSpec:
         qkv = (self.heads + 2 * self.kv_heads) * self.head_dim * h
         o = self.heads * self.head_dim * h
         mlp = 3 * h * f
+        if self.qkv_bias:
+            qkv += (self.heads + 2 * self.kv_heads) * self.head_dim
         per_layer = qkv + o + mlp + 2 * h
         emb = v * h * (1 if self.tie_embeddings else 2)
         return self.layers * per_layer + emb + h
@@ -60,6 +63,24 @@ _reg(ModelSpec("llama3:70b", hidden=8192, layers=80, heads=64, kv_heads=8,
                head_dim=128, intermediate=28672, vocab=128256,
                tie_embeddings=False),
      "llama3.1:70b", "llama3-70b")
+# Qwen2.5 family (the reference's own CI model is qwen2.5:0.5b,
+# .github/workflows/integration-tests.yml:28): llama-style GQA +
+# SwiGLU with q/k/v projection BIAS and rope theta 1e6
+_reg(ModelSpec("qwen2.5:0.5b", hidden=896, layers=24, heads=14, kv_heads=2,
+               head_dim=64, intermediate=4864, vocab=151936,
+               rope_theta=1e6, norm_eps=1e-6, max_seq_len=32768,
+               qkv_bias=True),
+     "qwen2.5-0.5b", "qwen2.5:0.5b-instruct")
+_reg(ModelSpec("qwen2.5:1.5b", hidden=1536, layers=28, heads=12, kv_heads=2,
+               head_dim=128, intermediate=8960, vocab=151936,
+               rope_theta=1e6, norm_eps=1e-6, max_seq_len=32768,
+               qkv_bias=True),
+     "qwen2.5-1.5b")
+_reg(ModelSpec("qwen2.5:7b", hidden=3584, layers=28, heads=28, kv_heads=4,
+               head_dim=128, intermediate=18944, vocab=152064,
+               rope_theta=1e6, norm_eps=1e-6, max_seq_len=32768,
+               qkv_bias=True, tie_embeddings=False),
+     "qwen2.5-7b", "qwen2.5", "qwen2.5:latest")
 # tiny configs for CPU tests and CI (tiny2: a SECOND architecture so
 # mixed-fleet routing tests have two distinct models on CPU)
 _reg(ModelSpec("tiny", hidden=256, layers=2, heads=4, kv_heads=2, head_dim=64,
@@ -68,6 +89,10 @@ _reg(ModelSpec("tiny", hidden=256, layers=2, heads=4, kv_heads=2, head_dim=64,
 _reg(ModelSpec("tiny2", hidden=128, layers=2, heads=2, kv_heads=2,
                head_dim=64, intermediate=256, vocab=512, max_seq_len=256),
      "tiny2:latest")
+_reg(ModelSpec("tiny-qwen", hidden=128, layers=2, heads=4, kv_heads=2,
+               head_dim=32, intermediate=256, vocab=512, max_seq_len=256,
+               rope_theta=1e6, norm_eps=1e-6, qkv_bias=True),
+     "tiny-qwen:latest")
 
 
 def get_spec(name: str, max_seq_len: int = 0) -> ModelSpec:
