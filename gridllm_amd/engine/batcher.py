@@ -114,10 +114,12 @@ class Batcher:
     def _prep(self, p: _Pending) -> tuple[str, dict]:
         """Render the prompt + merge stop sequences for a generate/chat row."""
         if p.kind == "chat":
-            prompt = chat_mod.render_chat(p.messages or [], think=p.think)
+            fam = chat_mod.family_of(self.engine.spec.name)
+            prompt = chat_mod.render_chat(p.messages or [], think=p.think,
+                                          family=fam)
             opts = dict(p.options)
             opts["stop"] = list(opts.get("stop") or []) + \
-                list(chat_mod.CHAT_STOPS)
+                chat_mod.chat_stops(fam)
         else:
             prompt = p.prompt
             opts = dict(p.options)
@@ -336,11 +338,12 @@ class Batcher:
         stops_all: Optional[list[str]] = None
         for p in gens:
             if p.kind == "chat":
+                fam = chat_mod.family_of(self.engine.spec.name)
                 prompt = chat_mod.render_chat(p.messages or [],
-                                              think=p.think)
+                                              think=p.think, family=fam)
                 opts = dict(p.options)
                 opts["stop"] = list(opts.get("stop") or []) + \
-                    list(chat_mod.CHAT_STOPS)
+                    chat_mod.chat_stops(fam)
             else:
                 prompt = p.prompt
                 opts = dict(p.options)

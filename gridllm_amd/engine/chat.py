@@ -1,25 +1,47 @@
-"""Chat templating (Llama-3 header style).
+"""Chat templating, per model family.
 
 The reference's Ollama path flattens chat to "role: content" text
 (reference: server/src/routes/ollama.ts:366-370) while its OpenAI path does
 real chat (openai.ts:651-654). We template properly on BOTH paths
 (SURVEY.md §7 hard-part 6 recommendation) while keeping wire shapes intact.
-"""
+Families: Llama-3 header style (default) and ChatML (Qwen2.5 — the
+reference's CI model family)."""
 
 from __future__ import annotations
 
 from typing import Iterable
 
 
-def render_chat(messages: Iterable[dict], think: bool = False) -> str:
+def _content_text(m: dict) -> str:
+    content = m.get("content", "")
+    if isinstance(content, list):  # OpenAI multimodal split: text parts only
+        content = "\n".join(p.get("text", "") for p in content
+                            if isinstance(p, dict) and p.get("type") == "text")
+    return content
+
+
+def family_of(model_name: str) -> str:
+    n = (model_name or "").lower()
+    return "qwen" if "qwen" in n else "llama"
+
+
+def render_chat(messages: Iterable[dict], think: bool = False,
+                family: str = "llama") -> str:
+    if family == "qwen":
+        # ChatML (Qwen2/2.5): <|im_start|>role\ncontent<|im_end|>
+        parts = []
+        for m in messages:
+            parts.append(f"<|im_start|>{m.get('role', 'user')}\n"
+                         f"{_content_text(m)}<|im_end|>\n")
+        parts.append("<|im_start|>assistant\n")
+        if think:
+            parts.append(THINK_OPEN + "\n")
+        return "".join(parts)
     parts = ["<|begin_of_text|>"]
     for m in messages:
         role = m.get("role", "user")
-        content = m.get("content", "")
-        if isinstance(content, list):  # OpenAI multimodal split: text parts only
-            content = "\n".join(p.get("text", "") for p in content
-                                if isinstance(p, dict) and p.get("type") == "text")
-        parts.append(f"<|start_header_id|>{role}<|end_header_id|>\n\n{content}<|eot_id|>")
+        parts.append(f"<|start_header_id|>{role}<|end_header_id|>"
+                     f"\n\n{_content_text(m)}<|eot_id|>")
     parts.append("<|start_header_id|>assistant<|end_header_id|>\n\n")
     if think:
         # thinking mode (Ollama `think: true`, reference forwards it at
@@ -30,6 +52,12 @@ def render_chat(messages: Iterable[dict], think: bool = False) -> str:
 
 
 CHAT_STOPS = ["<|eot_id|>", "<|start_header_id|>"]
+_CHAT_STOPS_BY_FAMILY = {"llama": CHAT_STOPS,
+                         "qwen": ["<|im_end|>", "<|im_start|>"]}
+
+
+def chat_stops(family: str = "llama") -> list:
+    return list(_CHAT_STOPS_BY_FAMILY.get(family, CHAT_STOPS))
 THINK_OPEN = "<think>"
 THINK_CLOSE = "</think>"
 

@@ -330,9 +330,10 @@ class InferenceEngine:
              on_token: Optional[Callable[[str, bool], None]] = None,
              cancel: Optional[threading.Event] = None,
              think: bool = False) -> GenerationResult:
-        prompt = chat_mod.render_chat(messages, think=think)
+        fam = chat_mod.family_of(self.spec.name)
+        prompt = chat_mod.render_chat(messages, think=think, family=fam)
         return self.generate(prompt, options, on_token, cancel,
-                             raw_stops=list(chat_mod.CHAT_STOPS))
+                             raw_stops=chat_mod.chat_stops(fam))
 
     def embed(self, inputs: list[str]) -> list[list[float]]:
         out: list[list[float]] = []

@@ -155,3 +155,26 @@ def test_batcher_deep_coalescing_thresholds():
     assert "len(live) >= 256" in src
     assert "32 if deep" in src
     assert "4 * self.admit_wait_s" in src
+
+
+def test_chat_template_per_family():
+    """Qwen models chat via ChatML, Llama via header style; stop sets
+    follow the family (the reference delegated templating to Ollama —
+    our engines must template per architecture)."""
+    from gridllm_amd.engine import chat as c
+    msgs = [{"role": "system", "content": "s"},
+            {"role": "user", "content": "u"}]
+    q = c.render_chat(msgs, family="qwen")
+    assert q.startswith("<|im_start|>system\ns<|im_end|>")
+    assert q.endswith("<|im_start|>assistant\n")
+    assert "<|start_header_id|>" not in q
+    l = c.render_chat(msgs)  # noqa: E741
+    assert "<|start_header_id|>user<|end_header_id|>" in l
+    assert c.family_of("qwen2.5:0.5b") == "qwen"
+    assert c.family_of("tiny-qwen") == "qwen"
+    assert c.family_of("llama3.2:3b") == "llama"
+    assert "<|im_end|>" in c.chat_stops("qwen")
+    assert "<|eot_id|>" in c.chat_stops("llama")
+    # think mode opens the think block in both families
+    assert c.render_chat(msgs, think=True, family="qwen").endswith(
+        c.THINK_OPEN + "\n")
