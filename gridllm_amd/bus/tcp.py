@@ -332,7 +332,8 @@ class TcpBus(Bus):
         return resp.get("value")
 
     async def identify(self, worker_id: str) -> None:
-        self._identities.append(worker_id)
+        if worker_id not in self._identities:   # reconnect re-identifies
+            self._identities.append(worker_id)
         await self._call("identify", worker_id=worker_id)
 
     async def publish(self, channel: str, payload: dict) -> None:
@@ -358,7 +359,11 @@ class TcpBus(Bus):
         ordering the submit-then-await pattern needs. The awaited ack
         cost 2 RTTs per request at the gateway (subscribe result+stream)."""
         first = not self._handlers[channel]
-        self._handlers[channel].append(handler)
+        # dedupe: after a reconnect the agent re-runs start() and
+        # re-subscribes the same bound methods — appending again would
+        # dispatch every push twice (bound methods compare equal)
+        if handler not in self._handlers[channel]:
+            self._handlers[channel].append(handler)
         if first:
             await self._send_nf("subscribe", channel=channel)
 

@@ -119,6 +119,8 @@ class WorkerConfig:
             max_concurrent_tasks=_env_int("MAX_CONCURRENT_TASKS", 1, 1, 4096),
             device=_env_int("WORKER_DEVICE", 0, 0, 63),
             tp_wedge_timeout_ms=_env_int("TP_WEDGE_TIMEOUT", 60_000, 1000),
+            reconnect_max_attempts=_env_int("RECONNECT_MAX_ATTEMPTS", 10, 1),
+            reconnect_base_delay_ms=_env_int("RECONNECT_BASE_DELAY", 500, 10),
         )
 
 

@@ -399,3 +399,83 @@ def test_replica_stream_survives_worker_sigkill(replica_stack, request):
             if p.poll() is None:
                 p.send_signal(signal.SIGKILL)
                 p.wait(timeout=10)
+
+
+def test_worker_reconnects_after_bus_restart(tmp_path):
+    """Kill -9 the standalone bus under a registered worker; restart the
+    bus (same snapshot) and the server: the worker's reconnect loop
+    (worker/main.py run_worker, exponential backoff — the reference's
+    RedisConnectionManager policy, C4) re-registers WITHOUT duplicate
+    subscriptions (bus/tcp.py subscribe dedupes bound methods) and the
+    fleet serves again."""
+    import httpx
+    bus_port, http_port = _free_port(), _free_port()
+    snap = str(tmp_path / "bus-rc.snap")
+    env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port)}
+    bus_args = [sys.executable, "-m", "gridllm_amd.bus",
+                "--port", str(bus_port), "--snapshot", snap,
+                "--snapshot-interval-ms", "200"]
+    procs = {}
+    procs["bus"] = _spawn(bus_args, env, "BUS_READY")
+    try:
+        procs["server"] = _spawn([sys.executable, "-u", "-c", SERVER_SRC],
+                                 env, "SERVER_READY")
+        wenv = {**os.environ, **env, "PYTHONPATH": REPO,
+                "ENGINE_MODEL": "tiny", "ENGINE_DEVICE": "cpu",
+                "ENGINE_DTYPE": "fp32", "BUS_HOST": "127.0.0.1",
+                "WORKER_ID": "w-reconnect", "WORKER_PORT": str(_free_port()),
+                "RECONNECT_BASE_DELAY": "100"}
+        procs["worker"] = subprocess.Popen(
+            [sys.executable, "-m", "gridllm_amd.worker"], env=wenv,
+            stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
+        base = f"http://127.0.0.1:{http_port}"
+        with httpx.Client(base_url=base, timeout=30) as c:
+            for _ in range(150):
+                try:
+                    r = c.get("/health/workers")
+                    if r.status_code == 200 and \
+                            r.json()["counts"]["total"] == 1:
+                        break
+                except Exception:  # noqa: BLE001
+                    pass
+                time.sleep(0.2)
+            else:
+                raise TimeoutError("worker never registered")
+            r = c.post("/api/generate", json={
+                "model": "tiny", "prompt": "before", "stream": False,
+                "options": {"num_predict": 3}})
+            assert r.status_code == 200 and r.json()["eval_count"] == 3
+
+        # kill the BUS (and the server whose connection died with it)
+        procs["bus"].send_signal(signal.SIGKILL)
+        procs["bus"].wait(timeout=10)
+        procs["server"].send_signal(signal.SIGKILL)
+        procs["server"].wait(timeout=10)
+        time.sleep(0.5)
+        procs["bus"] = _spawn(bus_args, env, "BUS_READY")
+        procs["server"] = _spawn([sys.executable, "-u", "-c", SERVER_SRC],
+                                 env, "SERVER_READY")
+
+        with httpx.Client(base_url=base, timeout=30) as c:
+            for _ in range(200):       # worker backoff + re-register
+                try:
+                    r = c.get("/health/workers")
+                    if r.status_code == 200 and \
+                            r.json()["counts"]["total"] == 1:
+                        break
+                except Exception:  # noqa: BLE001
+                    pass
+                time.sleep(0.2)
+            else:
+                raise TimeoutError("worker did not re-register")
+            r = c.post("/api/generate", json={
+                "model": "tiny", "prompt": "after reconnect",
+                "stream": False, "options": {"num_predict": 3}})
+            assert r.status_code == 200 and r.json()["eval_count"] == 3, \
+                r.text
+    finally:
+        for p in procs.values():
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs.values():
+            p.wait(timeout=10)
