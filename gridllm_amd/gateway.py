@@ -38,6 +38,12 @@ class GatewayReplica:
         self._stats_task = None
 
     async def start(self) -> None:
+        # bus reconnect supervisor: a restarted standalone bus must not
+        # strand detached replicas (the worker has the same loop in
+        # worker/main.py; in-flight requests fail, new ones serve —
+        # TcpBus.connect re-subscribes surviving handler channels)
+        disconnected = asyncio.Event()
+        self.bus.on_disconnect = disconnected.set
         await self.bus.connect()
         await self.stream_server.start()
         await self.registry.start()
@@ -45,15 +51,36 @@ class GatewayReplica:
 
         async def _stats_loop():
             while True:
-                await self.scheduler._refresh_stats()
+                try:
+                    await self.scheduler._refresh_stats()
+                except Exception:  # noqa: BLE001 — bus down: supervisor
+                    pass           # below reconnects; stats resume then
                 await asyncio.sleep(1.0)
 
+        async def _reconnect_loop():
+            while True:
+                await disconnected.wait()
+                disconnected.clear()
+                log.logger.warning("bus connection lost; reconnecting")
+                delay = 0.5
+                while True:
+                    try:
+                        await self.bus.connect()
+                        log.worker("bus reconnected")
+                        break
+                    except OSError:
+                        await asyncio.sleep(delay)
+                        delay = min(10.0, delay * 2)
+
+        self._reconnect_task = loop.create_task(_reconnect_loop())
         self._stats_task = loop.create_task(_stats_loop())
         log.worker("gateway replica started", port=self.cfg.port,
                    bus=f"{self.cfg.bus_host}:{self.cfg.bus_port}",
                    stream_addr=self.stream_server.addr)
 
     async def shutdown(self) -> None:
+        if getattr(self, "_reconnect_task", None):
+            self._reconnect_task.cancel()
         if self._stats_task:
             self._stats_task.cancel()
         await self.registry.stop()

@@ -479,3 +479,69 @@ def test_worker_reconnects_after_bus_restart(tmp_path):
                 p.send_signal(signal.SIGKILL)
         for p in procs.values():
             p.wait(timeout=10)
+
+
+def test_gateway_replica_survives_bus_restart(tmp_path):
+    """Kill -9 the standalone bus under a detached gateway replica;
+    restart bus + server: the replica's reconnect supervisor
+    (gateway.py) re-attaches and serves — the full durable-fleet story
+    (bus restart -> workers AND replicas reconnect, no process restarts
+    except the bus-owning server)."""
+    import httpx
+    bus_port, http_port, gw_port = _free_port(), _free_port(), _free_port()
+    snap = str(tmp_path / "bus-gw.snap")
+    env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port)}
+    bus_args = [sys.executable, "-m", "gridllm_amd.bus",
+                "--port", str(bus_port), "--snapshot", snap,
+                "--snapshot-interval-ms", "200"]
+    procs = {}
+    procs["bus"] = _spawn(bus_args, env, "BUS_READY")
+    try:
+        procs["server"] = _spawn([sys.executable, "-u", "-c", SERVER_SRC],
+                                 env, "SERVER_READY")
+        wenv = {**os.environ, **env, "PYTHONPATH": REPO,
+                "ENGINE_MODEL": "tiny", "ENGINE_DEVICE": "cpu",
+                "ENGINE_DTYPE": "fp32", "BUS_HOST": "127.0.0.1",
+                "WORKER_ID": "w-gwrc", "WORKER_PORT": str(_free_port()),
+                "RECONNECT_BASE_DELAY": "100"}
+        procs["worker"] = subprocess.Popen(
+            [sys.executable, "-m", "gridllm_amd.worker"], env=wenv,
+            stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
+        procs["gateway"] = _spawn(
+            [sys.executable, "-m", "gridllm_amd.gateway",
+             "--port", str(gw_port), "--host", "127.0.0.1",
+             "--bus", f"127.0.0.1:{bus_port}"], env, "GATEWAY_READY")
+        base = f"http://127.0.0.1:{gw_port}"
+
+        def wait_serving(msg):
+            with httpx.Client(base_url=base, timeout=30) as c:
+                for _ in range(200):
+                    try:
+                        r = c.post("/api/generate", json={
+                            "model": "tiny", "prompt": "p",
+                            "stream": False,
+                            "options": {"num_predict": 2}})
+                        if r.status_code == 200 and \
+                                r.json().get("eval_count") == 2:
+                            return
+                    except Exception:  # noqa: BLE001
+                        pass
+                    time.sleep(0.2)
+            raise TimeoutError(msg)
+
+        wait_serving("replica never served before restart")
+        procs["bus"].send_signal(signal.SIGKILL)
+        procs["bus"].wait(timeout=10)
+        procs["server"].send_signal(signal.SIGKILL)
+        procs["server"].wait(timeout=10)
+        time.sleep(0.5)
+        procs["bus"] = _spawn(bus_args, env, "BUS_READY")
+        procs["server"] = _spawn([sys.executable, "-u", "-c", SERVER_SRC],
+                                 env, "SERVER_READY")
+        wait_serving("replica did not recover after bus restart")
+    finally:
+        for p in procs.values():
+            if p.poll() is None:
+                p.send_signal(signal.SIGKILL)
+        for p in procs.values():
+            p.wait(timeout=10)
