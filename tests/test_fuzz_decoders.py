@@ -40,3 +40,35 @@ def test_byte_stream_decode_equals_batch_bytes_heavy(ids):
     dec = tok.decode_stream()
     out = "".join(dec.push(i) for i in ids) + dec.flush()
     assert out == tok.decode(ids)
+
+
+def test_hf_stream_decoder_fuzz_roundtrip(tmp_path_factory):
+    """HF span decoder: for any real text, streaming the encoded ids
+    reassembles the batch decode exactly (code points split across BPE
+    tokens included)."""
+    from tokenizers import (Tokenizer, decoders, models, pre_tokenizers,
+                            trainers)
+    tok = Tokenizer(models.BPE())
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    tr = trainers.BpeTrainer(
+        vocab_size=300, special_tokens=["<|end_of_text|>"],
+        initial_alphabet=pre_tokenizers.ByteLevel.alphabet())
+    tok.train_from_iterator(["hello wörld ✓ 日本語 {json: true}"] * 10, tr)
+    path = str(tmp_path_factory.mktemp("tokf") / "tokenizer.json")
+    tok.save(path)
+    from gridllm_amd.engine.tokenizer import HFTokenizer
+    hf = HFTokenizer(path)
+
+    from hypothesis import given as _given, settings as _settings
+    from hypothesis import strategies as _st
+
+    @_settings(max_examples=150, deadline=None)
+    @_given(_st.text(max_size=60))
+    def check(text):
+        ids = hf.encode(text)
+        dec = hf.decode_stream()
+        out = "".join(dec.push(i) for i in ids) + dec.flush()
+        assert out == hf.decode(ids)
+
+    check()
