@@ -52,11 +52,15 @@ class ByteTokenizer:
             if 0 <= i < 256:
                 run.append(i)
                 continue
+            if 256 <= i < 258:
+                continue               # BOS/EOS: text-invisible, and they
+                                       # must NOT split a byte run (stream
+                                       # decode equivalence — fuzz found
+                                       # [0xC2, BOS, 0x80] diverging)
             if run:
                 parts.append(bytes(run).decode("utf-8", errors="replace"))
                 run = []
-            if i >= 258:               # BOS/EOS stay invisible
-                parts.append(_pseudo_word(i))
+            parts.append(_pseudo_word(i))
         if run:
             parts.append(bytes(run).decode("utf-8", errors="replace"))
         return "".join(parts)
