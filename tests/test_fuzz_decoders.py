@@ -72,3 +72,33 @@ def test_hf_stream_decoder_fuzz_roundtrip(tmp_path_factory):
         assert out == hf.decode(ids)
 
     check()
+
+
+@settings(max_examples=400, deadline=None)
+@given(st.text(alphabet="<think/>abc\n ", max_size=60),
+       st.lists(st.integers(min_value=1, max_value=8), max_size=20))
+def test_think_splitter_chunking_invariant(text, cuts):
+    """ThinkSplitter: any chunking of the same text yields the same
+    (thinking, content) split as the batch split_think — the holdback
+    window for a partial close tag must not leak or drop characters.
+    Alphabet is tag-dense to hit partial-tag boundaries constantly."""
+    from gridllm_amd.engine.chat import ThinkSplitter, split_think
+    sp = ThinkSplitter()
+    th, co = [], []
+    pos = 0
+    for c in cuts:
+        piece, pos = text[pos:pos + c], pos + c
+        t, x = sp.push(piece)
+        th.append(t)
+        co.append(x)
+    t, x = sp.push(text[pos:])
+    th.append(t)
+    co.append(x)
+    t, x = sp.flush()
+    th.append(t)
+    co.append(x)
+    # both consume the GENERATION only (the opening tag is part of the
+    # prompt; generation starts inside the think block)
+    want_th, want_co = split_think(text)
+    assert "".join(th) == want_th, (text, cuts)
+    assert "".join(co) == want_co, (text, cuts)
