@@ -108,3 +108,33 @@ def test_random_walk_produces_conforming_json(schema, seed):
             return                     # pathological growth: skip sample
     doc = json.loads(bytes(out).decode("utf-8", errors="strict"))
     assert _conforms(doc, schema), (doc, schema)
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.integers(0, 2**32 - 1))
+def test_plain_json_constraint_random_walk(seed):
+    """format="json" (syntactic automaton): every random legal-byte walk
+    that reaches EOS parses as JSON."""
+    from gridllm_amd.engine.constraints import JsonByteConstraint
+    rng = random.Random(seed)
+    c = JsonByteConstraint()
+    out = bytearray()
+    for _ in range(400):
+        allowed, eos_ok = c.allowed()
+        if eos_ok and (not allowed or rng.random() < 0.3):
+            break
+        assert allowed, f"wedged after {bytes(out)[-40:]!r}"
+        b = rng.choice(sorted(allowed))
+        c.push(b)
+        out.append(b)
+    else:
+        for _ in range(2000):
+            allowed, eos_ok = c.allowed()
+            if eos_ok:
+                break
+            b = min(allowed)
+            c.push(b)
+            out.append(b)
+        if not c.allowed()[1]:
+            return
+    json.loads(bytes(out).decode("utf-8", errors="replace"))
