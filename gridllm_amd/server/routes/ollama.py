@@ -45,6 +45,11 @@ def _now_iso() -> str:
 def _clean_options(options: Optional[dict]) -> dict:
     if not isinstance(options, dict):
         return {}
+    from gridllm_amd.types import SamplingOptions
+    err = SamplingOptions.validate(options)
+    if err:                        # Ollama 400s wrongly-typed options at
+        raise HTTPException(       # unmarshal; without this the junk
+            status_code=400, detail=err)   # surfaced as an engine error
     out = {k: v for k, v in options.items() if k in _SAMPLING_KEYS}
     stop = out.get("stop")
     if isinstance(stop, str):

@@ -43,6 +43,11 @@ def _map_options(body: dict) -> dict:
         out["presence_penalty"] = body["presence_penalty"]
     if body.get("frequency_penalty") is not None:
         out["frequency_penalty"] = body["frequency_penalty"]
+    from gridllm_amd.types import SamplingOptions
+    err = SamplingOptions.validate(out)
+    if err:                        # OpenAI 400s invalid param types
+        from fastapi import HTTPException
+        raise HTTPException(status_code=400, detail=err)
     return out
 
 

@@ -74,6 +74,38 @@ class SamplingOptions:
         known = {f for f in cls.__dataclass_fields__}  # type: ignore[attr-defined]
         return cls(**{k: v for k, v in d.items() if k in known})
 
+    _FLOATS = ("temperature", "top_p", "min_p", "repeat_penalty",
+               "presence_penalty", "frequency_penalty")
+    _INTS = ("top_k", "seed", "num_predict", "num_ctx")
+
+    @classmethod
+    def validate(cls, d: Optional[dict]) -> Optional[str]:
+        """Type-check a wire options dict; returns an error message or
+        None. The gateway turns this into a 400 — Ollama rejects
+        wrongly-typed options at unmarshal time, and without this check
+        a junk value (temperature: "hot") surfaced as an opaque
+        engine-side TypeError in the job error."""
+        if d is None:
+            return None
+        if not isinstance(d, dict):
+            return "options must be an object"
+        for k in cls._FLOATS:
+            v = d.get(k)
+            if v is not None and (isinstance(v, bool)
+                                  or not isinstance(v, (int, float))):
+                return f"option {k!r} must be a number"
+        for k in cls._INTS:
+            v = d.get(k)
+            if v is not None and (isinstance(v, bool)
+                                  or not isinstance(v, int)):
+                return f"option {k!r} must be an integer"
+        stop = d.get("stop")
+        if stop is not None and not isinstance(stop, str) and not (
+                isinstance(stop, list)
+                and all(isinstance(s, str) for s in stop)):
+            return "option 'stop' must be a string or list of strings"
+        return None
+
     def to_dict(self) -> dict:
         return {k: v for k, v in asdict(self).items() if v is not None}
 

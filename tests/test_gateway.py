@@ -664,3 +664,24 @@ def test_default_num_predict_is_ctx_bounded(client):
     # decode until the ctx is full unless EOS was (legitimately) sampled
     if body["done_reason"] == "length":
         assert body["eval_count"] >= 100, body["eval_count"]
+
+
+def test_invalid_option_types_rejected_400(client):
+    """Wrongly-typed sampling options 400 at the gateway (Ollama rejects
+    them at unmarshal; they must not surface as opaque engine errors)."""
+    for opts in ({"temperature": "hot"}, {"num_predict": "many"},
+                 {"top_k": [1]}, {"seed": "x"}, {"stop": 5},
+                 {"top_p": {"a": 1}}, {"repeat_penalty": "no"},
+                 {"num_predict": True}):
+        r = client.post("/api/generate", json={
+            "model": "tiny", "prompt": "p", "stream": False,
+            "options": opts})
+        assert r.status_code == 400, (opts, r.status_code, r.text)
+    r = client.post("/v1/completions", json={
+        "model": "tiny", "prompt": "p", "temperature": "hot"})
+    assert r.status_code == 400, r.text
+    # valid ones still pass
+    r = client.post("/api/generate", json={
+        "model": "tiny", "prompt": "p", "stream": False,
+        "options": {"temperature": 0.5, "num_predict": 3, "stop": ["zz"]}})
+    assert r.status_code == 200, r.text
