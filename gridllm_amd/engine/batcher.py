@@ -219,9 +219,22 @@ class Batcher:
                 gens, hold = hold, []
             if gens:
                 items = []
+                ok_gens = []
                 for p in gens:
-                    prompt, opts = self._prep(p)
+                    # a malformed request (e.g. junk chat messages) must
+                    # fail ITS OWN future — an uncaught render error here
+                    # killed the batcher thread and turned the worker into
+                    # a heartbeating black hole (probe r02: every later
+                    # request hung)
+                    try:
+                        prompt, opts = self._prep(p)
+                    except Exception as e:  # noqa: BLE001
+                        if not p.future.done():
+                            p.future.set_exception(e)
+                        continue
+                    ok_gens.append(p)
                     items.append((prompt, opts, p.on_token, p.cancel, None))
+                gens = ok_gens
                 # async admission (no first-token sync) only while the
                 # pipeline is running and every new row will stay
                 # pipeline-eligible: no grammar constraint, no penalties
@@ -335,23 +348,21 @@ class Batcher:
         if not gens:
             return
         reqs, on_tokens, cancels = [], [], []
-        stops_all: Optional[list[str]] = None
+        ok_gens = []
         for p in gens:
-            if p.kind == "chat":
-                fam = chat_mod.family_of(self.engine.spec.name)
-                prompt = chat_mod.render_chat(p.messages or [],
-                                              think=p.think, family=fam)
-                opts = dict(p.options)
-                opts["stop"] = list(opts.get("stop") or []) + \
-                    chat_mod.chat_stops(fam)
-            else:
-                prompt = p.prompt
-                opts = dict(p.options)
-                if p.extra_stops:
-                    opts["stop"] = list(opts.get("stop") or []) + p.extra_stops
+            try:
+                prompt, opts = self._prep(p)
+            except Exception as e:  # noqa: BLE001 — malformed request:
+                if not p.future.done():   # fail it, keep the loop alive
+                    p.future.set_exception(e)
+                continue
+            ok_gens.append(p)
             reqs.append((prompt, opts))
             on_tokens.append(p.on_token)
             cancels.append(p.cancel)
+        gens = ok_gens
+        if not gens:
+            return
         try:
             results = self.engine.generate_batch(reqs, on_tokens=on_tokens,
                                                  cancels=cancels)

@@ -214,6 +214,14 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         if not isinstance(messages, list) or not messages:
             raise HTTPException(status_code=400,
                                 detail="messages must be a non-empty array")
+        for m in messages:             # Joi parity (ollama.ts:60-117): a
+            if not isinstance(m, dict) or \
+                    not isinstance(m.get("role", ""), str) or \
+                    not isinstance(m.get("content", ""), (str, list)):
+                raise HTTPException(   # junk message must 400 here, not
+                    status_code=400,   # crash inside the worker
+                    detail="each message needs a string role and "
+                           "string/array content")
         _validate_model(registry, model)
         stream = bool(body.get("stream", True))
         opts = _clean_options(body.get("options"))

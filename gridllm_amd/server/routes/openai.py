@@ -246,6 +246,15 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         if not isinstance(messages, list) or not messages:
             raise HTTPException(status_code=400,
                                 detail="messages must be a non-empty array")
+        for m in messages:
+            if not isinstance(m, dict) or \
+                    not isinstance(m.get("role", ""), str) or \
+                    not isinstance(m.get("content", ""),
+                                   (str, list, type(None))):
+                raise HTTPException(
+                    status_code=400,
+                    detail="each message needs a string role and "
+                           "string/array content")
         stream = bool(body.get("stream", False))
         cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         created = int(time.time())

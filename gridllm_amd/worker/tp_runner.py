@@ -105,9 +105,17 @@ class TPBatcher(Batcher):
                 else:
                     gens.append(p)
             items = []
+            ok_gens = []
             for p in gens:
-                prompt, opts = self._prep(p)
+                try:
+                    prompt, opts = self._prep(p)
+                except Exception as e:  # noqa: BLE001 — malformed
+                    if not p.future.done():   # request: fail it alone
+                        p.future.set_exception(e)
+                    continue
+                ok_gens.append(p)
                 items.append((prompt, opts))
+            gens = ok_gens
             _inject_seeds(items)
             embed_inputs = [p.inputs or [] for p in embeds]
             self._broadcast_iter(items, embed_inputs,

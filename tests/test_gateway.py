@@ -685,3 +685,46 @@ def test_invalid_option_types_rejected_400(client):
         "model": "tiny", "prompt": "p", "stream": False,
         "options": {"temperature": 0.5, "num_predict": 3, "stop": ["zz"]}})
     assert r.status_code == 200, r.text
+
+
+def test_malformed_chat_messages_rejected_and_worker_survives(client):
+    """Junk message shapes 400 at the gateway; and even a shape that
+    slips to the engine (defense tested via batcher isolation) must not
+    kill the worker — the batcher previously DIED on a render error,
+    turning the worker into a heartbeating black hole (every later
+    request hung)."""
+    for msgs in ("notalist", [], [5], [{"role": [], "content": 7}],
+                 [{"role": "user", "content": {"x": 1}}]):
+        r = client.post("/api/chat", json={
+            "model": "tiny", "messages": msgs, "stream": False,
+            "options": {"num_predict": 3}})
+        assert r.status_code == 400, (msgs, r.status_code, r.text)
+        r = client.post("/v1/chat/completions", json={
+            "model": "tiny", "messages": msgs})
+        assert r.status_code == 400, (msgs, r.status_code, r.text)
+    # gateway-accepted shapes still serve afterwards
+    r = client.post("/api/chat", json={
+        "model": "tiny", "stream": False,
+        "messages": [{"role": "user", "content": "hi"}],
+        "options": {"num_predict": 3}})
+    assert r.status_code == 200, r.text
+
+
+def test_batcher_survives_render_failure():
+    """Direct batcher-level check of the isolation: a _Pending whose
+    render raises fails ITS future; the next request still serves."""
+    from gridllm_amd.config import EngineConfig
+    from gridllm_amd.engine import InferenceEngine
+    from gridllm_amd.engine.batcher import Batcher
+    eng = InferenceEngine(EngineConfig(
+        model="tiny", device="cpu", dtype="fp32", max_seq_len=128,
+        max_batch_size=2, use_hipgraph=False))
+    b = Batcher(eng)
+    try:
+        bad = b.submit_chat([42], {"num_predict": 2})      # int message
+        with pytest.raises(Exception):
+            bad.result(timeout=10)
+        ok = b.submit_generate("still alive", {"num_predict": 2})
+        assert ok.result(timeout=10).eval_count == 2
+    finally:
+        b.shutdown()
