@@ -23,6 +23,9 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         body = await request.json()
         model = body.get("model", "")
         prompt = body.get("prompt", "")
+        if not isinstance(prompt, str):
+            raise HTTPException(status_code=400,
+                                detail="prompt must be a string")
         if not model or not prompt:
             raise HTTPException(status_code=400,
                                 detail="model and prompt are required")

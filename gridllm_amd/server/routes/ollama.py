@@ -365,6 +365,12 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         inp = body.get("input")
         if inp is None:
             raise HTTPException(status_code=400, detail="input is required")
+        if not isinstance(inp, str) and not (
+                isinstance(inp, list)
+                and all(isinstance(x, str) for x in inp)):
+            raise HTTPException(
+                status_code=400,
+                detail="input must be a string or array of strings")
         inputs = [inp] if isinstance(inp, str) else list(inp)
         _validate_model(registry, model)
         req = InferenceRequest(model=model, request_type="embedding",
@@ -386,6 +392,9 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         prompt = body.get("prompt")
         if prompt is None:
             raise HTTPException(status_code=400, detail="prompt is required")
+        if not isinstance(prompt, str):
+            raise HTTPException(status_code=400,
+                                detail="prompt must be a string")
         _validate_model(registry, model)
         req = InferenceRequest(model=model, request_type="embedding",
                                input=[prompt], timeout_ms=300_000)

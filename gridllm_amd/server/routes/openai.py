@@ -38,7 +38,15 @@ def _map_options(body: dict) -> dict:
         out["seed"] = body["seed"]
     if body.get("stop") is not None:
         stop = body["stop"]
-        out["stop"] = [stop] if isinstance(stop, str) else list(stop)
+        if isinstance(stop, str):
+            out["stop"] = [stop]
+        elif isinstance(stop, list):
+            out["stop"] = list(stop)
+        else:
+            from fastapi import HTTPException
+            raise HTTPException(
+                status_code=400,
+                detail="stop must be a string or array of strings")
     if body.get("presence_penalty") is not None:
         out["presence_penalty"] = body["presence_penalty"]
     if body.get("frequency_penalty") is not None:
@@ -186,6 +194,10 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         prompt = body.get("prompt", "")
         if isinstance(prompt, list):
             prompt = prompt[0] if prompt else ""
+        if not isinstance(prompt, str):
+            raise HTTPException(status_code=400,
+                                detail="prompt must be a string or "
+                                       "array of strings")
         stream = bool(body.get("stream", False))
         echo = bool(body.get("echo", False))
         cid = f"cmpl-{uuid.uuid4().hex[:24]}"

@@ -728,3 +728,25 @@ def test_batcher_survives_render_failure():
         assert ok.result(timeout=10).eval_count == 2
     finally:
         b.shutdown()
+
+
+def test_junk_payload_types_rejected_across_routes(client):
+    """Adversarial payload sweep: junk types 400 at every route instead
+    of surfacing as 500 engine errors (or crashing in the handler)."""
+    cases = [
+        ("/api/embed", {"model": "tiny", "input": 5}),
+        ("/api/embed", {"model": "tiny", "input": [5]}),
+        ("/api/embeddings", {"model": "tiny", "prompt": {"a": 1}}),
+        ("/inference", {"model": "tiny", "prompt": 5}),
+        ("/v1/completions", {"model": "tiny", "prompt": 5}),
+        ("/v1/completions", {"model": "tiny", "prompt": "p", "stop": 7}),
+    ]
+    for path, body in cases:
+        r = client.post(path, json=body)
+        assert r.status_code == 400, (path, body, r.status_code, r.text)
+    # the valid shapes still work
+    r = client.post("/api/embed", json={"model": "tiny", "input": ["ok"]})
+    assert r.status_code == 200, r.text
+    r = client.post("/v1/completions", json={
+        "model": "tiny", "prompt": ["p"], "max_tokens": 3})
+    assert r.status_code == 200, r.text
