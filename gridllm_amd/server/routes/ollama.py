@@ -132,6 +132,15 @@ def _stream_ndjson(scheduler: JobScheduler, req: InferenceRequest,
                     break
         finally:
             task.cancel()
+            if not task.done():
+                # client disconnected mid-stream: abort the job on the
+                # worker too, or it keeps generating to completion (the
+                # SSE path got this in round 1 — advisor finding — but
+                # the NDJSON path did not)
+                try:
+                    await scheduler.cancel_job(req.id)
+                except Exception:  # noqa: BLE001
+                    pass
 
     return StreamingResponse(gen(), media_type="application/x-ndjson")
 

@@ -545,3 +545,40 @@ def test_gateway_replica_survives_bus_restart(tmp_path):
                 p.send_signal(signal.SIGKILL)
         for p in procs.values():
             p.wait(timeout=10)
+
+
+def test_ndjson_client_disconnect_cancels_job(durable_stack):
+    """Abruptly close the TCP connection mid-/api/generate NDJSON
+    stream: the gateway must cancel the job on the worker (SSE got this
+    in round 1; the NDJSON path kept generating to completion). The
+    fixture worker decodes at 0.25 s/step, so an uncancelled job holds
+    active_jobs for minutes."""
+    import socket as _socket
+    base = durable_stack["base"]
+    port = durable_stack["http_port"]
+    body = json.dumps({"model": "tiny", "prompt": "disconnect probe",
+                       "stream": True,
+                       "options": {"temperature": 0.8, "seed": 9,
+                                   "num_predict": 400}})
+    req = ("POST /api/generate HTTP/1.1\r\nHost: x\r\n"
+           "Content-Type: application/json\r\n"
+           f"Content-Length: {len(body)}\r\n"
+           "Connection: close\r\n\r\n" + body)
+    s = _socket.create_connection(("127.0.0.1", port))
+    s.sendall(req.encode())
+    buf = b""
+    while b'"response"' not in buf:          # first streamed chunk
+        d = s.recv(65536)
+        assert d, f"stream ended early: {buf[:200]!r}"
+        buf += d
+    s.close()                                # walk away mid-stream
+    import httpx
+    with httpx.Client(base_url=base, timeout=30) as c:
+        for _ in range(100):                 # cancel beats the 100 s
+            r = c.get("/inference/queue")    # generation by a wide margin
+            if r.status_code == 200 and \
+                    (r.json().get("queue") or {}).get("active") == 0:
+                return
+            time.sleep(0.2)
+    raise AssertionError("job not cancelled after client disconnect: "
+                         + r.text[:200])

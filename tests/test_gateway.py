@@ -9,6 +9,7 @@ against golden Ollama/OpenAI response shapes, with no live Ollama needed.
 import asyncio
 import json
 import threading
+import time
 
 import pytest
 from fastapi.testclient import TestClient
@@ -750,3 +751,4 @@ def test_junk_payload_types_rejected_across_routes(client):
     r = client.post("/v1/completions", json={
         "model": "tiny", "prompt": ["p"], "max_tokens": 3})
     assert r.status_code == 200, r.text
+
