@@ -223,6 +223,7 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
         if not isinstance(messages, list) or not messages:
             raise HTTPException(status_code=400,
                                 detail="messages must be a non-empty array")
+        total = 0
         for m in messages:             # Joi parity (ollama.ts:60-117): a
             if not isinstance(m, dict) or \
                     not isinstance(m.get("role", ""), str) or \
@@ -231,6 +232,12 @@ def build_router(scheduler: JobScheduler, registry: WorkerRegistry) -> APIRouter
                     status_code=400,   # crash inside the worker
                     detail="each message needs a string role and "
                            "string/array content")
+            c = m.get("content", "")
+            total += len(c) if isinstance(c, str) else \
+                sum(len(str(p)) for p in c)
+        if total > 100_000:            # same cap the generate path has
+            raise HTTPException(status_code=400,
+                                detail="messages too long (max 100kB)")
         _validate_model(registry, model)
         stream = bool(body.get("stream", True))
         opts = _clean_options(body.get("options"))

@@ -752,3 +752,17 @@ def test_junk_payload_types_rejected_across_routes(client):
         "model": "tiny", "prompt": ["p"], "max_tokens": 3})
     assert r.status_code == 200, r.text
 
+
+
+def test_chat_messages_size_capped(client):
+    """The 100 kB cap applies to chat content too (the generate-path cap
+    was bypassable through /api/chat)."""
+    r = client.post("/api/chat", json={
+        "model": "tiny", "stream": False,
+        "messages": [{"role": "user", "content": "x" * 150_000}]})
+    assert r.status_code == 400, r.text
+    r = client.post("/api/chat", json={
+        "model": "tiny", "stream": False,
+        "messages": [{"role": "user", "content": "hello"}],
+        "options": {"num_predict": 2}})
+    assert r.status_code == 200, r.text
