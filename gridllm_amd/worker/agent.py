@@ -102,7 +102,14 @@ class WorkerAgent:
 
     async def start(self) -> None:
         await self.bus.identify(self.worker_id)
-        await self._register()
+        # subscribe BEFORE announcing registration: the bus processes this
+        # connection's frames in order and the scheduler only assigns
+        # after it sees worker:registered, so the job channel is always
+        # live by then. The reverse order LOST whole assignment bursts:
+        # a respawned worker with a queued backlog was re-registered and
+        # immediately assigned 16 jobs into the subscription gap — they
+        # sat "active" on a live idle worker until the job timeout
+        # (chaos repro: gpurun_out/lt_server_ready.log 10:46:29).
         await self.bus.subscribe(Channels.worker_job(self.worker_id),
                                  self._h_job_msg)
         await self.bus.subscribe(Channels.worker_reregister(self.worker_id),
@@ -110,6 +117,7 @@ class WorkerAgent:
         await self.bus.subscribe(Channels.worker_control(self.worker_id),
                                  self._h_control)
         await self.bus.subscribe(Channels.BROADCAST, self._h_control)
+        await self._register()
         loop = asyncio.get_running_loop()
         self._tasks.append(loop.create_task(self._heartbeat_loop()))
         self._tasks.append(loop.create_task(self._resource_loop()))

@@ -343,3 +343,41 @@ async def test_tp_wedge_watchdog_exits_group():
         engine.tp = real_tp
         batcher.busy = False
     await teardown(reg, sched, agent)
+
+
+@pytest.mark.asyncio
+async def test_agent_subscribes_before_registering():
+    """Ordering regression: the job-channel subscription must be live
+    BEFORE worker:registered is announced — the reverse order lost whole
+    assignment bursts to a respawned worker (scheduler assigns the
+    queued backlog the moment it sees the registration; chaos repro:
+    profiles/r02_logs/r02_chaos_lost_assignments.log, 16 jobs stuck
+    'active' on a live idle worker)."""
+    from gridllm_amd.types import Channels
+    core = MemoryBusCore()
+    bus = MemoryBus(core)
+    await bus.connect()
+    order = []
+    orig_sub, orig_pub = bus.subscribe, bus.publish
+
+    async def sub(ch, h):
+        order.append(("sub", ch))
+        return await orig_sub(ch, h)
+
+    async def pub(ch, payload):
+        order.append(("pub", ch))
+        return await orig_pub(ch, payload)
+
+    bus.subscribe, bus.publish = sub, pub
+    agent = WorkerAgent(bus, get_engine(),
+                        WorkerConfig(worker_id="w-order",
+                                     max_concurrent_tasks=1,
+                                     heartbeat_interval_ms=60_000))
+    await agent.start()
+    try:
+        job_ch = Channels.worker_job("w-order")
+        sub_i = order.index(("sub", job_ch))
+        reg_i = order.index(("pub", Channels.WORKER_REGISTERED))
+        assert sub_i < reg_i, order
+    finally:
+        await agent.stop()
