@@ -317,3 +317,70 @@ def test_module_entrypoints_e2e():
                     p.wait(timeout=10)
                 except Exception:  # noqa: BLE001
                     p.kill()
+
+
+def test_cancel_storm_drains_and_recovers():
+    """Half of 64 concurrent streams are abandoned mid-flight (client
+    walks away) for two rounds: no errors on the surviving half, the
+    queue drains to zero (disconnect-cancel + deferred KV-slot frees
+    under the pipelined decode don't leak), and the worker serves
+    afterwards."""
+    import asyncio
+    import aiohttp
+    import httpx
+    bus_port, http_port = _free_port(), _free_port()
+    env = {"BUS_PORT": str(bus_port), "HTTP_PORT": str(http_port)}
+    server = _spawn(SERVER_SRC, env, "SERVER_READY")
+    worker = _spawn(WORKER_SRC, {**env, "WORKER_ID": "w-cstorm"},
+                    "WORKER_READY")
+    try:
+        base = f"http://127.0.0.1:{http_port}"
+        with httpx.Client(base_url=base, timeout=30) as c:
+            for _ in range(100):
+                r = c.get("/health/workers")
+                if r.status_code == 200 and \
+                        r.json()["counts"]["total"] == 1:
+                    break
+                time.sleep(0.1)
+
+        async def storm():
+            stats = {"done": 0, "gone": 0, "err": 0}
+            async with aiohttp.ClientSession(base) as s:
+                async def gen(i):
+                    try:
+                        async with s.post("/api/generate", json={
+                                "model": "tiny", "prompt": f"storm {i}",
+                                "stream": True,
+                                "options": {"temperature": 0.8, "seed": i,
+                                            "num_predict": 48}},
+                                timeout=aiohttp.ClientTimeout(
+                                    total=60)) as r:
+                            async for raw in r.content:
+                                if raw.strip() and i % 2 == 1:
+                                    stats["gone"] += 1
+                                    return       # abandon mid-stream
+                            stats["done"] += 1
+                    except Exception:  # noqa: BLE001
+                        stats["err"] += 1
+                for rnd in range(2):
+                    await asyncio.gather(
+                        *(gen(rnd * 64 + i) for i in range(64)))
+            return stats
+
+        st = asyncio.run(storm())
+        assert st["err"] == 0 and st["done"] == 64 and st["gone"] == 64, st
+        with httpx.Client(base_url=base, timeout=30) as c:
+            for _ in range(150):
+                q = c.get("/inference/queue").json()["queue"]
+                if q["active"] == 0 and q["queued"] == 0:
+                    break
+                time.sleep(0.2)
+            else:
+                raise AssertionError(f"not drained: {q}")
+            r = c.post("/api/generate", json={
+                "model": "tiny", "prompt": "after", "stream": False,
+                "options": {"num_predict": 3}})
+            assert r.status_code == 200 and r.json()["eval_count"] == 3
+    finally:
+        server.kill()
+        worker.kill()
