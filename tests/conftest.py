@@ -30,3 +30,12 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip)
+
+
+def pytest_runtest_logreport(report):
+    """Append every failure to .pytest_failures.log — flaky failures in
+    CI loops otherwise vanish with the scrollback."""
+    if report.failed and report.when == "call":
+        import time as _t
+        with open(".pytest_failures.log", "a") as f:
+            f.write(f"{_t.strftime('%H:%M:%S')} {report.nodeid}\n")
